@@ -1,0 +1,113 @@
+"""Generic transport conformance test — one shared scenario instantiated per
+transport (the reference's tier-2 test, ``protocols/mod.rs:396-481``):
+bind -> accept/finalize -> bidirectional send/recv -> soft close."""
+
+import asyncio
+
+import pytest
+
+from pushcdn_amd.proto import message as m
+from pushcdn_amd.proto.errors import ConnectionError_
+from pushcdn_amd.proto.limiter import Bytes, Limiter
+from pushcdn_amd.proto.transports.memory import Memory, gen_testing_connection_pair
+from pushcdn_amd.proto.transports.tcp import Tcp
+
+
+def run(coro):
+    return asyncio.run(coro)
+
+
+async def _conformance(protocol, endpoint):
+    limiter = Limiter(global_memory_pool_size=1 << 20)
+    listener = await protocol.bind(endpoint, None, None)
+    bound_endpoint = endpoint
+    if hasattr(listener, "port"):
+        bound_endpoint = f"127.0.0.1:{listener.port}"
+
+    async def server():
+        unfinalized = await listener.accept()
+        conn = await unfinalized.finalize(limiter)
+        msg = await conn.recv_message()
+        assert msg == m.Direct(b"client", b"hello from client")
+        await conn.send_message(m.Direct(b"server", b"hello from server"))
+        # raw path too
+        await conn.send_message_raw(Bytes(m.serialize(m.Subscribe([1, 2]))))
+        await conn.soft_close()
+
+    async def client():
+        conn = await protocol.connect(bound_endpoint, True, limiter)
+        await conn.send_message(m.Direct(b"client", b"hello from client"))
+        reply = await conn.recv_message()
+        assert reply == m.Direct(b"server", b"hello from server")
+        raw = await conn.recv_message_raw()
+        assert m.deserialize(raw.data) == m.Subscribe([1, 2])
+        raw.drop()
+        await conn.soft_close()
+
+    await asyncio.wait_for(asyncio.gather(server(), client()), timeout=10)
+    await listener.close()
+
+
+def test_tcp_conformance():
+    run(_conformance(Tcp, "127.0.0.1:0"))
+
+
+def test_memory_conformance():
+    run(_conformance(Memory, "test-endpoint-1"))
+
+
+def test_connect_refused():
+    async def go():
+        limiter = Limiter()
+        with pytest.raises(ConnectionError_):
+            await Tcp.connect("127.0.0.1:1", False, limiter)
+        with pytest.raises(ConnectionError_):
+            await Memory.connect("nonexistent", False, limiter)
+
+    run(go())
+
+
+def test_memory_pair_send_recv():
+    async def go():
+        limiter = Limiter()
+        a, b = gen_testing_connection_pair(limiter)
+        await a.send_message(m.Broadcast([0], b"x" * 10))
+        got = await b.recv_message()
+        assert got == m.Broadcast([0], b"x" * 10)
+
+    run(go())
+
+
+def test_large_message_over_tcp():
+    async def go():
+        limiter = Limiter(global_memory_pool_size=1 << 24)
+        listener = await Tcp.bind("127.0.0.1:0", None, None)
+        endpoint = f"127.0.0.1:{listener.port}"
+        payload = bytes(range(256)) * 8192  # 2 MiB
+
+        async def server():
+            conn = await (await listener.accept()).finalize(limiter)
+            msg = await conn.recv_message()
+            assert msg.message == payload
+            await conn.soft_close()
+
+        async def client():
+            conn = await Tcp.connect(endpoint, True, limiter)
+            await conn.send_message(m.Broadcast([1], payload))
+            await conn.soft_close()
+
+        await asyncio.wait_for(asyncio.gather(server(), client()), timeout=15)
+        await listener.close()
+
+    run(go())
+
+
+def test_recv_on_dead_connection_errors():
+    async def go():
+        limiter = Limiter()
+        a, b = gen_testing_connection_pair(limiter)
+        await a.soft_close()
+        with pytest.raises(ConnectionError_):
+            await asyncio.wait_for(b.recv_message(), timeout=5)
+
+    run(go())
