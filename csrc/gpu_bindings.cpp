@@ -1,0 +1,165 @@
+// Torch extension bindings for the MI355X broker data-plane kernels
+// (csrc/hip/dataplane.hip). PyTorch owns every HBM buffer; these calls
+// invoke the extern "C" launchers defined next to the kernels.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+struct ParseOut {
+    int32_t* disc;
+    int64_t* payload_off;
+    int32_t* payload_len;
+    int64_t* topics_off;
+    int32_t* topics_cnt;
+    uint64_t* recip_hash;
+    uint64_t* timestamp;
+};
+
+extern "C" {
+void launch_k4_parse(const uint8_t*, const int64_t*, int32_t, ParseOut, hipStream_t);
+void launch_k2a_topic_mask(const uint64_t*, const uint8_t*, const int64_t*, const int32_t*,
+                           const int32_t*, uint64_t*, int32_t, int32_t, hipStream_t);
+void launch_k2b_count(const uint64_t*, int32_t, int32_t, int32_t, int32_t*, hipStream_t);
+void launch_k2b_emit(const uint64_t*, const int64_t*, const int32_t*, const int32_t*, int32_t,
+                     int32_t, int32_t, int64_t, uint64_t*, int32_t*, int32_t*, int64_t*,
+                     uint32_t*, hipStream_t);
+void launch_k3_fanout(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
+                      const int32_t*, const int64_t*, const uint32_t*, int32_t, uint8_t*,
+                      hipStream_t);
+void launch_k5_direct_lookup(const uint64_t*, const int32_t*, int64_t, const uint64_t*, int32_t,
+                             int32_t*, hipStream_t);
+void launch_k2c_apply_subs(uint64_t*, const uint8_t*, const int64_t*, const int32_t*,
+                           const int32_t*, const int32_t*, int32_t, int32_t, hipStream_t);
+}
+
+#define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+static inline hipStream_t cur_stream() {
+    return at::hip::getCurrentHIPStream().stream();
+}
+
+std::vector<torch::Tensor> parse_batch(torch::Tensor buf, torch::Tensor offsets) {
+    CHECK_DEV(buf); CHECK_CONTIG(buf); CHECK_DEV(offsets); CHECK_CONTIG(offsets);
+    TORCH_CHECK(buf.dtype() == torch::kUInt8 && offsets.dtype() == torch::kInt64);
+    int32_t M = (int32_t)offsets.size(0) - 1;
+    TORCH_CHECK(M >= 0);
+    auto o32 = torch::TensorOptions().dtype(torch::kInt32).device(buf.device());
+    auto o64 = torch::TensorOptions().dtype(torch::kInt64).device(buf.device());
+    auto disc = torch::empty({M}, o32);
+    auto payload_off = torch::empty({M}, o64);
+    auto payload_len = torch::empty({M}, o32);
+    auto topics_off = torch::empty({M}, o64);
+    auto topics_cnt = torch::empty({M}, o32);
+    auto recip_hash = torch::empty({M}, o64);  // bit-cast u64
+    auto timestamp = torch::empty({M}, o64);
+    if (M > 0) {
+        ParseOut out{disc.data_ptr<int32_t>(), payload_off.data_ptr<int64_t>(),
+                     payload_len.data_ptr<int32_t>(), topics_off.data_ptr<int64_t>(),
+                     topics_cnt.data_ptr<int32_t>(),
+                     (uint64_t*)recip_hash.data_ptr<int64_t>(),
+                     (uint64_t*)timestamp.data_ptr<int64_t>()};
+        launch_k4_parse(buf.data_ptr<uint8_t>(), offsets.data_ptr<int64_t>(), M, out,
+                        cur_stream());
+    }
+    return {disc, payload_off, payload_len, topics_off, topics_cnt, recip_hash, timestamp};
+}
+
+torch::Tensor topic_mask(torch::Tensor sub_bitmap, torch::Tensor buf, torch::Tensor topics_off,
+                         torch::Tensor topics_cnt, torch::Tensor disc) {
+    CHECK_DEV(sub_bitmap); CHECK_CONTIG(sub_bitmap);
+    TORCH_CHECK(sub_bitmap.dim() == 2 && sub_bitmap.size(0) == 256);
+    int32_t W = (int32_t)sub_bitmap.size(1);
+    int32_t M = (int32_t)disc.size(0);
+    auto mask = torch::empty({M, (int64_t)W},
+                             torch::TensorOptions().dtype(torch::kInt64).device(buf.device()));
+    if (M > 0) {
+        launch_k2a_topic_mask((const uint64_t*)sub_bitmap.data_ptr<int64_t>(),
+                              buf.data_ptr<uint8_t>(), topics_off.data_ptr<int64_t>(),
+                              topics_cnt.data_ptr<int32_t>(), disc.data_ptr<int32_t>(),
+                              (uint64_t*)mask.data_ptr<int64_t>(), M, W, cur_stream());
+    }
+    return mask;
+}
+
+std::vector<torch::Tensor> assign_emit(torch::Tensor mask, torch::Tensor payload_off,
+                                       torch::Tensor payload_len, torch::Tensor ring_wpos,
+                                       int64_t ring_bytes, int64_t n_users) {
+    CHECK_DEV(mask); CHECK_CONTIG(mask);
+    int32_t M = (int32_t)mask.size(0);
+    int32_t W = (int32_t)mask.size(1);
+    TORCH_CHECK(ring_bytes % 16 == 0, "ring_bytes must be a multiple of 16");
+    auto o32 = torch::TensorOptions().dtype(torch::kInt32).device(mask.device());
+    auto counts = torch::zeros({n_users}, o32);
+    launch_k2b_count((const uint64_t*)mask.data_ptr<int64_t>(), M, W, (int32_t)n_users,
+                     counts.data_ptr<int32_t>(), cur_stream());
+    auto cum = counts.cumsum(0).to(torch::kInt32);
+    auto pair_base = (cum - counts).contiguous();  // exclusive scan
+    int64_t total = cum[n_users - 1].item<int32_t>();  // one small D2H sync
+    auto pair_user = torch::empty({total}, o32);
+    auto pair_msg = torch::empty({total}, o32);
+    auto pair_dst = torch::empty({total},
+                                 torch::TensorOptions().dtype(torch::kInt64).device(mask.device()));
+    auto drops = torch::zeros({1}, o32);
+    launch_k2b_emit((const uint64_t*)mask.data_ptr<int64_t>(), payload_off.data_ptr<int64_t>(),
+                    payload_len.data_ptr<int32_t>(), pair_base.data_ptr<int32_t>(), M, W,
+                    (int32_t)n_users, ring_bytes, (uint64_t*)ring_wpos.data_ptr<int64_t>(),
+                    pair_user.data_ptr<int32_t>(), pair_msg.data_ptr<int32_t>(),
+                    pair_dst.data_ptr<int64_t>(), (uint32_t*)drops.data_ptr<int32_t>(),
+                    cur_stream());
+    return {pair_user, pair_msg, pair_dst, drops};
+}
+
+void fanout(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_len,
+            torch::Tensor pair_user, torch::Tensor pair_msg, torch::Tensor pair_dst,
+            torch::Tensor msg_seq, torch::Tensor egress) {
+    CHECK_DEV(egress); CHECK_CONTIG(egress);
+    int32_t n_pairs = (int32_t)pair_user.size(0);
+    if (n_pairs == 0) return;
+    launch_k3_fanout(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
+                     payload_len.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
+                     pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
+                     (const uint32_t*)msg_seq.data_ptr<int32_t>(), n_pairs,
+                     egress.data_ptr<uint8_t>(), cur_stream());
+}
+
+torch::Tensor direct_lookup(torch::Tensor table_keys, torch::Tensor table_vals,
+                            torch::Tensor query) {
+    CHECK_DEV(table_keys); CHECK_DEV(query);
+    int64_t S = table_keys.size(0);
+    TORCH_CHECK((S & (S - 1)) == 0, "table size must be a power of two");
+    int32_t N = (int32_t)query.size(0);
+    auto owner = torch::empty({N},
+                              torch::TensorOptions().dtype(torch::kInt32).device(query.device()));
+    if (N > 0) {
+        launch_k5_direct_lookup((const uint64_t*)table_keys.data_ptr<int64_t>(),
+                                table_vals.data_ptr<int32_t>(), S,
+                                (const uint64_t*)query.data_ptr<int64_t>(), N,
+                                owner.data_ptr<int32_t>(), cur_stream());
+    }
+    return owner;
+}
+
+void apply_subs(torch::Tensor sub_bitmap, torch::Tensor buf, torch::Tensor topics_off,
+                torch::Tensor topics_cnt, torch::Tensor disc, torch::Tensor user_idx) {
+    CHECK_DEV(sub_bitmap); CHECK_CONTIG(sub_bitmap);
+    int32_t W = (int32_t)sub_bitmap.size(1);
+    int32_t M = (int32_t)disc.size(0);
+    if (M == 0) return;
+    launch_k2c_apply_subs((uint64_t*)sub_bitmap.data_ptr<int64_t>(), buf.data_ptr<uint8_t>(),
+                          topics_off.data_ptr<int64_t>(), topics_cnt.data_ptr<int32_t>(),
+                          disc.data_ptr<int32_t>(), user_idx.data_ptr<int32_t>(), M, W,
+                          cur_stream());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("parse_batch", &parse_batch, "K4: on-device capnp parse of a message batch");
+    m.def("topic_mask", &topic_mask, "K2a: per-message subscriber mask");
+    m.def("assign_emit", &assign_emit, "K2b: per-user FIFO ring assignment + pair list");
+    m.def("fanout", &fanout, "K3: N-way payload fan-out into egress rings");
+    m.def("direct_lookup", &direct_lookup, "K5: batched direct-route hash probe");
+    m.def("apply_subs", &apply_subs, "K2c: apply subscribe/unsubscribe batch to bitmap");
+}

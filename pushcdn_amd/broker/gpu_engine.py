@@ -1,0 +1,257 @@
+"""The GPU broker data-plane engine (one instance per MI355X device).
+
+Owns the HBM-resident state and drives the kernel pipeline per tick:
+
+    ingest (H2D pinned staging)          [reference: per-conn reader tasks]
+      -> K4 parse_batch                  [message.rs deserialize]
+      -> K2a topic_mask                  [connections get_interested_by_topic]
+      -> (mesh all-gather via RCCL)      [try_send_to_brokers over xGMI]
+      -> K2b assign_emit                 [per-conn FIFO channel push]
+      -> K3 fanout                       [sender.rs raw-bytes fan-out copy]
+      -> K5 direct_lookup                [DirectMap get]
+
+State tensors (all owned by PyTorch, sized for 288 GB HBM):
+  sub_bitmap  int64 [256][W]      subscription bitmap (W = ceil(n_users/64))
+  egress      uint8 [n_users*ring_bytes]  per-user egress rings
+  ring_wpos   int64 [n_users]     ring write cursors (device)
+  direct_keys/vals                open-addressing DirectMap (u64 hash -> owner)
+
+Delivery record format in a ring: {u32 len, u32 seq, u64 pad} + payload,
+16-byte aligned.  A drain (the socket-write analog) reads the cursors back,
+consumes records, and resets the cursors.
+
+CPU mode (device="cpu") runs the pure-Python mirrors from ops.reference so
+the engine's semantics are testable without a GPU.
+"""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from ..utils.keyhash import fnv1a64
+
+
+@dataclass
+class TickStats:
+    n_messages: int = 0
+    n_broadcast: int = 0
+    n_direct: int = 0
+    n_deliveries: int = 0
+    n_drops: int = 0
+
+
+class GpuBrokerEngine:
+    def __init__(
+        self,
+        device: str = "cuda:0",
+        n_users: int = 10_000,
+        ring_bytes: int = 1 << 21,
+        direct_table_size: int = 1 << 16,
+        use_gpu_ops: Optional[bool] = None,
+    ) -> None:
+        self.device = torch.device(device)
+        self.is_cuda = self.device.type == "cuda"
+        if use_gpu_ops is None:
+            use_gpu_ops = self.is_cuda
+        self.use_gpu_ops = use_gpu_ops
+        if self.use_gpu_ops:
+            from ..ops import get_gpu_ops
+
+            self._ops = get_gpu_ops()
+        else:
+            self._ops = None
+
+        assert ring_bytes % 16 == 0
+        self.n_users = n_users
+        self.ring_bytes = ring_bytes
+        self.W = (n_users + 63) // 64
+
+        dev = self.device
+        self.sub_bitmap = torch.zeros((256, self.W), dtype=torch.int64, device=dev)
+        self.egress = torch.zeros(n_users * ring_bytes, dtype=torch.uint8, device=dev)
+        self.ring_wpos = torch.zeros(n_users, dtype=torch.int64, device=dev)
+        self.direct_keys = torch.zeros(direct_table_size, dtype=torch.int64, device=dev)
+        self.direct_vals = torch.zeros(direct_table_size, dtype=torch.int32, device=dev)
+        self._direct_entries: Dict[int, int] = {}
+        self.seq = 0
+        self.total = TickStats()
+
+    # ---------------- subscription management (host-driven) ----------------
+
+    def subscribe(self, user_idx: int, topics: List[int]) -> None:
+        w, bit = user_idx >> 6, 1 << (user_idx & 63)
+        bit = bit - (1 << 64) if bit >= (1 << 63) else bit
+        for t in topics:
+            self.sub_bitmap[t & 0xFF, w] |= bit
+
+    def unsubscribe(self, user_idx: int, topics: List[int]) -> None:
+        w, bit = user_idx >> 6, 1 << (user_idx & 63)
+        bit = bit - (1 << 64) if bit >= (1 << 63) else bit
+        for t in topics:
+            self.sub_bitmap[t & 0xFF, w] &= ~bit
+
+    def subscribe_all(self, topics: List[int]) -> None:
+        """Subscribe every user to the given topics (bulk, for benches)."""
+        full = torch.full((self.W,), -1, dtype=torch.int64)
+        tail = self.n_users & 63
+        if tail:
+            full[-1] = (1 << tail) - 1
+        for t in topics:
+            self.sub_bitmap[t & 0xFF] = full.to(self.device)
+
+    def register_direct(self, pubkey: bytes, owner: int) -> None:
+        """owner >= 0: local user index. owner < 0: -(broker_rank+2)."""
+        h = fnv1a64(pubkey)
+        self._direct_entries[h] = owner
+        self._rebuild_direct_table()
+
+    def _rebuild_direct_table(self) -> None:
+        from ..ops.reference import build_direct_table
+
+        keys, vals = build_direct_table(
+            list(self._direct_entries.items()), self.direct_keys.shape[0]
+        )
+        self.direct_keys.copy_(keys.to(self.device))
+        self.direct_vals.copy_(vals.to(self.device))
+
+    # ---------------------------- the hot tick ----------------------------
+
+    def ingest(self, batch: bytes, offsets: List[int]) -> Tuple[torch.Tensor, torch.Tensor]:
+        """H2D-copy one batch of serialized messages. Returns device (buf, offsets)."""
+        host = torch.frombuffer(bytearray(batch), dtype=torch.uint8)
+        off = torch.tensor(offsets, dtype=torch.int64)
+        if self.is_cuda:
+            return host.to(self.device, non_blocking=True), off.to(self.device, non_blocking=True)
+        return host, off
+
+    def tick(self, buf: torch.Tensor, offsets: torch.Tensor,
+             host_batch: Optional[bytes] = None,
+             host_offsets: Optional[List[int]] = None) -> TickStats:
+        """Run the full pipeline on one ingested batch already on device."""
+        if self.use_gpu_ops:
+            return self._tick_gpu(buf, offsets)
+        assert host_batch is not None and host_offsets is not None
+        return self._tick_cpu(host_batch, host_offsets)
+
+    def _tick_gpu(self, buf: torch.Tensor, offsets: torch.Tensor) -> TickStats:
+        ops = self._ops
+        M = offsets.shape[0] - 1
+        disc, payload_off, payload_len, topics_off, topics_cnt, recip_hash, _ts = ops.parse_batch(
+            buf, offsets
+        )
+        mask = ops.topic_mask(self.sub_bitmap, buf, topics_off, topics_cnt, disc)
+        pair_user, pair_msg, pair_dst, drops = ops.assign_emit(
+            mask, payload_off, payload_len, self.ring_wpos, self.ring_bytes, self.n_users
+        )
+        seq = torch.arange(self.seq, self.seq + M, dtype=torch.int32, device=self.device)
+        self.seq += M
+        ops.fanout(buf, payload_off, payload_len, pair_user, pair_msg, pair_dst, seq, self.egress)
+        # Direct messages: route via the DirectMap (local deliveries only here;
+        # remote ones are the mesh's job).
+        owner = ops.direct_lookup(self.direct_keys, self.direct_vals, recip_hash)
+        stats = TickStats(n_messages=M, n_deliveries=int(pair_user.shape[0]),
+                          n_drops=int(drops.item()) if drops.numel() else 0)
+        self._route_direct_gpu(buf, payload_off, payload_len, disc, owner, seq)
+        return stats
+
+    def _route_direct_gpu(self, buf, payload_off, payload_len, disc, owner, seq) -> None:
+        """Deliver direct messages to local users by synthesizing delivery
+        pairs for the K3 copy kernel (single-recipient fan-out)."""
+        ops = self._ops
+        is_direct = disc == 3
+        local = (owner >= 0) & is_direct
+        if not bool(local.any()):
+            return
+        # Direct traffic is low-rate vs broadcast fan-out: assign ring space
+        # in a small host loop (handles several directs to one user in one
+        # tick with correct FIFO offsets), then reuse K3 for the copies.
+        idx = torch.nonzero(local).flatten()
+        users_h = owner[idx].to("cpu").tolist()
+        lens_h = payload_len[idx].to("cpu").tolist()
+        wpos_h = {u: int(self.ring_wpos[u]) for u in set(users_h)}
+        pair_user: List[int] = []
+        pair_msg: List[int] = []
+        pair_dst: List[int] = []
+        for j, (u, length) in enumerate(zip(users_h, lens_h)):
+            rec = 16 + ((length + 15) & ~15)
+            if wpos_h[u] + rec > self.ring_bytes:
+                pair_user.append(-1)
+                pair_msg.append(int(idx[j]))
+                pair_dst.append(0)
+                continue
+            pair_user.append(u)
+            pair_msg.append(int(idx[j]))
+            pair_dst.append(u * self.ring_bytes + wpos_h[u])
+            wpos_h[u] += rec
+        for u, w in wpos_h.items():
+            self.ring_wpos[u] = w
+        dev = self.device
+        ops.fanout(
+            buf, payload_off, payload_len,
+            torch.tensor(pair_user, dtype=torch.int32, device=dev),
+            torch.tensor(pair_msg, dtype=torch.int32, device=dev),
+            torch.tensor(pair_dst, dtype=torch.int64, device=dev),
+            seq, self.egress,
+        )
+
+    def _tick_cpu(self, batch: bytes, offsets: List[int]) -> TickStats:
+        from ..ops import reference as ref
+
+        pr = ref.parse_batch(batch, offsets)
+        M = len(offsets) - 1
+        mask = ref.topic_mask(self.sub_bitmap, batch, pr.topics_off, pr.topics_cnt, pr.disc)
+        pair_user, pair_msg, pair_dst, drops = ref.assign_emit(
+            mask, pr.payload_len, self.ring_wpos, self.ring_bytes, self.n_users
+        )
+        seq = torch.arange(self.seq, self.seq + M, dtype=torch.int32)
+        self.seq += M
+        arr = bytearray(self.egress.numpy().tobytes())  # copy-in; copied back below
+        ref.fanout(batch, pr.payload_off, pr.payload_len, pair_user, pair_msg, pair_dst, seq, arr)
+        owner = ref.direct_lookup(self.direct_keys, self.direct_vals, pr.recip_hash)
+        for i in range(M):
+            if int(pr.disc[i]) == 3 and int(owner[i]) >= 0:
+                u = int(owner[i])
+                length = int(pr.payload_len[i])
+                rec = 16 + ((length + 15) & ~15)
+                wpos = int(self.ring_wpos[u])
+                if wpos + rec <= self.ring_bytes:
+                    dst = u * self.ring_bytes + wpos
+                    import struct
+
+                    struct.pack_into("<IIII", arr, dst, length, int(seq[i]) & 0xFFFFFFFF, 0, 0)
+                    arr[dst + 16 : dst + 16 + length] = batch[
+                        int(pr.payload_off[i]) : int(pr.payload_off[i]) + length
+                    ]
+                    self.ring_wpos[u] = wpos + rec
+        self.egress.copy_(torch.frombuffer(arr, dtype=torch.uint8))
+        return TickStats(n_messages=M, n_deliveries=int(pair_user.shape[0]), n_drops=drops)
+
+    # ------------------------------- drain ---------------------------------
+
+    def drain_cursors(self) -> torch.Tensor:
+        """Read back ring cursors (the per-tick notify) and reset them."""
+        wpos = self.ring_wpos.detach().clone().to("cpu")
+        self.ring_wpos.zero_()
+        return wpos
+
+    def read_ring(self, user_idx: int, nbytes: Optional[int] = None) -> bytes:
+        n = self.ring_bytes if nbytes is None else nbytes
+        start = user_idx * self.ring_bytes
+        return bytes(self.egress[start : start + n].to("cpu").numpy().tobytes())
+
+
+def parse_ring_records(ring: bytes, wpos: int) -> List[Tuple[int, bytes]]:
+    """Parse delivery records out of a drained ring: [(seq, payload), ...]."""
+    out = []
+    pos = 0
+    while pos + 16 <= wpos:
+        length = int.from_bytes(ring[pos : pos + 4], "little")
+        seq = int.from_bytes(ring[pos + 4 : pos + 8], "little")
+        payload = ring[pos + 16 : pos + 16 + length]
+        out.append((seq, payload))
+        pos += 16 + ((length + 15) & ~15)
+    return out
