@@ -153,11 +153,11 @@ extern "C" __global__ void k4_parse_batch(
         out.timestamp[i] = ld_u64(seg + it * 8);
         int64_t ko; int32_t kl;
         if (!read_byte_list(seg, nwords, it + idw, &ko, &kl)) return;
-        out.payload_off[i] = seg_base + ko;  // public key bytes
+        out.payload_off[i] = kl ? seg_base + ko : 0;  // public key bytes
         out.payload_len[i] = kl;
         int64_t so; int32_t sl;
         if (!read_byte_list(seg, nwords, it + idw + 1, &so, &sl)) return;
-        out.topics_off[i] = seg_base + so;   // signature bytes (reused slot)
+        out.topics_off[i] = sl ? seg_base + so : 0;   // signature bytes (reused slot)
         out.topics_cnt[i] = sl;
         break;
     }
@@ -175,7 +175,7 @@ extern "C" __global__ void k4_parse_batch(
         out.timestamp[i] = ld_u64(seg + it * 8);  // permit
         int64_t co; int32_t cl;
         if (!read_byte_list(seg, nwords, it + idw, &co, &cl)) return;
-        out.payload_off[i] = seg_base + co; out.payload_len[i] = cl;
+        out.payload_off[i] = cl ? seg_base + co : 0; out.payload_len[i] = cl;
         break;
     }
     case 3: {  // Direct
@@ -187,7 +187,7 @@ extern "C" __global__ void k4_parse_batch(
         out.recip_hash[i] = fnv1a64(seg + ro, rl);
         int64_t mo; int32_t ml;
         if (!read_byte_list(seg, nwords, it + idw + 1, &mo, &ml)) return;
-        out.payload_off[i] = seg_base + mo; out.payload_len[i] = ml;
+        out.payload_off[i] = ml ? seg_base + mo : 0; out.payload_len[i] = ml;
         break;
     }
     case 4: {  // Broadcast
@@ -196,22 +196,22 @@ extern "C" __global__ void k4_parse_batch(
         if (ipw < 2) return;
         int64_t to; int32_t tc;
         if (!read_byte_list(seg, nwords, it + idw, &to, &tc)) return;
-        out.topics_off[i] = seg_base + to; out.topics_cnt[i] = tc;
+        out.topics_off[i] = tc ? seg_base + to : 0; out.topics_cnt[i] = tc;
         int64_t mo; int32_t ml;
         if (!read_byte_list(seg, nwords, it + idw + 1, &mo, &ml)) return;
-        out.payload_off[i] = seg_base + mo; out.payload_len[i] = ml;
+        out.payload_off[i] = ml ? seg_base + mo : 0; out.payload_len[i] = ml;
         break;
     }
     case 5: case 6: {  // Subscribe / Unsubscribe
         int64_t to; int32_t tc;
         if (!read_byte_list(seg, nwords, up, &to, &tc)) return;
-        out.topics_off[i] = seg_base + to; out.topics_cnt[i] = tc;
+        out.topics_off[i] = tc ? seg_base + to : 0; out.topics_cnt[i] = tc;
         break;
     }
     case 7: case 8: {  // UserSync / TopicSync
         int64_t dof; int32_t dl;
         if (!read_byte_list(seg, nwords, up, &dof, &dl)) return;
-        out.payload_off[i] = seg_base + dof; out.payload_len[i] = dl;
+        out.payload_off[i] = dl ? seg_base + dof : 0; out.payload_len[i] = dl;
         break;
     }
     default:

@@ -52,59 +52,20 @@ def parse_batch(buf: bytes, offsets: List[int]) -> ParseResult:
 
     for i in range(M):
         raw = buf[offsets[i] : offsets[i + 1]]
+        base = offsets[i]
         try:
-            parsed = msglib.deserialize(raw)
+            r = msglib.parse_offsets(raw)
         except Exception:
             continue
-        base = offsets[i]
-        d = msglib._DISCRIMINANT[type(parsed)]
-        if isinstance(parsed, msglib.AuthenticateWithKey):
-            timestamp[i] = _i64(parsed.timestamp)
-            off = raw.index(parsed.public_key) if parsed.public_key else 0
-            payload_off[i] = base + off
-            payload_len[i] = len(parsed.public_key)
-            soff = raw.index(parsed.signature) if parsed.signature else 0
-            topics_off[i] = base + soff
-            topics_cnt[i] = len(parsed.signature)
-        elif isinstance(parsed, msglib.AuthenticateWithPermit):
-            timestamp[i] = _i64(parsed.permit)
-        elif isinstance(parsed, msglib.AuthenticateResponse):
-            timestamp[i] = _i64(parsed.permit)
-            ctx = parsed.context.encode() + b"\x00"
-            off = raw.index(ctx) if ctx.strip(b"\x00") else 0
-            payload_off[i] = base + (off if ctx.strip(b"\x00") else 0)
-            payload_len[i] = len(ctx)
-        elif isinstance(parsed, msglib.Direct):
-            recip_hash[i] = _i64(fnv1a64(parsed.recipient))
-            off = _find_payload(raw, parsed.message)
-            payload_off[i] = base + off
-            payload_len[i] = len(parsed.message)
-        elif isinstance(parsed, msglib.Broadcast):
-            tbytes = bytes(bytearray(parsed.topics))
-            toff = _find_payload(raw, tbytes)
-            topics_off[i] = base + toff
-            topics_cnt[i] = len(parsed.topics)
-            off = _find_payload(raw, parsed.message)
-            payload_off[i] = base + off
-            payload_len[i] = len(parsed.message)
-        elif isinstance(parsed, (msglib.Subscribe, msglib.Unsubscribe)):
-            tbytes = bytes(bytearray(parsed.topics))
-            toff = _find_payload(raw, tbytes)
-            topics_off[i] = base + toff
-            topics_cnt[i] = len(parsed.topics)
-        elif isinstance(parsed, (msglib.UserSync, msglib.TopicSync)):
-            off = _find_payload(raw, parsed.data)
-            payload_off[i] = base + off
-            payload_len[i] = len(parsed.data)
-        disc[i] = d
+        disc[i] = r["disc"]
+        payload_off[i] = base + r["payload_off"] if r["payload_len"] else 0
+        payload_len[i] = r["payload_len"]
+        topics_off[i] = base + r["topics_off"] if r["topics_cnt"] else 0
+        topics_cnt[i] = r["topics_cnt"]
+        timestamp[i] = _i64(r["timestamp"])
+        if r["disc"] == 3:
+            recip_hash[i] = _i64(fnv1a64(r["recipient"]))
     return ParseResult(disc, payload_off, payload_len, topics_off, topics_cnt, recip_hash, timestamp)
-
-
-def _find_payload(raw: bytes, needle: bytes) -> int:
-    if not needle:
-        return 0
-    idx = raw.find(needle)
-    return idx if idx >= 0 else 0
 
 
 def topic_mask(

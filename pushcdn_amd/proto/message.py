@@ -328,3 +328,81 @@ def deserialize(data: bytes) -> Message:
     if disc == 8:
         return TopicSync(seg.read_byte_list(union_ptr))
     raise DeserializeError(f"unknown discriminant {disc}")
+
+
+def parse_offsets(data: bytes):
+    """Structural parse returning byte offsets into ``data`` — the host mirror
+    of the K4 device kernel (csrc/hip/dataplane.hip k4_parse_batch).
+
+    Returns dict with: disc, payload_off, payload_len, topics_off, topics_cnt,
+    recipient (bytes), timestamp.  Offsets are 0 whenever the length is 0
+    (canonicalized, matching the kernel).  Raises DeserializeError on garbage.
+    """
+
+    def span(seg: _SegmentReader, seg_base: int, ptr_word: int):
+        val = seg.u64(ptr_word)
+        if val == 0:
+            return 0, 0
+        raw = seg.read_byte_list(ptr_word)
+        if not raw:
+            return 0, 0
+        b = (val >> 2) & 0x3FFFFFFF
+        if b & 0x20000000:
+            b -= 0x40000000
+        tgt = ptr_word + 1 + b
+        return seg_base + tgt * 8, len(raw)
+
+    out = {
+        "disc": -1, "payload_off": 0, "payload_len": 0,
+        "topics_off": 0, "topics_cnt": 0, "recipient": b"", "timestamp": 0,
+    }
+    if len(data) < 16:
+        raise DeserializeError("short buffer")
+    seg_count_m1, nwords = struct.unpack_from("<II", data, 0)
+    if seg_count_m1 != 0 or 8 + nwords * 8 > len(data):
+        raise DeserializeError("bad stream header")
+    seg = _SegmentReader(data[8 : 8 + nwords * 8])
+    seg_base = 8
+    tgt, dw, pw = seg.read_struct_ptr(0)
+    if dw < 1 or pw < 1:
+        raise DeserializeError("malformed Message struct")
+    disc = struct.unpack_from("<H", seg.data, tgt * 8)[0]
+    up = tgt + dw
+    if disc == 0:
+        it, idw, ipw = seg.read_struct_ptr(up)
+        if idw < 1 or ipw < 2:
+            raise DeserializeError("malformed AuthenticateWithKey")
+        out["timestamp"] = seg.u64(it)
+        out["payload_off"], out["payload_len"] = span(seg, seg_base, it + idw)
+        out["topics_off"], out["topics_cnt"] = span(seg, seg_base, it + idw + 1)
+    elif disc == 1:
+        it, idw, _ = seg.read_struct_ptr(up)
+        if idw < 1:
+            raise DeserializeError("malformed AuthenticateWithPermit")
+        out["timestamp"] = seg.u64(it)
+    elif disc == 2:
+        it, idw, ipw = seg.read_struct_ptr(up)
+        if idw < 1 or ipw < 1:
+            raise DeserializeError("malformed AuthenticateResponse")
+        out["timestamp"] = seg.u64(it)
+        out["payload_off"], out["payload_len"] = span(seg, seg_base, it + idw)
+    elif disc == 3:
+        it, idw, ipw = seg.read_struct_ptr(up)
+        if ipw < 2:
+            raise DeserializeError("malformed Direct")
+        out["recipient"] = seg.read_byte_list(it + idw)
+        out["payload_off"], out["payload_len"] = span(seg, seg_base, it + idw + 1)
+    elif disc == 4:
+        it, idw, ipw = seg.read_struct_ptr(up)
+        if ipw < 2:
+            raise DeserializeError("malformed Broadcast")
+        out["topics_off"], out["topics_cnt"] = span(seg, seg_base, it + idw)
+        out["payload_off"], out["payload_len"] = span(seg, seg_base, it + idw + 1)
+    elif disc in (5, 6):
+        out["topics_off"], out["topics_cnt"] = span(seg, seg_base, up)
+    elif disc in (7, 8):
+        out["payload_off"], out["payload_len"] = span(seg, seg_base, up)
+    else:
+        raise DeserializeError(f"unknown discriminant {disc}")
+    out["disc"] = disc
+    return out
