@@ -1,0 +1,277 @@
+// BN254 optimal-ate pairing — shared host/device header.
+//
+// Tower: Fp2 = Fp[u]/(u^2+1), Fp6 = Fp2[v]/(v^3 - xi), Fp12 = Fp6[w]/(w^2 - v)
+// with xi = 9 + u.  D-type sextic twist.  Miller loop over the NAF of
+// 6x+2, followed by the two Frobenius line additions, then final
+// exponentiation (easy part structured, hard part by generic exponent —
+// correctness-first; the x-chain hard part is a later optimization).
+//
+// Used by the host BLS library (csrc/bls/bls.h) and the K1 batched-verify
+// kernel (csrc/hip/bls_kernels.hip).
+
+#pragma once
+#include "bn254.h"
+
+namespace bn254 {
+
+// ---------------------------------------------------------------------------
+// Fp6 = Fp2[v] / (v^3 - xi)
+// ---------------------------------------------------------------------------
+struct Fp6 {
+    Fp2 c0, c1, c2;
+
+    BN_INLINE static Fp6 zero() { return {Fp2::zero(), Fp2::zero(), Fp2::zero()}; }
+    BN_INLINE static Fp6 one() { return {Fp2::one(), Fp2::zero(), Fp2::zero()}; }
+    BN_INLINE bool is_zero() const { return c0.is_zero() && c1.is_zero() && c2.is_zero(); }
+    BN_INLINE bool operator==(const Fp6& o) const {
+        return c0 == o.c0 && c1 == o.c1 && c2 == o.c2;
+    }
+
+    BN_INLINE static Fp6 add(const Fp6& a, const Fp6& b) {
+        return {Fp2::add(a.c0, b.c0), Fp2::add(a.c1, b.c1), Fp2::add(a.c2, b.c2)};
+    }
+    BN_INLINE static Fp6 sub(const Fp6& a, const Fp6& b) {
+        return {Fp2::sub(a.c0, b.c0), Fp2::sub(a.c1, b.c1), Fp2::sub(a.c2, b.c2)};
+    }
+    BN_INLINE static Fp6 neg(const Fp6& a) {
+        return {Fp2::neg(a.c0), Fp2::neg(a.c1), Fp2::neg(a.c2)};
+    }
+
+    BN_INLINE static Fp6 mul(const Fp6& a, const Fp6& b) {
+        // Toom/Karatsuba (devegili): v0=a0b0, v1=a1b1, v2=a2b2
+        Fp2 v0 = Fp2::mul(a.c0, b.c0);
+        Fp2 v1 = Fp2::mul(a.c1, b.c1);
+        Fp2 v2 = Fp2::mul(a.c2, b.c2);
+        Fp2 t0 = Fp2::mul(Fp2::add(a.c1, a.c2), Fp2::add(b.c1, b.c2));
+        t0 = Fp2::sub(Fp2::sub(t0, v1), v2);          // a1b2 + a2b1
+        Fp2 r0 = Fp2::add(v0, Fp2::mul_xi(t0));
+        Fp2 t1 = Fp2::mul(Fp2::add(a.c0, a.c1), Fp2::add(b.c0, b.c1));
+        t1 = Fp2::sub(Fp2::sub(t1, v0), v1);          // a0b1 + a1b0
+        Fp2 r1 = Fp2::add(t1, Fp2::mul_xi(v2));
+        Fp2 t2 = Fp2::mul(Fp2::add(a.c0, a.c2), Fp2::add(b.c0, b.c2));
+        t2 = Fp2::sub(Fp2::sub(t2, v0), v2);          // a0b2 + a2b0
+        Fp2 r2 = Fp2::add(t2, v1);
+        return {r0, r1, r2};
+    }
+
+    BN_INLINE static Fp6 sqr(const Fp6& a) { return mul(a, a); }
+
+    BN_INLINE static Fp6 mul_fp2(const Fp6& a, const Fp2& b) {
+        return {Fp2::mul(a.c0, b), Fp2::mul(a.c1, b), Fp2::mul(a.c2, b)};
+    }
+
+    // multiply by v: (c0 + c1 v + c2 v^2) * v = xi c2 + c0 v + c1 v^2
+    BN_INLINE static Fp6 mul_v(const Fp6& a) {
+        return {Fp2::mul_xi(a.c2), a.c0, a.c1};
+    }
+
+    BN_INLINE Fp6 inv() const {
+        // standard: A = c0^2 - xi c1 c2, B = xi c2^2 - c0 c1, C = c1^2 - c0 c2
+        Fp2 A = Fp2::sub(Fp2::sqr(c0), Fp2::mul_xi(Fp2::mul(c1, c2)));
+        Fp2 B = Fp2::sub(Fp2::mul_xi(Fp2::sqr(c2)), Fp2::mul(c0, c1));
+        Fp2 C = Fp2::sub(Fp2::sqr(c1), Fp2::mul(c0, c2));
+        Fp2 den = Fp2::add(Fp2::mul(c0, A),
+                           Fp2::mul_xi(Fp2::add(Fp2::mul(c2, B), Fp2::mul(c1, C))));
+        Fp2 di = den.inv();
+        return {Fp2::mul(A, di), Fp2::mul(B, di), Fp2::mul(C, di)};
+    }
+};
+
+// ---------------------------------------------------------------------------
+// Fp12 = Fp6[w] / (w^2 - v)
+// ---------------------------------------------------------------------------
+struct Fp12 {
+    Fp6 c0, c1;
+
+    BN_INLINE static Fp12 one() { return {Fp6::one(), Fp6::zero()}; }
+    BN_INLINE bool operator==(const Fp12& o) const { return c0 == o.c0 && c1 == o.c1; }
+    BN_INLINE bool is_one() const { return *this == one(); }
+
+    BN_INLINE static Fp12 mul(const Fp12& a, const Fp12& b) {
+        Fp6 v0 = Fp6::mul(a.c0, b.c0);
+        Fp6 v1 = Fp6::mul(a.c1, b.c1);
+        Fp6 t = Fp6::mul(Fp6::add(a.c0, a.c1), Fp6::add(b.c0, b.c1));
+        return {Fp6::add(v0, Fp6::mul_v(v1)), Fp6::sub(Fp6::sub(t, v0), v1)};
+    }
+
+    BN_INLINE static Fp12 sqr(const Fp12& a) {
+        // complex squaring: (c0 + c1 w)^2 = (c0^2 + v c1^2) + 2 c0 c1 w
+        Fp6 v0 = Fp6::mul(a.c0, a.c1);
+        Fp6 t = Fp6::mul(Fp6::add(a.c0, a.c1), Fp6::add(a.c0, Fp6::mul_v(a.c1)));
+        Fp6 c0n = Fp6::sub(Fp6::sub(t, v0), Fp6::mul_v(v0));
+        return {c0n, Fp6::add(v0, v0)};
+    }
+
+    BN_INLINE static Fp12 conj(const Fp12& a) { return {a.c0, Fp6::neg(a.c1)}; }
+
+    BN_INLINE Fp12 inv() const {
+        // 1/(c0 + c1 w) = (c0 - c1 w) / (c0^2 - v c1^2)
+        Fp6 d = Fp6::sub(Fp6::sqr(c0), Fp6::mul_v(Fp6::sqr(c1)));
+        Fp6 di = d.inv();
+        return {Fp6::mul(c0, di), Fp6::neg(Fp6::mul(c1, di))};
+    }
+
+    // sparse multiply by a line element  l = a0 + (a3 + a4 v) w
+    // (positions: c0.c0 = a0, c1.c0 = a3, c1.c1 = a4)
+    BN_INLINE static Fp12 mul_by_034(const Fp12& f, const Fp2& a0, const Fp2& a3,
+                                     const Fp2& a4) {
+        Fp12 l{{a0, Fp2::zero(), Fp2::zero()}, {a3, a4, Fp2::zero()}};
+        return mul(f, l);
+    }
+
+    // Frobenius^2: c_ij -> c_ij * gamma2 factors (Fp scalars, no conjugation)
+    BN_INLINE static Fp12 frobenius2(const Fp12& a) {
+        Fp g1 = Fp::from_u256(from_limbs(bn254c::GAMMA2_1));
+        Fp g2 = Fp::from_u256(from_limbs(bn254c::GAMMA2_2));
+        Fp g3 = Fp::from_u256(from_limbs(bn254c::GAMMA2_3));
+        Fp g4 = Fp::from_u256(from_limbs(bn254c::GAMMA2_4));
+        Fp g5 = Fp::from_u256(from_limbs(bn254c::GAMMA2_5));
+        return {{a.c0.c0, Fp2::mul_fp(a.c0.c1, g2), Fp2::mul_fp(a.c0.c2, g4)},
+                {Fp2::mul_fp(a.c1.c0, g1), Fp2::mul_fp(a.c1.c1, g3),
+                 Fp2::mul_fp(a.c1.c2, g5)}};
+    }
+
+    // generic pow by a little-endian multi-limb exponent (standard form)
+    BN_INLINE static Fp12 pow_limbs(const Fp12& a, const uint64_t* limbs, int n) {
+        Fp12 result = one();
+        bool started = false;
+        for (int i = n - 1; i >= 0; --i) {
+            for (int b = 63; b >= 0; --b) {
+                if (started) result = sqr(result);
+                if ((limbs[i] >> b) & 1) {
+                    if (started) result = mul(result, a);
+                    else { result = a; started = true; }
+                }
+            }
+        }
+        return started ? result : one();
+    }
+};
+
+// ---------------------------------------------------------------------------
+// Miller loop (optimal ate, D-type twist, NAF of 6x+2)
+// ---------------------------------------------------------------------------
+struct G2Affine {
+    Fp2 x, y;
+};
+
+struct G2Proj {
+    Fp2 x, y, z;  // homogeneous projective
+};
+
+struct LineCoeffs {
+    Fp2 c0, c1, c2;
+};
+
+// doubling step (arkworks models/bn/g2.rs shape, D-twist coefficients)
+BN_INLINE LineCoeffs doubling_step(G2Proj& r, const Fp& two_inv) {
+    Fp2 a = Fp2::mul_fp(Fp2::mul(r.x, r.y), two_inv);
+    Fp2 b = Fp2::sqr(r.y);
+    Fp2 c = Fp2::sqr(r.z);
+    Fp2 b_twist{Fp::from_u256(from_limbs(bn254c::B2_C0)),
+                Fp::from_u256(from_limbs(bn254c::B2_C1))};
+    Fp2 e = Fp2::mul(b_twist, Fp2::add(Fp2::dbl(c), c));
+    Fp2 f = Fp2::add(Fp2::dbl(e), e);
+    Fp2 g = Fp2::mul_fp(Fp2::add(b, f), two_inv);
+    Fp2 h = Fp2::sub(Fp2::sqr(Fp2::add(r.y, r.z)), Fp2::add(b, c));
+    Fp2 i = Fp2::sub(e, b);
+    Fp2 j = Fp2::sqr(r.x);
+    Fp2 e2 = Fp2::sqr(e);
+    r.x = Fp2::mul(a, Fp2::sub(b, f));
+    r.y = Fp2::sub(Fp2::sqr(g), Fp2::add(Fp2::dbl(e2), e2));
+    r.z = Fp2::mul(b, h);
+    return {Fp2::neg(h), Fp2::add(Fp2::dbl(j), j), i};
+}
+
+// mixed addition step
+BN_INLINE LineCoeffs addition_step(G2Proj& r, const G2Affine& q) {
+    Fp2 theta = Fp2::sub(r.y, Fp2::mul(q.y, r.z));
+    Fp2 lambda = Fp2::sub(r.x, Fp2::mul(q.x, r.z));
+    Fp2 c = Fp2::sqr(theta);
+    Fp2 d = Fp2::sqr(lambda);
+    Fp2 e = Fp2::mul(lambda, d);
+    Fp2 f = Fp2::mul(r.z, c);
+    Fp2 g = Fp2::mul(r.x, d);
+    Fp2 h = Fp2::add(Fp2::sub(e, Fp2::dbl(g)), f);
+    r.x = Fp2::mul(lambda, h);
+    r.y = Fp2::sub(Fp2::mul(theta, Fp2::sub(g, h)), Fp2::mul(e, r.y));
+    r.z = Fp2::mul(r.z, e);
+    Fp2 j = Fp2::sub(Fp2::mul(theta, q.x), Fp2::mul(lambda, q.y));
+    return {lambda, Fp2::neg(theta), j};
+}
+
+// evaluate a line at P = (px, py) and fold into f (D-twist: c0 *= py, c1 *= px)
+BN_INLINE void ell(Fp12& f, const LineCoeffs& l, const Fp& px, const Fp& py) {
+    Fp2 c0 = Fp2::mul_fp(l.c0, py);
+    Fp2 c1 = Fp2::mul_fp(l.c1, px);
+    f = Fp12::mul_by_034(f, c0, c1, l.c2);
+}
+
+// Frobenius endomorphism on the twist: pi(x, y) = (conj(x) FROB_X, conj(y) FROB_Y)
+BN_INLINE G2Affine g2_frobenius(const G2Affine& q) {
+    Fp2 fx{Fp::from_u256(from_limbs(bn254c::FROB_X_C0)),
+           Fp::from_u256(from_limbs(bn254c::FROB_X_C1))};
+    Fp2 fy{Fp::from_u256(from_limbs(bn254c::FROB_Y_C0)),
+           Fp::from_u256(from_limbs(bn254c::FROB_Y_C1))};
+    return {Fp2::mul(Fp2::conj(q.x), fx), Fp2::mul(Fp2::conj(q.y), fy)};
+}
+
+// Miller loop for one pair (P in G1 affine, Q in G2 affine). Both must be
+// non-infinity (callers handle degenerate cases).
+BN_INLINE Fp12 miller_loop(const Fp& px, const Fp& py, const G2Affine& q) {
+    Fp two_inv = Fp::from_u64(2).inv();
+    G2Proj r{q.x, q.y, Fp2::one()};
+    G2Affine negq{q.x, Fp2::neg(q.y)};
+    Fp12 f = Fp12::one();
+    for (int i = bn254c::ATE_NAF_LEN - 2; i >= 0; --i) {
+        f = Fp12::sqr(f);
+        LineCoeffs l = doubling_step(r, two_inv);
+        ell(f, l, px, py);
+        int8_t d = bn254c::ATE_NAF[i];
+        if (d == 1) {
+            l = addition_step(r, q);
+            ell(f, l, px, py);
+        } else if (d == -1) {
+            l = addition_step(r, negq);
+            ell(f, l, px, py);
+        }
+    }
+    // Frobenius corrections: add pi(Q), then subtract pi^2(Q)
+    G2Affine q1 = g2_frobenius(q);
+    G2Affine q2 = g2_frobenius(q1);
+    q2.y = Fp2::neg(q2.y);
+    LineCoeffs l = addition_step(r, q1);
+    ell(f, l, px, py);
+    l = addition_step(r, q2);
+    ell(f, l, px, py);
+    return f;
+}
+
+// ---------------------------------------------------------------------------
+// Final exponentiation: f^((p^12-1)/r)
+// easy part structured; hard part = generic pow by (p^4-p^2+1)/r
+// ---------------------------------------------------------------------------
+BN_INLINE Fp12 final_exponentiation(const Fp12& f) {
+    // easy part: f^(p^6 - 1) = conj(f) * f^-1 ; then ^(p^2 + 1)
+    Fp12 f1 = Fp12::conj(f);
+    Fp12 f2 = f.inv();
+    Fp12 r = Fp12::mul(f1, f2);          // f^(p^6 - 1)
+    r = Fp12::mul(Fp12::frobenius2(r), r);  // ^(p^2 + 1)
+    // hard part
+    return Fp12::pow_limbs(r, bn254c::HARD_EXP, bn254c::HARD_EXP_LIMBS);
+}
+
+// full pairing e(P, Q); P affine G1, Q affine G2
+BN_INLINE Fp12 pairing(const Fp& px, const Fp& py, const G2Affine& q) {
+    return final_exponentiation(miller_loop(px, py, q));
+}
+
+// product-of-pairings check: e(P1,Q1) * e(P2,Q2) == 1
+// (one shared final exponentiation — the BLS verification shape)
+BN_INLINE bool pairing_product_is_one(const Fp& p1x, const Fp& p1y, const G2Affine& q1,
+                                      const Fp& p2x, const Fp& p2y, const G2Affine& q2) {
+    Fp12 f = Fp12::mul(miller_loop(p1x, p1y, q1), miller_loop(p2x, p2y, q2));
+    return final_exponentiation(f).is_one();
+}
+
+}  // namespace bn254

@@ -1,0 +1,185 @@
+// pushcdn_core — host C++ module: BLS-over-BN254 (keygen/sign/verify) and
+// low-level self-test hooks used by the Python test-suite to cross-check the
+// field arithmetic against Python bignums.
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "bls/bls.h"
+
+namespace py = pybind11;
+using namespace bn254;
+
+static std::vector<uint8_t> to_vec(const py::bytes& b) {
+    std::string s = b;
+    return std::vector<uint8_t>(s.begin(), s.end());
+}
+
+static py::bytes to_bytes(const uint8_t* p, size_t n) {
+    return py::bytes(reinterpret_cast<const char*>(p), n);
+}
+
+// ---------------------------------------------------------------------------
+// keygen: deterministic from a u64 seed (reference broker.rs --key-seed)
+// sk = SHA256("pushcdn-bls-keygen" || seed_le || ctr) mod r  (rejection: != 0)
+// ---------------------------------------------------------------------------
+static Fr sk_from_seed(uint64_t seed) {
+    uint8_t buf[18 + 8 + 1];
+    memcpy(buf, "pushcdn-bls-keygen", 18);
+    for (int i = 0; i < 8; ++i) buf[18 + i] = (uint8_t)(seed >> (8 * i));
+    for (uint32_t ctr = 0;; ++ctr) {
+        buf[26] = (uint8_t)ctr;
+        uint8_t d[32];
+        sha256(buf, 27, d);
+        U256 v = bls::u256_mod(bls::u256_from_le(d), from_limbs(bn254c::R_MOD));
+        if (!u256_is_zero(v)) return Fr::from_u256(v);
+    }
+}
+
+static py::tuple keygen(uint64_t seed) {
+    Fr sk = sk_from_seed(seed);
+    U256 sk_std = sk.to_u256();
+    G2 pk = G2::scalar_mul(g2_generator(), sk_std);
+    Fp2 px, py_;
+    pk.to_affine(px, py_);
+    bls::VerKey vk{px, py_};
+    uint8_t skb[32], vkb[128];
+    bls::u256_to_le(sk_std, skb);
+    bls::verkey_serialize(vk, vkb);
+    return py::make_tuple(to_bytes(skb, 32), to_bytes(vkb, 128));
+}
+
+static std::vector<uint8_t> namespaced(const std::string& ns, const std::vector<uint8_t>& msg) {
+    std::vector<uint8_t> out(ns.begin(), ns.end());
+    out.insert(out.end(), msg.begin(), msg.end());
+    out.push_back(0);  // spare byte for the hash counter
+    return out;
+}
+
+static py::bytes sign(const py::bytes& sk_bytes, const std::string& ns, const py::bytes& message) {
+    auto skv = to_vec(sk_bytes);
+    if (skv.size() != 32) throw std::invalid_argument("sk must be 32 bytes");
+    U256 sk_std = bls::u256_from_le(skv.data());
+    if (u256_gte(sk_std, from_limbs(bn254c::R_MOD)))
+        throw std::invalid_argument("sk out of range");
+    auto msg = to_vec(message);
+    auto scratch = namespaced(ns, msg);
+    Fp hx, hy;
+    if (!bls::hash_to_g1_with_scratch(scratch.data(), (uint32_t)scratch.size() - 1, hx, hy))
+        throw std::runtime_error("hash_to_g1 failed");
+    G1 h{hx, hy, Fp::one()};
+    G1 sig = G1::scalar_mul(h, sk_std);
+    Fp sx, sy;
+    sig.to_affine(sx, sy);
+    uint8_t out[64];
+    bls::sig_serialize(sx, sy, out);
+    return to_bytes(out, 64);
+}
+
+static bool verify(const py::bytes& vk_bytes, const std::string& ns, const py::bytes& message,
+                   const py::bytes& sig_bytes) {
+    auto vkv = to_vec(vk_bytes);
+    auto sigv = to_vec(sig_bytes);
+    if (vkv.size() != 128 || sigv.size() != 64) return false;
+    bls::VerKey vk;
+    if (!bls::verkey_deserialize(vkv.data(), vk)) return false;
+    Fp sx, sy;
+    if (!bls::sig_deserialize(sigv.data(), sx, sy)) return false;
+    auto msg = to_vec(message);
+    auto scratch = namespaced(ns, msg);
+    return bls::verify_core(vk, scratch.data(), (uint32_t)scratch.size() - 1, sx, sy);
+}
+
+// ---------------------------------------------------------------------------
+// self-test hooks (Python cross-checks these against bignum arithmetic)
+// ---------------------------------------------------------------------------
+static py::bytes fp_mul_test(const py::bytes& a, const py::bytes& b) {
+    auto av = to_vec(a), bv = to_vec(b);
+    Fp fa = Fp::from_u256(bls::u256_from_le(av.data()));
+    Fp fb = Fp::from_u256(bls::u256_from_le(bv.data()));
+    uint8_t out[32];
+    bls::u256_to_le(Fp::mul(fa, fb).to_u256(), out);
+    return to_bytes(out, 32);
+}
+
+static py::bytes fp_inv_test(const py::bytes& a) {
+    auto av = to_vec(a);
+    Fp fa = Fp::from_u256(bls::u256_from_le(av.data()));
+    uint8_t out[32];
+    bls::u256_to_le(fa.inv().to_u256(), out);
+    return to_bytes(out, 32);
+}
+
+static py::bytes g1_mul_test(uint64_t k) {
+    G1 p = G1::scalar_mul(g1_generator(), U256{{k, 0, 0, 0}});
+    Fp x, y;
+    p.to_affine(x, y);
+    uint8_t out[64];
+    bls::sig_serialize(x, y, out);
+    return to_bytes(out, 64);
+}
+
+static bool pairing_bilinearity_test(uint64_t a, uint64_t b) {
+    // e(aP, bQ) == e(abP, Q) and != 1
+    G1 P = g1_generator();
+    G2 Q = g2_generator();
+    G1 aP = G1::scalar_mul(P, U256{{a, 0, 0, 0}});
+    G2 bQ = G2::scalar_mul(Q, U256{{b, 0, 0, 0}});
+    unsigned __int128 ab128 = (unsigned __int128)a * b;
+    G1 abP = G1::scalar_mul(P, U256{{(u64)ab128, (u64)(ab128 >> 64), 0, 0}});
+    Fp ax, ay, abx, aby;
+    aP.to_affine(ax, ay);
+    abP.to_affine(abx, aby);
+    Fp2 qx, qy, bqx, bqy;
+    Q.to_affine(qx, qy);
+    bQ.to_affine(bqx, bqy);
+    Fp12 e1 = pairing(ax, ay, G2Affine{bqx, bqy});
+    Fp12 e2 = pairing(abx, aby, G2Affine{qx, qy});
+    if (e1.is_one()) return false;  // degenerate
+    return e1 == e2;
+}
+
+static bool subgroup_test() {
+    // r * G1 == infinity and r * G2 == infinity
+    U256 r = from_limbs(bn254c::R_MOD);
+    return G1::scalar_mul(g1_generator(), r).is_infinity() &&
+           G2::scalar_mul(g2_generator(), r).is_infinity();
+}
+
+static py::bytes sha256_test(const py::bytes& data) {
+    auto v = to_vec(data);
+    uint8_t d[32];
+    sha256(v.data(), (uint32_t)v.size(), d);
+    return to_bytes(d, 32);
+}
+
+static py::bytes hash_to_g1_test(const std::string& ns, const py::bytes& message) {
+    auto msg = to_vec(message);
+    auto scratch = namespaced(ns, msg);
+    Fp hx, hy;
+    if (!bls::hash_to_g1_with_scratch(scratch.data(), (uint32_t)scratch.size() - 1, hx, hy))
+        throw std::runtime_error("hash failed");
+    uint8_t out[64];
+    bls::sig_serialize(hx, hy, out);
+    return to_bytes(out, 64);
+}
+
+PYBIND11_MODULE(pushcdn_core, m) {
+    m.doc() = "pushcdn host core: BLS-over-BN254";
+    m.def("keygen", &keygen, "deterministic BLS keypair from a u64 seed -> (sk, vk)");
+    m.def("sign", &sign, "sign(sk, namespace, message) -> 64B signature");
+    m.def("verify", &verify, "verify(vk, namespace, message, sig) -> bool");
+    // self-test hooks
+    m.def("_fp_mul", &fp_mul_test);
+    m.def("_fp_inv", &fp_inv_test);
+    m.def("_g1_mul", &g1_mul_test);
+    m.def("_pairing_bilinear", &pairing_bilinearity_test);
+    m.def("_subgroup_ok", &subgroup_test);
+    m.def("_sha256", &sha256_test);
+    m.def("_hash_to_g1", &hash_to_g1_test);
+}

@@ -66,3 +66,37 @@ def load_gpu_prebuilt():
     spec.loader.exec_module(mod)
     sys.modules["pushcdn_gpu"] = mod
     return mod
+
+
+CORE_SOURCES = [CSRC / "core_bindings.cpp"]
+
+
+def build_core(verbose: bool = False):
+    """Compile (if needed) and load the host C++ core (pybind11, no torch)."""
+    import importlib.util
+    import pybind11
+    import subprocess
+    import sysconfig
+
+    BUILD_DIR.mkdir(exist_ok=True)
+    so = BUILD_DIR / "pushcdn_core.so"
+    srcs = [str(s) for s in CORE_SOURCES]
+    deps = list((CSRC / "common").glob("*.h")) + list((CSRC / "bls").glob("*.h")) + CORE_SOURCES
+    newest_dep = max(p.stat().st_mtime for p in deps)
+    if not so.exists() or so.stat().st_mtime < newest_dep:
+        cmd = [
+            "g++", "-O2", "-std=c++17", "-shared", "-fPIC",
+            f"-I{pybind11.get_include()}",
+            f"-I{sysconfig.get_paths()['include']}",
+            f"-I{CSRC}",
+            *srcs,
+            "-o", str(so),
+        ]
+        if verbose:
+            print(" ".join(cmd))
+        subprocess.run(cmd, check=True)
+    spec = importlib.util.spec_from_file_location("pushcdn_core", so)
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    sys.modules["pushcdn_core"] = mod
+    return mod
