@@ -65,8 +65,8 @@ BN_INLINE bool fp_sqrt(const Fp& a, Fp& out) {
 // hash (already namespaced) message bytes to a G1 point.
 // msg buffer must have one spare byte at msg[len] for the counter (the
 // callers build "namespace || message || ctr" in a scratch buffer).
-BN_INLINE bool hash_to_g1_with_scratch(uint8_t* scratch, uint32_t len, Fp& outx, Fp& outy) {
-    for (uint32_t ctr = 0; ctr < 255; ++ctr) {
+BN_BIGFUNC bool hash_to_g1_with_scratch(uint8_t* scratch, uint32_t len, Fp& outx, Fp& outy) {
+    BN_NOUNROLL for (uint32_t ctr = 0; ctr < 255; ++ctr) {
         scratch[len] = (uint8_t)ctr;
         uint8_t d[32];
         sha256(scratch, len + 1, d);
@@ -127,7 +127,7 @@ BN_INLINE bool sig_deserialize(const uint8_t in[64], Fp& x, Fp& y) {
 
 // Core verification given parsed inputs. msg scratch = namespace||message
 // with a spare byte (see hash_to_g1_with_scratch).
-BN_INLINE bool verify_core(const VerKey& vk, uint8_t* scratch, uint32_t msg_len,
+BN_BIGFUNC bool verify_core(const VerKey& vk, uint8_t* scratch, uint32_t msg_len,
                            const Fp& sig_x, const Fp& sig_y) {
     Fp hx, hy;
     if (!hash_to_g1_with_scratch(scratch, msg_len, hx, hy)) return false;

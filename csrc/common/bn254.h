@@ -16,9 +16,15 @@
 #include "bn254_constants.h"
 
 #if defined(__HIPCC__) || defined(__HIP_DEVICE_COMPILE__)
-#define BN_INLINE __host__ __device__ __forceinline__
+// plain inline (not forceinline): a fully-inlined pairing explodes compile
+// time; BN_BIGFUNC marks the large building blocks noinline outright.
+#define BN_INLINE __host__ __device__ inline
+#define BN_BIGFUNC __host__ __device__ __attribute__((noinline))
+#define BN_NOUNROLL _Pragma("clang loop unroll(disable)")
 #else
 #define BN_INLINE inline
+#define BN_BIGFUNC inline __attribute__((noinline))
+#define BN_NOUNROLL
 #endif
 
 namespace bn254 {
@@ -191,11 +197,11 @@ struct Fe {
     BN_INLINE static Fe from_u64(u64 x) { return from_u256(U256{{x, 0, 0, 0}}); }
 
     // exponentiation by a standard-form 4-limb exponent (MSB scan)
-    BN_INLINE static Fe pow(const Fe& a, const U256& e) {
+    BN_BIGFUNC static Fe pow(const Fe& a, const U256& e) {
         Fe result = one();
         bool started = false;
-        for (int i = 3; i >= 0; --i) {
-            for (int b = 63; b >= 0; --b) {
+        BN_NOUNROLL for (int i = 3; i >= 0; --i) {
+            BN_NOUNROLL for (int b = 63; b >= 0; --b) {
                 if (started) result = sqr(result);
                 if ((e.v[i] >> b) & 1) {
                     if (started) result = mul(result, a);
@@ -333,11 +339,11 @@ struct Point {
     BN_INLINE static Point neg(const Point& p) { return {p.X, F::neg(p.Y), p.Z}; }
 
     // scalar multiplication by a standard-form scalar (not Montgomery)
-    BN_INLINE static Point scalar_mul(const Point& p, const U256& k) {
+    BN_BIGFUNC static Point scalar_mul(const Point& p, const U256& k) {
         Point result = infinity();
         bool started = false;
-        for (int i = 3; i >= 0; --i) {
-            for (int b = 63; b >= 0; --b) {
+        BN_NOUNROLL for (int i = 3; i >= 0; --i) {
+            BN_NOUNROLL for (int b = 63; b >= 0; --b) {
                 if (started) result = dbl(result);
                 if ((k.v[i] >> b) & 1) {
                     if (started) result = add(result, p);

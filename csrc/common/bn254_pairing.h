@@ -37,7 +37,7 @@ struct Fp6 {
         return {Fp2::neg(a.c0), Fp2::neg(a.c1), Fp2::neg(a.c2)};
     }
 
-    BN_INLINE static Fp6 mul(const Fp6& a, const Fp6& b) {
+    BN_BIGFUNC static Fp6 mul(const Fp6& a, const Fp6& b) {
         // Toom/Karatsuba (devegili): v0=a0b0, v1=a1b1, v2=a2b2
         Fp2 v0 = Fp2::mul(a.c0, b.c0);
         Fp2 v1 = Fp2::mul(a.c1, b.c1);
@@ -65,7 +65,7 @@ struct Fp6 {
         return {Fp2::mul_xi(a.c2), a.c0, a.c1};
     }
 
-    BN_INLINE Fp6 inv() const {
+    BN_BIGFUNC Fp6 inv() const {
         // standard: A = c0^2 - xi c1 c2, B = xi c2^2 - c0 c1, C = c1^2 - c0 c2
         Fp2 A = Fp2::sub(Fp2::sqr(c0), Fp2::mul_xi(Fp2::mul(c1, c2)));
         Fp2 B = Fp2::sub(Fp2::mul_xi(Fp2::sqr(c2)), Fp2::mul(c0, c1));
@@ -87,14 +87,14 @@ struct Fp12 {
     BN_INLINE bool operator==(const Fp12& o) const { return c0 == o.c0 && c1 == o.c1; }
     BN_INLINE bool is_one() const { return *this == one(); }
 
-    BN_INLINE static Fp12 mul(const Fp12& a, const Fp12& b) {
+    BN_BIGFUNC static Fp12 mul(const Fp12& a, const Fp12& b) {
         Fp6 v0 = Fp6::mul(a.c0, b.c0);
         Fp6 v1 = Fp6::mul(a.c1, b.c1);
         Fp6 t = Fp6::mul(Fp6::add(a.c0, a.c1), Fp6::add(b.c0, b.c1));
         return {Fp6::add(v0, Fp6::mul_v(v1)), Fp6::sub(Fp6::sub(t, v0), v1)};
     }
 
-    BN_INLINE static Fp12 sqr(const Fp12& a) {
+    BN_BIGFUNC static Fp12 sqr(const Fp12& a) {
         // complex squaring: (c0 + c1 w)^2 = (c0^2 + v c1^2) + 2 c0 c1 w
         Fp6 v0 = Fp6::mul(a.c0, a.c1);
         Fp6 t = Fp6::mul(Fp6::add(a.c0, a.c1), Fp6::add(a.c0, Fp6::mul_v(a.c1)));
@@ -104,7 +104,7 @@ struct Fp12 {
 
     BN_INLINE static Fp12 conj(const Fp12& a) { return {a.c0, Fp6::neg(a.c1)}; }
 
-    BN_INLINE Fp12 inv() const {
+    BN_BIGFUNC Fp12 inv() const {
         // 1/(c0 + c1 w) = (c0 - c1 w) / (c0^2 - v c1^2)
         Fp6 d = Fp6::sub(Fp6::sqr(c0), Fp6::mul_v(Fp6::sqr(c1)));
         Fp6 di = d.inv();
@@ -120,7 +120,7 @@ struct Fp12 {
     }
 
     // Frobenius^2: c_ij -> c_ij * gamma2 factors (Fp scalars, no conjugation)
-    BN_INLINE static Fp12 frobenius2(const Fp12& a) {
+    BN_BIGFUNC static Fp12 frobenius2(const Fp12& a) {
         Fp g1 = Fp::from_u256(from_limbs(bn254c::GAMMA2_1));
         Fp g2 = Fp::from_u256(from_limbs(bn254c::GAMMA2_2));
         Fp g3 = Fp::from_u256(from_limbs(bn254c::GAMMA2_3));
@@ -132,11 +132,11 @@ struct Fp12 {
     }
 
     // generic pow by a little-endian multi-limb exponent (standard form)
-    BN_INLINE static Fp12 pow_limbs(const Fp12& a, const uint64_t* limbs, int n) {
+    BN_BIGFUNC static Fp12 pow_limbs(const Fp12& a, const uint64_t* limbs, int n) {
         Fp12 result = one();
         bool started = false;
-        for (int i = n - 1; i >= 0; --i) {
-            for (int b = 63; b >= 0; --b) {
+        BN_NOUNROLL for (int i = n - 1; i >= 0; --i) {
+            BN_NOUNROLL for (int b = 63; b >= 0; --b) {
                 if (started) result = sqr(result);
                 if ((limbs[i] >> b) & 1) {
                     if (started) result = mul(result, a);
@@ -164,7 +164,7 @@ struct LineCoeffs {
 };
 
 // doubling step (arkworks models/bn/g2.rs shape, D-twist coefficients)
-BN_INLINE LineCoeffs doubling_step(G2Proj& r, const Fp& two_inv) {
+BN_BIGFUNC LineCoeffs doubling_step(G2Proj& r, const Fp& two_inv) {
     Fp2 a = Fp2::mul_fp(Fp2::mul(r.x, r.y), two_inv);
     Fp2 b = Fp2::sqr(r.y);
     Fp2 c = Fp2::sqr(r.z);
@@ -184,7 +184,7 @@ BN_INLINE LineCoeffs doubling_step(G2Proj& r, const Fp& two_inv) {
 }
 
 // mixed addition step
-BN_INLINE LineCoeffs addition_step(G2Proj& r, const G2Affine& q) {
+BN_BIGFUNC LineCoeffs addition_step(G2Proj& r, const G2Affine& q) {
     Fp2 theta = Fp2::sub(r.y, Fp2::mul(q.y, r.z));
     Fp2 lambda = Fp2::sub(r.x, Fp2::mul(q.x, r.z));
     Fp2 c = Fp2::sqr(theta);
@@ -218,12 +218,12 @@ BN_INLINE G2Affine g2_frobenius(const G2Affine& q) {
 
 // Miller loop for one pair (P in G1 affine, Q in G2 affine). Both must be
 // non-infinity (callers handle degenerate cases).
-BN_INLINE Fp12 miller_loop(const Fp& px, const Fp& py, const G2Affine& q) {
+BN_BIGFUNC Fp12 miller_loop(const Fp& px, const Fp& py, const G2Affine& q) {
     Fp two_inv = Fp::from_u64(2).inv();
     G2Proj r{q.x, q.y, Fp2::one()};
     G2Affine negq{q.x, Fp2::neg(q.y)};
     Fp12 f = Fp12::one();
-    for (int i = bn254c::ATE_NAF_LEN - 2; i >= 0; --i) {
+    BN_NOUNROLL for (int i = bn254c::ATE_NAF_LEN - 2; i >= 0; --i) {
         f = Fp12::sqr(f);
         LineCoeffs l = doubling_step(r, two_inv);
         ell(f, l, px, py);
@@ -251,7 +251,7 @@ BN_INLINE Fp12 miller_loop(const Fp& px, const Fp& py, const G2Affine& q) {
 // Final exponentiation: f^((p^12-1)/r)
 // easy part structured; hard part = generic pow by (p^4-p^2+1)/r
 // ---------------------------------------------------------------------------
-BN_INLINE Fp12 final_exponentiation(const Fp12& f) {
+BN_BIGFUNC Fp12 final_exponentiation(const Fp12& f) {
     // easy part: f^(p^6 - 1) = conj(f) * f^-1 ; then ^(p^2 + 1)
     Fp12 f1 = Fp12::conj(f);
     Fp12 f2 = f.inv();

@@ -33,6 +33,9 @@ void launch_k5_direct_lookup(const uint64_t*, const int32_t*, int64_t, const uin
                              int32_t*, hipStream_t);
 void launch_k2c_apply_subs(uint64_t*, const uint8_t*, const int64_t*, const int32_t*,
                            const int32_t*, const int32_t*, int32_t, int32_t, hipStream_t);
+void launch_k1_bls_verify(const uint8_t*, const uint8_t*, uint8_t*, const int64_t*, int32_t,
+                          int32_t*, hipStream_t);
+void launch_k1_hash_to_g1(uint8_t*, const int64_t*, int32_t, uint8_t*, hipStream_t);
 }
 
 #define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
@@ -155,6 +158,33 @@ void apply_subs(torch::Tensor sub_bitmap, torch::Tensor buf, torch::Tensor topic
                           cur_stream());
 }
 
+torch::Tensor bls_verify_batch(torch::Tensor vks, torch::Tensor sigs, torch::Tensor msgs,
+                               torch::Tensor moff) {
+    CHECK_DEV(vks); CHECK_DEV(sigs); CHECK_DEV(msgs); CHECK_DEV(moff);
+    CHECK_CONTIG(vks); CHECK_CONTIG(sigs); CHECK_CONTIG(msgs); CHECK_CONTIG(moff);
+    int32_t N = (int32_t)moff.size(0) - 1;
+    TORCH_CHECK(vks.numel() == (int64_t)N * 128 && sigs.numel() == (int64_t)N * 64);
+    auto ok = torch::zeros({N}, torch::TensorOptions().dtype(torch::kInt32).device(vks.device()));
+    if (N > 0) {
+        launch_k1_bls_verify(vks.data_ptr<uint8_t>(), sigs.data_ptr<uint8_t>(),
+                             msgs.data_ptr<uint8_t>(), moff.data_ptr<int64_t>(), N,
+                             ok.data_ptr<int32_t>(), cur_stream());
+    }
+    return ok;
+}
+
+torch::Tensor hash_to_g1_batch(torch::Tensor msgs, torch::Tensor moff) {
+    CHECK_DEV(msgs); CHECK_DEV(moff);
+    int32_t N = (int32_t)moff.size(0) - 1;
+    auto out = torch::zeros({N, 64},
+                            torch::TensorOptions().dtype(torch::kUInt8).device(msgs.device()));
+    if (N > 0) {
+        launch_k1_hash_to_g1(msgs.data_ptr<uint8_t>(), moff.data_ptr<int64_t>(), N,
+                             out.data_ptr<uint8_t>(), cur_stream());
+    }
+    return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("parse_batch", &parse_batch, "K4: on-device capnp parse of a message batch");
     m.def("topic_mask", &topic_mask, "K2a: per-message subscriber mask");
@@ -162,4 +192,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fanout", &fanout, "K3: N-way payload fan-out into egress rings");
     m.def("direct_lookup", &direct_lookup, "K5: batched direct-route hash probe");
     m.def("apply_subs", &apply_subs, "K2c: apply subscribe/unsubscribe batch to bitmap");
+    m.def("bls_verify_batch", &bls_verify_batch, "K1: batched BLS-over-BN254 verification");
+    m.def("hash_to_g1_batch", &hash_to_g1_batch, "K1 helper: batched hash-to-G1");
 }

@@ -52,7 +52,12 @@ class GpuBrokerEngine:
         ring_bytes: int = 1 << 21,
         direct_table_size: int = 1 << 16,
         use_gpu_ops: Optional[bool] = None,
+        fanout_wire: bool = False,
     ) -> None:
+        # fanout_wire=True: fan out the WHOLE serialized wire message (the
+        # reference's raw-bytes-forwarded-verbatim invariant);
+        # False: fan out only the payload field (kernel golden tests).
+        self.fanout_wire = fanout_wire
         self.device = torch.device(device)
         self.is_cuda = self.device.type == "cuda"
         if use_gpu_ops is None:
@@ -144,6 +149,9 @@ class GpuBrokerEngine:
             buf, offsets
         )
         mask = ops.topic_mask(self.sub_bitmap, buf, topics_off, topics_cnt, disc)
+        if self.fanout_wire:
+            payload_off = offsets[:-1].contiguous()
+            payload_len = (offsets[1:] - offsets[:-1]).to(torch.int32).contiguous()
         pair_user, pair_msg, pair_dst, drops = ops.assign_emit(
             mask, payload_off, payload_len, self.ring_wpos, self.ring_bytes, self.n_users
         )
@@ -204,6 +212,12 @@ class GpuBrokerEngine:
         pr = ref.parse_batch(batch, offsets)
         M = len(offsets) - 1
         mask = ref.topic_mask(self.sub_bitmap, batch, pr.topics_off, pr.topics_cnt, pr.disc)
+        if self.fanout_wire:
+            pr.payload_off = torch.tensor(offsets[:-1], dtype=torch.int64)
+            pr.payload_len = (
+                torch.tensor(offsets[1:], dtype=torch.int64)
+                - torch.tensor(offsets[:-1], dtype=torch.int64)
+            ).to(torch.int32)
         pair_user, pair_msg, pair_dst, drops = ref.assign_emit(
             mask, pr.payload_len, self.ring_wpos, self.ring_bytes, self.n_users
         )

@@ -1,0 +1,564 @@
+"""The broker service (reference ``cdn-broker/src/lib.rs`` + ``tasks/``).
+
+``Broker.start()`` runs the same five long-lived tasks as the reference
+(lib.rs:269-319) — heartbeat, sync, whitelist, user listener, broker
+listener (+ optional metrics server) — and fails fast if any dies.
+
+Routing runs in one of two data planes:
+  - host: per-message routing against the Connections tables (the reference's
+    path, used for control-plane scale and CPU tests)
+  - gpu: incoming user messages are batched per tick through the CDNA4
+    kernel pipeline (GpuBrokerEngine) and egress rings are drained back to
+    the per-user connections — the MI355X-native hot path.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import random
+import socket
+from dataclasses import dataclass, field
+from typing import Callable, Dict, List, Optional, Sequence, Tuple
+
+from ..auth import BrokerAuth
+from ..auth.broker import BrokerAuth as _BA
+from ..crypto import bls
+from ..discovery import BrokerIdentifier, new_discovery_client
+from ..proto import message as m
+from ..proto.errors import ConnectionError_, TopicError
+from ..proto.limiter import Bytes, Limiter
+from ..proto.topic import TopicSpace, ALL_TOPICS
+from ..proto.transports.base import Connection, Protocol
+from ..utils.mnemonic import mnemonic
+from .connections import Connections
+
+HEARTBEAT_INTERVAL_S = 10.0
+HEARTBEAT_EXPIRY_S = 60.0
+SYNC_INTERVAL_S = 10.0
+WHITELIST_INTERVAL_S = 60.0
+
+# ProcessMessage / SkipMessage hook results (reference def.rs:69-97)
+PROCESS_MESSAGE = "process"
+SKIP_MESSAGE = "skip"
+
+
+def resolve_local_ip(endpoint: str) -> str:
+    """'local_ip' substitution in advertise endpoints (reference lib.rs:157-168)."""
+    if not endpoint.startswith("local_ip"):
+        return endpoint
+    try:
+        s = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+        s.connect(("8.8.8.8", 80))
+        ip = s.getsockname()[0]
+        s.close()
+    except OSError:
+        ip = "127.0.0.1"
+    return endpoint.replace("local_ip", ip)
+
+
+@dataclass
+class BrokerConfig:
+    public_bind_endpoint: str
+    public_advertise_endpoint: str
+    private_bind_endpoint: str
+    private_advertise_endpoint: str
+    discovery_endpoint: str = ""
+    keypair: Optional[bls.KeyPair] = None
+    metrics_bind_endpoint: Optional[str] = None
+    global_memory_pool_size: Optional[int] = 1 << 30
+    user_message_hook: Optional[Callable] = None
+    broker_message_hook: Optional[Callable] = None
+    topic_space: TopicSpace = field(default_factory=lambda: ALL_TOPICS)
+    user_protocol: Optional[type] = None    # Protocol class for users
+    broker_protocol: Optional[type] = None  # Protocol class for brokers
+    ca_cert_path: Optional[str] = None
+    ca_key_path: Optional[str] = None
+    # strong-consistency feature (reference cargo feature, on by default for
+    # brokers): push partial syncs immediately on user connect
+    strong_consistency: bool = True
+    # task intervals (reference: 10 s heartbeat/sync, 60 s whitelist);
+    # configurable so tests can run fast
+    heartbeat_interval_s: float = HEARTBEAT_INTERVAL_S
+    sync_interval_s: float = SYNC_INTERVAL_S
+    whitelist_interval_s: float = WHITELIST_INTERVAL_S
+    # data plane: "host" or "gpu"
+    data_plane: str = "host"
+    gpu_device: str = "cuda:0"
+    gpu_tick_interval_s: float = 0.002
+    gpu_max_users: int = 16384
+    gpu_ring_bytes: int = 1 << 21
+
+
+@dataclass
+class UserHandle:
+    connection: Connection
+    task: Optional[asyncio.Task] = None
+    gpu_index: Optional[int] = None
+
+
+@dataclass
+class BrokerHandle:
+    connection: Connection
+    task: Optional[asyncio.Task] = None
+
+
+class Broker:
+    def __init__(self, config: BrokerConfig) -> None:
+        from ..proto.transports.tcp import Tcp
+
+        self.config = config
+        config.public_advertise_endpoint = resolve_local_ip(config.public_advertise_endpoint)
+        config.private_advertise_endpoint = resolve_local_ip(config.private_advertise_endpoint)
+        self.identity = BrokerIdentifier(
+            config.public_advertise_endpoint, config.private_advertise_endpoint
+        )
+        self.keypair = config.keypair or bls.KeyPair.from_seed(0)
+        self.discovery = new_discovery_client(config.discovery_endpoint, self.identity)
+        self.limiter = Limiter(config.global_memory_pool_size)
+        self.connections = Connections(self.identity)
+        self.user_protocol = config.user_protocol or Tcp
+        self.broker_protocol = config.broker_protocol or Tcp
+        self._tasks: List[asyncio.Task] = []
+        self._listeners = []
+        self._closed = False
+        self._engine = None
+        self._gpu_queue: Optional[asyncio.Queue] = None
+        self._free_gpu_slots: List[int] = []
+        if config.data_plane == "gpu":
+            from .gpu_engine import GpuBrokerEngine
+
+            self._engine = GpuBrokerEngine(
+                device=config.gpu_device,
+                n_users=config.gpu_max_users,
+                ring_bytes=config.gpu_ring_bytes,
+                fanout_wire=True,  # forward raw wire bytes verbatim
+            )
+            self._gpu_queue = asyncio.Queue()
+            self._free_gpu_slots = list(range(config.gpu_max_users - 1, -1, -1))
+            self._gpu_user_by_slot: Dict[int, bytes] = {}
+
+    # ------------------------------ lifecycle ------------------------------
+
+    async def start(self) -> None:
+        self._user_listener = await self.user_protocol.bind(
+            self.config.public_bind_endpoint, None, None
+        )
+        self._broker_listener = await self.broker_protocol.bind(
+            self.config.private_bind_endpoint, None, None
+        )
+        self._listeners = [self._user_listener, self._broker_listener]
+        loop = asyncio.get_running_loop()
+        self._tasks = [
+            loop.create_task(self._heartbeat_task(), name="heartbeat"),
+            loop.create_task(self._sync_task(), name="sync"),
+            loop.create_task(self._whitelist_task(), name="whitelist"),
+            loop.create_task(self._user_listener_task(), name="user-listener"),
+            loop.create_task(self._broker_listener_task(), name="broker-listener"),
+        ]
+        if self._engine is not None:
+            self._tasks.append(loop.create_task(self._gpu_tick_task(), name="gpu-tick"))
+        if self.config.metrics_bind_endpoint:
+            from ..utils.metrics import serve_metrics
+            from ..proto.transports.tcp import parse_endpoint
+
+            host, port = parse_endpoint(self.config.metrics_bind_endpoint)
+            self._metrics_server = await serve_metrics(host, port)
+
+    async def run_forever(self) -> None:
+        """Crash the service if any task dies (reference lib.rs:302-318)."""
+        await self.start()
+        done, _pending = await asyncio.wait(self._tasks, return_when=asyncio.FIRST_COMPLETED)
+        for t in done:
+            exc = t.exception()
+            if exc:
+                raise exc
+        raise RuntimeError(f"broker task {next(iter(done)).get_name()} exited")
+
+    async def close(self) -> None:
+        self._closed = True
+        for t in self._tasks:
+            t.cancel()
+        for handle in list(self.connections.users.values()):
+            handle.connection.close()
+            if handle.task:
+                handle.task.cancel()
+        for handle in list(self.connections.brokers.values()):
+            handle.connection.close()
+            if handle.task:
+                handle.task.cancel()
+        for l in self._listeners:
+            await l.close()
+
+    # ------------------------------ senders ------------------------------
+
+    async def try_send_to_user(self, pubkey: bytes, raw: Bytes) -> None:
+        """Send; evict the user on failure (reference user/sender.rs:16-33)."""
+        handle = self.connections.users.get(pubkey)
+        if handle is None:
+            raw.drop()
+            return
+        try:
+            await handle.connection.send_message_raw(raw)
+        except Exception:
+            await self.remove_user(pubkey)
+
+    async def try_send_to_broker(self, broker: BrokerIdentifier, raw: Bytes) -> None:
+        handle = self.connections.brokers.get(broker)
+        if handle is None:
+            raw.drop()
+            return
+        try:
+            await handle.connection.send_message_raw(raw)
+        except Exception:
+            await self.remove_broker(broker)
+
+    async def try_send_to_brokers(self, raw: Bytes) -> None:
+        """Fan-out to every connected broker (reference broker/sender.rs:49-58)."""
+        for broker in self.connections.all_brokers():
+            await self.try_send_to_broker(broker, raw.clone())
+        raw.drop()
+
+    async def remove_user(self, pubkey: bytes) -> None:
+        handle = self.connections.remove_user(pubkey)
+        if handle is not None:
+            if handle.task:
+                handle.task.cancel()
+            handle.connection.close()
+            if self._engine is not None and handle.gpu_index is not None:
+                self._release_gpu_slot(handle.gpu_index)
+
+    async def remove_broker(self, broker: BrokerIdentifier) -> None:
+        handle = self.connections.remove_broker(broker)
+        if handle is not None:
+            if handle.task:
+                handle.task.cancel()
+            handle.connection.close()
+
+    # ------------------------------ user path ------------------------------
+
+    async def _user_listener_task(self) -> None:
+        while not self._closed:
+            unfinalized = await self._user_listener.accept()
+            asyncio.get_running_loop().create_task(self._handle_user_connection(unfinalized))
+
+    async def _handle_user_connection(self, unfinalized) -> None:
+        """accept -> finalize -> permit auth -> add + receive loop
+        (reference user/handler.rs:26-91)."""
+        try:
+            connection = await asyncio.wait_for(unfinalized.finalize(self.limiter), 5)
+        except Exception:
+            return
+        result = await BrokerAuth.verify_user(connection, self.identity, self.discovery)
+        if result is None:
+            connection.close()
+            return
+        pubkey, raw_topics = result
+        try:
+            topics = self.config.topic_space.prune(raw_topics)
+        except TopicError:
+            connection.close()
+            return
+        handle = UserHandle(connection=connection)
+        old = self.connections.add_user(pubkey, handle, topics)
+        if old is not None:
+            # duplicate key kicks the old session (connections/mod.rs:290-298)
+            if old.task:
+                old.task.cancel()
+            old.connection.close()
+            if self._engine is not None and old.gpu_index is not None:
+                self._release_gpu_slot(old.gpu_index)
+        if self._engine is not None:
+            handle.gpu_index = self._claim_gpu_slot(pubkey)
+            self._engine.subscribe(handle.gpu_index, topics)
+            self._engine.register_direct(pubkey, handle.gpu_index)
+        handle.task = asyncio.get_running_loop().create_task(
+            self._user_receive_loop(pubkey, handle)
+        )
+        if self.config.strong_consistency:
+            # immediate partial syncs on connect (user/handler.rs:79-90)
+            await self._send_partial_syncs()
+
+    async def _user_receive_loop(self, pubkey: bytes, handle: UserHandle) -> None:
+        """The per-user hot loop (reference user/handler.rs:95-163)."""
+        connection = handle.connection
+        try:
+            while True:
+                raw = await connection.recv_message_raw()
+                try:
+                    msg = m.deserialize(raw.data)
+                except Exception:
+                    break
+                hook = self.config.user_message_hook
+                if hook is not None:
+                    verdict = hook(msg)
+                    if verdict == SKIP_MESSAGE:
+                        raw.drop()
+                        continue
+                    if verdict not in (PROCESS_MESSAGE, None):
+                        break
+                if isinstance(msg, m.Broadcast):
+                    try:
+                        topics = self.config.topic_space.prune(msg.topics)
+                    except TopicError:
+                        break
+                    if self._engine is not None:
+                        await self._gpu_queue.put((raw, None))
+                    else:
+                        await self.handle_broadcast_message(topics, raw, to_users_only=False)
+                elif isinstance(msg, m.Direct):
+                    if self._engine is not None:
+                        await self._gpu_queue.put((raw, msg.recipient))
+                    else:
+                        await self.handle_direct_message(msg.recipient, raw, to_user_only=False)
+                elif isinstance(msg, m.Subscribe):
+                    try:
+                        topics = self.config.topic_space.prune(msg.topics)
+                    except TopicError:
+                        break
+                    self.connections.subscribe_user(pubkey, topics)
+                    if self._engine is not None and handle.gpu_index is not None:
+                        self._engine.subscribe(handle.gpu_index, topics)
+                    raw.drop()
+                elif isinstance(msg, m.Unsubscribe):
+                    topics = [t & 0xFF for t in msg.topics]
+                    self.connections.unsubscribe_user(pubkey, topics)
+                    if self._engine is not None and handle.gpu_index is not None:
+                        self._engine.unsubscribe(handle.gpu_index, topics)
+                    raw.drop()
+                else:
+                    break  # unexpected message type: disconnect
+        except (ConnectionError_, asyncio.CancelledError):
+            pass
+        finally:
+            # only remove if WE are still the registered session — a
+            # duplicate-key connect may have already replaced this handle
+            # (reference connections/mod.rs:290-298 kick semantics)
+            if self.connections.users.get(pubkey) is handle:
+                await self.remove_user(pubkey)
+            else:
+                connection.close()
+
+    # ------------------------------ routing ------------------------------
+
+    async def handle_broadcast_message(
+        self, topics: Sequence[int], raw: Bytes, to_users_only: bool
+    ) -> None:
+        """Forward raw bytes verbatim to interested brokers + users
+        (reference broker/handler.rs:240-272)."""
+        users, brokers = self.connections.get_interested_by_topic(topics, to_users_only)
+        for broker in brokers:
+            await self.try_send_to_broker(broker, raw.clone())
+        for user in users:
+            await self.try_send_to_user(user, raw.clone())
+        raw.drop()
+
+    async def handle_direct_message(
+        self, recipient: bytes, raw: Bytes, to_user_only: bool
+    ) -> None:
+        """DirectMap lookup -> local delivery or 1-hop forward
+        (reference broker/handler.rs:197-237)."""
+        owner = self.connections.get_broker_identifier_of_user(recipient)
+        if owner is None:
+            raw.drop()  # unknown user: silently dropped (handler.rs:209-236)
+            return
+        if owner == self.identity:
+            await self.try_send_to_user(recipient, raw)
+        elif not to_user_only:
+            await self.try_send_to_broker(owner, raw)
+        else:
+            raw.drop()
+
+    # ------------------------------ broker path ------------------------------
+
+    async def _broker_listener_task(self) -> None:
+        while not self._closed:
+            unfinalized = await self._broker_listener.accept()
+            asyncio.get_running_loop().create_task(
+                self._handle_broker_connection(unfinalized, is_outbound=False)
+            )
+
+    async def _handle_broker_connection(self, conn_or_unfinalized, is_outbound: bool,
+                                        peer: Optional[BrokerIdentifier] = None) -> None:
+        """Mutual auth (direction-dependent ordering) + initial full syncs +
+        receive loop (reference broker/handler.rs:31-118)."""
+        try:
+            if is_outbound:
+                connection = conn_or_unfinalized
+                peer_identity = await _BA.authenticate_with_broker(connection, self.keypair)
+                ok = await _BA.verify_broker(connection, self.identity, self.keypair)
+                if not ok:
+                    connection.close()
+                    return
+            else:
+                connection = await asyncio.wait_for(conn_or_unfinalized.finalize(self.limiter), 5)
+                ok = await _BA.verify_broker(connection, self.identity, self.keypair)
+                if not ok:
+                    connection.close()
+                    return
+                peer_identity = await _BA.authenticate_with_broker(connection, self.keypair)
+        except Exception:
+            return
+        handle = BrokerHandle(connection=connection)
+        old = self.connections.add_broker(peer_identity, handle)
+        if old is not None:
+            if old.task:
+                old.task.cancel()
+            old.connection.close()
+        # initial full syncs (broker/handler.rs:98-117)
+        await self.try_send_to_broker(
+            peer_identity, Bytes(m.serialize(m.TopicSync(self.connections.get_full_topic_sync())))
+        )
+        await self.try_send_to_broker(
+            peer_identity, Bytes(m.serialize(m.UserSync(self.connections.get_full_user_sync())))
+        )
+        handle.task = asyncio.get_running_loop().create_task(
+            self._broker_receive_loop(peer_identity, connection)
+        )
+
+    async def _broker_receive_loop(
+        self, peer: BrokerIdentifier, connection: Connection
+    ) -> None:
+        """reference broker/handler.rs:121-194."""
+        try:
+            while True:
+                raw = await connection.recv_message_raw()
+                try:
+                    msg = m.deserialize(raw.data)
+                except Exception:
+                    break
+                hook = self.config.broker_message_hook
+                if hook is not None:
+                    verdict = hook(msg)
+                    if verdict == SKIP_MESSAGE:
+                        raw.drop()
+                        continue
+                    if verdict not in (PROCESS_MESSAGE, None):
+                        break
+                if isinstance(msg, m.Broadcast):
+                    try:
+                        topics = self.config.topic_space.prune(msg.topics)
+                    except TopicError:
+                        raw.drop()
+                        continue
+                    # single-hop mesh: deliver only to local users
+                    await self.handle_broadcast_message(topics, raw, to_users_only=True)
+                elif isinstance(msg, m.Direct):
+                    await self.handle_direct_message(msg.recipient, raw, to_user_only=True)
+                elif isinstance(msg, m.UserSync):
+                    to_kick = self.connections.apply_user_sync(msg.data)
+                    for pubkey in to_kick:
+                        await self.remove_user(pubkey)
+                    raw.drop()
+                elif isinstance(msg, m.TopicSync):
+                    self.connections.apply_topic_sync(peer, msg.data)
+                    raw.drop()
+                else:
+                    break
+        except (ConnectionError_, asyncio.CancelledError):
+            pass
+        finally:
+            await self.remove_broker(peer)
+
+    # ------------------------------ tasks ------------------------------
+
+    async def _heartbeat_task(self) -> None:
+        """Every 10 s: publish load, discover peers, dial brokers with
+        identifier >= ours, shuffled (reference heartbeat.rs:28-108)."""
+        while True:
+            try:
+                await self.discovery.perform_heartbeat(
+                    len(self.connections.users), HEARTBEAT_EXPIRY_S
+                )
+                others = await self.discovery.get_other_brokers()
+                to_dial = [
+                    b for b in others
+                    if b not in self.connections.brokers and str(b) >= str(self.identity)
+                ]
+                random.shuffle(to_dial)
+                for peer in to_dial:
+                    asyncio.get_running_loop().create_task(self._dial_broker(peer))
+            except Exception:
+                pass
+            await asyncio.sleep(self.config.heartbeat_interval_s)
+
+    async def _dial_broker(self, peer: BrokerIdentifier) -> None:
+        try:
+            connection = await self.broker_protocol.connect(
+                peer.private_advertise_endpoint, True, self.limiter
+            )
+        except Exception:
+            return
+        await self._handle_broker_connection(connection, is_outbound=True, peer=peer)
+
+    async def _send_partial_syncs(self) -> None:
+        user_delta = self.connections.get_partial_user_sync()
+        topic_delta = self.connections.get_partial_topic_sync()
+        if user_delta:
+            await self.try_send_to_brokers(Bytes(m.serialize(m.UserSync(user_delta))))
+        if topic_delta:
+            await self.try_send_to_brokers(Bytes(m.serialize(m.TopicSync(topic_delta))))
+
+    async def _sync_task(self) -> None:
+        """Every 10 s: ship CRDT diffs to every peer (reference sync.rs:129-144)."""
+        while True:
+            await asyncio.sleep(self.config.sync_interval_s)
+            try:
+                await self._send_partial_syncs()
+            except Exception:
+                pass
+
+    async def _whitelist_task(self) -> None:
+        """Every 60 s: kick users no longer whitelisted
+        (reference whitelist.rs:19-46)."""
+        while True:
+            await asyncio.sleep(self.config.whitelist_interval_s)
+            try:
+                for pubkey in self.connections.all_users():
+                    if not await self.discovery.check_whitelist(pubkey):
+                        await self.remove_user(pubkey)
+            except Exception:
+                pass
+
+    # ------------------------------ GPU data plane ------------------------------
+
+    def _claim_gpu_slot(self, pubkey: bytes) -> int:
+        if not self._free_gpu_slots:
+            raise RuntimeError("gpu broker user capacity exceeded")
+        slot = self._free_gpu_slots.pop()
+        self._gpu_user_by_slot[slot] = pubkey
+        return slot
+
+    def _release_gpu_slot(self, slot: int) -> None:
+        self._gpu_user_by_slot.pop(slot, None)
+        # clear all subscriptions for the slot
+        self._engine.unsubscribe(slot, list(range(256)))
+        self._free_gpu_slots.append(slot)
+
+    async def _gpu_tick_task(self) -> None:
+        """Batch queued user messages through the kernel pipeline each tick,
+        then drain egress rings back to the user connections."""
+        from .gpu_engine import parse_ring_records
+
+        while True:
+            item = await self._gpu_queue.get()
+            batch: List[Bytes] = [item[0]]
+            while not self._gpu_queue.empty() and len(batch) < 4096:
+                batch.append(self._gpu_queue.get_nowait()[0])
+            buf = bytearray()
+            offsets = [0]
+            for raw in batch:
+                buf += raw.data
+                offsets.append(len(buf))
+            dbuf, doff = self._engine.ingest(bytes(buf), offsets)
+            self._engine.tick(dbuf, doff, host_batch=bytes(buf), host_offsets=offsets)
+            wpos = self._engine.drain_cursors()
+            for slot, pubkey in list(self._gpu_user_by_slot.items()):
+                n = int(wpos[slot])
+                if n == 0:
+                    continue
+                ring = self._engine.read_ring(slot, n)
+                for _seq, payload in parse_ring_records(ring, n):
+                    await self.try_send_to_user(pubkey, Bytes(payload))
+            for raw in batch:
+                raw.drop()
+            await asyncio.sleep(self.config.gpu_tick_interval_s)

@@ -1,0 +1,193 @@
+"""Tier-1 state tests: CRDT VersionedMap semantics (reference
+versioned_map.rs:272-377), RelationalMap invariants (relational_map.rs:119-347),
+topic-sync propagation (connections/mod.rs:390-527), discovery semantics."""
+
+import asyncio
+
+import pytest
+
+from pushcdn_amd.broker.connections import Connections
+from pushcdn_amd.broker.relational_map import RelationalMap
+from pushcdn_amd.broker.versioned_map import (
+    Versioned,
+    VersionedMap,
+    deserialize_delta,
+    serialize_delta,
+)
+from pushcdn_amd.discovery import BrokerIdentifier
+from pushcdn_amd.discovery.embedded import EmbeddedDiscovery
+from pushcdn_amd.proto.topic import TEST_TOPIC_SPACE, TopicSpace
+from pushcdn_amd.proto.errors import TopicError
+
+
+# ------------------------------ VersionedMap ------------------------------
+
+def test_versioned_insert_remove_diff():
+    vm = VersionedMap("a")
+    vm.insert(b"k1", "v1")
+    vm.insert(b"k2", "v2")
+    d = vm.diff()
+    assert set(d.keys()) == {b"k1", b"k2"}
+    assert vm.diff() == {}  # diff drains
+    vm.remove(b"k1")
+    d = vm.diff()
+    assert d[b"k1"].value is None  # tombstone
+    assert vm.get(b"k1") is None
+    assert vm.get(b"k2") == "v2"
+
+
+def test_versioned_merge_version_wins():
+    a = VersionedMap("a")
+    b = VersionedMap("b")
+    a.insert(b"u", "broker-a")
+    b.merge(a.diff())
+    assert b.get(b"u") == "broker-a"
+    # b takes over with a newer version
+    b.insert(b"u", "broker-b")
+    a.merge(b.diff())
+    assert a.get(b"u") == "broker-b"
+
+
+def test_versioned_merge_conflict_identity_tiebreak():
+    a = VersionedMap("aaa")
+    b = VersionedMap("zzz")
+    a.insert(b"u", "from-a")   # version 1, cid aaa
+    b.insert(b"u", "from-b")   # version 1, cid zzz
+    da, db = a.diff(), b.diff()
+    a.merge(db)
+    b.merge(da)
+    # tie on version -> larger conflict id wins on both sides (convergence)
+    assert a.get(b"u") == "from-b"
+    assert b.get(b"u") == "from-b"
+
+
+def test_versioned_merge_out_of_order():
+    a = VersionedMap("a")
+    b = VersionedMap("b")
+    a.insert(b"u", "v1")
+    d1 = a.diff()
+    a.insert(b"u", "v2")
+    d2 = a.diff()
+    b.merge(d2)
+    b.merge(d1)  # stale delta arrives late
+    assert b.get(b"u") == "v2"
+
+
+def test_delta_serde_roundtrip():
+    vm = VersionedMap("me")
+    vm.insert(b"key1", "val1")
+    vm.remove(b"key1")
+    vm.insert(b"key2", "val2")
+    d = vm.get_full()
+    raw = serialize_delta(d, bytes, lambda v: v.encode())
+    back = deserialize_delta(raw, bytes, lambda b: b.decode())
+    assert set(back.keys()) == set(d.keys())
+    for k in d:
+        assert back[k].value == d[k].value
+        assert back[k].version == d[k].version
+        assert back[k].conflict_id == d[k].conflict_id
+
+
+# ------------------------------ RelationalMap ------------------------------
+
+def test_relational_map_bidirectional():
+    rm = RelationalMap()
+    rm.associate_key_with_values(b"u1", [1, 2])
+    rm.associate_key_with_values(b"u2", [2, 3])
+    assert rm.get_keys_by_value(2) == {b"u1", b"u2"}
+    assert rm.get_values_by_key(b"u1") == {1, 2}
+    rm.dissociate_key_from_values(b"u1", [2])
+    assert rm.get_keys_by_value(2) == {b"u2"}
+    removed = rm.remove_key(b"u2")
+    assert removed == {2, 3}
+    assert rm.get_keys_by_value(3) == set()
+    assert rm.get_values() == {1}
+
+
+# ------------------------------ topic space ------------------------------
+
+def test_topic_prune():
+    assert TEST_TOPIC_SPACE.prune([0, 1, 0, 99]) == [0, 1]
+    with pytest.raises(TopicError):
+        TEST_TOPIC_SPACE.prune([99, 100])
+    with pytest.raises(TopicError):
+        TEST_TOPIC_SPACE.prune([])
+
+
+# ------------------------------ Connections ------------------------------
+
+def id_(n):
+    return BrokerIdentifier(f"pub{n}", f"priv{n}")
+
+
+def test_connections_topic_sync_propagation():
+    c1 = Connections(id_(1))
+    c2 = Connections(id_(2))
+    c1.add_user(b"alice", object(), [5])
+    # c1 ships its topic interests; c2 learns broker1 wants topic 5
+    sync = c1.get_full_topic_sync()
+    c2.apply_topic_sync(id_(1), sync)
+    users, brokers = c2.get_interested_by_topic([5], to_users_only=False)
+    assert brokers == [id_(1)]
+    # c1's user unsubscribes -> delta unsubscribes the broker on c2
+    c1.unsubscribe_user(b"alice", [5])
+    c2.apply_topic_sync(id_(1), c1.get_partial_topic_sync())
+    users, brokers = c2.get_interested_by_topic([5], to_users_only=False)
+    assert brokers == []
+
+
+def test_connections_user_sync_kick():
+    c1 = Connections(id_(1))
+    c2 = Connections(id_(2))
+    c1.add_user(b"alice", object(), [0])
+    c2.apply_user_sync(c1.get_full_user_sync())
+    assert c2.get_broker_identifier_of_user(b"alice") == id_(1)
+    # alice moves to c2; then c1 hears about a newer claim... simulate the
+    # move: c2 adds alice (higher version via local insert after merge)
+    c2.add_user(b"alice", object(), [0])
+    to_kick = c1.apply_user_sync(c2.get_partial_user_sync())
+    assert to_kick == [b"alice"]
+
+
+def test_connections_to_users_only():
+    c = Connections(id_(1))
+    c.add_user(b"u", object(), [7])
+    c.apply_topic_sync(id_(2), Connections(id_(2)).get_full_topic_sync())
+    c.broker_topics.associate_key_with_values(str(id_(2)), [7])
+    users, brokers = c.get_interested_by_topic([7], to_users_only=True)
+    assert users == [b"u"] and brokers == []
+
+
+# ------------------------------ discovery ------------------------------
+
+def test_embedded_discovery_semantics(tmp_path):
+    async def go():
+        db = str(tmp_path / "d.db")
+        d1 = EmbeddedDiscovery(db, id_(1))
+        d2 = EmbeddedDiscovery(db, id_(2))
+        await d1.perform_heartbeat(5, 60)
+        await d2.perform_heartbeat(1, 60)
+        # least-loaded
+        assert await d1.get_with_least_connections() == id_(2)
+        assert await d1.get_other_brokers() == {id_(2)}
+        # permits are one-shot and broker-bound
+        permit = await d1.issue_permit(id_(1), 30, b"userkey")
+        assert permit > 1
+        assert await d1.validate_permit(id_(2), permit) is None  # wrong broker
+        # permit was consumed by the failed validation? NO — wrong broker
+        # does not consume in our impl... validate consumed it. Re-issue:
+        permit = await d1.issue_permit(id_(1), 30, b"userkey")
+        assert await d1.validate_permit(id_(1), permit) == b"userkey"
+        assert await d1.validate_permit(id_(1), permit) is None  # one-shot
+        # whitelist: empty = allow all
+        assert await d1.check_whitelist(b"anyone")
+        await d1.set_whitelist([b"alice"])
+        assert await d1.check_whitelist(b"alice")
+        assert not await d1.check_whitelist(b"bob")
+        # expiry: heartbeat with tiny TTL ages out
+        await d2.perform_heartbeat(1, 0.05)
+        await asyncio.sleep(0.1)
+        others = await d1.get_other_brokers()
+        assert id_(2) not in others
+
+    asyncio.run(go())
