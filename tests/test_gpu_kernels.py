@@ -202,3 +202,62 @@ def test_engine_gpu_matches_cpu_end_to_end(ops):
         rc = parse_ring_records(cpu.read_ring(u), int(wpos_c[u]))
         rg = parse_ring_records(gpu.read_ring(u), int(wpos_g[u]))
         assert rc == rg, f"user {u}"
+
+
+# ------------------------------ K1: BLS verify ------------------------------
+
+def _namespaced(ns: str, msg: bytes) -> bytes:
+    # namespace || message || spare counter byte (csrc/bls/bls.h contract)
+    return ns.encode() + msg + b"\x00"
+
+
+def test_k1_bls_verify_batch_matches_host(ops):
+    from pushcdn_amd.crypto import bls
+    from pushcdn_amd.ops.build import build_core
+
+    core = build_core()
+    ns = bls.USER_MARSHAL_NAMESPACE
+    N = 64
+    vks, sigs, msgs, offsets, want = [], [], bytearray(), [0], []
+    for i in range(N):
+        kp = bls.KeyPair.from_seed(i)
+        msg = f"timestamp-{i}".encode()
+        sig = bls.sign(kp.private_key, ns, msg)
+        good = i % 3 != 0
+        if not good:
+            sig = bytearray(sig)
+            sig[1] ^= 0x40  # corrupt
+            sig = bytes(sig)
+        vks.append(kp.public_key)
+        sigs.append(sig)
+        msgs += _namespaced(ns, msg)
+        offsets.append(len(msgs))
+        want.append(1 if core.verify(kp.public_key, ns, msg, sig) else 0)
+        assert want[-1] == (1 if good else 0)
+
+    vks_t = torch.frombuffer(bytearray(b"".join(vks)), dtype=torch.uint8).to("cuda")
+    sigs_t = torch.frombuffer(bytearray(b"".join(sigs)), dtype=torch.uint8).to("cuda")
+    msgs_t = torch.frombuffer(bytearray(msgs), dtype=torch.uint8).to("cuda")
+    moff_t = torch.tensor(offsets, dtype=torch.int64, device="cuda")
+    ok = ops.bls_verify_batch(vks_t, sigs_t, msgs_t, moff_t)
+    torch.cuda.synchronize()
+    assert ok.cpu().tolist() == want
+
+
+def test_k1_hash_to_g1_matches_host(ops):
+    from pushcdn_amd.crypto import bls
+    from pushcdn_amd.ops.build import build_core
+
+    core = build_core()
+    msgs, offsets, want = bytearray(), [0], []
+    for i in range(32):
+        msg = f"htg-{i}".encode()
+        msgs += _namespaced("ns", msg)
+        offsets.append(len(msgs))
+        want.append(core._hash_to_g1("ns", msg))
+    msgs_t = torch.frombuffer(bytearray(msgs), dtype=torch.uint8).to("cuda")
+    moff_t = torch.tensor(offsets, dtype=torch.int64, device="cuda")
+    out = ops.hash_to_g1_batch(msgs_t, moff_t)
+    torch.cuda.synchronize()
+    got = out.cpu().numpy().tobytes()
+    assert got == b"".join(want)
