@@ -101,8 +101,10 @@ def main() -> None:
 
     n_local_users = (args.subscribers + world_size - 1) // world_size
     ring_bytes = 1 << 9
-    # Ring must hold one step's worth: world_size * batch * (16 + padded payload)
-    need = world_size * args.batch * (16 + ((args.payload + 15) & ~15))
+    # Ring must hold one step's worth of WIRE messages (payload + ~64 B capnp
+    # envelope), each as a 16 B record header + 16-aligned body.
+    wire_est = 16 + ((args.payload + 64 + 15) & ~15)
+    need = world_size * args.batch * wire_est
     while ring_bytes < need * 2:
         ring_bytes <<= 1
 
@@ -111,6 +113,9 @@ def main() -> None:
         n_users=n_local_users,
         ring_bytes=ring_bytes,
         use_gpu_ops=not use_cpu,
+        fanout_wire=True,       # forward raw wire bytes verbatim (reference semantics)
+        direct_enabled=False,   # broadcast-only benchmark (BASELINE configs 2-3)
+        pair_capacity=max(1 << 20, world_size * args.batch * n_local_users),
     )
     eng.subscribe_all(list(range(args.topics)))
 

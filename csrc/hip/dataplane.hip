@@ -309,7 +309,10 @@ extern "C" __global__ void k2b_emit(
         // 16-byte record header + payload padded to 16 so every payload copy
         // in K3 is uint4-aligned (ring_bytes must be a multiple of 16).
         uint64_t rec = 16 + (uint64_t)((len + 15) & ~15);
-        if (wpos + rec > (uint64_t)ring_bytes) { pair_user[slot] = -1; slot++; dropped++; continue; }
+        if (wpos + rec > (uint64_t)ring_bytes) {
+            pair_user[slot] = -1; pair_msg[slot] = m; pair_dst[slot] = 0;
+            slot++; dropped++; continue;
+        }
         pair_user[slot] = u;
         pair_msg[slot] = m;
         pair_dst[slot] = (int64_t)u * ring_bytes + wpos;
@@ -487,6 +490,146 @@ void launch_k2c_apply_subs(uint64_t* sub_bitmap, const uint8_t* buf, const int64
     int threads = 256, blocks = (M + threads - 1) / threads;
     hipLaunchKernelGGL(k2c_apply_subs, dim3(blocks), dim3(threads), 0, s, sub_bitmap, buf,
                        topics_off, topics_cnt, disc, user_idx, M, W);
+}
+
+}  // extern "C"
+
+// ---------------------------------------------------------------------------
+// K3v2: wave-per-pair fan-out. 256-thread workgroups = 4 waves; each wave
+// owns one delivery pair per grid-stride iteration (64 lanes x 16 B = 1 KiB
+// per pass — matches the 1 KiB-payload sweet spot; a 64 KiB payload takes 64
+// passes). n_pairs is read from a device pointer so the host never syncs on
+// the pair count. NT=1 uses non-temporal stores for the egress payload
+// (written once, consumed by the D2H drain) to keep the per-XCD L2 for the
+// hot message source bytes instead of the streaming egress.
+// ---------------------------------------------------------------------------
+template <int NT>
+__global__ void __launch_bounds__(256) k3_fanout_wave_t(
+    const uint8_t* __restrict__ buf,
+    const int64_t* __restrict__ payload_off,
+    const int32_t* __restrict__ payload_len,
+    const int32_t* __restrict__ pair_user,
+    const int32_t* __restrict__ pair_msg,
+    const int64_t* __restrict__ pair_dst,
+    const uint32_t* __restrict__ msg_seq,
+    const int32_t* __restrict__ n_pairs_ptr,
+    uint8_t* __restrict__ egress)
+{
+    const int n_pairs = *n_pairs_ptr;
+    const int lane = threadIdx.x & 63;
+    const int wave_in_wg = threadIdx.x >> 6;
+    const int waves_total = gridDim.x * 4;
+    for (int p = blockIdx.x * 4 + wave_in_wg; p < n_pairs; p += waves_total) {
+        const int u = pair_user[p];
+        if (u < 0) continue;
+        const int mi = pair_msg[p];
+        const int32_t len = payload_len[mi];
+        const uint8_t* src = buf + payload_off[mi];
+        uint8_t* dst = egress + pair_dst[p];
+        if (lane == 0) {
+            uint32_t hdr[4] = {(uint32_t)len, msg_seq[mi], 0, 0};
+            if (NT) {
+                typedef unsigned int v4u __attribute__((ext_vector_type(4)));
+                v4u h; memcpy(&h, hdr, 16);
+                __builtin_nontemporal_store(h, (v4u*)dst);
+            } else {
+                memcpy(dst, hdr, 16);
+            }
+        }
+        dst += 16;
+        const int32_t nvec = len >> 4;
+        const bool src16 = (((uintptr_t)src) & 15) == 0;
+        if (src16) {
+            typedef unsigned int v4u __attribute__((ext_vector_type(4)));
+            const v4u* s4 = (const v4u*)src;
+            v4u* d4 = (v4u*)dst;
+            for (int k = lane; k < nvec; k += 64) {
+                v4u v = s4[k];
+                if (NT) __builtin_nontemporal_store(v, d4 + k);
+                else d4[k] = v;
+            }
+        } else {
+            for (int k = lane; k < nvec; k += 64) {
+                uint8_t tmp[16];
+                memcpy(tmp, src + (size_t)k * 16, 16);
+                memcpy(dst + (size_t)k * 16, tmp, 16);
+            }
+        }
+        for (int k = (nvec << 4) + lane; k < len; k += 64) dst[k] = src[k];
+    }
+}
+
+// K2b capacity-guarded emit: same per-user ordered scan, but drops (and
+// counts) deliveries whose pair slot would exceed `capacity` so the host can
+// preallocate fixed buffers and never sync on the exact count.
+extern "C" __global__ void k2b_emit_capped(
+    const uint64_t* __restrict__ mask,
+    const int64_t* __restrict__ payload_off,
+    const int32_t* __restrict__ payload_len,
+    const int32_t* __restrict__ pair_base,
+    int32_t M, int32_t W, int32_t n_users,
+    int64_t ring_bytes, int32_t capacity,
+    uint64_t* __restrict__ ring_wpos,
+    int32_t* __restrict__ pair_user,
+    int32_t* __restrict__ pair_msg,
+    int64_t* __restrict__ pair_dst,
+    uint32_t* __restrict__ drops)
+{
+    int u = blockIdx.x * blockDim.x + threadIdx.x;
+    if (u >= n_users) return;
+    int w = u >> 6;
+    uint64_t bit = 1ull << (u & 63);
+    int slot = pair_base[u];
+    uint64_t wpos = ring_wpos[u];
+    uint32_t dropped = 0;
+    for (int m = 0; m < M; ++m) {
+        if (!(mask[(int64_t)m * W + w] & bit)) continue;
+        int32_t len = payload_len[m];
+        uint64_t rec = 16 + (uint64_t)((len + 15) & ~15);
+        bool fits_ring = (wpos + rec <= (uint64_t)ring_bytes);
+        bool fits_cap = (slot < capacity);
+        if (!fits_ring || !fits_cap) {
+            if (fits_cap) { pair_user[slot] = -1; pair_msg[slot] = m; pair_dst[slot] = 0; slot++; }
+            dropped++;
+            continue;
+        }
+        pair_user[slot] = u;
+        pair_msg[slot] = m;
+        pair_dst[slot] = (int64_t)u * ring_bytes + wpos;
+        slot++;
+        wpos += rec;
+    }
+    ring_wpos[u] = wpos;
+    if (dropped) atomicAdd(drops, dropped);
+}
+
+extern "C" {
+
+void launch_k3_fanout_wave(const uint8_t* buf, const int64_t* payload_off,
+                           const int32_t* payload_len, const int32_t* pair_user,
+                           const int32_t* pair_msg, const int64_t* pair_dst,
+                           const uint32_t* msg_seq, const int32_t* n_pairs_ptr,
+                           uint8_t* egress, int nt, int grid, hipStream_t s) {
+    if (grid <= 0) grid = 4096;  // 4096 WGs x 4 waves = 16384 concurrent pairs
+    if (nt)
+        hipLaunchKernelGGL((k3_fanout_wave_t<1>), dim3(grid), dim3(256), 0, s, buf, payload_off,
+                           payload_len, pair_user, pair_msg, pair_dst, msg_seq, n_pairs_ptr,
+                           egress);
+    else
+        hipLaunchKernelGGL((k3_fanout_wave_t<0>), dim3(grid), dim3(256), 0, s, buf, payload_off,
+                           payload_len, pair_user, pair_msg, pair_dst, msg_seq, n_pairs_ptr,
+                           egress);
+}
+
+void launch_k2b_emit_capped(const uint64_t* mask, const int64_t* payload_off,
+                            const int32_t* payload_len, const int32_t* pair_base, int32_t M,
+                            int32_t W, int32_t n_users, int64_t ring_bytes, int32_t capacity,
+                            uint64_t* ring_wpos, int32_t* pair_user, int32_t* pair_msg,
+                            int64_t* pair_dst, uint32_t* drops, hipStream_t s) {
+    int threads = 256, blocks = (n_users + threads - 1) / threads;
+    hipLaunchKernelGGL(k2b_emit_capped, dim3(blocks), dim3(threads), 0, s, mask, payload_off,
+                       payload_len, pair_base, M, W, n_users, ring_bytes, capacity, ring_wpos,
+                       pair_user, pair_msg, pair_dst, drops);
 }
 
 }  // extern "C"

@@ -261,3 +261,77 @@ def test_k1_hash_to_g1_matches_host(ops):
     torch.cuda.synchronize()
     got = out.cpu().numpy().tobytes()
     assert got == b"".join(want)
+
+
+def test_assign_emit_into_and_fanout_wave(ops):
+    """The sync-free pipeline (preallocated buffers + device count + wave
+    fanout, NT and non-NT) must produce identical rings to the reference."""
+    rng = random.Random(13)
+    n_users = 257
+    ring_bytes = 1 << 13
+    W = (n_users + 63) // 64
+    sub = torch.zeros((256, W), dtype=torch.int64)
+    for u in range(n_users):
+        for t in rng.sample(range(6), 2):
+            sub[t, u >> 6] |= (1 << (u & 63)) - (1 << 64) if (u & 63) == 63 else 1 << (u & 63)
+    msgs = [m.Broadcast([rng.randrange(6)], bytes([rng.randrange(256)]) * rng.randrange(1, 700))
+            for _ in range(50)]
+    buf, offsets = make_batch(msgs)
+
+    pr = ref.parse_batch(buf, offsets)
+    maskr = ref.topic_mask(sub, buf, pr.topics_off, pr.topics_cnt, pr.disc)
+    wposr = torch.zeros(n_users, dtype=torch.int64)
+    pu_r, pm_r, pd_r, drops_r = ref.assign_emit(maskr, pr.payload_len, wposr, ring_bytes, n_users)
+    arr = bytearray(n_users * ring_bytes)
+    seq = torch.arange(0, len(msgs), dtype=torch.int32)
+    ref.fanout(buf, pr.payload_off, pr.payload_len, pu_r, pm_r, pd_r, seq, arr)
+
+    for nt in (0, 1):
+        dbuf, doff = to_dev(buf, offsets)
+        disc, poff, plen, toff, tcnt, rhash, ts = ops.parse_batch(dbuf, doff)
+        mask = ops.topic_mask(sub.to("cuda"), dbuf, toff, tcnt, disc)
+        wpos = torch.zeros(n_users, dtype=torch.int64, device="cuda")
+        cap = pu_r.shape[0] + 16
+        pu = torch.empty(cap, dtype=torch.int32, device="cuda")
+        pm = torch.empty(cap, dtype=torch.int32, device="cuda")
+        pd = torch.empty(cap, dtype=torch.int64, device="cuda")
+        drops = torch.zeros(1, dtype=torch.int32, device="cuda")
+        n_pairs = torch.zeros(1, dtype=torch.int32, device="cuda")
+        ops.assign_emit_into(mask, poff, plen, wpos, ring_bytes, n_users,
+                             pu, pm, pd, drops, n_pairs)
+        npc = int(n_pairs.cpu()[0])
+        assert npc == pu_r.shape[0]
+        assert torch.equal(pu[:npc].cpu(), pu_r)
+        assert torch.equal(pm[:npc].cpu(), pm_r)
+        assert torch.equal(pd[:npc].cpu(), pd_r)
+        assert int(drops.cpu()[0]) == drops_r
+        assert torch.equal(wpos.cpu(), wposr)
+        egress = torch.zeros(n_users * ring_bytes, dtype=torch.uint8, device="cuda")
+        ops.fanout_wave(dbuf, poff, plen, pu, pm, pd, seq.to("cuda"), n_pairs, egress, nt, 0)
+        torch.cuda.synchronize()
+        assert egress.cpu().numpy().tobytes() == bytes(arr), f"nt={nt}"
+
+
+def test_assign_emit_into_capacity_drop(ops):
+    """When the pair buffer is smaller than the delivery count, the excess is
+    dropped and counted, and n_pairs is clamped to capacity."""
+    n_users = 64
+    W = 1
+    sub = torch.zeros((256, W), dtype=torch.int64)
+    sub[0, 0] = -1  # all 64 users subscribe topic 0
+    msgs = [m.Broadcast([0], b"x" * 32)]
+    buf, offsets = make_batch(msgs)
+    dbuf, doff = to_dev(buf, offsets)
+    disc, poff, plen, toff, tcnt, rhash, ts = ops.parse_batch(dbuf, doff)
+    mask = ops.topic_mask(sub.to("cuda"), dbuf, toff, tcnt, disc)
+    wpos = torch.zeros(n_users, dtype=torch.int64, device="cuda")
+    cap = 40  # < 64 deliveries
+    pu = torch.empty(cap, dtype=torch.int32, device="cuda")
+    pm = torch.empty(cap, dtype=torch.int32, device="cuda")
+    pd = torch.empty(cap, dtype=torch.int64, device="cuda")
+    drops = torch.zeros(1, dtype=torch.int32, device="cuda")
+    n_pairs = torch.zeros(1, dtype=torch.int32, device="cuda")
+    ops.assign_emit_into(mask, poff, plen, wpos, 1 << 12, n_users, pu, pm, pd, drops, n_pairs)
+    torch.cuda.synchronize()
+    assert int(n_pairs.cpu()[0]) == cap
+    assert int(drops.cpu()[0]) == 64 - cap
