@@ -46,20 +46,26 @@ from pushcdn_amd.proto import message as msglib
 
 
 def build_batch(n_msgs: int, payload_bytes: int, n_topics: int, seed: int):
-    """Serialize one ingest batch; returns (buf, offsets) with every message
-    padded to equal wire size so batches are all-gather-able."""
+    """Serialize one ingest batch; returns (buf, offsets, wire_len).  Every
+    message start is 16-byte aligned (so the flat K3 kernel's 16 B units stay
+    vector-aligned) and every message has the same wire length."""
     import random
 
     rng = random.Random(seed)
     buf = bytearray()
     offsets = [0]
+    wire_len = None
     for i in range(n_msgs):
         payload = bytes(rng.randrange(256) for _ in range(payload_bytes))
         msg = msglib.Broadcast([i % n_topics], payload)
         raw = msglib.serialize(msg)
-        buf += raw
+        padded = (len(raw) + 15) & ~15
+        if wire_len is None:
+            wire_len = padded
+        assert padded == wire_len, "non-uniform wire sizes in batch"
+        buf += raw + b"\x00" * (padded - len(raw))
         offsets.append(len(buf))
-    return bytes(buf), offsets
+    return bytes(buf), offsets, wire_len
 
 
 def main() -> None:
@@ -126,11 +132,12 @@ def main() -> None:
         build_batch(args.batch, args.payload, args.topics, seed=rank * 1000 + v)
         for v in range(n_variants)
     ]
-    cap = max(len(b) for b, _ in host_batches)
+    wire_len = host_batches[0][2]
+    cap = max(len(b) for b, _, _ in host_batches)
     cap = (cap + 255) & ~255
     offsets_t = torch.tensor(host_batches[0][1], dtype=torch.int64)
     pinned = []
-    for b, off in host_batches:
+    for b, off, _ in host_batches:
         t = torch.zeros(cap, dtype=torch.uint8)
         t[: len(b)] = torch.frombuffer(bytearray(b), dtype=torch.uint8)
         if not use_cpu:
@@ -153,6 +160,7 @@ def main() -> None:
                     dev_offsets,
                     host_batch=None if not use_cpu else bytes(src[: len(host_batches[i % n_variants][0])].numpy().tobytes()),
                     host_offsets=None if not use_cpu else host_batches[i % n_variants][1],
+                    uniform_wire_len=wire_len,
                 )
         else:
             eng.tick(
@@ -160,6 +168,7 @@ def main() -> None:
                 dev_offsets,
                 host_batch=None if not use_cpu else host_batches[i % n_variants][0],
                 host_offsets=None if not use_cpu else host_batches[i % n_variants][1],
+                uniform_wire_len=wire_len,
             )
         eng.drain_cursors()
 

@@ -42,6 +42,9 @@ void launch_k3_fanout_wave(const uint8_t*, const int64_t*, const int32_t*, const
 void launch_k2b_emit_capped(const uint64_t*, const int64_t*, const int32_t*, const int32_t*,
                             int32_t, int32_t, int32_t, int64_t, int32_t, uint64_t*, int32_t*,
                             int32_t*, int64_t*, uint32_t*, hipStream_t);
+void launch_k3_fanout_flat(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
+                           const int32_t*, const int64_t*, const uint32_t*, const int32_t*,
+                           int32_t, uint8_t*, int, int, hipStream_t);
 }
 
 #define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
@@ -231,6 +234,19 @@ void fanout_wave(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor pay
                           (int)grid, cur_stream());
 }
 
+void fanout_flat(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_len,
+                 torch::Tensor pair_user, torch::Tensor pair_msg, torch::Tensor pair_dst,
+                 torch::Tensor msg_seq, torch::Tensor n_pairs, int64_t units_per_pair,
+                 torch::Tensor egress, int64_t nt, int64_t grid) {
+    CHECK_DEV(egress); CHECK_CONTIG(egress);
+    launch_k3_fanout_flat(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
+                          payload_len.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
+                          pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
+                          (const uint32_t*)msg_seq.data_ptr<int32_t>(),
+                          n_pairs.data_ptr<int32_t>(), (int32_t)units_per_pair,
+                          egress.data_ptr<uint8_t>(), (int)nt, (int)grid, cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("parse_batch", &parse_batch, "K4: on-device capnp parse of a message batch");
     m.def("topic_mask", &topic_mask, "K2a: per-message subscriber mask");
@@ -243,4 +259,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("assign_emit_into", &assign_emit_into,
           "K2b sync-free: emit pairs into preallocated buffers, count stays on device");
     m.def("fanout_wave", &fanout_wave, "K3v2: wave-per-pair fan-out (nt flag, device count)");
+    m.def("fanout_flat", &fanout_flat, "K3v3: flat-index fan-out for uniform record sizes");
 }

@@ -410,20 +410,25 @@ extern "C" __global__ void k2c_apply_subs(
     const int32_t* __restrict__ user_idx,    // [M] local user index per message
     int32_t M, int32_t W)
 {
-    int m = blockIdx.x * blockDim.x + threadIdx.x;
-    if (m >= M) return;
-    int d = disc[m];
-    if (d != 5 && d != 6) return;
-    int u = user_idx[m];
-    if (u < 0) return;
-    int w = u >> 6;
-    uint64_t bit = 1ull << (u & 63);
-    const uint8_t* topics = buf + topics_off[m];
-    int n = topics_cnt[m];
-    for (int t = 0; t < n; ++t) {
-        uint64_t* word = &sub_bitmap[(int64_t)topics[t] * W + w];
-        if (d == 5) atomicOr((unsigned long long*)word, (unsigned long long)bit);
-        else        atomicAnd((unsigned long long*)word, (unsigned long long)~bit);
+    // One thread applies the whole batch IN MESSAGE ORDER: a subscribe and
+    // a later unsubscribe of the same (user, topic) in one batch must land
+    // in order (subscription updates are control-plane-rate, so the serial
+    // walk is microseconds; the data-plane kernels stay fully parallel).
+    if (blockIdx.x != 0 || threadIdx.x != 0) return;
+    for (int m = 0; m < M; ++m) {
+        int d = disc[m];
+        if (d != 5 && d != 6) continue;
+        int u = user_idx[m];
+        if (u < 0) continue;
+        int w = u >> 6;
+        uint64_t bit = 1ull << (u & 63);
+        const uint8_t* topics = buf + topics_off[m];
+        int n = topics_cnt[m];
+        for (int t = 0; t < n; ++t) {
+            uint64_t* word = &sub_bitmap[(int64_t)topics[t] * W + w];
+            if (d == 5) *word |= bit;
+            else        *word &= ~bit;
+        }
     }
 }
 
@@ -633,3 +638,70 @@ void launch_k2b_emit_capped(const uint64_t* mask, const int64_t* payload_off,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// K3v3: flat-index fan-out for UNIFORM record sizes (every message in the
+// tick has the same wire length — the common case for batched pub/sub
+// traffic). Work unit = one 16 B chunk of one delivery record; lane ->
+// consecutive flat units, so lane utilization is ~100% regardless of
+// payload size (the wave-per-pair variant idles lanes on the tail pass).
+// units_per_pair = record_bytes / 16 (header unit + payload units).
+// ---------------------------------------------------------------------------
+template <int NT>
+__global__ void __launch_bounds__(256) k3_fanout_flat_t(
+    const uint8_t* __restrict__ buf,
+    const int64_t* __restrict__ payload_off,
+    const int32_t* __restrict__ payload_len,
+    const int32_t* __restrict__ pair_user,
+    const int32_t* __restrict__ pair_msg,
+    const int64_t* __restrict__ pair_dst,
+    const uint32_t* __restrict__ msg_seq,
+    const int32_t* __restrict__ n_pairs_ptr,
+    int32_t units_per_pair,
+    uint8_t* __restrict__ egress)
+{
+    typedef unsigned int v4u __attribute__((ext_vector_type(4)));
+    const int64_t n_units = (int64_t)(*n_pairs_ptr) * units_per_pair;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < n_units; f += stride) {
+        const int p = (int)(f / units_per_pair);
+        const int unit = (int)(f - (int64_t)p * units_per_pair);
+        const int u = pair_user[p];
+        if (u < 0) continue;
+        const int mi = pair_msg[p];
+        uint8_t* dst = egress + pair_dst[p] + (size_t)unit * 16;
+        if (unit == 0) {
+            uint32_t hdr[4] = {(uint32_t)payload_len[mi], msg_seq[mi], 0, 0};
+            v4u h; memcpy(&h, hdr, 16);
+            if (NT) __builtin_nontemporal_store(h, (v4u*)dst);
+            else memcpy(dst, hdr, 16);
+            continue;
+        }
+        const uint8_t* src = buf + payload_off[mi] + (size_t)(unit - 1) * 16;
+        const int32_t len = payload_len[mi];
+        const int32_t coff = (unit - 1) * 16;
+        if (coff + 16 <= len && (((uintptr_t)src) & 15) == 0) {
+            v4u v = *(const v4u*)src;
+            if (NT) __builtin_nontemporal_store(v, (v4u*)dst);
+            else *(v4u*)dst = v;
+        } else {
+            for (int b = 0; b < 16 && coff + b < len; ++b) dst[b] = src[b];
+        }
+    }
+}
+
+extern "C" void launch_k3_fanout_flat(
+    const uint8_t* buf, const int64_t* payload_off, const int32_t* payload_len,
+    const int32_t* pair_user, const int32_t* pair_msg, const int64_t* pair_dst,
+    const uint32_t* msg_seq, const int32_t* n_pairs_ptr, int32_t units_per_pair,
+    uint8_t* egress, int nt, int grid, hipStream_t s) {
+    if (grid <= 0) grid = 8192;
+    if (nt)
+        hipLaunchKernelGGL((k3_fanout_flat_t<1>), dim3(grid), dim3(256), 0, s, buf, payload_off,
+                           payload_len, pair_user, pair_msg, pair_dst, msg_seq, n_pairs_ptr,
+                           units_per_pair, egress);
+    else
+        hipLaunchKernelGGL((k3_fanout_flat_t<0>), dim3(grid), dim3(256), 0, s, buf, payload_off,
+                           payload_len, pair_user, pair_msg, pair_dst, msg_seq, n_pairs_ptr,
+                           units_per_pair, egress);
+}

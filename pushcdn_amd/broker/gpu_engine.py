@@ -152,14 +152,20 @@ class GpuBrokerEngine:
 
     def tick(self, buf: torch.Tensor, offsets: torch.Tensor,
              host_batch: Optional[bytes] = None,
-             host_offsets: Optional[List[int]] = None) -> TickStats:
-        """Run the full pipeline on one ingested batch already on device."""
+             host_offsets: Optional[List[int]] = None,
+             uniform_wire_len: Optional[int] = None) -> TickStats:
+        """Run the full pipeline on one ingested batch already on device.
+
+        uniform_wire_len: if the caller knows every message in the batch has
+        this exact wire length (and fanout_wire is set), the flat-index K3
+        variant runs at ~100% lane utilization."""
         if self.use_gpu_ops:
-            return self._tick_gpu(buf, offsets)
+            return self._tick_gpu(buf, offsets, uniform_wire_len)
         assert host_batch is not None and host_offsets is not None
         return self._tick_cpu(host_batch, host_offsets)
 
-    def _tick_gpu(self, buf: torch.Tensor, offsets: torch.Tensor) -> TickStats:
+    def _tick_gpu(self, buf: torch.Tensor, offsets: torch.Tensor,
+                  uniform_wire_len: Optional[int] = None) -> TickStats:
         ops = self._ops
         M = offsets.shape[0] - 1
         disc, payload_off, payload_len, topics_off, topics_cnt, recip_hash, _ts = ops.parse_batch(
@@ -176,9 +182,14 @@ class GpuBrokerEngine:
         )
         seq = torch.arange(self.seq, self.seq + M, dtype=torch.int32, device=self.device)
         self.seq += M
-        ops.fanout_wave(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
-                        self._pair_dst, seq, self._n_pairs, self.egress,
-                        1 if self.nt_fanout else 0, 0)
+        nt = 1 if self.nt_fanout else 0
+        if self.fanout_wire and uniform_wire_len is not None:
+            units = 1 + ((uniform_wire_len + 15) & ~15) // 16
+            ops.fanout_flat(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
+                            self._pair_dst, seq, self._n_pairs, units, self.egress, nt, 0)
+        else:
+            ops.fanout_wave(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
+                            self._pair_dst, seq, self._n_pairs, self.egress, nt, 0)
         stats = TickStats(n_messages=M)
         # K5 lookup always runs (cheap, no host sync); the host-side direct
         # pair synthesis (one tiny sync) is skipped for broadcast-only

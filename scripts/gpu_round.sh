@@ -15,29 +15,32 @@ import time, torch, random
 from pushcdn_amd.broker.gpu_engine import GpuBrokerEngine
 from pushcdn_amd.proto import message as msglib
 
-def bench_engine(nt, steps=20):
+def bench_engine(nt, flat, steps=20):
     eng = GpuBrokerEngine(device="cuda:0", n_users=10000, ring_bytes=1<<21,
                           fanout_wire=True, direct_enabled=False, nt_fanout=nt,
                           pair_capacity=4<<20)
     eng.subscribe_all(list(range(8)))
     rng = random.Random(0)
-    buf = bytearray(); offsets=[0]
+    buf = bytearray(); offsets=[0]; wl=None
     for i in range(256):
         raw = msglib.serialize(msglib.Broadcast([i % 8], bytes(rng.randrange(256) for _ in range(1024))))
-        buf += raw; offsets.append(len(buf))
+        padded = (len(raw)+15)&~15; wl = padded
+        buf += raw + b"\x00"*(padded-len(raw)); offsets.append(len(buf))
     dbuf, doff = eng.ingest(bytes(buf), offsets)
+    uw = wl if flat else None
     for _ in range(5):
-        eng.tick(dbuf, doff); eng.drain_cursors()
+        eng.tick(dbuf, doff, uniform_wire_len=uw); eng.drain_cursors()
     torch.cuda.synchronize()
     t0 = time.time()
     for _ in range(steps):
-        eng.tick(dbuf, doff); eng.drain_cursors()
+        eng.tick(dbuf, doff, uniform_wire_len=uw); eng.drain_cursors()
     torch.cuda.synchronize()
     return (time.time()-t0)/steps*1000
 
 for rnd in range(3):
     for nt in (0,1):
-        print(f"round {rnd} nt={nt}: {bench_engine(bool(nt)):.3f} ms/tick")
+        for flat in (0,1):
+            print(f"round {rnd} nt={nt} flat={flat}: {bench_engine(bool(nt), bool(flat)):.3f} ms/tick")
 PYEOF
 
 cd /tmp && export TMPDIR=/tmp && cd /root/repo

@@ -335,3 +335,56 @@ def test_assign_emit_into_capacity_drop(ops):
     torch.cuda.synchronize()
     assert int(n_pairs.cpu()[0]) == cap
     assert int(drops.cpu()[0]) == 64 - cap
+
+
+def test_fanout_flat_uniform_matches_reference(ops):
+    """Flat-index K3 (uniform 16-aligned records) == reference fanout."""
+    rng = random.Random(17)
+    n_users = 130
+    ring_bytes = 1 << 16
+    W = (n_users + 63) // 64
+    sub = torch.zeros((256, W), dtype=torch.int64)
+    sub[2, 0] = -1
+    sub[2, 1] = -1
+    # uniform wire messages, 16-aligned starts
+    buf = bytearray()
+    offsets = [0]
+    wire_len = None
+    for i in range(32):
+        raw = m.serialize(m.Broadcast([2], bytes([i]) * 1024))
+        padded = (len(raw) + 15) & ~15
+        wire_len = padded
+        buf += raw + b"\x00" * (padded - len(raw))
+        offsets.append(len(buf))
+    buf = bytes(buf)
+
+    # reference (wire-mode: payload = whole message)
+    woff = torch.tensor(offsets[:-1], dtype=torch.int64)
+    wlen = torch.full((32,), wire_len, dtype=torch.int32)
+    mask = torch.zeros((32, W), dtype=torch.int64)
+    mask[:, 0] = -1
+    mask[:, 1] = -1
+    wposr = torch.zeros(n_users, dtype=torch.int64)
+    pu_r, pm_r, pd_r, _ = ref.assign_emit(mask, wlen, wposr, ring_bytes, n_users)
+    arr = bytearray(n_users * ring_bytes)
+    seq = torch.arange(0, 32, dtype=torch.int32)
+    ref.fanout(buf, woff, wlen, pu_r, pm_r, pd_r, seq, arr)
+
+    for nt in (0, 1):
+        dmask = mask.to("cuda")
+        dbuf = torch.frombuffer(bytearray(buf), dtype=torch.uint8).to("cuda")
+        wpos = torch.zeros(n_users, dtype=torch.int64, device="cuda")
+        cap = pu_r.shape[0]
+        pu = torch.empty(cap, dtype=torch.int32, device="cuda")
+        pm = torch.empty(cap, dtype=torch.int32, device="cuda")
+        pd = torch.empty(cap, dtype=torch.int64, device="cuda")
+        drops = torch.zeros(1, dtype=torch.int32, device="cuda")
+        n_pairs = torch.zeros(1, dtype=torch.int32, device="cuda")
+        ops.assign_emit_into(dmask, woff.to("cuda"), wlen.to("cuda"), wpos, ring_bytes,
+                             n_users, pu, pm, pd, drops, n_pairs)
+        egress = torch.zeros(n_users * ring_bytes, dtype=torch.uint8, device="cuda")
+        units = 1 + ((wire_len + 15) & ~15) // 16
+        ops.fanout_flat(dbuf, woff.to("cuda"), wlen.to("cuda"), pu, pm, pd,
+                        seq.to("cuda"), n_pairs, units, egress, nt, 0)
+        torch.cuda.synchronize()
+        assert egress.cpu().numpy().tobytes() == bytes(arr), f"nt={nt}"
