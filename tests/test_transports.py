@@ -111,3 +111,41 @@ def test_recv_on_dead_connection_errors():
             await asyncio.wait_for(b.recv_message(), timeout=5)
 
     run(go())
+
+
+def test_tcp_tls_conformance():
+    from pushcdn_amd.proto.transports.tcp_tls import TcpTls
+
+    run(_conformance(TcpTls, "127.0.0.1:0"))
+
+
+def test_tls_rejects_untrusted_ca(tmp_path):
+    """A client trusting a DIFFERENT CA must fail the handshake."""
+    from pushcdn_amd.crypto import tls as tlslib
+    from pushcdn_amd.proto.transports.tcp_tls import TcpTls
+
+    async def go():
+        limiter = Limiter()
+        listener = await TcpTls.bind("127.0.0.1:0", None, None)
+        endpoint = f"127.0.0.1:{listener.port}"
+
+        other_ca_cert, _ = tlslib.generate_ca(str(tmp_path / "other-ca"))
+
+        class UntrustingClient(TcpTls):
+            ca_cert_path = other_ca_cert
+
+        async def server():
+            unfinalized = await listener.accept()
+            try:
+                await unfinalized.finalize(limiter)
+            except ConnectionError_:
+                pass
+
+        async def client():
+            with pytest.raises(ConnectionError_):
+                await UntrustingClient.connect(endpoint, False, limiter)
+
+        await asyncio.wait_for(asyncio.gather(server(), client()), timeout=15)
+        await listener.close()
+
+    run(go())
