@@ -134,18 +134,12 @@ def test_tls_rejects_untrusted_ca(tmp_path):
         class UntrustingClient(TcpTls):
             ca_cert_path = other_ca_cert
 
-        async def server():
-            unfinalized = await listener.accept()
-            try:
-                await unfinalized.finalize(limiter)
-            except ConnectionError_:
-                pass
-
-        async def client():
-            with pytest.raises(ConnectionError_):
-                await UntrustingClient.connect(endpoint, False, limiter)
-
-        await asyncio.wait_for(asyncio.gather(server(), client()), timeout=15)
+        # the handshake fails before accept() ever yields a connection, so
+        # only the client side needs checking
+        with pytest.raises(ConnectionError_):
+            await asyncio.wait_for(
+                UntrustingClient.connect(endpoint, False, limiter), timeout=15
+            )
         await listener.close()
 
     run(go())
