@@ -11,6 +11,7 @@
 #include <vector>
 
 #include "bls/bls.h"
+#include "wire/message.h"
 
 namespace py = pybind11;
 using namespace bn254;
@@ -169,6 +170,57 @@ static py::bytes hash_to_g1_test(const std::string& ns, const py::bytes& message
     return to_bytes(out, 64);
 }
 
+// ---------------------------------------------------------------------------
+// wire codec bindings (native host serde; byte-identical to the Python
+// reference implementation — cross-checked in tests/test_wire.py)
+// ---------------------------------------------------------------------------
+static py::bytes vec_bytes(const std::vector<uint8_t>& v) {
+    return py::bytes((const char*)v.data(), v.size());
+}
+
+static py::bytes w_ser_auth_key(const py::bytes& pk, uint64_t ts, const py::bytes& sig) {
+    auto p = to_vec(pk), s = to_vec(sig);
+    return vec_bytes(wire::serialize_authenticate_with_key(p.data(), p.size(), ts, s.data(),
+                                                           s.size()));
+}
+static py::bytes w_ser_auth_permit(uint64_t permit) {
+    return vec_bytes(wire::serialize_authenticate_with_permit(permit));
+}
+static py::bytes w_ser_auth_response(uint64_t permit, const std::string& ctx) {
+    return vec_bytes(wire::serialize_authenticate_response(permit, ctx));
+}
+static py::bytes w_ser_direct(const py::bytes& rcpt, const py::bytes& msg) {
+    auto r = to_vec(rcpt), p = to_vec(msg);
+    return vec_bytes(wire::serialize_direct(r.data(), r.size(), p.data(), p.size()));
+}
+static py::bytes w_ser_broadcast(const py::bytes& topics, const py::bytes& msg) {
+    auto t = to_vec(topics), p = to_vec(msg);
+    return vec_bytes(wire::serialize_broadcast(t.data(), t.size(), p.data(), p.size()));
+}
+static py::bytes w_ser_topics(uint16_t disc, const py::bytes& topics) {
+    auto t = to_vec(topics);
+    return vec_bytes(wire::serialize_topic_list(disc, t.data(), t.size()));
+}
+static py::bytes w_ser_sync(uint16_t disc, const py::bytes& data) {
+    auto d = to_vec(data);
+    return vec_bytes(wire::serialize_sync(disc, d.data(), d.size()));
+}
+static py::object w_deserialize(const py::bytes& raw) {
+    auto v = to_vec(raw);
+    wire::Parsed p;
+    if (!wire::deserialize(v.data(), v.size(), &p)) return py::none();
+    py::dict d;
+    d["disc"] = p.disc;
+    d["timestamp"] = p.timestamp;
+    d["public_key"] = vec_bytes(p.public_key);
+    d["signature"] = vec_bytes(p.signature);
+    d["context"] = p.context;
+    d["recipient"] = vec_bytes(p.recipient);
+    d["topics"] = vec_bytes(p.topics);
+    d["payload"] = vec_bytes(p.payload);
+    return d;
+}
+
 PYBIND11_MODULE(pushcdn_core, m) {
     m.doc() = "pushcdn host core: BLS-over-BN254";
     m.def("keygen", &keygen, "deterministic BLS keypair from a u64 seed -> (sk, vk)");
@@ -182,4 +234,13 @@ PYBIND11_MODULE(pushcdn_core, m) {
     m.def("_subgroup_ok", &subgroup_test);
     m.def("_sha256", &sha256_test);
     m.def("_hash_to_g1", &hash_to_g1_test);
+    // wire codec
+    m.def("wire_serialize_authenticate_with_key", &w_ser_auth_key);
+    m.def("wire_serialize_authenticate_with_permit", &w_ser_auth_permit);
+    m.def("wire_serialize_authenticate_response", &w_ser_auth_response);
+    m.def("wire_serialize_direct", &w_ser_direct);
+    m.def("wire_serialize_broadcast", &w_ser_broadcast);
+    m.def("wire_serialize_topics", &w_ser_topics);
+    m.def("wire_serialize_sync", &w_ser_sync);
+    m.def("wire_deserialize", &w_deserialize);
 }

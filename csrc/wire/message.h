@@ -1,0 +1,276 @@
+// Host C++ Cap'n Proto wire codec for the 9-variant Message union —
+// byte-identical to the Python reference implementation
+// (pushcdn_amd/proto/message.py) and to the encodings the reference's
+// capnp-generated Rust produces (cdn-proto/src/message.rs:116-312;
+// layouts verified against schema/messages_capnp.rs: Message{data 1, ptr 1},
+// AuthenticateWithKey{1,2}, AuthenticateWithPermit{1,0},
+// AuthenticateResponse{1,1}, Direct{0,2}, Broadcast{0,2}).
+//
+// This is the native data path for host-side services (client batch
+// serialization, broker control plane); the GPU path parses the same bytes
+// on-device (csrc/hip/dataplane.hip k4_parse_batch).
+
+#pragma once
+#include <stdint.h>
+#include <cstring>
+#include <optional>
+#include <string>
+#include <vector>
+
+namespace wire {
+
+enum Disc : uint16_t {
+    AUTHENTICATE_WITH_KEY = 0,
+    AUTHENTICATE_WITH_PERMIT = 1,
+    AUTHENTICATE_RESPONSE = 2,
+    DIRECT = 3,
+    BROADCAST = 4,
+    SUBSCRIBE = 5,
+    UNSUBSCRIBE = 6,
+    USER_SYNC = 7,
+    TOPIC_SYNC = 8,
+};
+
+struct Parsed {
+    uint16_t disc = 0xffff;
+    // AuthenticateWithKey
+    std::vector<uint8_t> public_key, signature;
+    uint64_t timestamp = 0;  // also permit for 1/2
+    std::string context;     // AuthenticateResponse
+    std::vector<uint8_t> recipient;                   // Direct
+    std::vector<uint8_t> topics;                      // Broadcast/Sub/Unsub
+    std::vector<uint8_t> payload;                     // message/Data bytes
+};
+
+class Builder {
+  public:
+    std::vector<uint8_t> words;  // segment bytes (multiple of 8)
+
+    size_t alloc(size_t nwords) {
+        size_t off = words.size() / 8;
+        words.resize(words.size() + nwords * 8, 0);
+        return off;
+    }
+    void put_u64(size_t w, uint64_t v) { memcpy(&words[w * 8], &v, 8); }
+    void put_u16(size_t w, size_t byte, uint16_t v) { memcpy(&words[w * 8 + byte], &v, 2); }
+    void put_bytes(size_t w, const uint8_t* d, size_t n) { memcpy(&words[w * 8], d, n); }
+
+    void struct_ptr(size_t pw, size_t tgt, uint16_t dw, uint16_t ptrw) {
+        int64_t b = (int64_t)tgt - (int64_t)(pw + 1);
+        uint64_t v = 0 | (((uint64_t)b & 0x3fffffff) << 2) | ((uint64_t)dw << 32) |
+                     ((uint64_t)ptrw << 48);
+        put_u64(pw, v);
+    }
+    void list_ptr(size_t pw, size_t tgt, uint32_t code, uint32_t count) {
+        int64_t b = (int64_t)tgt - (int64_t)(pw + 1);
+        uint64_t v = 1 | (((uint64_t)b & 0x3fffffff) << 2) | ((uint64_t)code << 32) |
+                     ((uint64_t)count << 35);
+        put_u64(pw, v);
+    }
+    void write_byte_list(size_t pw, const uint8_t* d, size_t n) {
+        size_t tgt = alloc((n + 7) / 8);
+        if (n) put_bytes(tgt, d, n);
+        list_ptr(pw, tgt, 2, (uint32_t)n);
+    }
+
+    // finish: prepend the stream header [u32 0][u32 nwords]
+    std::vector<uint8_t> finish() {
+        std::vector<uint8_t> out(8 + words.size());
+        uint32_t zero = 0, nwords = (uint32_t)(words.size() / 8);
+        memcpy(out.data(), &zero, 4);
+        memcpy(out.data() + 4, &nwords, 4);
+        memcpy(out.data() + 8, words.data(), words.size());
+        return out;
+    }
+
+    // common preamble: root ptr + Message struct; returns union ptr word
+    size_t preamble(uint16_t disc) {
+        size_t root = alloc(1);
+        size_t msg = alloc(2);
+        struct_ptr(root, msg, 1, 1);
+        put_u16(msg, 0, disc);
+        return msg + 1;
+    }
+};
+
+inline std::vector<uint8_t> serialize_authenticate_with_key(
+    const uint8_t* pk, size_t pk_len, uint64_t timestamp, const uint8_t* sig, size_t sig_len) {
+    Builder b;
+    size_t up = b.preamble(AUTHENTICATE_WITH_KEY);
+    size_t inner = b.alloc(3);
+    b.struct_ptr(up, inner, 1, 2);
+    b.write_byte_list(inner + 1, pk, pk_len);
+    b.put_u64(inner, timestamp);
+    b.write_byte_list(inner + 2, sig, sig_len);
+    return b.finish();
+}
+
+inline std::vector<uint8_t> serialize_authenticate_with_permit(uint64_t permit) {
+    Builder b;
+    size_t up = b.preamble(AUTHENTICATE_WITH_PERMIT);
+    size_t inner = b.alloc(1);
+    b.struct_ptr(up, inner, 1, 0);
+    b.put_u64(inner, permit);
+    return b.finish();
+}
+
+inline std::vector<uint8_t> serialize_authenticate_response(uint64_t permit,
+                                                            const std::string& context) {
+    Builder b;
+    size_t up = b.preamble(AUTHENTICATE_RESPONSE);
+    size_t inner = b.alloc(2);
+    b.struct_ptr(up, inner, 1, 1);
+    b.put_u64(inner, permit);
+    std::string text = context;
+    text.push_back('\0');
+    b.write_byte_list(inner + 1, (const uint8_t*)text.data(), text.size());
+    return b.finish();
+}
+
+inline std::vector<uint8_t> serialize_direct(const uint8_t* rcpt, size_t rcpt_len,
+                                             const uint8_t* msg, size_t msg_len) {
+    Builder b;
+    size_t up = b.preamble(DIRECT);
+    size_t inner = b.alloc(2);
+    b.struct_ptr(up, inner, 0, 2);
+    b.write_byte_list(inner, rcpt, rcpt_len);
+    b.write_byte_list(inner + 1, msg, msg_len);
+    return b.finish();
+}
+
+inline std::vector<uint8_t> serialize_broadcast(const uint8_t* topics, size_t n_topics,
+                                                const uint8_t* msg, size_t msg_len) {
+    Builder b;
+    size_t up = b.preamble(BROADCAST);
+    size_t inner = b.alloc(2);
+    b.struct_ptr(up, inner, 0, 2);
+    b.write_byte_list(inner, topics, n_topics);
+    b.write_byte_list(inner + 1, msg, msg_len);
+    return b.finish();
+}
+
+inline std::vector<uint8_t> serialize_topic_list(uint16_t disc, const uint8_t* topics,
+                                                 size_t n) {
+    Builder b;
+    size_t up = b.preamble(disc);
+    b.write_byte_list(up, topics, n);
+    return b.finish();
+}
+
+inline std::vector<uint8_t> serialize_sync(uint16_t disc, const uint8_t* data, size_t n) {
+    Builder b;
+    size_t up = b.preamble(disc);
+    b.write_byte_list(up, data, n);
+    return b.finish();
+}
+
+// ---------------------------------------------------------------------------
+// Deserialization (bounds-checked)
+// ---------------------------------------------------------------------------
+struct Reader {
+    const uint8_t* data;
+    size_t nwords;
+
+    bool u64(size_t w, uint64_t* out) const {
+        if (w >= nwords) return false;
+        memcpy(out, data + w * 8, 8);
+        return true;
+    }
+    bool struct_ptr(size_t pw, size_t* tgt, uint16_t* dw, uint16_t* ptrw) const {
+        uint64_t v;
+        if (!u64(pw, &v) || v == 0 || (v & 3) != 0) return false;
+        int64_t b = (v >> 2) & 0x3fffffff;
+        if (b & 0x20000000) b -= 0x40000000;
+        *dw = (uint16_t)(v >> 32);
+        *ptrw = (uint16_t)(v >> 48);
+        int64_t t = (int64_t)pw + 1 + b;
+        if (t < 0 || (uint64_t)t + *dw + *ptrw > nwords) return false;
+        *tgt = (size_t)t;
+        return true;
+    }
+    bool byte_list(size_t pw, std::vector<uint8_t>* out) const {
+        uint64_t v;
+        if (!u64(pw, &v)) return false;
+        if (v == 0) { out->clear(); return true; }
+        if ((v & 3) != 1) return false;
+        int64_t b = (v >> 2) & 0x3fffffff;
+        if (b & 0x20000000) b -= 0x40000000;
+        uint32_t code = (v >> 32) & 7;
+        uint64_t count = (v >> 35) & 0x1fffffff;
+        if (code != 2) return false;
+        int64_t t = (int64_t)pw + 1 + b;
+        if (t < 0 || (uint64_t)t * 8 + count > nwords * 8) return false;
+        out->assign(data + t * 8, data + t * 8 + count);
+        return true;
+    }
+};
+
+inline bool deserialize(const uint8_t* buf, size_t len, Parsed* out) {
+    if (len < 16) return false;
+    uint32_t seg_m1, nw;
+    memcpy(&seg_m1, buf, 4);
+    memcpy(&nw, buf + 4, 4);
+    if (seg_m1 != 0 || 8 + (uint64_t)nw * 8 > len) return false;
+    Reader r{buf + 8, nw};
+    size_t mt;
+    uint16_t mdw, mpw;
+    if (!r.struct_ptr(0, &mt, &mdw, &mpw) || mdw < 1 || mpw < 1) return false;
+    uint64_t w0;
+    r.u64(mt, &w0);
+    uint16_t disc = (uint16_t)w0;
+    size_t up = mt + mdw;
+    switch (disc) {
+    case AUTHENTICATE_WITH_KEY: {
+        size_t it; uint16_t idw, ipw;
+        if (!r.struct_ptr(up, &it, &idw, &ipw) || idw < 1 || ipw < 2) return false;
+        r.u64(it, &out->timestamp);
+        if (!r.byte_list(it + idw, &out->public_key)) return false;
+        if (!r.byte_list(it + idw + 1, &out->signature)) return false;
+        break;
+    }
+    case AUTHENTICATE_WITH_PERMIT: {
+        size_t it; uint16_t idw, ipw;
+        if (!r.struct_ptr(up, &it, &idw, &ipw) || idw < 1) return false;
+        r.u64(it, &out->timestamp);
+        break;
+    }
+    case AUTHENTICATE_RESPONSE: {
+        size_t it; uint16_t idw, ipw;
+        if (!r.struct_ptr(up, &it, &idw, &ipw) || idw < 1 || ipw < 1) return false;
+        r.u64(it, &out->timestamp);
+        std::vector<uint8_t> text;
+        if (!r.byte_list(it + idw, &text)) return false;
+        if (!text.empty() && text.back() == 0) text.pop_back();
+        out->context.assign(text.begin(), text.end());
+        break;
+    }
+    case DIRECT: {
+        size_t it; uint16_t idw, ipw;
+        if (!r.struct_ptr(up, &it, &idw, &ipw) || ipw < 2) return false;
+        if (!r.byte_list(it + idw, &out->recipient)) return false;
+        if (!r.byte_list(it + idw + 1, &out->payload)) return false;
+        break;
+    }
+    case BROADCAST: {
+        size_t it; uint16_t idw, ipw;
+        if (!r.struct_ptr(up, &it, &idw, &ipw) || ipw < 2) return false;
+        if (!r.byte_list(it + idw, &out->topics)) return false;
+        if (!r.byte_list(it + idw + 1, &out->payload)) return false;
+        break;
+    }
+    case SUBSCRIBE:
+    case UNSUBSCRIBE:
+        if (!r.byte_list(up, &out->topics)) return false;
+        break;
+    case USER_SYNC:
+    case TOPIC_SYNC:
+        if (!r.byte_list(up, &out->payload)) return false;
+        break;
+    default:
+        return false;
+    }
+    out->disc = disc;
+    return true;
+}
+
+}  // namespace wire

@@ -158,8 +158,84 @@ class _SegmentBuilder:
         self.list_ptr(ptr_word, tgt, 2, len(raw))
 
 
+_core = None
+_core_failed = False
+
+
+def _get_core():
+    """Lazy-load the native C++ wire codec (pushcdn_core); None if unavailable."""
+    global _core, _core_failed
+    if _core is None and not _core_failed:
+        try:
+            from ..ops.build import build_core
+
+            _core = build_core()
+        except Exception:
+            _core_failed = True
+    return _core
+
+
 def serialize(msg: Message) -> bytes:
-    """Serialize to the capnp stream format (single segment).
+    """Serialize via the native C++ codec when available (csrc/wire/message.h,
+    byte-identical to serialize_py — cross-checked in tests/test_wire.py)."""
+    core = _get_core()
+    if core is None:
+        return serialize_py(msg)
+    t = type(msg)
+    if t is Broadcast:
+        return core.wire_serialize_broadcast(bytes(bytearray(x & 0xFF for x in msg.topics)),
+                                             bytes(msg.message))
+    if t is Direct:
+        return core.wire_serialize_direct(bytes(msg.recipient), bytes(msg.message))
+    if t is AuthenticateWithKey:
+        return core.wire_serialize_authenticate_with_key(
+            bytes(msg.public_key), msg.timestamp & 0xFFFFFFFFFFFFFFFF, bytes(msg.signature))
+    if t is AuthenticateWithPermit:
+        return core.wire_serialize_authenticate_with_permit(msg.permit & 0xFFFFFFFFFFFFFFFF)
+    if t is AuthenticateResponse:
+        return core.wire_serialize_authenticate_response(
+            msg.permit & 0xFFFFFFFFFFFFFFFF, msg.context)
+    if t is Subscribe:
+        return core.wire_serialize_topics(5, bytes(bytearray(x & 0xFF for x in msg.topics)))
+    if t is Unsubscribe:
+        return core.wire_serialize_topics(6, bytes(bytearray(x & 0xFF for x in msg.topics)))
+    if t is UserSync:
+        return core.wire_serialize_sync(7, bytes(msg.data))
+    if t is TopicSync:
+        return core.wire_serialize_sync(8, bytes(msg.data))
+    raise SerializeError(f"unknown message type {t!r}")
+
+
+def deserialize(data: bytes) -> Message:
+    """Parse via the native C++ codec when available."""
+    core = _get_core()
+    if core is None:
+        return deserialize_py(data)
+    d = core.wire_deserialize(bytes(data))
+    if d is None:
+        raise DeserializeError("malformed message")
+    disc = d["disc"]
+    if disc == 0:
+        return AuthenticateWithKey(d["public_key"], d["timestamp"], d["signature"])
+    if disc == 1:
+        return AuthenticateWithPermit(d["timestamp"])
+    if disc == 2:
+        return AuthenticateResponse(d["timestamp"], d["context"])
+    if disc == 3:
+        return Direct(d["recipient"], d["payload"])
+    if disc == 4:
+        return Broadcast(list(d["topics"]), d["payload"])
+    if disc == 5:
+        return Subscribe(list(d["topics"]))
+    if disc == 6:
+        return Unsubscribe(list(d["topics"]))
+    if disc == 7:
+        return UserSync(d["payload"])
+    return TopicSync(d["payload"])
+
+
+def serialize_py(msg: Message) -> bytes:
+    """Pure-Python reference serializer (single capnp segment).
 
     Mirrors reference ``Message::serialize`` (message.rs:116-204): stream
     header ``[u32 segcount-1 = 0][u32 nwords]`` then the segment.
@@ -268,8 +344,8 @@ class _SegmentReader:
             raise DeserializeError(f"invalid utf-8 in Text: {e}") from e
 
 
-def deserialize(data: bytes) -> Message:
-    """Parse the capnp stream format back into a ``Message``.
+def deserialize_py(data: bytes) -> Message:
+    """Pure-Python reference parser.
 
     Mirrors reference ``Message::deserialize`` (message.rs:212-312); the
     traversal limit there equals the buffer length, which bounds work the same

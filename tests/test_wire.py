@@ -101,3 +101,30 @@ def test_large_payload_roundtrip():
 def test_max_message_size_constant():
     # reference cdn-proto/src/lib.rs:25
     assert m.MAX_MESSAGE_SIZE == (2**32 - 1) // 8
+
+
+def test_cpp_codec_byte_identical_to_python():
+    """The native C++ wire codec (csrc/wire/message.h) must produce the
+    exact bytes of the Python reference serializer, and parse them back."""
+    from pushcdn_amd.proto.message import _get_core, serialize_py, deserialize_py
+
+    core = _get_core()
+    assert core is not None, "native codec must build in this environment"
+    for msg in ALL_MESSAGES:
+        py_bytes = serialize_py(msg)
+        cpp_bytes = m.serialize(msg)
+        assert cpp_bytes == py_bytes, type(msg).__name__
+        assert m.deserialize(cpp_bytes) == msg
+        assert deserialize_py(cpp_bytes) == msg
+
+
+def test_cpp_codec_rejects_garbage():
+    from pushcdn_amd.proto.message import _get_core
+
+    core = _get_core()
+    assert core.wire_deserialize(b"") is None
+    assert core.wire_deserialize(b"\xff" * 64) is None
+    data = m.serialize(m.Broadcast([1], b"x" * 50))
+    assert core.wire_deserialize(bytes(data)) is not None
+    for cut in (9, 17, len(data) - 8):
+        assert core.wire_deserialize(bytes(data[:cut])) is None
