@@ -225,3 +225,15 @@ def test_two_broker_mesh_broadcast_and_direct(tmp_path):
         await stop_stack([b1, b2], marshal, alice, bob)
 
     run(go())
+
+
+def test_cli_parser_smoke():
+    """CLI surface exists with the reference's flag set (no daemon start)."""
+    from pushcdn_amd import cli
+
+    p_err = None
+    try:
+        cli.main(["broker", "--help"])
+    except SystemExit as e:
+        p_err = e.code
+    assert p_err == 0
