@@ -1,0 +1,133 @@
+"""Tier-3 deterministic broker routing tests via the injection harness
+(reference cdn-broker/src/tests/{broadcast,direct}.rs scenarios):
+  - broadcast from a user: delivered to subscribed users + interested
+    brokers, not to unsubscribed ones, no duplicates
+  - broadcast from a broker: delivered to local users ONLY (1-hop mesh)
+  - direct to self / same-broker user / remote user / from-broker no-bounce
+"""
+
+import asyncio
+
+import pytest
+
+from pushcdn_amd.broker.testing import (
+    TestBroker,
+    TestDefinition,
+    TestUser,
+    assert_not_received,
+    assert_received,
+    at_index,
+)
+from pushcdn_amd.proto import message as m
+
+
+def run(coro):
+    return asyncio.run(asyncio.wait_for(coro, timeout=30))
+
+
+def test_broadcast_from_user():
+    async def go():
+        td = TestDefinition(
+            connected_users=[TestUser(topics=[0]), TestUser(topics=[0]), TestUser(topics=[1])],
+            connected_brokers=[TestBroker(connected_users=[], topics=[0]),
+                               TestBroker(connected_users=[], topics=[1])],
+        )
+        tr = await td.into_run()
+        msg = m.Broadcast([0], b"to-topic-0")
+        await tr.users[0].send_message(msg)
+        # subscribed users receive (including the sender — reference echoes
+        # to the sender if subscribed)
+        await assert_received(tr.users[0], msg, 1)
+        await assert_received(tr.users[1], msg, 1)
+        await assert_not_received(tr.users[2])
+        # interested broker 0 receives, broker 1 does not
+        await assert_received(tr.brokers[0], msg, 1)
+        await assert_not_received(tr.brokers[1])
+        await tr.close()
+
+    run(go())
+
+
+def test_broadcast_from_broker_is_single_hop():
+    async def go():
+        td = TestDefinition(
+            connected_users=[TestUser(topics=[0])],
+            connected_brokers=[TestBroker(connected_users=[], topics=[0]),
+                               TestBroker(connected_users=[], topics=[0])],
+        )
+        tr = await td.into_run()
+        msg = m.Broadcast([0], b"from-peer")
+        await tr.brokers[0].send_message(msg)
+        # local user gets it; the OTHER broker must NOT (no re-forwarding)
+        await assert_received(tr.users[0], msg, 1)
+        await assert_not_received(tr.brokers[1])
+        await assert_not_received(tr.brokers[0])  # no echo either
+        await tr.close()
+
+    run(go())
+
+
+def test_direct_local_and_self():
+    async def go():
+        td = TestDefinition(connected_users=[TestUser(topics=[]), TestUser(topics=[])])
+        tr = await td.into_run()
+        to_self = m.Direct(at_index(0), b"to-self")
+        await tr.users[0].send_message(to_self)
+        await assert_received(tr.users[0], to_self, 1)
+        to_other = m.Direct(at_index(1), b"to-other")
+        await tr.users[0].send_message(to_other)
+        await assert_received(tr.users[1], to_other, 1)
+        await assert_not_received(tr.users[0])
+        await tr.close()
+
+    run(go())
+
+
+def test_direct_to_remote_user_forwards_to_owner():
+    async def go():
+        td = TestDefinition(
+            connected_users=[TestUser(topics=[])],
+            connected_brokers=[TestBroker(connected_users=[7]),
+                               TestBroker(connected_users=[8])],
+        )
+        tr = await td.into_run()
+        msg = m.Direct(at_index(7), b"cross")
+        await tr.users[0].send_message(msg)
+        # forwarded to the owning broker only
+        await assert_received(tr.brokers[0], msg, 1)
+        await assert_not_received(tr.brokers[1])
+        await tr.close()
+
+    run(go())
+
+
+def test_direct_from_broker_no_bounce():
+    async def go():
+        td = TestDefinition(
+            connected_users=[TestUser(topics=[])],
+            connected_brokers=[TestBroker(connected_users=[5])],
+        )
+        tr = await td.into_run()
+        # a peer broker sends a direct for a user owned by ANOTHER broker:
+        # with to_user_only semantics it must not bounce back out
+        msg = m.Direct(at_index(5), b"bounce?")
+        await tr.brokers[0].send_message(msg)
+        await assert_not_received(tr.brokers[0])
+        # but a direct for OUR local user is delivered
+        msg2 = m.Direct(at_index(0), b"deliver")
+        await tr.brokers[0].send_message(msg2)
+        await assert_received(tr.users[0], msg2, 1)
+        await tr.close()
+
+    run(go())
+
+
+def test_direct_unknown_user_dropped():
+    async def go():
+        td = TestDefinition(connected_users=[TestUser(topics=[])])
+        tr = await td.into_run()
+        await tr.users[0].send_message(m.Direct(b"who-is-this", b"x"))
+        await assert_not_received(tr.users[0])
+        await tr.close()
+
+    run(go())
