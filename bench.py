@@ -45,6 +45,27 @@ import torch
 from pushcdn_amd.proto import message as msglib
 
 
+def build_mixed_batch(n_msgs, payload_bytes, n_topics, n_users, seed):
+    """Alternating Direct/Broadcast batch (config 4). Non-uniform wire sizes
+    -> the wave-per-pair K3 path. Message starts stay 16-aligned."""
+    import random
+
+    rng = random.Random(seed)
+    buf = bytearray()
+    offsets = [0]
+    for i in range(n_msgs):
+        payload = bytes(rng.randrange(256) for _ in range(payload_bytes))
+        if i % 2 == 0:
+            msg = msglib.Broadcast([rng.randrange(n_topics)], payload)
+        else:
+            msg = msglib.Direct(f"user-{rng.randrange(n_users)}".encode(), payload)
+        raw = msglib.serialize(msg)
+        padded = (len(raw) + 15) & ~15
+        buf += raw + b"\x00" * (padded - len(raw))
+        offsets.append(len(buf))
+    return bytes(buf), offsets, None
+
+
 def build_batch(n_msgs: int, payload_bytes: int, n_topics: int, seed: int):
     """Serialize one ingest batch; returns (buf, offsets, wire_len).  Every
     message start is 16-byte aligned (so the flat K3 kernel's 16 B units stay
@@ -78,7 +99,14 @@ def main() -> None:
     p.add_argument("--subscribers", type=int, default=10_000, help="total subscriber population")
     p.add_argument("--topics", type=int, default=8)
     p.add_argument("--device", default=None, help="cpu to force the CPU reference path")
+    p.add_argument("--mode", choices=["broadcast", "mixed"], default="broadcast",
+                   help="broadcast: every subscriber gets every message (configs 2-3); "
+                        "mixed: 50%% direct + 50%% broadcast, users spread over topics "
+                        "(config 4: 64 KiB payloads at -100k clients)")
+    p.add_argument("--ring-kb", type=int, default=0, help="override per-user ring size (KiB)")
     args = p.parse_args()
+    if args.mode == "mixed" and args.payload == 1024:
+        args.payload = 65536  # config-4 default
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -114,24 +142,43 @@ def main() -> None:
     while ring_bytes < need * 2:
         ring_bytes <<= 1
 
+    if args.mode == "mixed":
+        # expected per-user bytes/tick is small (users spread over topics);
+        # ring sized for bursts
+        ring_bytes = (args.ring_kb or 1024) << 10
+        pair_cap = 1 << 22
+    else:
+        pair_cap = max(1 << 20, world_size * args.batch * n_local_users)
+        if args.ring_kb:
+            ring_bytes = args.ring_kb << 10
     eng = GpuBrokerEngine(
         device=device,
         n_users=n_local_users,
         ring_bytes=ring_bytes,
         use_gpu_ops=not use_cpu,
         fanout_wire=True,       # forward raw wire bytes verbatim (reference semantics)
-        direct_enabled=False,   # broadcast-only benchmark (BASELINE configs 2-3)
-        pair_capacity=max(1 << 20, world_size * args.batch * n_local_users),
+        direct_enabled=(args.mode == "mixed"),
+        pair_capacity=pair_cap,
+        direct_table_size=1 << max(10, (n_local_users * 2).bit_length()),
     )
-    eng.subscribe_all(list(range(args.topics)))
+    if args.mode == "mixed":
+        eng.subscribe_modulo(args.topics)
+        eng.register_direct_bulk(
+            (f"user-{u}".encode(), u) for u in range(n_local_users)
+        )
+    else:
+        eng.subscribe_all(list(range(args.topics)))
 
     # Pre-serialize a few distinct wire batches (client-side work in the real
     # system); the H2D copy + full GPU pipeline still runs fresh every step.
     n_variants = 4
-    host_batches = [
-        build_batch(args.batch, args.payload, args.topics, seed=rank * 1000 + v)
-        for v in range(n_variants)
-    ]
+    builder = (
+        (lambda **kw: build_mixed_batch(args.batch, args.payload, args.topics,
+                                        n_local_users, kw["seed"]))
+        if args.mode == "mixed"
+        else (lambda **kw: build_batch(args.batch, args.payload, args.topics, kw["seed"]))
+    )
+    host_batches = [builder(seed=rank * 1000 + v) for v in range(n_variants)]
     wire_len = host_batches[0][2]
     cap = max(len(b) for b, _, _ in host_batches)
     cap = (cap + 255) & ~255
@@ -206,11 +253,19 @@ def main() -> None:
     msgs_per_sec = total_msgs / elapsed
     ms_per_step = elapsed / args.steps * 1000
     p50_ms = statistics.median(step_times) * 1000
-    deliveries_per_step = world_size * args.batch * n_local_users
+    if args.mode == "mixed":
+        # ~half the messages are broadcasts to n_local/topics subscribers,
+        # half are directs to one user
+        deliveries_per_step = world_size * (
+            (args.batch // 2) * (n_local_users // max(1, args.topics)) + args.batch // 2
+        )
+    else:
+        deliveries_per_step = world_size * args.batch * n_local_users
 
     if rank == 0:
         result = {
-            "metric": "broadcast_msgs_per_sec",
+            "metric": "broadcast_msgs_per_sec" if args.mode == "broadcast"
+                      else "mixed_direct_broadcast_msgs_per_sec",
             "value": msgs_per_sec,
             "unit": "msgs/s",
             "n_gpus": world_size,
@@ -232,6 +287,9 @@ def main() -> None:
                 "deliveries_per_step_node": deliveries_per_step,
                 "deliveries_per_sec_node": deliveries_per_step * args.steps / elapsed,
                 "p50_e2e_latency_ms": p50_ms,
+                "mode": args.mode,
+                "drops": int(eng._drops.cpu()[0]) if not use_cpu else 0,
+                "egress_hbm_gb": round(n_local_users * ring_bytes / 2**30, 1),
                 "parallelism": f"mesh{world_size} (RCCL all-gather over xGMI)" if world_size > 1 else "single-broker",
                 "global_batch": world_size * args.batch,
                 "seq_len": args.payload,

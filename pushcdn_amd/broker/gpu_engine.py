@@ -131,6 +131,27 @@ class GpuBrokerEngine:
         self._direct_entries[h] = owner
         self._rebuild_direct_table()
 
+    def register_direct_bulk(self, entries) -> None:
+        """Register many (pubkey, owner) pairs with one table rebuild."""
+        for pubkey, owner in entries:
+            self._direct_entries[fnv1a64(pubkey)] = owner
+        self._rebuild_direct_table()
+
+    def subscribe_modulo(self, n_topics: int) -> None:
+        """Bulk: user u subscribes to topic (u % n_topics) — the mixed-bench
+        population shape. Vectorized bitmap build on host, one H2D copy."""
+        import numpy as np
+
+        bitmap = np.zeros((256, self.W), dtype=np.uint64)
+        users = np.arange(self.n_users)
+        for t in range(n_topics):
+            sel = users[users % n_topics == t]
+            words = sel >> 6
+            bits = np.zeros(self.W, dtype=np.uint64)
+            np.bitwise_or.at(bits, words, np.uint64(1) << (sel & 63).astype(np.uint64))
+            bitmap[t] = bits
+        self.sub_bitmap.copy_(torch.from_numpy(bitmap.view(np.int64)).to(self.device))
+
     def _rebuild_direct_table(self) -> None:
         from ..ops.reference import build_direct_table
 
