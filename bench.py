@@ -54,7 +54,7 @@ def build_mixed_batch(n_msgs, payload_bytes, n_topics, n_users, seed):
     buf = bytearray()
     offsets = [0]
     for i in range(n_msgs):
-        payload = bytes(rng.randrange(256) for _ in range(payload_bytes))
+        payload = rng.randbytes(payload_bytes)
         if i % 2 == 0:
             msg = msglib.Broadcast([rng.randrange(n_topics)], payload)
         else:
@@ -77,7 +77,7 @@ def build_batch(n_msgs: int, payload_bytes: int, n_topics: int, seed: int):
     offsets = [0]
     wire_len = None
     for i in range(n_msgs):
-        payload = bytes(rng.randrange(256) for _ in range(payload_bytes))
+        payload = rng.randbytes(payload_bytes)
         msg = msglib.Broadcast([i % n_topics], payload)
         raw = msglib.serialize(msg)
         padded = (len(raw) + 15) & ~15
