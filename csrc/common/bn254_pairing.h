@@ -119,6 +119,28 @@ struct Fp12 {
         return mul(f, l);
     }
 
+    // Frobenius^1: conjugate each Fp2 coefficient, multiply by gamma1 factors
+    // (v^p = v * GAMMA1_2, w^p = w * GAMMA1_1; see scripts/gen_bn254_constants.py)
+    BN_BIGFUNC static Fp12 frobenius1(const Fp12& a) {
+        Fp2 g1{Fp::from_u256(from_limbs(bn254c::GAMMA1_1_C0)),
+               Fp::from_u256(from_limbs(bn254c::GAMMA1_1_C1))};
+        Fp2 g2{Fp::from_u256(from_limbs(bn254c::GAMMA1_2_C0)),
+               Fp::from_u256(from_limbs(bn254c::GAMMA1_2_C1))};
+        Fp2 g3{Fp::from_u256(from_limbs(bn254c::GAMMA1_3_C0)),
+               Fp::from_u256(from_limbs(bn254c::GAMMA1_3_C1))};
+        Fp2 g4{Fp::from_u256(from_limbs(bn254c::GAMMA1_4_C0)),
+               Fp::from_u256(from_limbs(bn254c::GAMMA1_4_C1))};
+        Fp2 g5{Fp::from_u256(from_limbs(bn254c::GAMMA1_5_C0)),
+               Fp::from_u256(from_limbs(bn254c::GAMMA1_5_C1))};
+        return {{Fp2::conj(a.c0.c0), Fp2::mul(Fp2::conj(a.c0.c1), g2),
+                 Fp2::mul(Fp2::conj(a.c0.c2), g4)},
+                {Fp2::mul(Fp2::conj(a.c1.c0), g1), Fp2::mul(Fp2::conj(a.c1.c1), g3),
+                 Fp2::mul(Fp2::conj(a.c1.c2), g5)}};
+    }
+
+    // Frobenius^3 = Frobenius^1 o Frobenius^2
+    BN_INLINE static Fp12 frobenius3(const Fp12& a) { return frobenius1(frobenius2(a)); }
+
     // Frobenius^2: c_ij -> c_ij * gamma2 factors (Fp scalars, no conjugation)
     BN_BIGFUNC static Fp12 frobenius2(const Fp12& a) {
         Fp g1 = Fp::from_u256(from_limbs(bn254c::GAMMA2_1));
@@ -249,16 +271,67 @@ BN_BIGFUNC Fp12 miller_loop(const Fp& px, const Fp& py, const G2Affine& q) {
 
 // ---------------------------------------------------------------------------
 // Final exponentiation: f^((p^12-1)/r)
-// easy part structured; hard part = generic pow by (p^4-p^2+1)/r
+// easy part structured; hard part via the Fuentes-Castaneda x-chain (the
+// standard BN chain, ~300 Fp12 ops vs ~1140 for the generic 762-bit pow —
+// hard_exponentiation_generic is kept and cross-checked in tests).
 // ---------------------------------------------------------------------------
-BN_BIGFUNC Fp12 final_exponentiation(const Fp12& f) {
-    // easy part: f^(p^6 - 1) = conj(f) * f^-1 ; then ^(p^2 + 1)
+BN_INLINE Fp12 easy_part(const Fp12& f) {
     Fp12 f1 = Fp12::conj(f);
     Fp12 f2 = f.inv();
-    Fp12 r = Fp12::mul(f1, f2);          // f^(p^6 - 1)
-    r = Fp12::mul(Fp12::frobenius2(r), r);  // ^(p^2 + 1)
-    // hard part
+    Fp12 r = Fp12::mul(f1, f2);             // f^(p^6 - 1)
+    return Fp12::mul(Fp12::frobenius2(r), r);  // ^(p^2 + 1)
+}
+
+BN_BIGFUNC Fp12 hard_exponentiation_generic(const Fp12& r) {
     return Fp12::pow_limbs(r, bn254c::HARD_EXP, bn254c::HARD_EXP_LIMBS);
+}
+
+// f^x for the 64-bit BN parameter x (element is in the cyclotomic subgroup
+// after the easy part, so conj(f) == f^-1; plain squaring kept for safety)
+BN_BIGFUNC Fp12 pow_by_x(const Fp12& a) {
+    uint64_t e = bn254c::BN_X;
+    Fp12 result = Fp12::one();
+    bool started = false;
+    BN_NOUNROLL for (int b = 63; b >= 0; --b) {
+        if (started) result = Fp12::sqr(result);
+        if ((e >> b) & 1) {
+            if (started) result = Fp12::mul(result, a);
+            else { result = a; started = true; }
+        }
+    }
+    return result;
+}
+
+BN_BIGFUNC Fp12 hard_exponentiation_chain(const Fp12& r_in) {
+    // arkworks models/bn/mod.rs shape; BN254's x is positive, so
+    // exp_by_neg_x(f) = conj(f^x).
+    Fp12 r = r_in;
+    Fp12 y0 = Fp12::conj(pow_by_x(r));
+    Fp12 y1 = Fp12::sqr(y0);
+    Fp12 y2 = Fp12::sqr(y1);
+    Fp12 y3 = Fp12::mul(y2, y1);
+    Fp12 y4 = Fp12::conj(pow_by_x(y3));
+    Fp12 y5 = Fp12::sqr(y4);
+    Fp12 y6 = Fp12::conj(pow_by_x(y5));
+    y3 = Fp12::conj(y3);
+    y6 = Fp12::conj(y6);
+    Fp12 y7 = Fp12::mul(y6, y4);
+    Fp12 y8 = Fp12::mul(y7, y3);
+    Fp12 y9 = Fp12::mul(y8, y1);
+    Fp12 y10 = Fp12::mul(y8, y4);
+    Fp12 y11 = Fp12::mul(y10, r);
+    Fp12 y12 = Fp12::frobenius1(y9);
+    Fp12 y13 = Fp12::mul(y12, y11);
+    y8 = Fp12::frobenius2(y8);
+    Fp12 y14 = Fp12::mul(y8, y13);
+    r = Fp12::conj(r);
+    Fp12 y15 = Fp12::mul(r, y9);
+    y15 = Fp12::frobenius3(y15);
+    return Fp12::mul(y15, y14);
+}
+
+BN_BIGFUNC Fp12 final_exponentiation(const Fp12& f) {
+    return hard_exponentiation_chain(easy_part(f));
 }
 
 // full pairing e(P, Q); P affine G1, Q affine G2

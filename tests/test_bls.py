@@ -122,3 +122,10 @@ def test_timestamp_helpers():
     sig = bls.sign_timestamp(kp.private_key, bls.USER_MARSHAL_NAMESPACE, ts)
     assert bls.verify_timestamp(kp.public_key, bls.USER_MARSHAL_NAMESPACE, ts, sig)
     assert not bls.verify_timestamp(kp.public_key, bls.USER_MARSHAL_NAMESPACE, ts + 1, sig)
+
+
+def test_hard_exponentiation_chain_matches_generic(core):
+    """The Fuentes-Castaneda x-chain must equal generic^c (c = the F-C
+    multiple, verified symbolically in scripts/gen_bn254_constants.py lineage)."""
+    for a, b in [(1, 1), (3, 5), (123456789, 987654321)]:
+        assert core._hard_exp_chain_ok(a, b)
