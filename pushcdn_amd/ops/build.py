@@ -36,6 +36,15 @@ def build_gpu(verbose: bool = False):
 
     BUILD_DIR.mkdir(exist_ok=True)
     sources = [str(s) for s in GPU_SOURCES if s.exists()]
+    # ninja's rules for hipcc lack header depfiles: if any shared header is
+    # newer than a built object, touch the sources so the rebuild triggers.
+    headers = list((CSRC / "common").glob("*.h")) + list((CSRC / "bls").glob("*.h")) \
+        + list((CSRC / "wire").glob("*.h"))
+    if headers:
+        newest_h = max(h.stat().st_mtime for h in headers)
+        for src in GPU_SOURCES:
+            if src.exists() and src.stat().st_mtime < newest_h:
+                os.utime(src)
     mod = load(
         name="pushcdn_gpu",
         sources=sources,
