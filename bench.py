@@ -46,8 +46,10 @@ from pushcdn_amd.proto import message as msglib
 
 
 def build_mixed_batch(n_msgs, payload_bytes, n_topics, n_users, seed):
-    """Alternating Direct/Broadcast batch (config 4). Non-uniform wire sizes
-    -> the wave-per-pair K3 path. Message starts stay 16-aligned."""
+    """Alternating Direct/Broadcast batch (config 4). Per-position wire sizes
+    are deterministic (fixed-width recipient keys), so the offsets table is
+    identical on every rank and one device offsets tensor serves every
+    mesh-exchanged batch. Non-uniform across positions -> wave-per-pair K3."""
     import random
 
     rng = random.Random(seed)
@@ -58,7 +60,7 @@ def build_mixed_batch(n_msgs, payload_bytes, n_topics, n_users, seed):
         if i % 2 == 0:
             msg = msglib.Broadcast([rng.randrange(n_topics)], payload)
         else:
-            msg = msglib.Direct(f"user-{rng.randrange(n_users)}".encode(), payload)
+            msg = msglib.Direct(f"user-{rng.randrange(n_users):08d}".encode(), payload)
         raw = msglib.serialize(msg)
         padded = (len(raw) + 15) & ~15
         buf += raw + b"\x00" * (padded - len(raw))
@@ -117,14 +119,6 @@ def main() -> None:
         print("ERROR: no GPU available and --device cpu not requested", file=sys.stderr)
         sys.exit(1)
 
-    dist = None
-    if world_size > 1:
-        import torch.distributed as dist_mod
-
-        dist = dist_mod
-        backend = "gloo" if use_cpu else "nccl"
-        dist.init_process_group(backend=backend)
-
     if use_cpu:
         device = "cpu"
     else:
@@ -164,7 +158,7 @@ def main() -> None:
     if args.mode == "mixed":
         eng.subscribe_modulo(args.topics)
         eng.register_direct_bulk(
-            (f"user-{u}".encode(), u) for u in range(n_local_users)
+            (f"user-{u:08d}".encode(), u) for u in range(n_local_users)
         )
     else:
         eng.subscribe_all(list(range(args.topics)))
@@ -192,36 +186,27 @@ def main() -> None:
         pinned.append(t)
 
     dev_offsets = offsets_t.to(device)
-    gathered = None
-    if world_size > 1:
-        gathered = torch.zeros(world_size * cap, dtype=torch.uint8, device=device)
+    from pushcdn_amd.parallel.mesh import RcclMesh
+
+    mesh = RcclMesh(torch.device(device), batch_capacity=cap)
 
     def step(i: int) -> None:
-        src = pinned[i % n_variants]
+        v = i % n_variants
+        src = pinned[v]
         buf = src.to(device, non_blocking=True)
-        if dist is not None:
-            dist.all_gather_into_tensor(gathered, buf)
-            for r in range(world_size):
-                eng.tick(
-                    gathered[r * cap : (r + 1) * cap],
-                    dev_offsets,
-                    host_batch=None if not use_cpu else bytes(src[: len(host_batches[i % n_variants][0])].numpy().tobytes()),
-                    host_offsets=None if not use_cpu else host_batches[i % n_variants][1],
-                    uniform_wire_len=wire_len,
-                )
-        else:
+        # broker->broker mesh: all-gather this tick's batches over xGMI
+        for r, view, n_msgs, nbytes in mesh.exchange(buf, args.batch, len(host_batches[v][0])):
             eng.tick(
-                buf,
+                view,
                 dev_offsets,
-                host_batch=None if not use_cpu else host_batches[i % n_variants][0],
-                host_offsets=None if not use_cpu else host_batches[i % n_variants][1],
+                host_batch=None if not use_cpu else host_batches[v][0],
+                host_offsets=None if not use_cpu else host_batches[v][1],
                 uniform_wire_len=wire_len,
             )
         eng.drain_cursors()
 
     def barrier_sync() -> None:
-        if dist is not None:
-            dist.barrier()
+        mesh.barrier()
         if not use_cpu:
             torch.cuda.synchronize()
 
@@ -242,12 +227,7 @@ def main() -> None:
     elapsed = t1 - t0
 
     # MAX elapsed over ranks (the slowest rank defines the job)
-    if dist is not None:
-        et = torch.tensor([elapsed], dtype=torch.float64)
-        if not use_cpu:
-            et = et.to(device)
-        dist.all_reduce(et, op=dist.ReduceOp.MAX)
-        elapsed = float(et.cpu()[0])
+    elapsed = mesh.max_over_ranks(elapsed)
 
     total_msgs = world_size * args.batch * args.steps
     msgs_per_sec = total_msgs / elapsed
@@ -297,8 +277,8 @@ def main() -> None:
         }
         print(json.dumps(result))
 
-    if dist is not None:
-        dist.destroy_process_group()
+    if mesh.enabled:
+        mesh.dist.destroy_process_group()
 
 
 if __name__ == "__main__":

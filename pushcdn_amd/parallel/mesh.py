@@ -1,0 +1,91 @@
+"""RCCL broker mesh over xGMI (one process per GPU via torch.distributed).
+
+Replaces the reference's broker↔broker TCP fan-out (try_send_to_brokers,
+broker/sender.rs:49-58) with collective exchange on the 8-GPU communicator:
+each tick, every broker contributes its ingest batch and receives every
+peer's batch (the 1-hop mesh: remote batches are processed with
+to_users_only semantics, exactly like the reference's
+broker_receive_loop -> handle_broadcast_message(..., to_users_only=true)).
+
+On MI355X the 7 xGMI links are point-to-point (~153 GB/s each), so the
+all-gather of B-byte batches moves (N-1)*B per GPU spread across links —
+for the bench shapes (≤16 MiB per tick) this is far from link-bound.
+
+Membership changes (broker kill/rejoin) require communicator rebuild; the
+driver of this process group (torchrun / the service supervisor) owns that —
+`rebuild()` tears down and re-inits from the environment.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional, Tuple
+
+import torch
+
+
+class RcclMesh:
+    def __init__(self, device: torch.device, batch_capacity: int) -> None:
+        import torch.distributed as dist
+
+        self.dist = dist
+        self.device = device
+        self.capacity = batch_capacity
+        self.world_size = int(os.environ.get("WORLD_SIZE", "1"))
+        self.rank = int(os.environ.get("RANK", "0"))
+        self.enabled = self.world_size > 1
+        if self.enabled and not dist.is_initialized():
+            backend = "nccl" if device.type == "cuda" else "gloo"
+            dist.init_process_group(backend=backend)
+        if self.enabled:
+            self._gathered = torch.zeros(
+                self.world_size * batch_capacity, dtype=torch.uint8, device=device
+            )
+            # per-rank metadata: [n_messages, batch_bytes]
+            self._meta = torch.zeros(self.world_size * 2, dtype=torch.int64, device=device)
+
+    def exchange(
+        self, batch: torch.Tensor, n_messages: int, batch_bytes: int
+    ) -> List[Tuple[int, torch.Tensor, int, int]]:
+        """All-gather this tick's batch. `batch` must be a device tensor of
+        exactly `capacity` bytes (zero-padded). Returns
+        [(rank, batch_view, n_messages, batch_bytes), ...] for every rank
+        (including self, so local and remote batches route identically)."""
+        if not self.enabled:
+            return [(0, batch, n_messages, batch_bytes)]
+        assert batch.numel() == self.capacity
+        meta_local = torch.tensor([n_messages, batch_bytes], dtype=torch.int64,
+                                  device=self.device)
+        self.dist.all_gather_into_tensor(self._meta, meta_local)
+        self.dist.all_gather_into_tensor(self._gathered, batch)
+        meta = self._meta.to("cpu")
+        out = []
+        for r in range(self.world_size):
+            out.append((
+                r,
+                self._gathered[r * self.capacity : (r + 1) * self.capacity],
+                int(meta[r * 2]),
+                int(meta[r * 2 + 1]),
+            ))
+        return out
+
+    def barrier(self) -> None:
+        if self.enabled:
+            self.dist.barrier()
+
+    def max_over_ranks(self, value: float) -> float:
+        if not self.enabled:
+            return value
+        t = torch.tensor([value], dtype=torch.float64)
+        if self.device.type == "cuda":
+            t = t.to(self.device)
+        self.dist.all_reduce(t, op=self.dist.ReduceOp.MAX)
+        return float(t.cpu()[0])
+
+    def rebuild(self) -> None:
+        """Communicator teardown/rebuild on membership change (the xGMI
+        analog of a TCP reconnect — SURVEY §5.3)."""
+        if self.dist.is_initialized():
+            self.dist.destroy_process_group()
+        backend = "nccl" if self.device.type == "cuda" else "gloo"
+        self.dist.init_process_group(backend=backend)
