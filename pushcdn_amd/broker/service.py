@@ -29,8 +29,11 @@ from ..proto.errors import ConnectionError_, TopicError
 from ..proto.limiter import Bytes, Limiter
 from ..proto.topic import TopicSpace, ALL_TOPICS
 from ..proto.transports.base import Connection, Protocol
+from ..utils.log import get_logger, ident
 from ..utils.mnemonic import mnemonic
 from .connections import Connections
+
+log = get_logger("broker")
 
 HEARTBEAT_INTERVAL_S = 10.0
 HEARTBEAT_EXPIRY_S = 60.0
@@ -221,6 +224,7 @@ class Broker:
     async def remove_user(self, pubkey: bytes) -> None:
         handle = self.connections.remove_user(pubkey)
         if handle is not None:
+            log.info("user disconnected: %s", ident(pubkey))
             if handle.task:
                 handle.task.cancel()
             handle.connection.close()
@@ -230,6 +234,7 @@ class Broker:
     async def remove_broker(self, broker: BrokerIdentifier) -> None:
         handle = self.connections.remove_broker(broker)
         if handle is not None:
+            log.info("broker disconnected: %s", broker)
             if handle.task:
                 handle.task.cancel()
             handle.connection.close()
@@ -260,6 +265,8 @@ class Broker:
             return
         handle = UserHandle(connection=connection)
         old = self.connections.add_user(pubkey, handle, topics)
+        log.info("user connected: %s topics=%s%s", ident(pubkey), topics,
+                 " (kicked old session)" if old else "")
         if old is not None:
             # duplicate key kicks the old session (connections/mod.rs:290-298)
             if old.task:
@@ -400,6 +407,8 @@ class Broker:
             return
         handle = BrokerHandle(connection=connection)
         old = self.connections.add_broker(peer_identity, handle)
+        log.info("broker connected: %s (outbound=%s)%s", peer_identity, is_outbound,
+                 " (replaced old connection)" if old else "")
         if old is not None:
             if old.task:
                 old.task.cancel()

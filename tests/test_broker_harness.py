@@ -131,3 +131,40 @@ def test_direct_unknown_user_dropped():
         await tr.close()
 
     run(go())
+
+
+def test_message_hooks_skip_and_disconnect():
+    """MessageHook extension point (reference def.rs:69-97): SkipMessage
+    drops the message; a non-process verdict disconnects the user."""
+    import asyncio as aio
+
+    from pushcdn_amd.broker.service import SKIP_MESSAGE
+
+    async def go():
+        seen = []
+
+        def hook(msg):
+            seen.append(type(msg).__name__)
+            if isinstance(msg, m.Broadcast) and msg.message == b"skip-me":
+                return SKIP_MESSAGE
+            if isinstance(msg, m.Broadcast) and msg.message == b"kill-me":
+                return "disconnect"
+            return "process"
+
+        td = TestDefinition(connected_users=[TestUser(topics=[0]), TestUser(topics=[0])])
+        tr = await td.into_run()
+        tr.broker.config.user_message_hook = hook
+
+        await tr.users[0].send_message(m.Broadcast([0], b"normal"))
+        await assert_received(tr.users[1], m.Broadcast([0], b"normal"), 1)
+
+        await tr.users[0].send_message(m.Broadcast([0], b"skip-me"))
+        await assert_not_received(tr.users[1])
+
+        await tr.users[0].send_message(m.Broadcast([0], b"kill-me"))
+        await aio.sleep(0.2)
+        assert len(tr.broker.connections.users) == 1  # sender disconnected
+        assert "Broadcast" in seen
+        await tr.close()
+
+    run(go())
