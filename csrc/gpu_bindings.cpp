@@ -45,6 +45,12 @@ void launch_k2b_emit_capped(const uint64_t*, const int64_t*, const int32_t*, con
 void launch_k3_fanout_flat(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
                            const int32_t*, const int64_t*, const uint32_t*, const int32_t*,
                            int32_t, uint8_t*, int, int, hipStream_t);
+void launch_k2b_fused(const uint64_t*, const int32_t*, int32_t, int32_t, int32_t, int64_t,
+                      int32_t, uint64_t*, int32_t*, int32_t*, int32_t*, int64_t*, uint32_t*,
+                      hipStream_t);
+void launch_k3_fanout_flat2(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
+                            const int32_t*, const int64_t*, uint32_t, const int32_t*, int32_t,
+                            int32_t, uint8_t*, int, int, hipStream_t);
 }
 
 #define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
@@ -247,6 +253,37 @@ void fanout_flat(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor pay
                           egress.data_ptr<uint8_t>(), (int)nt, (int)grid, cur_stream());
 }
 
+void assign_emit_fused(torch::Tensor mask, torch::Tensor payload_len, torch::Tensor ring_wpos,
+                       int64_t ring_bytes, int64_t n_users, torch::Tensor pair_user,
+                       torch::Tensor pair_msg, torch::Tensor pair_dst, torch::Tensor drops,
+                       torch::Tensor n_pairs) {
+    CHECK_DEV(mask); CHECK_CONTIG(mask);
+    int32_t M = (int32_t)mask.size(0);
+    int32_t W = (int32_t)mask.size(1);
+    TORCH_CHECK(ring_bytes % 16 == 0);
+    int32_t capacity = (int32_t)pair_user.size(0);
+    launch_k2b_fused((const uint64_t*)mask.data_ptr<int64_t>(), payload_len.data_ptr<int32_t>(),
+                     M, W, (int32_t)n_users, ring_bytes, capacity,
+                     (uint64_t*)ring_wpos.data_ptr<int64_t>(), n_pairs.data_ptr<int32_t>(),
+                     pair_user.data_ptr<int32_t>(), pair_msg.data_ptr<int32_t>(),
+                     pair_dst.data_ptr<int64_t>(), (uint32_t*)drops.data_ptr<int32_t>(),
+                     cur_stream());
+}
+
+void fanout_flat2(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_len,
+                  torch::Tensor pair_user, torch::Tensor pair_msg, torch::Tensor pair_dst,
+                  int64_t seq_base, torch::Tensor n_pairs, int64_t units_per_pair,
+                  torch::Tensor egress, int64_t nt, int64_t grid) {
+    CHECK_DEV(egress); CHECK_CONTIG(egress);
+    int32_t capacity = (int32_t)pair_user.size(0);
+    launch_k3_fanout_flat2(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
+                           payload_len.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
+                           pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
+                           (uint32_t)seq_base, n_pairs.data_ptr<int32_t>(), capacity,
+                           (int32_t)units_per_pair, egress.data_ptr<uint8_t>(), (int)nt,
+                           (int)grid, cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("parse_batch", &parse_batch, "K4: on-device capnp parse of a message batch");
     m.def("topic_mask", &topic_mask, "K2a: per-message subscriber mask");
@@ -260,4 +297,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "K2b sync-free: emit pairs into preallocated buffers, count stays on device");
     m.def("fanout_wave", &fanout_wave, "K3v2: wave-per-pair fan-out (nt flag, device count)");
     m.def("fanout_flat", &fanout_flat, "K3v3: flat-index fan-out for uniform record sizes");
+    m.def("assign_emit_fused", &assign_emit_fused,
+          "K2b-fused: count+claim+emit in one kernel (atomic slot claim)");
+    m.def("fanout_flat2", &fanout_flat2, "K3v4: flat fan-out, seq from base, capacity clamp");
 }

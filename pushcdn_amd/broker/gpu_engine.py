@@ -196,19 +196,22 @@ class GpuBrokerEngine:
         if self.fanout_wire:
             payload_off = offsets[:-1].contiguous()
             payload_len = (offsets[1:] - offsets[:-1]).to(torch.int32).contiguous()
-        # sync-free pipeline: pair count stays on device
-        ops.assign_emit_into(
-            mask, payload_off, payload_len, self.ring_wpos, self.ring_bytes, self.n_users,
+        # fused sync-free pipeline: one kernel counts + claims slots
+        # atomically + emits; the pair count stays on device
+        self._n_pairs.zero_()
+        ops.assign_emit_fused(
+            mask, payload_len, self.ring_wpos, self.ring_bytes, self.n_users,
             self._pair_user, self._pair_msg, self._pair_dst, self._drops, self._n_pairs,
         )
-        seq = torch.arange(self.seq, self.seq + M, dtype=torch.int32, device=self.device)
+        seq_base = self.seq
         self.seq += M
         nt = 1 if self.nt_fanout else 0
         if self.fanout_wire and uniform_wire_len is not None:
             units = 1 + ((uniform_wire_len + 15) & ~15) // 16
-            ops.fanout_flat(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
-                            self._pair_dst, seq, self._n_pairs, units, self.egress, nt, 0)
+            ops.fanout_flat2(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
+                             self._pair_dst, seq_base, self._n_pairs, units, self.egress, nt, 0)
         else:
+            seq = torch.arange(seq_base, seq_base + M, dtype=torch.int32, device=self.device)
             ops.fanout_wave(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
                             self._pair_dst, seq, self._n_pairs, self.egress, nt, 0)
         stats = TickStats(n_messages=M)
@@ -217,7 +220,8 @@ class GpuBrokerEngine:
         # workloads via direct_enabled=False.
         owner = ops.direct_lookup(self.direct_keys, self.direct_vals, recip_hash)
         if self.direct_enabled:
-            self._route_direct_gpu(buf, payload_off, payload_len, disc, owner, seq)
+            dseq = torch.arange(seq_base, seq_base + M, dtype=torch.int32, device=self.device)
+            self._route_direct_gpu(buf, payload_off, payload_len, disc, owner, dseq)
         return stats
 
     def _route_direct_gpu(self, buf, payload_off, payload_len, disc, owner, seq) -> None:

@@ -388,3 +388,49 @@ def test_fanout_flat_uniform_matches_reference(ops):
                         seq.to("cuda"), n_pairs, units, egress, nt, 0)
         torch.cuda.synchronize()
         assert egress.cpu().numpy().tobytes() == bytes(arr), f"nt={nt}"
+
+
+def test_fused_pipeline_rings_match_reference(ops):
+    """The fused K2b (atomic slot claim) + flat2 K3 must produce the SAME
+    ring contents as the reference (pair-list order differs — that's fine,
+    only per-user ring order is semantic)."""
+    rng = random.Random(23)
+    n_users = 200
+    ring_bytes = 1 << 16
+    W = (n_users + 63) // 64
+    sub = torch.zeros((256, W), dtype=torch.int64)
+    for u in range(n_users):
+        t = u % 5
+        sub[t, u >> 6] |= (1 << (u & 63)) - (1 << 64) if (u & 63) == 63 else 1 << (u & 63)
+    buf = bytearray()
+    offsets = [0]
+    for i in range(40):
+        raw = m.serialize(m.Broadcast([i % 5], bytes([i]) * 512))
+        padded = (len(raw) + 15) & ~15
+        wire_len = padded
+        buf += raw + b"\x00" * (padded - len(raw))
+        offsets.append(len(buf))
+    buf = bytes(buf)
+
+    # reference (wire mode)
+    woff = torch.tensor(offsets[:-1], dtype=torch.int64)
+    wlen = torch.full((40,), wire_len, dtype=torch.int32)
+    pr = ref.parse_batch(buf, offsets)
+    maskr = ref.topic_mask(sub, buf, pr.topics_off, pr.topics_cnt, pr.disc)
+    wposr = torch.zeros(n_users, dtype=torch.int64)
+    pu_r, pm_r, pd_r, _ = ref.assign_emit(maskr, wlen, wposr, ring_bytes, n_users)
+    arr = bytearray(n_users * ring_bytes)
+    seq = torch.arange(0, 40, dtype=torch.int32)
+    ref.fanout(buf, woff, wlen, pu_r, pm_r, pd_r, seq, arr)
+
+    # fused GPU path via the engine
+    from pushcdn_amd.broker.gpu_engine import GpuBrokerEngine
+
+    eng = GpuBrokerEngine(device="cuda:0", n_users=n_users, ring_bytes=ring_bytes,
+                          fanout_wire=True, direct_enabled=False, pair_capacity=1 << 16)
+    eng.sub_bitmap.copy_(sub.to("cuda"))
+    dbuf, doff = eng.ingest(buf, offsets)
+    eng.tick(dbuf, doff, uniform_wire_len=wire_len)
+    torch.cuda.synchronize()
+    assert torch.equal(eng.ring_wpos.cpu(), wposr)
+    assert eng.egress.cpu().numpy().tobytes() == bytes(arr)
