@@ -1,0 +1,10 @@
+#!/bin/bash
+# ASan+UBSan lane for the native host core (the C++ rewrite
+# replaces Rust's compile-time safety with sanitizer lanes).
+set -e
+cd "$(dirname "$0")/.."
+mkdir -p build
+g++ -O1 -g -std=c++17 -fsanitize=address,undefined -fno-omit-frame-pointer \
+    -I csrc csrc/tests/test_native.cpp -o build/test_native_asan
+./build/test_native_asan
+echo "sanitizer lane OK"

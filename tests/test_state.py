@@ -191,3 +191,15 @@ def test_embedded_discovery_semantics(tmp_path):
         assert id_(2) not in others
 
     asyncio.run(go())
+
+
+def test_native_sanitizer_lane():
+    """Build + run the C++ core under ASan/UBSan (SURVEY §5.2 safety lane)."""
+    import subprocess
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    out = subprocess.run(["bash", str(repo / "scripts" / "native_sanitize.sh")],
+                         capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "sanitizer lane OK" in out.stdout

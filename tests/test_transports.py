@@ -143,3 +143,28 @@ def test_tls_rejects_untrusted_ca(tmp_path):
         await listener.close()
 
     run(go())
+
+
+def test_metrics_endpoint():
+    """Prometheus /metrics endpoint parity (reference metrics.rs:18-39)."""
+    import urllib.request
+
+    from pushcdn_amd.utils.metrics import serve_metrics, BYTES_SENT
+
+    async def go():
+        server = await serve_metrics("127.0.0.1", 0)
+        port = server.sockets[0].getsockname()[1]
+        BYTES_SENT.inc(17)
+
+        def fetch():
+            with urllib.request.urlopen(f"http://127.0.0.1:{port}/metrics", timeout=5) as r:
+                return r.read().decode()
+
+        body = await asyncio.get_running_loop().run_in_executor(None, fetch)
+        for metric in ("total_bytes_sent", "total_bytes_recv", "running_latency",
+                       "num_users_connected", "num_brokers_connected"):
+            assert metric in body, metric
+        server.close()
+        await server.wait_closed()
+
+    run(go())
