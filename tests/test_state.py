@@ -174,8 +174,9 @@ def test_embedded_discovery_semantics(tmp_path):
         permit = await d1.issue_permit(id_(1), 30, b"userkey")
         assert permit > 1
         assert await d1.validate_permit(id_(2), permit) is None  # wrong broker
-        # permit was consumed by the failed validation? NO — wrong broker
-        # does not consume in our impl... validate consumed it. Re-issue:
+        # a validation attempt consumes the permit either way (GETDEL
+        # semantics, reference redis.rs:246-265) — re-issue for the
+        # happy-path check:
         permit = await d1.issue_permit(id_(1), 30, b"userkey")
         assert await d1.validate_permit(id_(1), permit) == b"userkey"
         assert await d1.validate_permit(id_(1), permit) is None  # one-shot
