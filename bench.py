@@ -106,6 +106,8 @@ def main() -> None:
                         "mixed: 50%% direct + 50%% broadcast, users spread over topics "
                         "(config 4: 64 KiB payloads at -100k clients)")
     p.add_argument("--ring-kb", type=int, default=0, help="override per-user ring size (KiB)")
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable hipGraph capture of the tick (A/B)")
     args = p.parse_args()
     if args.mode == "mixed" and args.payload == 1024:
         args.payload = 65536  # config-4 default
@@ -189,20 +191,31 @@ def main() -> None:
     from pushcdn_amd.parallel.mesh import RcclMesh
 
     mesh = RcclMesh(torch.device(device), batch_capacity=cap)
+    dev_buf = None if use_cpu else torch.zeros(cap, dtype=torch.uint8, device=device)
+    use_graph = (not use_cpu) and (not args.no_graph) and args.mode == "broadcast"
 
     def step(i: int) -> None:
         v = i % n_variants
         src = pinned[v]
-        buf = src.to(device, non_blocking=True)
+        if use_cpu:
+            buf = src
+        else:
+            dev_buf.copy_(src, non_blocking=True)  # fixed-address ingest buffer
+            buf = dev_buf
         # broker->broker mesh: all-gather this tick's batches over xGMI
         for r, view, n_msgs, nbytes in mesh.exchange(buf, args.batch, len(host_batches[v][0])):
-            eng.tick(
-                view,
-                dev_offsets,
-                host_batch=None if not use_cpu else host_batches[v][0],
-                host_offsets=None if not use_cpu else host_batches[v][1],
-                uniform_wire_len=wire_len,
-            )
+            if use_graph:
+                # hipGraph-captured tick (captured per fixed buffer; the
+                # gathered views and dev_buf are stable addresses)
+                eng.tick_graphed(view, dev_offsets, wire_len)
+            else:
+                eng.tick(
+                    view,
+                    dev_offsets,
+                    host_batch=None if not use_cpu else host_batches[v][0],
+                    host_offsets=None if not use_cpu else host_batches[v][1],
+                    uniform_wire_len=wire_len,
+                )
         eng.drain_cursors()
 
     def barrier_sync() -> None:

@@ -51,6 +51,10 @@ void launch_k2b_fused(const uint64_t*, const int32_t*, int32_t, int32_t, int32_t
 void launch_k3_fanout_flat2(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
                             const int32_t*, const int64_t*, uint32_t, const int32_t*, int32_t,
                             int32_t, uint8_t*, int, int, hipStream_t);
+void launch_k3_fanout_flat3(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
+                            const int32_t*, const int64_t*, const uint32_t*, const int32_t*,
+                            int32_t, int32_t, uint8_t*, int, int, hipStream_t);
+void launch_k_seq_advance(uint32_t*, int32_t, hipStream_t);
 }
 
 #define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
@@ -284,6 +288,24 @@ void fanout_flat2(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor pa
                            (int)grid, cur_stream());
 }
 
+void fanout_flat3(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_len,
+                  torch::Tensor pair_user, torch::Tensor pair_msg, torch::Tensor pair_dst,
+                  torch::Tensor seq_state, torch::Tensor n_pairs, int64_t units_per_pair,
+                  torch::Tensor egress, int64_t nt, int64_t grid) {
+    CHECK_DEV(egress); CHECK_CONTIG(egress);
+    int32_t capacity = (int32_t)pair_user.size(0);
+    launch_k3_fanout_flat3(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
+                           payload_len.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
+                           pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
+                           (const uint32_t*)seq_state.data_ptr<int32_t>(),
+                           n_pairs.data_ptr<int32_t>(), capacity, (int32_t)units_per_pair,
+                           egress.data_ptr<uint8_t>(), (int)nt, (int)grid, cur_stream());
+}
+
+void seq_advance(torch::Tensor seq_state, int64_t m) {
+    launch_k_seq_advance((uint32_t*)seq_state.data_ptr<int32_t>(), (int32_t)m, cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("parse_batch", &parse_batch, "K4: on-device capnp parse of a message batch");
     m.def("topic_mask", &topic_mask, "K2a: per-message subscriber mask");
@@ -300,4 +322,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("assign_emit_fused", &assign_emit_fused,
           "K2b-fused: count+claim+emit in one kernel (atomic slot claim)");
     m.def("fanout_flat2", &fanout_flat2, "K3v4: flat fan-out, seq from base, capacity clamp");
+    m.def("fanout_flat3", &fanout_flat3, "K3v5: graph-capturable (device seq counter)");
+    m.def("seq_advance", &seq_advance, "bump the device seq counter (inside the graph)");
 }

@@ -734,6 +734,7 @@ extern "C" __global__ void k2b_fused(
     int w = u >> 6;
     uint64_t bit = 1ull << (u & 63);
     int count = 0;
+#pragma unroll 8
     for (int m = 0; m < M; ++m) count += (mask[(int64_t)m * W + w] & bit) ? 1 : 0;
     if (count == 0) return;
     int slot = atomicAdd(n_pairs, count);
@@ -835,6 +836,88 @@ void launch_k3_fanout_flat2(const uint8_t* buf, const int64_t* payload_off,
         hipLaunchKernelGGL((k3_fanout_flat2_t<0>), dim3(grid), dim3(256), 0, s, buf,
                            payload_off, payload_len, pair_user, pair_msg, pair_dst, seq_base,
                            n_pairs_ptr, capacity, units_per_pair, egress);
+}
+
+}  // extern "C"
+
+// ---------------------------------------------------------------------------
+// hipGraph-capturable variants: the per-tick message sequence base lives in
+// a DEVICE counter (seq_state[0]) so the whole tick can be captured once and
+// replayed (a host-passed seq base would be frozen into the graph).
+// k_seq_advance bumps the counter at the end of the tick.
+// ---------------------------------------------------------------------------
+template <int NT>
+__global__ void __launch_bounds__(256) k3_fanout_flat3_t(
+    const uint8_t* __restrict__ buf,
+    const int64_t* __restrict__ payload_off,
+    const int32_t* __restrict__ payload_len,
+    const int32_t* __restrict__ pair_user,
+    const int32_t* __restrict__ pair_msg,
+    const int64_t* __restrict__ pair_dst,
+    const uint32_t* __restrict__ seq_state,   // [0] = seq base (device counter)
+    const int32_t* __restrict__ n_pairs_ptr,
+    int32_t capacity,
+    int32_t units_per_pair,
+    uint8_t* __restrict__ egress)
+{
+    typedef unsigned int v4u __attribute__((ext_vector_type(4)));
+    const uint32_t seq_base = seq_state[0];
+    int np = *n_pairs_ptr;
+    if (np > capacity) np = capacity;
+    const int64_t n_units = (int64_t)np * units_per_pair;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < n_units; f += stride) {
+        const int p = (int)(f / units_per_pair);
+        const int unit = (int)(f - (int64_t)p * units_per_pair);
+        const int u = pair_user[p];
+        if (u < 0) continue;
+        const int mi = pair_msg[p];
+        uint8_t* dst = egress + pair_dst[p] + (size_t)unit * 16;
+        if (unit == 0) {
+            uint32_t hdr[4] = {(uint32_t)payload_len[mi], seq_base + (uint32_t)mi, 0, 0};
+            v4u h; memcpy(&h, hdr, 16);
+            if (NT) __builtin_nontemporal_store(h, (v4u*)dst);
+            else memcpy(dst, hdr, 16);
+            continue;
+        }
+        const uint8_t* src = buf + payload_off[mi] + (size_t)(unit - 1) * 16;
+        const int32_t len = payload_len[mi];
+        const int32_t coff = (unit - 1) * 16;
+        if (coff + 16 <= len && (((uintptr_t)src) & 15) == 0) {
+            v4u v = *(const v4u*)src;
+            if (NT) __builtin_nontemporal_store(v, (v4u*)dst);
+            else *(v4u*)dst = v;
+        } else {
+            for (int b = 0; b < 16 && coff + b < len; ++b) dst[b] = src[b];
+        }
+    }
+}
+
+extern "C" __global__ void k_seq_advance(uint32_t* seq_state, int32_t m) {
+    if (blockIdx.x == 0 && threadIdx.x == 0) seq_state[0] += (uint32_t)m;
+}
+
+extern "C" {
+
+void launch_k3_fanout_flat3(const uint8_t* buf, const int64_t* payload_off,
+                            const int32_t* payload_len, const int32_t* pair_user,
+                            const int32_t* pair_msg, const int64_t* pair_dst,
+                            const uint32_t* seq_state, const int32_t* n_pairs_ptr,
+                            int32_t capacity, int32_t units_per_pair, uint8_t* egress, int nt,
+                            int grid, hipStream_t s) {
+    if (grid <= 0) grid = 8192;
+    if (nt)
+        hipLaunchKernelGGL((k3_fanout_flat3_t<1>), dim3(grid), dim3(256), 0, s, buf,
+                           payload_off, payload_len, pair_user, pair_msg, pair_dst, seq_state,
+                           n_pairs_ptr, capacity, units_per_pair, egress);
+    else
+        hipLaunchKernelGGL((k3_fanout_flat3_t<0>), dim3(grid), dim3(256), 0, s, buf,
+                           payload_off, payload_len, pair_user, pair_msg, pair_dst, seq_state,
+                           n_pairs_ptr, capacity, units_per_pair, egress);
+}
+
+void launch_k_seq_advance(uint32_t* seq_state, int32_t m, hipStream_t s) {
+    hipLaunchKernelGGL(k_seq_advance, dim3(1), dim3(1), 0, s, seq_state, m);
 }
 
 }  // extern "C"

@@ -434,3 +434,41 @@ def test_fused_pipeline_rings_match_reference(ops):
     torch.cuda.synchronize()
     assert torch.equal(eng.ring_wpos.cpu(), wposr)
     assert eng.egress.cpu().numpy().tobytes() == bytes(arr)
+
+
+def test_tick_graphed_matches_eager(ops):
+    """hipGraph-captured tick == eager tick (rings + cursors + seq)."""
+    from pushcdn_amd.broker.gpu_engine import GpuBrokerEngine
+
+    n_users, ring_bytes = 300, 1 << 17
+    buf = bytearray()
+    offsets = [0]
+    wire_len = None
+    for i in range(32):
+        raw = m.serialize(m.Broadcast([i % 3], bytes([i]) * 256))
+        padded = (len(raw) + 15) & ~15
+        wire_len = padded
+        buf += raw + b"\x00" * (padded - len(buf) + len(buf) - len(raw))
+        offsets.append(len(buf))
+    buf = bytes(buf)
+
+    def make(graph: bool):
+        eng = GpuBrokerEngine(device="cuda:0", n_users=n_users, ring_bytes=ring_bytes,
+                              fanout_wire=True, direct_enabled=False, pair_capacity=1 << 16)
+        for u in range(n_users):
+            eng.subscribe(u, [u % 3])
+        dbuf, doff = eng.ingest(buf, offsets)
+        dbuf = dbuf.clone()  # stable address
+        for _ in range(3):
+            if graph:
+                eng.tick_graphed(dbuf, doff, wire_len)
+            else:
+                eng.tick(dbuf, doff, uniform_wire_len=wire_len)
+        torch.cuda.synchronize()
+        return eng
+
+    eager = make(False)
+    graphed = make(True)
+    assert torch.equal(eager.ring_wpos.cpu(), graphed.ring_wpos.cpu())
+    assert torch.equal(eager.egress.cpu(), graphed.egress.cpu())
+    assert int(graphed._seq_dev.cpu()[0]) == 96  # 3 ticks x 32 msgs
