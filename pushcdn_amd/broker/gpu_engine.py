@@ -257,6 +257,9 @@ class GpuBrokerEngine:
         key = (buf.data_ptr(), offsets.data_ptr(), units)
         g = self._graphs.get(key)
         if g is None:
+            # warmup runs the body for real and mutates broker state —
+            # snapshot and restore ring cursors / seq / drop counter
+            saved = (self.ring_wpos.clone(), self._seq_dev.clone(), self._drops.clone())
             side = torch.cuda.Stream(device=self.device)
             side.wait_stream(torch.cuda.current_stream(self.device))
             with torch.cuda.stream(side):
@@ -267,6 +270,9 @@ class GpuBrokerEngine:
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
                 self._graph_tick_body(buf, offsets, units)
+            self.ring_wpos.copy_(saved[0])
+            self._seq_dev.copy_(saved[1])
+            self._drops.copy_(saved[2])
             self._graphs[key] = g
         g.replay()
 
