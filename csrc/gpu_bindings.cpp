@@ -55,6 +55,11 @@ void launch_k3_fanout_flat3(const uint8_t*, const int64_t*, const int32_t*, cons
                             const int32_t*, const int64_t*, const uint32_t*, const int32_t*,
                             int32_t, int32_t, uint8_t*, int, int, hipStream_t);
 void launch_k_seq_advance(uint32_t*, int32_t, hipStream_t);
+void launch_k2a_topic_mask_t(const uint64_t*, const uint8_t*, const int64_t*, const int32_t*,
+                             const int32_t*, uint64_t*, int32_t, int32_t, hipStream_t);
+void launch_k2b_fused_t(const uint64_t*, const int32_t*, int32_t, int32_t, int32_t, int64_t,
+                        int32_t, uint64_t*, int32_t*, int32_t*, int32_t*, int64_t*, uint32_t*,
+                        hipStream_t);
 }
 
 #define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
@@ -306,6 +311,40 @@ void seq_advance(torch::Tensor seq_state, int64_t m) {
     launch_k_seq_advance((uint32_t*)seq_state.data_ptr<int32_t>(), (int32_t)m, cur_stream());
 }
 
+torch::Tensor topic_mask_t(torch::Tensor sub_bitmap, torch::Tensor buf,
+                           torch::Tensor topics_off, torch::Tensor topics_cnt,
+                           torch::Tensor disc) {
+    CHECK_DEV(sub_bitmap); CHECK_CONTIG(sub_bitmap);
+    int32_t W = (int32_t)sub_bitmap.size(1);
+    int32_t M = (int32_t)disc.size(0);
+    auto mask_t = torch::empty({(int64_t)W, M},
+                               torch::TensorOptions().dtype(torch::kInt64).device(buf.device()));
+    if (M > 0) {
+        launch_k2a_topic_mask_t((const uint64_t*)sub_bitmap.data_ptr<int64_t>(),
+                                buf.data_ptr<uint8_t>(), topics_off.data_ptr<int64_t>(),
+                                topics_cnt.data_ptr<int32_t>(), disc.data_ptr<int32_t>(),
+                                (uint64_t*)mask_t.data_ptr<int64_t>(), M, W, cur_stream());
+    }
+    return mask_t;
+}
+
+void assign_emit_fused_t(torch::Tensor mask_t, torch::Tensor payload_len,
+                         torch::Tensor ring_wpos, int64_t ring_bytes, int64_t n_users,
+                         torch::Tensor pair_user, torch::Tensor pair_msg,
+                         torch::Tensor pair_dst, torch::Tensor drops, torch::Tensor n_pairs) {
+    CHECK_DEV(mask_t); CHECK_CONTIG(mask_t);
+    int32_t W = (int32_t)mask_t.size(0);
+    int32_t M = (int32_t)mask_t.size(1);
+    TORCH_CHECK(ring_bytes % 16 == 0);
+    int32_t capacity = (int32_t)pair_user.size(0);
+    launch_k2b_fused_t((const uint64_t*)mask_t.data_ptr<int64_t>(),
+                       payload_len.data_ptr<int32_t>(), M, W, (int32_t)n_users, ring_bytes,
+                       capacity, (uint64_t*)ring_wpos.data_ptr<int64_t>(),
+                       n_pairs.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
+                       pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
+                       (uint32_t*)drops.data_ptr<int32_t>(), cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("parse_batch", &parse_batch, "K4: on-device capnp parse of a message batch");
     m.def("topic_mask", &topic_mask, "K2a: per-message subscriber mask");
@@ -324,4 +363,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fanout_flat2", &fanout_flat2, "K3v4: flat fan-out, seq from base, capacity clamp");
     m.def("fanout_flat3", &fanout_flat3, "K3v5: graph-capturable (device seq counter)");
     m.def("seq_advance", &seq_advance, "bump the device seq counter (inside the graph)");
+    m.def("topic_mask_t", &topic_mask_t, "K2a transposed: mask[W][M]");
+    m.def("assign_emit_fused_t", &assign_emit_fused_t,
+          "K2b fused on transposed mask (contiguous per-user scans)");
 }

@@ -194,15 +194,15 @@ class GpuBrokerEngine:
         disc, payload_off, payload_len, topics_off, topics_cnt, recip_hash, _ts = ops.parse_batch(
             buf, offsets
         )
-        mask = ops.topic_mask(self.sub_bitmap, buf, topics_off, topics_cnt, disc)
+        mask_t = ops.topic_mask_t(self.sub_bitmap, buf, topics_off, topics_cnt, disc)
         if self.fanout_wire:
             payload_off = offsets[:-1].contiguous()
             payload_len = (offsets[1:] - offsets[:-1]).to(torch.int32).contiguous()
-        # fused sync-free pipeline: one kernel counts + claims slots
-        # atomically + emits; the pair count stays on device
+        # fused sync-free pipeline on the transposed mask: one kernel counts
+        # + claims slots atomically + emits; the pair count stays on device
         self._n_pairs.zero_()
-        ops.assign_emit_fused(
-            mask, payload_len, self.ring_wpos, self.ring_bytes, self.n_users,
+        ops.assign_emit_fused_t(
+            mask_t, payload_len, self.ring_wpos, self.ring_bytes, self.n_users,
             self._pair_user, self._pair_msg, self._pair_dst, self._drops, self._n_pairs,
         )
         seq_base = self.seq
@@ -233,12 +233,12 @@ class GpuBrokerEngine:
         ops = self._ops
         M = offsets.shape[0] - 1
         disc, _po, _pl, topics_off, topics_cnt, _rh, _ts = ops.parse_batch(buf, offsets)
-        mask = ops.topic_mask(self.sub_bitmap, buf, topics_off, topics_cnt, disc)
+        mask_t = ops.topic_mask_t(self.sub_bitmap, buf, topics_off, topics_cnt, disc)
         payload_off = offsets[:-1].contiguous()
         payload_len = (offsets[1:] - offsets[:-1]).to(torch.int32).contiguous()
         self._n_pairs.zero_()
-        ops.assign_emit_fused(
-            mask, payload_len, self.ring_wpos, self.ring_bytes, self.n_users,
+        ops.assign_emit_fused_t(
+            mask_t, payload_len, self.ring_wpos, self.ring_bytes, self.n_users,
             self._pair_user, self._pair_msg, self._pair_dst, self._drops, self._n_pairs,
         )
         ops.fanout_flat3(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
