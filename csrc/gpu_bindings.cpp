@@ -60,9 +60,6 @@ void launch_k2a_topic_mask_t(const uint64_t*, const uint8_t*, const int64_t*, co
 void launch_k2b_fused_t(const uint64_t*, const int32_t*, int32_t, int32_t, int32_t, int64_t,
                         int32_t, int32_t, uint64_t*, int32_t*, int32_t*, int32_t*, int64_t*,
                         uint32_t*, hipStream_t);
-void launch_k3_fanout_shard(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
-                            const int64_t*, const uint32_t*, const int32_t*, int32_t, int32_t,
-                            int32_t, uint8_t*, int, int, hipStream_t);
 }
 
 #define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
@@ -340,9 +337,7 @@ void assign_emit_fused_t(torch::Tensor mask_t, torch::Tensor payload_len,
     int32_t W = (int32_t)mask_t.size(0);
     int32_t M = (int32_t)mask_t.size(1);
     TORCH_CHECK(ring_bytes % 16 == 0);
-    TORCH_CHECK(n_pairs.numel() >= 9, "n_pairs must hold 8 shard counters + 1 general");
     int32_t capacity = (int32_t)pair_user.size(0);
-    TORCH_CHECK(capacity % 8 == 0, "pair capacity must be a multiple of 8");
     launch_k2b_fused_t((const uint64_t*)mask_t.data_ptr<int64_t>(),
                        payload_len.data_ptr<int32_t>(), M, W, (int32_t)n_users, ring_bytes,
                        capacity, (int32_t)uniform_rec,
@@ -350,21 +345,6 @@ void assign_emit_fused_t(torch::Tensor mask_t, torch::Tensor payload_len,
                        n_pairs.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
                        pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
                        (uint32_t*)drops.data_ptr<int32_t>(), cur_stream());
-}
-
-void fanout_shard(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor pair_user,
-                  torch::Tensor pair_msg, torch::Tensor pair_dst, torch::Tensor seq_state,
-                  torch::Tensor n_pairs, int64_t wire_len, int64_t units_per_pair,
-                  torch::Tensor egress, int64_t nt, int64_t grid) {
-    CHECK_DEV(egress); CHECK_CONTIG(egress);
-    int32_t capacity = (int32_t)pair_user.size(0);
-    launch_k3_fanout_shard(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
-                           pair_user.data_ptr<int32_t>(), pair_msg.data_ptr<int32_t>(),
-                           pair_dst.data_ptr<int64_t>(),
-                           (const uint32_t*)seq_state.data_ptr<int32_t>(),
-                           n_pairs.data_ptr<int32_t>(), capacity, (int32_t)wire_len,
-                           (int32_t)units_per_pair, egress.data_ptr<uint8_t>(), (int)nt,
-                           (int)grid, cur_stream());
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -388,5 +368,4 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("topic_mask_t", &topic_mask_t, "K2a transposed: mask[W][M]");
     m.def("assign_emit_fused_t", &assign_emit_fused_t,
           "K2b fused on transposed mask (sharded counters, uniform-rec fast path)");
-    m.def("fanout_shard", &fanout_shard, "K3v6: fan-out over the sharded pair buffer");
 }

@@ -951,48 +951,57 @@ extern "C" __global__ void k2a_topic_mask_t(
 
 // uniform_rec != 0: every record in this tick is uniform_rec bytes (the
 // uniform-wire broadcast shape) — ring math is closed-form and payload_len
-// is never loaded. The slot counter is SHARDED 8 ways (one per XCD-ish
-// group): a single counter word saturates at ~88 atomics/us, which alone
-// costs ~100 us for 10k users (MI355X_MICROARCH 'dequeue' row).
+// is never loaded. Slot claims are WAVE-AGGREGATED: a wave64 prefix-scan of
+// per-lane counts and ONE atomicAdd per wave (157 atomics for 10k users
+// instead of 10k — a single counter word saturates at ~88 atomics/us,
+// MI355X_MICROARCH 'dequeue' row) keep the pair list contiguous.
 extern "C" __global__ void k2b_fused_t(
     const uint64_t* __restrict__ mask_t,      // [W][M]
     const int32_t* __restrict__ payload_len,
     int32_t M, int32_t W, int32_t n_users,
     int64_t ring_bytes, int32_t capacity, int32_t uniform_rec,
     uint64_t* __restrict__ ring_wpos,
-    int32_t* __restrict__ n_pairs,            // [9]: 8 shard counters + total
+    int32_t* __restrict__ n_pairs,            // [1] global pair counter (zeroed)
     int32_t* __restrict__ pair_user,
     int32_t* __restrict__ pair_msg,
     int64_t* __restrict__ pair_dst,
     uint32_t* __restrict__ drops)
 {
-    int u = blockIdx.x * blockDim.x + threadIdx.x;
-    if (u >= n_users) return;
-    int w = u >> 6;
-    uint64_t bit = 1ull << (u & 63);
-    const uint64_t* col = mask_t + (int64_t)w * M;  // contiguous per thread
+    const int u = blockIdx.x * blockDim.x + threadIdx.x;
+    const bool active = u < n_users;
+    const int w = u >> 6;
+    const uint64_t bit = 1ull << (u & 63);
+    const uint64_t* col = mask_t + (int64_t)w * M;
     int count = 0;
+    if (active) {
 #pragma unroll 8
-    for (int m = 0; m < M; ++m) count += (col[m] & bit) ? 1 : 0;
-    if (count == 0) return;
-    int slot;
+        for (int m = 0; m < M; ++m) count += (col[m] & bit) ? 1 : 0;
+    }
+    // wave64 inclusive scan of counts; one atomic per wave claims the span
+    const int lane = threadIdx.x & 63;
+    int incl = count;
+    for (int d = 1; d < 64; d <<= 1) {
+        int ngh = __shfl_up(incl, d);
+        if (lane >= d) incl += ngh;
+    }
+    int wave_total = __shfl(incl, 63);
+    int base = 0;
+    if (lane == 0 && wave_total > 0) base = atomicAdd(n_pairs, wave_total);
+    base = __shfl(base, 0);
+    int slot = base + incl - count;
+    if (!active || count == 0) return;
+
     if (uniform_rec) {
-        // shard the pair buffer: shard s owns [s*cap8, (s+1)*cap8)
-        const int32_t cap8 = capacity >> 3;
-        const int shard = u & 7;
-        int local = atomicAdd(&n_pairs[shard], count);
-        slot = shard * cap8 + local;
-        const int32_t shard_cap = (shard + 1) * cap8;
         uint64_t wpos = ring_wpos[u];
-        int32_t fit = (int32_t)((ring_bytes - wpos) / (uint64_t)uniform_rec);
+        const int32_t fit = (int32_t)((ring_bytes - wpos) / (uint64_t)uniform_rec);
         uint32_t dropped = 0;
         int emitted = 0;
-        int64_t dst_base = (int64_t)u * ring_bytes + (int64_t)wpos;
+        const int64_t dst_base = (int64_t)u * ring_bytes + (int64_t)wpos;
         for (int m = 0; m < M; ++m) {
             if (!(col[m] & bit)) continue;
-            bool ok = (emitted < fit) && (slot < shard_cap);
+            const bool ok = (emitted < fit) && (slot < capacity);
             if (!ok) {
-                if (slot < shard_cap) { pair_user[slot] = -1; pair_msg[slot] = m; pair_dst[slot] = 0; slot++; }
+                if (slot < capacity) { pair_user[slot] = -1; pair_msg[slot] = m; pair_dst[slot] = 0; slot++; }
                 dropped++;
                 continue;
             }
@@ -1006,11 +1015,7 @@ extern "C" __global__ void k2b_fused_t(
         if (dropped) atomicAdd(drops, dropped);
         return;
     }
-    {
-    // general (mixed-size) branch: single counter in n_pairs[8], unsharded
-    // contiguous pair list (consumed by fanout_wave/flat2)
-    slot = atomicAdd(&n_pairs[8], count);
-    const int32_t gen_cap = capacity;
+
     uint64_t wpos = ring_wpos[u];
     uint32_t dropped = 0;
     for (int m = 0; m < M; ++m) {
@@ -1018,7 +1023,7 @@ extern "C" __global__ void k2b_fused_t(
         int32_t len = payload_len[m];
         uint64_t rec = 16 + (uint64_t)((len + 15) & ~15);
         bool fits_ring = (wpos + rec <= (uint64_t)ring_bytes);
-        bool fits_cap = (slot < gen_cap);
+        bool fits_cap = (slot < capacity);
         if (!fits_ring || !fits_cap) {
             if (fits_cap) { pair_user[slot] = -1; pair_msg[slot] = m; pair_dst[slot] = 0; slot++; }
             dropped++;
@@ -1032,87 +1037,7 @@ extern "C" __global__ void k2b_fused_t(
     }
     ring_wpos[u] = wpos;
     if (dropped) atomicAdd(drops, dropped);
-    }
 }
-
-// K3 over the SHARDED pair buffer: 8 regions of capacity/8, each with its
-// own count in n_pairs[0..7]; flat over the sum of per-shard unit spans.
-template <int NT>
-__global__ void __launch_bounds__(256) k3_fanout_shard_t(
-    const uint8_t* __restrict__ buf,
-    const int64_t* __restrict__ payload_off,
-    const int32_t* __restrict__ pair_user,
-    const int32_t* __restrict__ pair_msg,
-    const int64_t* __restrict__ pair_dst,
-    const uint32_t* __restrict__ seq_state,
-    const int32_t* __restrict__ n_pairs,     // [8] shard counts
-    int32_t capacity, int32_t wire_len,
-    int32_t units_per_pair,
-    uint8_t* __restrict__ egress)
-{
-    typedef unsigned int v4u __attribute__((ext_vector_type(4)));
-    const uint32_t seq_base = seq_state[0];
-    const int32_t cap8 = capacity >> 3;
-    // per-shard clamped counts and exclusive prefix (small, recomputed per block)
-    int cnt[8], pre[9];
-    pre[0] = 0;
-    for (int s = 0; s < 8; ++s) {
-        int c = n_pairs[s];
-        cnt[s] = c > cap8 ? cap8 : c;
-        pre[s + 1] = pre[s] + cnt[s];
-    }
-    const int total_pairs = pre[8];
-    const int64_t n_units = (int64_t)total_pairs * units_per_pair;
-    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < n_units; f += stride) {
-        int vp = (int)(f / units_per_pair);           // virtual pair index
-        const int unit = (int)(f - (int64_t)vp * units_per_pair);
-        int s = 0;
-        while (vp >= pre[s + 1]) ++s;                 // ≤8 iterations
-        const int p = s * cap8 + (vp - pre[s]);
-        const int u = pair_user[p];
-        if (u < 0) continue;
-        const int mi = pair_msg[p];
-        uint8_t* dst = egress + pair_dst[p] + (size_t)unit * 16;
-        if (unit == 0) {
-            uint32_t hdr[4] = {(uint32_t)wire_len, seq_base + (uint32_t)mi, 0, 0};
-            v4u h; memcpy(&h, hdr, 16);
-            if (NT) __builtin_nontemporal_store(h, (v4u*)dst);
-            else memcpy(dst, hdr, 16);
-            continue;
-        }
-        const uint8_t* src = buf + payload_off[mi] + (size_t)(unit - 1) * 16;
-        const int32_t coff = (unit - 1) * 16;
-        if (coff + 16 <= wire_len && (((uintptr_t)src) & 15) == 0) {
-            v4u v = *(const v4u*)src;
-            if (NT) __builtin_nontemporal_store(v, (v4u*)dst);
-            else *(v4u*)dst = v;
-        } else {
-            for (int b = 0; b < 16 && coff + b < wire_len; ++b) dst[b] = src[b];
-        }
-    }
-}
-
-extern "C" {
-
-void launch_k3_fanout_shard(const uint8_t* buf, const int64_t* payload_off,
-                            const int32_t* pair_user, const int32_t* pair_msg,
-                            const int64_t* pair_dst, const uint32_t* seq_state,
-                            const int32_t* n_pairs, int32_t capacity, int32_t wire_len,
-                            int32_t units_per_pair, uint8_t* egress, int nt, int grid,
-                            hipStream_t s) {
-    if (grid <= 0) grid = 8192;
-    if (nt)
-        hipLaunchKernelGGL((k3_fanout_shard_t<1>), dim3(grid), dim3(256), 0, s, buf,
-                           payload_off, pair_user, pair_msg, pair_dst, seq_state, n_pairs,
-                           capacity, wire_len, units_per_pair, egress);
-    else
-        hipLaunchKernelGGL((k3_fanout_shard_t<0>), dim3(grid), dim3(256), 0, s, buf,
-                           payload_off, pair_user, pair_msg, pair_dst, seq_state, n_pairs,
-                           capacity, wire_len, units_per_pair, egress);
-}
-
-}  // extern "C"
 
 extern "C" {
 

@@ -100,9 +100,7 @@ class GpuBrokerEngine:
             self._pair_msg = torch.empty(pair_capacity, **o32)
             self._pair_dst = torch.empty(pair_capacity, dtype=torch.int64, device=dev)
             self._drops = torch.zeros(1, **o32)
-            # [0..7] = per-shard pair counters (uniform path), [8] = the
-            # single counter of the general (mixed-size) path
-            self._n_pairs = torch.zeros(9, **o32)
+            self._n_pairs = torch.zeros(1, **o32)
             self._seq_dev = torch.zeros(1, **o32)  # device seq counter (graph path)
             self._graphs: Dict[Tuple[int, int, int], object] = {}
 
@@ -216,14 +214,13 @@ class GpuBrokerEngine:
         nt = 1 if self.nt_fanout else 0
         if uniform:
             units = rec // 16
-            self._seq_dev.fill_(seq_base)
-            ops.fanout_shard(buf, payload_off, self._pair_user, self._pair_msg,
-                             self._pair_dst, self._seq_dev, self._n_pairs,
-                             (uniform_wire_len + 15) & ~15, units, self.egress, nt, 0)
+            ops.fanout_flat2(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
+                             self._pair_dst, seq_base, self._n_pairs, units, self.egress,
+                             nt, 0)
         else:
             seq = torch.arange(seq_base, seq_base + M, dtype=torch.int32, device=self.device)
             ops.fanout_wave(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
-                            self._pair_dst, seq, self._n_pairs[8:9], self.egress, nt, 0)
+                            self._pair_dst, seq, self._n_pairs, self.egress, nt, 0)
         stats = TickStats(n_messages=M)
         # K5 lookup always runs (cheap, no host sync); the host-side direct
         # pair synthesis (one tiny sync) is skipped for broadcast-only
@@ -245,9 +242,11 @@ class GpuBrokerEngine:
         payload_off = offsets[:-1].contiguous()
         payload_len = (offsets[1:] - offsets[:-1]).to(torch.int32).contiguous()
         self._n_pairs.zero_()
+        rec = units * 16
         ops.assign_emit_fused_t(
             mask_t, payload_len, self.ring_wpos, self.ring_bytes, self.n_users,
             self._pair_user, self._pair_msg, self._pair_dst, self._drops, self._n_pairs,
+            rec,
         )
         ops.fanout_flat3(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
                          self._pair_dst, self._seq_dev, self._n_pairs, units, self.egress,
