@@ -24,10 +24,13 @@ async def _fail(connection: Connection, context: str) -> None:
 class MarshalAuth:
     @staticmethod
     async def verify_user(
-        connection: Connection, discovery: DiscoveryClient
+        connection: Connection, discovery: DiscoveryClient, verifier=None
     ) -> Optional[bytes]:
         """Run the marshal side of user auth on a fresh connection. Returns
-        the verified pubkey (after the response is sent) or None on failure."""
+        the verified pubkey (after the response is sent) or None on failure.
+
+        verifier: optional async batch verifier (GpuBatchVerifier) — auth
+        storms verify on the K1 kernel; None = host BLS."""
         try:
             msg = await connection.recv_message()
         except Exception:
@@ -37,9 +40,16 @@ class MarshalAuth:
             return None
 
         # signature + 5 s freshness window (marshal.rs:66-83)
-        if not bls.verify_timestamp(
-            msg.public_key, bls.USER_MARSHAL_NAMESPACE, msg.timestamp, msg.signature
-        ):
+        ts_bytes = (msg.timestamp & 0xFFFFFFFFFFFFFFFF).to_bytes(8, "little")
+        if verifier is not None:
+            sig_ok = await verifier.verify(
+                msg.public_key, bls.USER_MARSHAL_NAMESPACE, ts_bytes, msg.signature
+            )
+        else:
+            sig_ok = bls.verify_timestamp(
+                msg.public_key, bls.USER_MARSHAL_NAMESPACE, msg.timestamp, msg.signature
+            )
+        if not sig_ok:
             await _fail(connection, "failed to verify signature")
             return None
         if int(time.time()) - msg.timestamp > TIMESTAMP_WINDOW_S:

@@ -28,6 +28,9 @@ class MarshalConfig:
     ca_cert_path: Optional[str] = None
     ca_key_path: Optional[str] = None
     protocol: Optional[type] = None
+    # verify auth signatures in K1 batches on the GPU (auth-storm path)
+    gpu_verify: bool = False
+    gpu_device: str = "cuda:0"
 
 
 class Marshal:
@@ -40,6 +43,11 @@ class Marshal:
         self.protocol = config.protocol or Tcp
         self._tasks: List[asyncio.Task] = []
         self._closed = False
+        self._verifier = None
+        if config.gpu_verify:
+            from ..crypto.gpu_verify import GpuBatchVerifier
+
+            self._verifier = GpuBatchVerifier(device=config.gpu_device)
 
     async def start(self) -> None:
         self._listener = await self.protocol.bind(self.config.bind_endpoint, None, None)
@@ -69,7 +77,8 @@ class Marshal:
         except Exception:
             return
         try:
-            await asyncio.wait_for(MarshalAuth.verify_user(connection, self.discovery), 5)
+            await asyncio.wait_for(
+                MarshalAuth.verify_user(connection, self.discovery, self._verifier), 5)
         except asyncio.TimeoutError:
             pass
         finally:
