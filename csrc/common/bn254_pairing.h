@@ -60,6 +60,19 @@ struct Fp6 {
         return {Fp2::mul(a.c0, b), Fp2::mul(a.c1, b), Fp2::mul(a.c2, b)};
     }
 
+    // sparse multiply by (b0 + b1 v): 
+    //   c0 = f0 b0 + xi f2 b1, c1 = f0 b1 + f1 b0, c2 = f1 b1 + f2 b0
+    BN_BIGFUNC static Fp6 mul_by_01(const Fp6& f, const Fp2& b0, const Fp2& b1) {
+        Fp2 f0b0 = Fp2::mul(f.c0, b0);
+        Fp2 f1b1 = Fp2::mul(f.c1, b1);
+        Fp2 f2b0 = Fp2::mul(f.c2, b0);
+        Fp2 f2b1 = Fp2::mul(f.c2, b1);
+        Fp2 f0b1 = Fp2::mul(f.c0, b1);
+        Fp2 f1b0 = Fp2::mul(f.c1, b0);
+        return {Fp2::add(f0b0, Fp2::mul_xi(f2b1)), Fp2::add(f0b1, f1b0),
+                Fp2::add(f1b1, f2b0)};
+    }
+
     // multiply by v: (c0 + c1 v + c2 v^2) * v = xi c2 + c0 v + c1 v^2
     BN_INLINE static Fp6 mul_v(const Fp6& a) {
         return {Fp2::mul_xi(a.c2), a.c0, a.c1};
@@ -111,12 +124,54 @@ struct Fp12 {
         return {Fp6::mul(c0, di), Fp6::neg(Fp6::mul(c1, di))};
     }
 
-    // sparse multiply by a line element  l = a0 + (a3 + a4 v) w
-    // (positions: c0.c0 = a0, c1.c0 = a3, c1.c1 = a4)
-    BN_INLINE static Fp12 mul_by_034(const Fp12& f, const Fp2& a0, const Fp2& a3,
-                                     const Fp2& a4) {
-        Fp12 l{{a0, Fp2::zero(), Fp2::zero()}, {a3, a4, Fp2::zero()}};
-        return mul(f, l);
+    // sparse multiply by a line element  l = s0 + s1 w with s0 = a0 (Fp2 in
+    // the c0 slot) and s1 = a3 + a4 v  — ~15 Fp2 mults vs 18 for a full mul.
+    // (Equivalent to multiplying by the dense Fp12 with positions
+    // c0.c0 = a0, c1.c0 = a3, c1.c1 = a4 — cross-checked in the native
+    // sanitizer tests.)
+    BN_BIGFUNC static Fp12 mul_by_034(const Fp12& f, const Fp2& a0, const Fp2& a3,
+                                      const Fp2& a4) {
+        Fp6 x = Fp6::mul_fp2(f.c0, a0);                     // a0 * s0
+        Fp6 y = Fp6::mul_by_01(f.c1, a3, a4);               // a1 * s1
+        Fp2 s03 = Fp2::add(a0, a3);
+        Fp6 e = Fp6::mul_by_01(Fp6::add(f.c0, f.c1), s03, a4);  // (a0+a1)(s0+s1)
+        return {Fp6::add(x, Fp6::mul_v(y)),
+                Fp6::sub(e, Fp6::add(x, y))};
+    }
+
+    // Granger-Scott cyclotomic squaring — valid ONLY for elements of the
+    // cyclotomic subgroup (anything after the easy part of the final
+    // exponentiation); ~half the cost of a generic square. Verified against
+    // Fp12::sqr at runtime in the host test-suite.
+    BN_BIGFUNC static Fp12 cyclotomic_sqr(const Fp12& f) {
+        // fp4_square(a, b) with Fp4 = Fp2[v]/(v^2 - xi):
+        //   out0 = a^2 + xi b^2, out1 = (a+b)^2 - a^2 - b^2
+        Fp2 z0 = f.c0.c0, z4 = f.c0.c1, z3 = f.c0.c2;
+        Fp2 z2 = f.c1.c0, z1 = f.c1.c1, z5 = f.c1.c2;
+        Fp2 t0, t1, t2, t3;
+        {
+            Fp2 a2 = Fp2::sqr(z0), b2 = Fp2::sqr(z1);
+            t0 = Fp2::add(a2, Fp2::mul_xi(b2));
+            t1 = Fp2::sub(Fp2::sub(Fp2::sqr(Fp2::add(z0, z1)), a2), b2);
+        }
+        Fp2 r0 = Fp2::add(Fp2::dbl(Fp2::sub(t0, z0)), t0);   // 3 t0 - 2 z0
+        Fp2 r1 = Fp2::add(Fp2::dbl(Fp2::add(t1, z1)), t1);   // 3 t1 + 2 z1
+        {
+            Fp2 a2 = Fp2::sqr(z2), b2 = Fp2::sqr(z3);
+            t0 = Fp2::add(a2, Fp2::mul_xi(b2));
+            t1 = Fp2::sub(Fp2::sub(Fp2::sqr(Fp2::add(z2, z3)), a2), b2);
+        }
+        {
+            Fp2 a2 = Fp2::sqr(z4), b2 = Fp2::sqr(z5);
+            t2 = Fp2::add(a2, Fp2::mul_xi(b2));
+            t3 = Fp2::sub(Fp2::sub(Fp2::sqr(Fp2::add(z4, z5)), a2), b2);
+        }
+        Fp2 r4 = Fp2::add(Fp2::dbl(Fp2::sub(t0, z4)), t0);   // 3 t0 - 2 z4
+        Fp2 r5 = Fp2::add(Fp2::dbl(Fp2::add(t1, z5)), t1);   // 3 t1 + 2 z5
+        Fp2 xt3 = Fp2::mul_xi(t3);
+        Fp2 r2 = Fp2::add(Fp2::dbl(Fp2::add(xt3, z2)), xt3); // 3 xi t3 + 2 z2
+        Fp2 r3 = Fp2::add(Fp2::dbl(Fp2::sub(t2, z3)), t2);   // 3 t2 - 2 z3
+        return {{r0, r4, r3}, {r2, r1, r5}};
     }
 
     // Frobenius^1: conjugate each Fp2 coefficient, multiply by gamma1 factors
@@ -293,7 +348,7 @@ BN_BIGFUNC Fp12 pow_by_x(const Fp12& a) {
     Fp12 result = Fp12::one();
     bool started = false;
     BN_NOUNROLL for (int b = 63; b >= 0; --b) {
-        if (started) result = Fp12::sqr(result);
+        if (started) result = Fp12::cyclotomic_sqr(result);
         if ((e >> b) & 1) {
             if (started) result = Fp12::mul(result, a);
             else { result = a; started = true; }
@@ -307,11 +362,11 @@ BN_BIGFUNC Fp12 hard_exponentiation_chain(const Fp12& r_in) {
     // exp_by_neg_x(f) = conj(f^x).
     Fp12 r = r_in;
     Fp12 y0 = Fp12::conj(pow_by_x(r));
-    Fp12 y1 = Fp12::sqr(y0);
-    Fp12 y2 = Fp12::sqr(y1);
+    Fp12 y1 = Fp12::cyclotomic_sqr(y0);
+    Fp12 y2 = Fp12::cyclotomic_sqr(y1);
     Fp12 y3 = Fp12::mul(y2, y1);
     Fp12 y4 = Fp12::conj(pow_by_x(y3));
-    Fp12 y5 = Fp12::sqr(y4);
+    Fp12 y5 = Fp12::cyclotomic_sqr(y4);
     Fp12 y6 = Fp12::conj(pow_by_x(y5));
     y3 = Fp12::conj(y3);
     y6 = Fp12::conj(y6);

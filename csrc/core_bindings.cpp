@@ -191,6 +191,24 @@ static py::tuple frob_consistency(uint64_t a, uint64_t b) {
     return py::make_tuple(f11_eq_f2, f12_id, f1_pow);
 }
 
+static bool fp12_fastpath_consistency(uint64_t a, uint64_t b) {
+    G1 P = G1::scalar_mul(g1_generator(), U256{{a, 0, 0, 0}});
+    G2 Q = G2::scalar_mul(g2_generator(), U256{{b, 0, 0, 0}});
+    Fp px, py_;
+    P.to_affine(px, py_);
+    Fp2 qx, qy;
+    Q.to_affine(qx, qy);
+    Fp12 f = miller_loop(px, py_, G2Affine{qx, qy});
+    // sparse mul_by_034 vs dense multiply (arbitrary nonzero Fp2 operands)
+    Fp2 c0 = qx, c3 = qy, c4 = Fp2::add(qx, qy);
+    Fp12 dense{{c0, Fp2::zero(), Fp2::zero()}, {c3, c4, Fp2::zero()}};
+    if (!(Fp12::mul_by_034(f, c0, c3, c4) == Fp12::mul(f, dense))) return false;
+    // cyclotomic square vs generic square on a cyclotomic element
+    Fp12 e = easy_part(f);
+    if (!(Fp12::cyclotomic_sqr(e) == Fp12::sqr(e))) return false;
+    return true;
+}
+
 static py::bytes sha256_test(const py::bytes& data) {
     auto v = to_vec(data);
     uint8_t d[32];
@@ -275,6 +293,7 @@ PYBIND11_MODULE(pushcdn_core, m) {
     m.def("_hash_to_g1", &hash_to_g1_test);
     m.def("_hard_exp_chain_ok", &hard_exp_chain_matches_generic);
     m.def("_frob_consistency", &frob_consistency);
+    m.def("_fp12_fastpath_ok", &fp12_fastpath_consistency);
     // wire codec
     m.def("wire_serialize_authenticate_with_key", &w_ser_auth_key);
     m.def("wire_serialize_authenticate_with_permit", &w_ser_auth_permit);

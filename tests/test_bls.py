@@ -129,3 +129,10 @@ def test_hard_exponentiation_chain_matches_generic(core):
     multiple, verified symbolically in scripts/gen_bn254_constants.py lineage)."""
     for a, b in [(1, 1), (3, 5), (123456789, 987654321)]:
         assert core._hard_exp_chain_ok(a, b)
+
+
+def test_fp12_fast_paths_consistent(core):
+    """Sparse line multiplication (mul_by_034) and Granger-Scott cyclotomic
+    squaring must agree with the dense/general implementations."""
+    for a, b in [(3, 5), (7, 11), (123456, 654321)]:
+        assert core._fp12_fastpath_ok(a, b)
