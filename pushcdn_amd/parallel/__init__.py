@@ -1,0 +1,3 @@
+"""Distributed mesh communication (RCCL over xGMI)."""
+
+from .mesh import RcclMesh  # noqa: F401

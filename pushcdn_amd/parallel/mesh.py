@@ -69,6 +69,56 @@ class RcclMesh:
             ))
         return out
 
+    def exchange_p2p(self, batch: torch.Tensor, n_messages: int, batch_bytes: int,
+                     targets) -> list:
+        """Interest-targeted exchange: send this tick's batch ONLY to the
+        peers in `targets` (the brokers with >=1 subscriber on the batch's
+        topics — reference handler.rs:262-265 sends per interested broker).
+        On the 7-link xGMI topology each peer send rides its own
+        point-to-point link (grouped isend/irecv), which beats a ring
+        broadcast for k<=7 fan-out.
+
+        Every rank must call this each tick. Returns the same structure as
+        exchange(); peers that did not target us contribute empty batches.
+        Metadata is still all-gathered (tiny) so receive counts are known."""
+        if not self.enabled:
+            return [(0, batch, n_messages, batch_bytes)]
+        assert batch.numel() == self.capacity
+        # metadata: for each rank: [n_messages, batch_bytes, targets_bitmap]
+        tbits = 0
+        for t in targets:
+            tbits |= 1 << t
+        meta_local = torch.tensor([n_messages, batch_bytes, tbits], dtype=torch.int64,
+                                  device=self.device)
+        meta = torch.zeros(self.world_size * 3, dtype=torch.int64, device=self.device)
+        self.dist.all_gather_into_tensor(meta, meta_local)
+        meta_h = meta.to("cpu")
+        ops = []
+        for r in range(self.world_size):
+            if r == self.rank:
+                continue
+            if tbits & (1 << r):
+                ops.append(self.dist.P2POp(self.dist.isend, batch, r))
+            if int(meta_h[r * 3 + 2]) & (1 << self.rank):
+                ops.append(self.dist.P2POp(
+                    self.dist.irecv,
+                    self._gathered[r * self.capacity : (r + 1) * self.capacity], r))
+        if ops:
+            for req in self.dist.batch_isend_irecv(ops):
+                req.wait()
+        out = [(self.rank, batch, n_messages, batch_bytes)]
+        for r in range(self.world_size):
+            if r == self.rank:
+                continue
+            if int(meta_h[r * 3 + 2]) & (1 << self.rank):
+                out.append((
+                    r,
+                    self._gathered[r * self.capacity : (r + 1) * self.capacity],
+                    int(meta_h[r * 3]),
+                    int(meta_h[r * 3 + 1]),
+                ))
+        return out
+
     def barrier(self) -> None:
         if self.enabled:
             self.dist.barrier()
