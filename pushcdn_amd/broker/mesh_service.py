@@ -1,0 +1,135 @@
+"""MeshBroker — the GPU broker with its broker-plane on RCCL over xGMI.
+
+One process per GPU (torchrun); the broker↔broker transport is the
+RcclMesh collective exchange instead of framed TCP connections: each tick,
+every broker contributes its ingest batch (wire bytes + offsets packed into
+a fixed-capacity buffer) and receives every peer's batch over the
+all-gather, then routes ALL batches through the local kernel pipeline —
+remote batches deliver to local subscribers only, which is exactly the
+reference's single-hop to_users_only semantics
+(broker/handler.rs:151-161).
+
+The user plane (TCP/TLS/memory transports, auth, permits) is unchanged from
+the base Broker; the framed broker-mesh tasks (dialing, sync blasts) are
+disabled — the communicator IS the mesh, and every broker sees every
+message, so topic-interest sync is unnecessary for routing correctness
+(exchange_p2p + interest maps are the bandwidth optimization for sparse
+topologies).
+
+Batch wire layout (what travels over the collective):
+    [i64 n_messages][i64 offsets[n+1]  (absolute byte offsets)]
+    [pad to 16-byte boundary][message bytes, each start 16-aligned]
+"""
+
+from __future__ import annotations
+
+import asyncio
+from typing import List, Optional, Tuple
+
+import torch
+
+from ..parallel.mesh import RcclMesh
+from ..proto.limiter import Bytes
+from .service import Broker, BrokerConfig
+
+
+def pack_mesh_batch(msgs: List[bytes], capacity: int) -> Tuple[torch.Tensor, int, int]:
+    """Pack wire messages into the travel layout. Returns
+    (host uint8 tensor of `capacity`, n_messages, used_bytes)."""
+    n = len(msgs)
+    header_words = n + 2  # n + offsets[n+1]
+    base = (header_words * 8 + 15) & ~15
+    offsets = [base]
+    total = base
+    for raw in msgs:
+        total += (len(raw) + 15) & ~15
+        offsets.append(total)
+    if total > capacity:
+        raise ValueError(f"mesh batch overflow: {total} > {capacity}")
+    buf = torch.zeros(capacity, dtype=torch.uint8)
+    header = torch.tensor([n] + offsets, dtype=torch.int64)
+    buf[: header_words * 8] = header.view(torch.uint8)
+    pos = base
+    for raw in msgs:
+        buf[pos : pos + len(raw)] = torch.frombuffer(bytearray(raw), dtype=torch.uint8)
+        pos += (len(raw) + 15) & ~15
+    return buf, n, total
+
+
+def unpack_mesh_offsets(view: torch.Tensor, n_messages: int) -> torch.Tensor:
+    """Device (or host) int64 offsets tensor [n+1] out of a packed batch."""
+    header_words = n_messages + 2
+    return view[: header_words * 8].view(torch.int64)[1 : n_messages + 2].contiguous()
+
+
+class MeshBroker(Broker):
+    def __init__(self, config: BrokerConfig, batch_capacity: int = 1 << 22) -> None:
+        assert config.data_plane == "gpu", "MeshBroker is the GPU data-plane broker"
+        super().__init__(config)
+        self.batch_capacity = batch_capacity
+        self.mesh: Optional[RcclMesh] = None
+
+    # the framed broker mesh is replaced by the collective: no dialing, no
+    # framed sync blasts; heartbeats still publish load for the marshal
+    async def _heartbeat_task(self) -> None:
+        while True:
+            try:
+                await self.discovery.perform_heartbeat(len(self.connections.users), 60.0)
+            except Exception:
+                pass
+            await asyncio.sleep(self.config.heartbeat_interval_s)
+
+    async def _sync_task(self) -> None:
+        while True:  # nothing to sync: every broker sees every message
+            await asyncio.sleep(3600)
+
+    async def _send_partial_syncs(self) -> None:
+        pass
+
+    async def _gpu_tick_task(self) -> None:
+        """Fixed-cadence mesh tick: pack queued local messages (possibly
+        zero), exchange with every peer, route every rank's batch locally,
+        drain egress back to user connections."""
+        from .gpu_engine import parse_ring_records
+
+        self.mesh = RcclMesh(self._engine.device, self.batch_capacity)
+        dev_buf = (
+            torch.zeros(self.batch_capacity, dtype=torch.uint8, device=self._engine.device)
+            if self._engine.is_cuda
+            else None
+        )
+        while True:
+            batch: List[Bytes] = []
+            while not self._gpu_queue.empty() and len(batch) < 4096:
+                batch.append(self._gpu_queue.get_nowait()[0])
+            msgs = [raw.data for raw in batch]
+            host_buf, n_local, _used = pack_mesh_batch(msgs, self.batch_capacity)
+            if dev_buf is not None:
+                dev_buf.copy_(host_buf, non_blocking=True)
+                send_buf = dev_buf
+            else:
+                send_buf = host_buf
+            for rank, view, n_msgs, _nbytes in self.mesh.exchange(send_buf, n_local, 0):
+                if n_msgs == 0:
+                    continue
+                offsets = unpack_mesh_offsets(view, n_msgs)
+                if self._engine.use_gpu_ops:
+                    self._engine.tick(view, offsets)
+                else:
+                    host_bytes = bytes(view.numpy().tobytes())
+                    self._engine.tick(
+                        view, offsets,
+                        host_batch=host_bytes,
+                        host_offsets=[int(x) for x in offsets],
+                    )
+            wpos = self._engine.drain_cursors()
+            for slot, pubkey in list(self._gpu_user_by_slot.items()):
+                nbytes = int(wpos[slot])
+                if nbytes == 0:
+                    continue
+                ring = self._engine.read_ring(slot, nbytes)
+                for _seq, payload in parse_ring_records(ring, nbytes):
+                    await self.try_send_to_user(pubkey, Bytes(payload))
+            for raw in batch:
+                raw.drop()
+            await asyncio.sleep(self.config.gpu_tick_interval_s)

@@ -56,6 +56,35 @@ def cmd_broker(args) -> None:
     asyncio.run(Broker(cfg).run_forever())
 
 
+def cmd_mesh_broker(args) -> None:
+    """GPU broker with the RCCL/xGMI broker-plane — launch one per GPU:
+    torchrun --nproc-per-node 8 -m pushcdn_amd.cli mesh-broker ..."""
+    import os
+
+    from .broker.mesh_service import MeshBroker
+    from .broker.service import BrokerConfig
+    from .crypto import bls
+
+    rank = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", "0")))
+
+    def port_shift(ep: str) -> str:
+        host, _, port = ep.rpartition(":")
+        return f"{host}:{int(port) + 2 * rank}"
+
+    cfg = BrokerConfig(
+        public_bind_endpoint=port_shift(args.public_bind_endpoint),
+        public_advertise_endpoint=port_shift(args.public_advertise_endpoint),
+        private_bind_endpoint=port_shift(args.private_bind_endpoint),
+        private_advertise_endpoint=port_shift(args.private_advertise_endpoint),
+        discovery_endpoint=args.discovery_endpoint,
+        keypair=bls.KeyPair.from_seed(args.key_seed),
+        global_memory_pool_size=args.global_memory_pool_size,
+        data_plane="gpu",
+        gpu_device=f"cuda:{rank}",
+    )
+    asyncio.run(MeshBroker(cfg).run_forever())
+
+
 def cmd_marshal(args) -> None:
     from .marshal import Marshal, MarshalConfig
 
@@ -185,6 +214,10 @@ def main(argv=None) -> None:
     b = sub.add_parser("broker")
     _broker_args(b)
     b.set_defaults(fn=cmd_broker)
+
+    mb = sub.add_parser("mesh-broker")
+    _broker_args(mb)
+    mb.set_defaults(fn=cmd_mesh_broker)
 
     ms = sub.add_parser("marshal")
     ms.add_argument("-d", "--discovery-endpoint", default="/tmp/pushcdn-discovery.db")
