@@ -55,6 +55,9 @@ void launch_k3_fanout_flat3(const uint8_t*, const int64_t*, const int32_t*, cons
                             const int32_t*, const int64_t*, const uint32_t*, const int32_t*,
                             int32_t, int32_t, uint8_t*, int, int, hipStream_t);
 void launch_k_seq_advance(uint32_t*, int32_t, hipStream_t);
+void launch_k5b_emit_direct(const int32_t*, const int32_t*, const int64_t*, const int32_t*,
+                            int32_t, int64_t, int32_t, uint64_t*, int32_t*, int32_t*, int32_t*,
+                            int64_t*, uint32_t*, hipStream_t);
 void launch_k2a_topic_mask_t(const uint64_t*, const uint8_t*, const int64_t*, const int32_t*,
                              const int32_t*, uint64_t*, int32_t, int32_t, hipStream_t);
 void launch_k2b_fused_t(const uint64_t*, const int32_t*, int32_t, int32_t, int32_t, int64_t,
@@ -347,6 +350,21 @@ void assign_emit_fused_t(torch::Tensor mask_t, torch::Tensor payload_len,
                        (uint32_t*)drops.data_ptr<int32_t>(), cur_stream());
 }
 
+void emit_direct(torch::Tensor disc, torch::Tensor owner, torch::Tensor payload_off,
+                 torch::Tensor payload_len, int64_t ring_bytes, torch::Tensor ring_wpos,
+                 torch::Tensor n_pairs, torch::Tensor pair_user, torch::Tensor pair_msg,
+                 torch::Tensor pair_dst, torch::Tensor drops) {
+    int32_t M = (int32_t)disc.size(0);
+    if (M == 0) return;
+    int32_t capacity = (int32_t)pair_user.size(0);
+    launch_k5b_emit_direct(disc.data_ptr<int32_t>(), owner.data_ptr<int32_t>(),
+                           payload_off.data_ptr<int64_t>(), payload_len.data_ptr<int32_t>(),
+                           M, ring_bytes, capacity, (uint64_t*)ring_wpos.data_ptr<int64_t>(),
+                           n_pairs.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
+                           pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
+                           (uint32_t*)drops.data_ptr<int32_t>(), cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("parse_batch", &parse_batch, "K4: on-device capnp parse of a message batch");
     m.def("topic_mask", &topic_mask, "K2a: per-message subscriber mask");
@@ -366,6 +384,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fanout_flat3", &fanout_flat3, "K3v5: graph-capturable (device seq counter)");
     m.def("seq_advance", &seq_advance, "bump the device seq counter (inside the graph)");
     m.def("topic_mask_t", &topic_mask_t, "K2a transposed: mask[W][M]");
+    m.def("emit_direct", &emit_direct,
+          "K5b: on-device direct-delivery pair emission (no host sync)");
     m.def("assign_emit_fused_t", &assign_emit_fused_t,
           "K2b fused on transposed mask (sharded counters, uniform-rec fast path)");
 }

@@ -1064,3 +1064,61 @@ void launch_k2b_fused_t(const uint64_t* mask_t, const int32_t* payload_len, int3
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// K5b: on-device direct-message delivery pair emission. For each Direct
+// message whose K5 owner lookup resolved to a LOCAL user, claim ring space
+// (atomic on the user's cursor — per-sender order within a tick is
+// preserved by message index only per thread; cross-sender order is
+// unspecified, as in the reference's independent per-conn tasks) and append
+// a delivery pair. Replaces the host-side direct routing loop (which cost a
+// D2H sync per tick).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void k5b_emit_direct(
+    const int32_t* __restrict__ disc,
+    const int32_t* __restrict__ owner,        // K5 output; >=0 = local user
+    const int64_t* __restrict__ payload_off,  // wire offsets (fanout_wire)
+    const int32_t* __restrict__ payload_len,
+    int32_t M,
+    int64_t ring_bytes, int32_t capacity,
+    uint64_t* __restrict__ ring_wpos,
+    int32_t* __restrict__ n_pairs,
+    int32_t* __restrict__ pair_user,
+    int32_t* __restrict__ pair_msg,
+    int64_t* __restrict__ pair_dst,
+    uint32_t* __restrict__ drops)
+{
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= M) return;
+    if (disc[i] != 3) return;
+    int u = owner[i];
+    if (u < 0) return;
+    int32_t len = payload_len[i];
+    uint64_t rec = 16 + (uint64_t)((len + 15) & ~15);
+    uint64_t old = atomicAdd((unsigned long long*)&ring_wpos[u], (unsigned long long)rec);
+    int slot = atomicAdd(n_pairs, 1);
+    if (old + rec > (uint64_t)ring_bytes || slot >= capacity) {
+        // roll the cursor back (transient overshoot may make a concurrent
+        // claim drop spuriously — benign; never corrupts a neighbor ring)
+        if (old + rec > (uint64_t)ring_bytes)
+            atomicAdd((unsigned long long*)&ring_wpos[u],
+                      (unsigned long long)(-(long long)rec));
+        if (slot < capacity) { pair_user[slot] = -1; pair_msg[slot] = i; pair_dst[slot] = 0; }
+        atomicAdd(drops, 1u);
+        return;
+    }
+    pair_user[slot] = u;
+    pair_msg[slot] = i;
+    pair_dst[slot] = (int64_t)u * ring_bytes + (int64_t)old;
+}
+
+extern "C" void launch_k5b_emit_direct(
+    const int32_t* disc, const int32_t* owner, const int64_t* payload_off,
+    const int32_t* payload_len, int32_t M, int64_t ring_bytes, int32_t capacity,
+    uint64_t* ring_wpos, int32_t* n_pairs, int32_t* pair_user, int32_t* pair_msg,
+    int64_t* pair_dst, uint32_t* drops, hipStream_t s) {
+    int threads = 256, blocks = (M + threads - 1) / threads;
+    hipLaunchKernelGGL(k5b_emit_direct, dim3(blocks), dim3(threads), 0, s, disc, owner,
+                       payload_off, payload_len, M, ring_bytes, capacity, ring_wpos, n_pairs,
+                       pair_user, pair_msg, pair_dst, drops);
+}

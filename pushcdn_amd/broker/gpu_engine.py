@@ -199,8 +199,8 @@ class GpuBrokerEngine:
             payload_off = offsets[:-1].contiguous()
             payload_len = (offsets[1:] - offsets[:-1]).to(torch.int32).contiguous()
         # fused sync-free pipeline on the transposed mask: one kernel counts
-        # + claims slots atomically (sharded 8 ways in the uniform case) +
-        # emits; the pair counts stay on device
+        # + claims slots atomically (wave-aggregated) + emits; the pair
+        # count stays on device
         self._n_pairs.zero_()
         uniform = self.fanout_wire and uniform_wire_len is not None
         rec = (16 + ((uniform_wire_len + 15) & ~15)) if uniform else 0
@@ -209,6 +209,16 @@ class GpuBrokerEngine:
             self._pair_user, self._pair_msg, self._pair_dst, self._drops, self._n_pairs,
             rec,
         )
+        if self.direct_enabled:
+            # K5 lookup + K5b on-device delivery-pair emission: direct pairs
+            # append to the same pair list, all consumed by the single
+            # fan-out below — no host sync anywhere in the tick.
+            # (uniform_wire_len callers promise Direct messages share the
+            # same padded wire length as broadcasts.)
+            owner = ops.direct_lookup(self.direct_keys, self.direct_vals, recip_hash)
+            ops.emit_direct(disc, owner, payload_off, payload_len, self.ring_bytes,
+                            self.ring_wpos, self._n_pairs, self._pair_user,
+                            self._pair_msg, self._pair_dst, self._drops)
         seq_base = self.seq
         self.seq += M
         nt = 1 if self.nt_fanout else 0
@@ -221,15 +231,7 @@ class GpuBrokerEngine:
             seq = torch.arange(seq_base, seq_base + M, dtype=torch.int32, device=self.device)
             ops.fanout_wave(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
                             self._pair_dst, seq, self._n_pairs, self.egress, nt, 0)
-        stats = TickStats(n_messages=M)
-        # K5 lookup always runs (cheap, no host sync); the host-side direct
-        # pair synthesis (one tiny sync) is skipped for broadcast-only
-        # workloads via direct_enabled=False.
-        owner = ops.direct_lookup(self.direct_keys, self.direct_vals, recip_hash)
-        if self.direct_enabled:
-            dseq = torch.arange(seq_base, seq_base + M, dtype=torch.int32, device=self.device)
-            self._route_direct_gpu(buf, payload_off, payload_len, disc, owner, dseq)
-        return stats
+        return TickStats(n_messages=M)
 
     def _graph_tick_body(self, buf: torch.Tensor, offsets: torch.Tensor, units: int) -> None:
         """The capturable broadcast-tick body (uniform wire records, no host
@@ -282,46 +284,6 @@ class GpuBrokerEngine:
             self._drops.copy_(saved[2])
             self._graphs[key] = g
         g.replay()
-
-    def _route_direct_gpu(self, buf, payload_off, payload_len, disc, owner, seq) -> None:
-        """Deliver direct messages to local users by synthesizing delivery
-        pairs for the K3 copy kernel (single-recipient fan-out)."""
-        ops = self._ops
-        is_direct = disc == 3
-        local = (owner >= 0) & is_direct
-        if not bool(local.any()):
-            return
-        # Direct traffic is low-rate vs broadcast fan-out: assign ring space
-        # in a small host loop (handles several directs to one user in one
-        # tick with correct FIFO offsets), then reuse K3 for the copies.
-        idx = torch.nonzero(local).flatten()
-        users_h = owner[idx].to("cpu").tolist()
-        lens_h = payload_len[idx].to("cpu").tolist()
-        wpos_h = {u: int(self.ring_wpos[u]) for u in set(users_h)}
-        pair_user: List[int] = []
-        pair_msg: List[int] = []
-        pair_dst: List[int] = []
-        for j, (u, length) in enumerate(zip(users_h, lens_h)):
-            rec = 16 + ((length + 15) & ~15)
-            if wpos_h[u] + rec > self.ring_bytes:
-                pair_user.append(-1)
-                pair_msg.append(int(idx[j]))
-                pair_dst.append(0)
-                continue
-            pair_user.append(u)
-            pair_msg.append(int(idx[j]))
-            pair_dst.append(u * self.ring_bytes + wpos_h[u])
-            wpos_h[u] += rec
-        for u, w in wpos_h.items():
-            self.ring_wpos[u] = w
-        dev = self.device
-        ops.fanout(
-            buf, payload_off, payload_len,
-            torch.tensor(pair_user, dtype=torch.int32, device=dev),
-            torch.tensor(pair_msg, dtype=torch.int32, device=dev),
-            torch.tensor(pair_dst, dtype=torch.int64, device=dev),
-            seq, self.egress,
-        )
 
     def _tick_cpu(self, batch: bytes, offsets: List[int]) -> TickStats:
         from ..ops import reference as ref
