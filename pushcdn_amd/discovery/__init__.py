@@ -62,12 +62,14 @@ class DiscoveryClient(ABC):
     async def check_whitelist(self, user: bytes) -> bool: ...
 
 
-def new_discovery_client(path: str, identity: Optional[BrokerIdentifier]) -> DiscoveryClient:
-    """Factory: redis:// URLs get the Redis client, anything else Embedded."""
+def new_discovery_client(path: str, identity: Optional[BrokerIdentifier],
+                         global_permits: bool = False) -> DiscoveryClient:
+    """Factory: redis:// URLs get the Redis client, anything else Embedded.
+    global_permits mirrors the reference's cargo feature of the same name."""
     if path.startswith("redis://") or path.startswith("rediss://"):
         from .redis import RedisDiscovery
 
-        return RedisDiscovery(path, identity)
+        return RedisDiscovery(path, identity, global_permits)
     from .embedded import EmbeddedDiscovery
 
-    return EmbeddedDiscovery(path, identity)
+    return EmbeddedDiscovery(path, identity, global_permits)

@@ -39,9 +39,13 @@ CREATE TABLE IF NOT EXISTS whitelist (
 
 
 class EmbeddedDiscovery(DiscoveryClient):
-    def __init__(self, path: str, identity: Optional[BrokerIdentifier]) -> None:
+    def __init__(self, path: str, identity: Optional[BrokerIdentifier],
+                 global_permits: bool = False) -> None:
+        # global_permits: permits are valid at ANY broker (the reference's
+        # `global-permits` cargo feature, discovery/mod.rs:50-63)
         self.path = path
         self.identity = identity
+        self.global_permits = global_permits
         self._conn = sqlite3.connect(path, timeout=10, check_same_thread=False)
         self._conn.executescript(_SCHEMA)
         self._conn.commit()
@@ -124,9 +128,9 @@ class EmbeddedDiscovery(DiscoveryClient):
             self._conn.execute("DELETE FROM permits WHERE permit = ?", (permit,))
             self._conn.commit()
         pubkey, issued_broker = row
-        # permits are broker-bound (the non-global-permits default,
-        # reference redis.rs:246-265)
-        if issued_broker != str(broker):
+        # permits are broker-bound unless global_permits is on
+        # (reference redis.rs:219-265)
+        if not self.global_permits and issued_broker != str(broker):
             return None
         return bytes(pubkey)
 

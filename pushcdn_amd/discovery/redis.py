@@ -73,10 +73,12 @@ class _Resp:
 
 
 class RedisDiscovery(DiscoveryClient):
-    def __init__(self, url: str, identity: Optional[BrokerIdentifier]) -> None:
+    def __init__(self, url: str, identity: Optional[BrokerIdentifier],
+                 global_permits: bool = False) -> None:
         u = urlparse(url)
         self._r = _Resp(u.hostname or "127.0.0.1", u.port or 6379)
         self.identity = identity
+        self.global_permits = global_permits
 
     async def perform_heartbeat(self, num_connections: int, expiry_s: float) -> None:
         if self.identity is None:
@@ -121,7 +123,8 @@ class RedisDiscovery(DiscoveryClient):
         self, broker: BrokerIdentifier, expiry_s: float, user_pubkey: bytes
     ) -> int:
         permit = random.randrange(2, 2**63)
-        await self._r.cmd("SET", f"permit:{broker}:{permit}", user_pubkey,
+        scope = "any" if self.global_permits else str(broker)
+        await self._r.cmd("SET", f"permit:{scope}:{permit}", user_pubkey,
                           "EX", int(max(1, expiry_s)))
         await self._r.cmd("SADD", f"permits:{broker}", permit)
         await self._r.cmd("EXPIRE", f"permits:{broker}", int(max(1, expiry_s)))
@@ -130,7 +133,8 @@ class RedisDiscovery(DiscoveryClient):
     async def validate_permit(
         self, broker: BrokerIdentifier, permit: int
     ) -> Optional[bytes]:
-        key = f"permit:{broker}:{permit}"
+        scope = "any" if self.global_permits else str(broker)
+        key = f"permit:{scope}:{permit}"
         raw = await self._r.cmd("GETDEL", key)
         await self._r.cmd("SREM", f"permits:{broker}", permit)
         return bytes(raw) if raw is not None else None

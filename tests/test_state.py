@@ -204,3 +204,18 @@ def test_native_sanitizer_lane():
                          capture_output=True, text=True, timeout=600)
     assert out.returncode == 0, out.stderr[-2000:]
     assert "sanitizer lane OK" in out.stdout
+
+
+def test_global_permits_feature(tmp_path):
+    """global-permits (reference cargo feature): a permit issued for one
+    broker validates at any broker."""
+    async def go():
+        db = str(tmp_path / "gp.db")
+        d = EmbeddedDiscovery(db, id_(1), global_permits=True)
+        await d.perform_heartbeat(0, 60)
+        permit = await d.issue_permit(id_(1), 30, b"ukey")
+        assert await d.validate_permit(id_(2), permit) == b"ukey"  # other broker OK
+        # and still one-shot
+        assert await d.validate_permit(id_(2), permit) is None
+
+    asyncio.run(go())
