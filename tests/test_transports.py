@@ -168,3 +168,52 @@ def test_metrics_endpoint():
         await server.wait_closed()
 
     run(go())
+
+
+def test_oversize_frame_rejected():
+    """A frame header larger than MAX_MESSAGE_SIZE must error, not allocate
+    (reference protocols/mod.rs:322-325)."""
+    import struct
+
+    async def go():
+        limiter = Limiter()
+        listener = await Tcp.bind("127.0.0.1:0", None, None)
+        endpoint_port = listener.port
+
+        async def server():
+            conn = await (await listener.accept()).finalize(limiter)
+            with pytest.raises(ConnectionError_):
+                await asyncio.wait_for(conn.recv_message_raw(), timeout=5)
+
+        async def client():
+            reader, writer = await asyncio.open_connection("127.0.0.1", endpoint_port)
+            writer.write(struct.pack(">I", m.MAX_MESSAGE_SIZE + 1))
+            await writer.drain()
+            await asyncio.sleep(0.2)
+            writer.close()
+
+        await asyncio.wait_for(asyncio.gather(server(), client()), timeout=15)
+        await listener.close()
+
+    run(go())
+
+
+def test_limiter_backpressure_on_connection():
+    """With a tiny global pool, a second message blocks until the first's
+    allocation is released (reference protocols/mod.rs:328)."""
+
+    async def go():
+        limiter = Limiter(global_memory_pool_size=64)
+        a, b = gen_testing_connection_pair(limiter)
+        await a.send_message_raw(Bytes(b"x" * 60))
+        await a.send_message_raw(Bytes(b"y" * 60))
+        first = await asyncio.wait_for(b.recv_message_raw(), timeout=5)
+        # second recv must block while the first allocation is held
+        second_task = asyncio.ensure_future(b.recv_message_raw())
+        await asyncio.sleep(0.1)
+        assert not second_task.done(), "backpressure did not hold"
+        first.drop()  # release 60 bytes
+        second = await asyncio.wait_for(second_task, timeout=5)
+        second.drop()
+
+    run(go())
