@@ -101,12 +101,10 @@ class Connection:
                 while True:
                     item = await send_q.get()
                     if item is None:  # soft close: flush then stop
-                        with_flush = getattr(writer, "drain", None)
-                        if with_flush:
-                            try:
-                                await writer.drain()
-                            except Exception:
-                                pass
+                        try:
+                            await writer.drain()
+                        except Exception:
+                            pass
                         return
                     try:
                         await write_length_delimited(writer, item)
