@@ -37,7 +37,7 @@ struct Fp6 {
         return {Fp2::neg(a.c0), Fp2::neg(a.c1), Fp2::neg(a.c2)};
     }
 
-    BN_BIGFUNC static Fp6 mul(const Fp6& a, const Fp6& b) {
+    BN_INLINE static Fp6 mul(const Fp6& a, const Fp6& b) {
         // Toom/Karatsuba (devegili): v0=a0b0, v1=a1b1, v2=a2b2
         Fp2 v0 = Fp2::mul(a.c0, b.c0);
         Fp2 v1 = Fp2::mul(a.c1, b.c1);
@@ -62,7 +62,7 @@ struct Fp6 {
 
     // sparse multiply by (b0 + b1 v): 
     //   c0 = f0 b0 + xi f2 b1, c1 = f0 b1 + f1 b0, c2 = f1 b1 + f2 b0
-    BN_BIGFUNC static Fp6 mul_by_01(const Fp6& f, const Fp2& b0, const Fp2& b1) {
+    BN_INLINE static Fp6 mul_by_01(const Fp6& f, const Fp2& b0, const Fp2& b1) {
         Fp2 f0b0 = Fp2::mul(f.c0, b0);
         Fp2 f1b1 = Fp2::mul(f.c1, b1);
         Fp2 f2b0 = Fp2::mul(f.c2, b0);
@@ -78,7 +78,7 @@ struct Fp6 {
         return {Fp2::mul_xi(a.c2), a.c0, a.c1};
     }
 
-    BN_BIGFUNC Fp6 inv() const {
+    BN_INLINE Fp6 inv() const {
         // standard: A = c0^2 - xi c1 c2, B = xi c2^2 - c0 c1, C = c1^2 - c0 c2
         Fp2 A = Fp2::sub(Fp2::sqr(c0), Fp2::mul_xi(Fp2::mul(c1, c2)));
         Fp2 B = Fp2::sub(Fp2::mul_xi(Fp2::sqr(c2)), Fp2::mul(c0, c1));
@@ -100,14 +100,14 @@ struct Fp12 {
     BN_INLINE bool operator==(const Fp12& o) const { return c0 == o.c0 && c1 == o.c1; }
     BN_INLINE bool is_one() const { return *this == one(); }
 
-    BN_BIGFUNC static Fp12 mul(const Fp12& a, const Fp12& b) {
+    BN_INLINE static Fp12 mul(const Fp12& a, const Fp12& b) {
         Fp6 v0 = Fp6::mul(a.c0, b.c0);
         Fp6 v1 = Fp6::mul(a.c1, b.c1);
         Fp6 t = Fp6::mul(Fp6::add(a.c0, a.c1), Fp6::add(b.c0, b.c1));
         return {Fp6::add(v0, Fp6::mul_v(v1)), Fp6::sub(Fp6::sub(t, v0), v1)};
     }
 
-    BN_BIGFUNC static Fp12 sqr(const Fp12& a) {
+    BN_INLINE static Fp12 sqr(const Fp12& a) {
         // complex squaring: (c0 + c1 w)^2 = (c0^2 + v c1^2) + 2 c0 c1 w
         Fp6 v0 = Fp6::mul(a.c0, a.c1);
         Fp6 t = Fp6::mul(Fp6::add(a.c0, a.c1), Fp6::add(a.c0, Fp6::mul_v(a.c1)));
@@ -117,7 +117,7 @@ struct Fp12 {
 
     BN_INLINE static Fp12 conj(const Fp12& a) { return {a.c0, Fp6::neg(a.c1)}; }
 
-    BN_BIGFUNC Fp12 inv() const {
+    BN_INLINE Fp12 inv() const {
         // 1/(c0 + c1 w) = (c0 - c1 w) / (c0^2 - v c1^2)
         Fp6 d = Fp6::sub(Fp6::sqr(c0), Fp6::mul_v(Fp6::sqr(c1)));
         Fp6 di = d.inv();
@@ -129,7 +129,7 @@ struct Fp12 {
     // (Equivalent to multiplying by the dense Fp12 with positions
     // c0.c0 = a0, c1.c0 = a3, c1.c1 = a4 — cross-checked in the native
     // sanitizer tests.)
-    BN_BIGFUNC static Fp12 mul_by_034(const Fp12& f, const Fp2& a0, const Fp2& a3,
+    BN_INLINE static Fp12 mul_by_034(const Fp12& f, const Fp2& a0, const Fp2& a3,
                                       const Fp2& a4) {
         Fp6 x = Fp6::mul_fp2(f.c0, a0);                     // a0 * s0
         Fp6 y = Fp6::mul_by_01(f.c1, a3, a4);               // a1 * s1
@@ -143,7 +143,7 @@ struct Fp12 {
     // cyclotomic subgroup (anything after the easy part of the final
     // exponentiation); ~half the cost of a generic square. Verified against
     // Fp12::sqr at runtime in the host test-suite.
-    BN_BIGFUNC static Fp12 cyclotomic_sqr(const Fp12& f) {
+    BN_INLINE static Fp12 cyclotomic_sqr(const Fp12& f) {
         // fp4_square(a, b) with Fp4 = Fp2[v]/(v^2 - xi):
         //   out0 = a^2 + xi b^2, out1 = (a+b)^2 - a^2 - b^2
         Fp2 z0 = f.c0.c0, z4 = f.c0.c1, z3 = f.c0.c2;
@@ -176,7 +176,7 @@ struct Fp12 {
 
     // Frobenius^1: conjugate each Fp2 coefficient, multiply by gamma1 factors
     // (v^p = v * GAMMA1_2, w^p = w * GAMMA1_1; see scripts/gen_bn254_constants.py)
-    BN_BIGFUNC static Fp12 frobenius1(const Fp12& a) {
+    BN_INLINE static Fp12 frobenius1(const Fp12& a) {
         Fp2 g1{Fp::from_u256(from_limbs(bn254c::GAMMA1_1_C0)),
                Fp::from_u256(from_limbs(bn254c::GAMMA1_1_C1))};
         Fp2 g2{Fp::from_u256(from_limbs(bn254c::GAMMA1_2_C0)),
@@ -197,7 +197,7 @@ struct Fp12 {
     BN_INLINE static Fp12 frobenius3(const Fp12& a) { return frobenius1(frobenius2(a)); }
 
     // Frobenius^2: c_ij -> c_ij * gamma2 factors (Fp scalars, no conjugation)
-    BN_BIGFUNC static Fp12 frobenius2(const Fp12& a) {
+    BN_INLINE static Fp12 frobenius2(const Fp12& a) {
         Fp g1 = Fp::from_u256(from_limbs(bn254c::GAMMA2_1));
         Fp g2 = Fp::from_u256(from_limbs(bn254c::GAMMA2_2));
         Fp g3 = Fp::from_u256(from_limbs(bn254c::GAMMA2_3));
@@ -241,7 +241,7 @@ struct LineCoeffs {
 };
 
 // doubling step (arkworks models/bn/g2.rs shape, D-twist coefficients)
-BN_BIGFUNC LineCoeffs doubling_step(G2Proj& r, const Fp& two_inv) {
+BN_INLINE LineCoeffs doubling_step(G2Proj& r, const Fp& two_inv) {
     Fp2 a = Fp2::mul_fp(Fp2::mul(r.x, r.y), two_inv);
     Fp2 b = Fp2::sqr(r.y);
     Fp2 c = Fp2::sqr(r.z);
@@ -261,7 +261,7 @@ BN_BIGFUNC LineCoeffs doubling_step(G2Proj& r, const Fp& two_inv) {
 }
 
 // mixed addition step
-BN_BIGFUNC LineCoeffs addition_step(G2Proj& r, const G2Affine& q) {
+BN_INLINE LineCoeffs addition_step(G2Proj& r, const G2Affine& q) {
     Fp2 theta = Fp2::sub(r.y, Fp2::mul(q.y, r.z));
     Fp2 lambda = Fp2::sub(r.x, Fp2::mul(q.x, r.z));
     Fp2 c = Fp2::sqr(theta);

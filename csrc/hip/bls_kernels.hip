@@ -21,7 +21,12 @@
 
 using namespace bn254;
 
-extern "C" __global__ void __launch_bounds__(64)
+// __launch_bounds__(64, 1): one wave per SIMD unlocks the full 512-VGPR
+// budget per lane — the pairing's working set (~110 u64 live values) spills
+// ~5.4 KB/lane to scratch at the default budget, and scratch waits were
+// ~74% of wave time (profiles/pmc_r01.txt). Occupancy is irrelevant here:
+// the kernel is latency-bound per lane, not bandwidth- or wave-limited.
+extern "C" __global__ void __launch_bounds__(64, 1)
 k1_bls_verify(
     const uint8_t* __restrict__ vks,     // [N][128]
     const uint8_t* __restrict__ sigs,    // [N][64]
