@@ -106,6 +106,9 @@ def main() -> None:
                         "mixed: 50%% direct + 50%% broadcast, users spread over topics "
                         "(config 4: 64 KiB payloads at -100k clients)")
     p.add_argument("--ring-kb", type=int, default=0, help="override per-user ring size (KiB)")
+    p.add_argument("--population", choices=["all", "modulo"], default="all",
+                   help="all: every subscriber on every benched topic (dense fan-out); "
+                        "modulo: user u subscribes only to topic u%%topics (sparse)")
     p.add_argument("--graph", action="store_true",
                    help="hipGraph-capture the tick (measured ~2%% slower than "
                         "eager at this kernel count; kept for A/B)")
@@ -163,6 +166,8 @@ def main() -> None:
         eng.register_direct_bulk(
             (f"user-{u:08d}".encode(), u) for u in range(n_local_users)
         )
+    elif args.population == "modulo":
+        eng.subscribe_modulo(args.topics)
     else:
         eng.subscribe_all(list(range(args.topics)))
 
@@ -253,6 +258,8 @@ def main() -> None:
         deliveries_per_step = world_size * (
             (args.batch // 2) * (n_local_users // max(1, args.topics)) + args.batch // 2
         )
+    elif args.population == "modulo":
+        deliveries_per_step = world_size * args.batch * (n_local_users // max(1, args.topics))
     else:
         deliveries_per_step = world_size * args.batch * n_local_users
 
@@ -282,6 +289,7 @@ def main() -> None:
                 "deliveries_per_sec_node": deliveries_per_step * args.steps / elapsed,
                 "p50_e2e_latency_ms": p50_ms,
                 "mode": args.mode,
+                "population": args.population,
                 "drops": int(eng._drops.cpu()[0]) if not use_cpu else 0,
                 "egress_hbm_gb": round(n_local_users * ring_bytes / 2**30, 1),
                 "parallelism": f"mesh{world_size} (RCCL all-gather over xGMI)" if world_size > 1 else "single-broker",
