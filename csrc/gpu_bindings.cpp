@@ -39,15 +39,6 @@ void launch_k1_hash_to_g1(uint8_t*, const int64_t*, int32_t, uint8_t*, hipStream
 void launch_k3_fanout_wave(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
                            const int32_t*, const int64_t*, const uint32_t*, const int32_t*,
                            uint8_t*, int, int, hipStream_t);
-void launch_k2b_emit_capped(const uint64_t*, const int64_t*, const int32_t*, const int32_t*,
-                            int32_t, int32_t, int32_t, int64_t, int32_t, uint64_t*, int32_t*,
-                            int32_t*, int64_t*, uint32_t*, hipStream_t);
-void launch_k3_fanout_flat(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
-                           const int32_t*, const int64_t*, const uint32_t*, const int32_t*,
-                           int32_t, uint8_t*, int, int, hipStream_t);
-void launch_k2b_fused(const uint64_t*, const int32_t*, int32_t, int32_t, int32_t, int64_t,
-                      int32_t, uint64_t*, int32_t*, int32_t*, int32_t*, int64_t*, uint32_t*,
-                      hipStream_t);
 void launch_k3_fanout_flat2(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
                             const int32_t*, const int64_t*, uint32_t, const int32_t*, int32_t,
                             int32_t, uint8_t*, int, int, hipStream_t);
@@ -212,33 +203,6 @@ torch::Tensor hash_to_g1_batch(torch::Tensor msgs, torch::Tensor moff) {
     return out;
 }
 
-// Sync-free pipeline: counts -> cumsum -> capacity-guarded emit into
-// PREALLOCATED pair buffers; the pair count stays on-device (n_pairs_out).
-void assign_emit_into(torch::Tensor mask, torch::Tensor payload_off, torch::Tensor payload_len,
-                      torch::Tensor ring_wpos, int64_t ring_bytes, int64_t n_users,
-                      torch::Tensor pair_user, torch::Tensor pair_msg, torch::Tensor pair_dst,
-                      torch::Tensor drops, torch::Tensor n_pairs_out) {
-    CHECK_DEV(mask); CHECK_CONTIG(mask);
-    int32_t M = (int32_t)mask.size(0);
-    int32_t W = (int32_t)mask.size(1);
-    TORCH_CHECK(ring_bytes % 16 == 0, "ring_bytes must be a multiple of 16");
-    int32_t capacity = (int32_t)pair_user.size(0);
-    auto o32 = torch::TensorOptions().dtype(torch::kInt32).device(mask.device());
-    auto counts = torch::zeros({n_users}, o32);
-    launch_k2b_count((const uint64_t*)mask.data_ptr<int64_t>(), M, W, (int32_t)n_users,
-                     counts.data_ptr<int32_t>(), cur_stream());
-    auto cum = counts.cumsum(0).to(torch::kInt32);
-    auto pair_base = (cum - counts).contiguous();
-    n_pairs_out.copy_(cum.narrow(0, n_users - 1, 1).clamp_max((int64_t)capacity));
-    launch_k2b_emit_capped((const uint64_t*)mask.data_ptr<int64_t>(),
-                           payload_off.data_ptr<int64_t>(), payload_len.data_ptr<int32_t>(),
-                           pair_base.data_ptr<int32_t>(), M, W, (int32_t)n_users, ring_bytes,
-                           capacity, (uint64_t*)ring_wpos.data_ptr<int64_t>(),
-                           pair_user.data_ptr<int32_t>(), pair_msg.data_ptr<int32_t>(),
-                           pair_dst.data_ptr<int64_t>(), (uint32_t*)drops.data_ptr<int32_t>(),
-                           cur_stream());
-}
-
 void fanout_wave(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_len,
                  torch::Tensor pair_user, torch::Tensor pair_msg, torch::Tensor pair_dst,
                  torch::Tensor msg_seq, torch::Tensor n_pairs, torch::Tensor egress,
@@ -250,36 +214,6 @@ void fanout_wave(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor pay
                           (const uint32_t*)msg_seq.data_ptr<int32_t>(),
                           n_pairs.data_ptr<int32_t>(), egress.data_ptr<uint8_t>(), (int)nt,
                           (int)grid, cur_stream());
-}
-
-void fanout_flat(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_len,
-                 torch::Tensor pair_user, torch::Tensor pair_msg, torch::Tensor pair_dst,
-                 torch::Tensor msg_seq, torch::Tensor n_pairs, int64_t units_per_pair,
-                 torch::Tensor egress, int64_t nt, int64_t grid) {
-    CHECK_DEV(egress); CHECK_CONTIG(egress);
-    launch_k3_fanout_flat(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
-                          payload_len.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
-                          pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
-                          (const uint32_t*)msg_seq.data_ptr<int32_t>(),
-                          n_pairs.data_ptr<int32_t>(), (int32_t)units_per_pair,
-                          egress.data_ptr<uint8_t>(), (int)nt, (int)grid, cur_stream());
-}
-
-void assign_emit_fused(torch::Tensor mask, torch::Tensor payload_len, torch::Tensor ring_wpos,
-                       int64_t ring_bytes, int64_t n_users, torch::Tensor pair_user,
-                       torch::Tensor pair_msg, torch::Tensor pair_dst, torch::Tensor drops,
-                       torch::Tensor n_pairs) {
-    CHECK_DEV(mask); CHECK_CONTIG(mask);
-    int32_t M = (int32_t)mask.size(0);
-    int32_t W = (int32_t)mask.size(1);
-    TORCH_CHECK(ring_bytes % 16 == 0);
-    int32_t capacity = (int32_t)pair_user.size(0);
-    launch_k2b_fused((const uint64_t*)mask.data_ptr<int64_t>(), payload_len.data_ptr<int32_t>(),
-                     M, W, (int32_t)n_users, ring_bytes, capacity,
-                     (uint64_t*)ring_wpos.data_ptr<int64_t>(), n_pairs.data_ptr<int32_t>(),
-                     pair_user.data_ptr<int32_t>(), pair_msg.data_ptr<int32_t>(),
-                     pair_dst.data_ptr<int64_t>(), (uint32_t*)drops.data_ptr<int32_t>(),
-                     cur_stream());
 }
 
 void fanout_flat2(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_len,
@@ -374,12 +308,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("apply_subs", &apply_subs, "K2c: apply subscribe/unsubscribe batch to bitmap");
     m.def("bls_verify_batch", &bls_verify_batch, "K1: batched BLS-over-BN254 verification");
     m.def("hash_to_g1_batch", &hash_to_g1_batch, "K1 helper: batched hash-to-G1");
-    m.def("assign_emit_into", &assign_emit_into,
-          "K2b sync-free: emit pairs into preallocated buffers, count stays on device");
     m.def("fanout_wave", &fanout_wave, "K3v2: wave-per-pair fan-out (nt flag, device count)");
-    m.def("fanout_flat", &fanout_flat, "K3v3: flat-index fan-out for uniform record sizes");
-    m.def("assign_emit_fused", &assign_emit_fused,
-          "K2b-fused: count+claim+emit in one kernel (atomic slot claim)");
     m.def("fanout_flat2", &fanout_flat2, "K3v4: flat fan-out, seq from base, capacity clamp");
     m.def("fanout_flat3", &fanout_flat3, "K3v5: graph-capturable (device seq counter)");
     m.def("seq_advance", &seq_advance, "bump the device seq counter (inside the graph)");

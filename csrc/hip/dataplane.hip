@@ -564,50 +564,6 @@ __global__ void __launch_bounds__(256) k3_fanout_wave_t(
     }
 }
 
-// K2b capacity-guarded emit: same per-user ordered scan, but drops (and
-// counts) deliveries whose pair slot would exceed `capacity` so the host can
-// preallocate fixed buffers and never sync on the exact count.
-extern "C" __global__ void k2b_emit_capped(
-    const uint64_t* __restrict__ mask,
-    const int64_t* __restrict__ payload_off,
-    const int32_t* __restrict__ payload_len,
-    const int32_t* __restrict__ pair_base,
-    int32_t M, int32_t W, int32_t n_users,
-    int64_t ring_bytes, int32_t capacity,
-    uint64_t* __restrict__ ring_wpos,
-    int32_t* __restrict__ pair_user,
-    int32_t* __restrict__ pair_msg,
-    int64_t* __restrict__ pair_dst,
-    uint32_t* __restrict__ drops)
-{
-    int u = blockIdx.x * blockDim.x + threadIdx.x;
-    if (u >= n_users) return;
-    int w = u >> 6;
-    uint64_t bit = 1ull << (u & 63);
-    int slot = pair_base[u];
-    uint64_t wpos = ring_wpos[u];
-    uint32_t dropped = 0;
-    for (int m = 0; m < M; ++m) {
-        if (!(mask[(int64_t)m * W + w] & bit)) continue;
-        int32_t len = payload_len[m];
-        uint64_t rec = 16 + (uint64_t)((len + 15) & ~15);
-        bool fits_ring = (wpos + rec <= (uint64_t)ring_bytes);
-        bool fits_cap = (slot < capacity);
-        if (!fits_ring || !fits_cap) {
-            if (fits_cap) { pair_user[slot] = -1; pair_msg[slot] = m; pair_dst[slot] = 0; slot++; }
-            dropped++;
-            continue;
-        }
-        pair_user[slot] = u;
-        pair_msg[slot] = m;
-        pair_dst[slot] = (int64_t)u * ring_bytes + wpos;
-        slot++;
-        wpos += rec;
-    }
-    ring_wpos[u] = wpos;
-    if (dropped) atomicAdd(drops, dropped);
-}
-
 extern "C" {
 
 void launch_k3_fanout_wave(const uint8_t* buf, const int64_t* payload_off,
@@ -626,28 +582,20 @@ void launch_k3_fanout_wave(const uint8_t* buf, const int64_t* payload_off,
                            egress);
 }
 
-void launch_k2b_emit_capped(const uint64_t* mask, const int64_t* payload_off,
-                            const int32_t* payload_len, const int32_t* pair_base, int32_t M,
-                            int32_t W, int32_t n_users, int64_t ring_bytes, int32_t capacity,
-                            uint64_t* ring_wpos, int32_t* pair_user, int32_t* pair_msg,
-                            int64_t* pair_dst, uint32_t* drops, hipStream_t s) {
-    int threads = 256, blocks = (n_users + threads - 1) / threads;
-    hipLaunchKernelGGL(k2b_emit_capped, dim3(blocks), dim3(threads), 0, s, mask, payload_off,
-                       payload_len, pair_base, M, W, n_users, ring_bytes, capacity, ring_wpos,
-                       pair_user, pair_msg, pair_dst, drops);
-}
 
 }  // extern "C"
 
 // ---------------------------------------------------------------------------
-// K3v3: flat-index fan-out for UNIFORM record sizes (every message in the
-// tick has the same wire length — the common case for batched pub/sub
-// traffic). Work unit = one 16 B chunk of one delivery record; lane ->
-// consecutive flat units, so lane utilization is ~100% regardless of
-// payload size (the wave-per-pair variant idles lanes on the tail pass).
-// units_per_pair = record_bytes / 16 (header unit + payload units).
+// K3 flat fan-out for UNIFORM record sizes: work unit = one 16 B chunk of
+// one delivery record, lane -> consecutive flat units (~100% lane
+// utilization at any payload size; the wave-per-pair variant idles lanes on
+// the tail pass). Two entry points share the template:
+//   flat2: seq base passed by value (eager path)
+//   flat3: seq base read from a device counter (hipGraph-capturable; a
+//          host-passed base would be frozen into the captured graph)
+// n_pairs is read from a device pointer and clamped to the buffer capacity.
 // ---------------------------------------------------------------------------
-template <int NT>
+template <int NT, bool SEQ_FROM_PTR>
 __global__ void __launch_bounds__(256) k3_fanout_flat_t(
     const uint8_t* __restrict__ buf,
     const int64_t* __restrict__ payload_off,
@@ -655,13 +603,18 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
     const int32_t* __restrict__ pair_user,
     const int32_t* __restrict__ pair_msg,
     const int64_t* __restrict__ pair_dst,
-    const uint32_t* __restrict__ msg_seq,
+    uint32_t seq_base_val,
+    const uint32_t* __restrict__ seq_state,
     const int32_t* __restrict__ n_pairs_ptr,
+    int32_t capacity,
     int32_t units_per_pair,
     uint8_t* __restrict__ egress)
 {
     typedef unsigned int v4u __attribute__((ext_vector_type(4)));
-    const int64_t n_units = (int64_t)(*n_pairs_ptr) * units_per_pair;
+    const uint32_t seq_base = SEQ_FROM_PTR ? seq_state[0] : seq_base_val;
+    int np = *n_pairs_ptr;
+    if (np > capacity) np = capacity;
+    const int64_t n_units = (int64_t)np * units_per_pair;
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < n_units; f += stride) {
         const int p = (int)(f / units_per_pair);
@@ -671,7 +624,7 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
         const int mi = pair_msg[p];
         uint8_t* dst = egress + pair_dst[p] + (size_t)unit * 16;
         if (unit == 0) {
-            uint32_t hdr[4] = {(uint32_t)payload_len[mi], msg_seq[mi], 0, 0};
+            uint32_t hdr[4] = {(uint32_t)payload_len[mi], seq_base + (uint32_t)mi, 0, 0};
             v4u h; memcpy(&h, hdr, 16);
             if (NT) __builtin_nontemporal_store(h, (v4u*)dst);
             else memcpy(dst, hdr, 16);
@@ -690,207 +643,36 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
     }
 }
 
-extern "C" void launch_k3_fanout_flat(
+extern "C" void launch_k3_fanout_flat2(
     const uint8_t* buf, const int64_t* payload_off, const int32_t* payload_len,
     const int32_t* pair_user, const int32_t* pair_msg, const int64_t* pair_dst,
-    const uint32_t* msg_seq, const int32_t* n_pairs_ptr, int32_t units_per_pair,
+    uint32_t seq_base, const int32_t* n_pairs_ptr, int32_t capacity, int32_t units_per_pair,
     uint8_t* egress, int nt, int grid, hipStream_t s) {
     if (grid <= 0) grid = 8192;
     if (nt)
-        hipLaunchKernelGGL((k3_fanout_flat_t<1>), dim3(grid), dim3(256), 0, s, buf, payload_off,
-                           payload_len, pair_user, pair_msg, pair_dst, msg_seq, n_pairs_ptr,
-                           units_per_pair, egress);
+        hipLaunchKernelGGL((k3_fanout_flat_t<1, false>), dim3(grid), dim3(256), 0, s, buf,
+                           payload_off, payload_len, pair_user, pair_msg, pair_dst, seq_base,
+                           nullptr, n_pairs_ptr, capacity, units_per_pair, egress);
     else
-        hipLaunchKernelGGL((k3_fanout_flat_t<0>), dim3(grid), dim3(256), 0, s, buf, payload_off,
-                           payload_len, pair_user, pair_msg, pair_dst, msg_seq, n_pairs_ptr,
-                           units_per_pair, egress);
+        hipLaunchKernelGGL((k3_fanout_flat_t<0, false>), dim3(grid), dim3(256), 0, s, buf,
+                           payload_off, payload_len, pair_user, pair_msg, pair_dst, seq_base,
+                           nullptr, n_pairs_ptr, capacity, units_per_pair, egress);
 }
 
-// ---------------------------------------------------------------------------
-// K2b-fused: count + slot claim + emit in ONE kernel (replaces
-// k2b_count + torch cumsum + k2b_emit_capped — pair-list GROUPING is
-// irrelevant to K3 correctness, only per-user ring order matters, so a
-// global atomic slot claim replaces the exclusive scan).
-//   pass 1: popcount this user's mask column
-//   claim:  base = atomicAdd(n_pairs, count)
-//   pass 2: walk messages in order, assign ring offsets, write pairs
-// n_pairs must be zeroed by the host before the launch; K3 clamps its read
-// of n_pairs to the buffer capacity.
-// ---------------------------------------------------------------------------
-extern "C" __global__ void k2b_fused(
-    const uint64_t* __restrict__ mask,
-    const int32_t* __restrict__ payload_len,
-    int32_t M, int32_t W, int32_t n_users,
-    int64_t ring_bytes, int32_t capacity,
-    uint64_t* __restrict__ ring_wpos,
-    int32_t* __restrict__ n_pairs,          // [1] global pair counter (zeroed)
-    int32_t* __restrict__ pair_user,
-    int32_t* __restrict__ pair_msg,
-    int64_t* __restrict__ pair_dst,
-    uint32_t* __restrict__ drops)
-{
-    int u = blockIdx.x * blockDim.x + threadIdx.x;
-    if (u >= n_users) return;
-    int w = u >> 6;
-    uint64_t bit = 1ull << (u & 63);
-    int count = 0;
-#pragma unroll 8
-    for (int m = 0; m < M; ++m) count += (mask[(int64_t)m * W + w] & bit) ? 1 : 0;
-    if (count == 0) return;
-    int slot = atomicAdd(n_pairs, count);
-    uint64_t wpos = ring_wpos[u];
-    uint32_t dropped = 0;
-    for (int m = 0; m < M; ++m) {
-        if (!(mask[(int64_t)m * W + w] & bit)) continue;
-        int32_t len = payload_len[m];
-        uint64_t rec = 16 + (uint64_t)((len + 15) & ~15);
-        bool fits_ring = (wpos + rec <= (uint64_t)ring_bytes);
-        bool fits_cap = (slot < capacity);
-        if (!fits_ring || !fits_cap) {
-            if (fits_cap) { pair_user[slot] = -1; pair_msg[slot] = m; pair_dst[slot] = 0; slot++; }
-            dropped++;
-            continue;
-        }
-        pair_user[slot] = u;
-        pair_msg[slot] = m;
-        pair_dst[slot] = (int64_t)u * ring_bytes + wpos;
-        slot++;
-        wpos += rec;
-    }
-    ring_wpos[u] = wpos;
-    if (dropped) atomicAdd(drops, dropped);
-}
-
-// K3 flat variant with seq derived from a base (no msg_seq tensor) and
-// n_pairs clamped to capacity.
-template <int NT>
-__global__ void __launch_bounds__(256) k3_fanout_flat2_t(
-    const uint8_t* __restrict__ buf,
-    const int64_t* __restrict__ payload_off,
-    const int32_t* __restrict__ payload_len,
-    const int32_t* __restrict__ pair_user,
-    const int32_t* __restrict__ pair_msg,
-    const int64_t* __restrict__ pair_dst,
-    uint32_t seq_base,
-    const int32_t* __restrict__ n_pairs_ptr,
-    int32_t capacity,
-    int32_t units_per_pair,
-    uint8_t* __restrict__ egress)
-{
-    typedef unsigned int v4u __attribute__((ext_vector_type(4)));
-    int np = *n_pairs_ptr;
-    if (np > capacity) np = capacity;
-    const int64_t n_units = (int64_t)np * units_per_pair;
-    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < n_units; f += stride) {
-        const int p = (int)(f / units_per_pair);
-        const int unit = (int)(f - (int64_t)p * units_per_pair);
-        const int u = pair_user[p];
-        if (u < 0) continue;
-        const int mi = pair_msg[p];
-        uint8_t* dst = egress + pair_dst[p] + (size_t)unit * 16;
-        if (unit == 0) {
-            uint32_t hdr[4] = {(uint32_t)payload_len[mi], seq_base + (uint32_t)mi, 0, 0};
-            v4u h; memcpy(&h, hdr, 16);
-            if (NT) __builtin_nontemporal_store(h, (v4u*)dst);
-            else memcpy(dst, hdr, 16);
-            continue;
-        }
-        const uint8_t* src = buf + payload_off[mi] + (size_t)(unit - 1) * 16;
-        const int32_t len = payload_len[mi];
-        const int32_t coff = (unit - 1) * 16;
-        if (coff + 16 <= len && (((uintptr_t)src) & 15) == 0) {
-            v4u v = *(const v4u*)src;
-            if (NT) __builtin_nontemporal_store(v, (v4u*)dst);
-            else *(v4u*)dst = v;
-        } else {
-            for (int b = 0; b < 16 && coff + b < len; ++b) dst[b] = src[b];
-        }
-    }
-}
-
-extern "C" {
-
-void launch_k2b_fused(const uint64_t* mask, const int32_t* payload_len, int32_t M, int32_t W,
-                      int32_t n_users, int64_t ring_bytes, int32_t capacity,
-                      uint64_t* ring_wpos, int32_t* n_pairs, int32_t* pair_user,
-                      int32_t* pair_msg, int64_t* pair_dst, uint32_t* drops, hipStream_t s) {
-    int threads = 256, blocks = (n_users + threads - 1) / threads;
-    hipLaunchKernelGGL(k2b_fused, dim3(blocks), dim3(threads), 0, s, mask, payload_len, M, W,
-                       n_users, ring_bytes, capacity, ring_wpos, n_pairs, pair_user, pair_msg,
-                       pair_dst, drops);
-}
-
-void launch_k3_fanout_flat2(const uint8_t* buf, const int64_t* payload_off,
-                            const int32_t* payload_len, const int32_t* pair_user,
-                            const int32_t* pair_msg, const int64_t* pair_dst, uint32_t seq_base,
-                            const int32_t* n_pairs_ptr, int32_t capacity,
-                            int32_t units_per_pair, uint8_t* egress, int nt, int grid,
-                            hipStream_t s) {
+extern "C" void launch_k3_fanout_flat3(
+    const uint8_t* buf, const int64_t* payload_off, const int32_t* payload_len,
+    const int32_t* pair_user, const int32_t* pair_msg, const int64_t* pair_dst,
+    const uint32_t* seq_state, const int32_t* n_pairs_ptr, int32_t capacity,
+    int32_t units_per_pair, uint8_t* egress, int nt, int grid, hipStream_t s) {
     if (grid <= 0) grid = 8192;
     if (nt)
-        hipLaunchKernelGGL((k3_fanout_flat2_t<1>), dim3(grid), dim3(256), 0, s, buf,
-                           payload_off, payload_len, pair_user, pair_msg, pair_dst, seq_base,
-                           n_pairs_ptr, capacity, units_per_pair, egress);
+        hipLaunchKernelGGL((k3_fanout_flat_t<1, true>), dim3(grid), dim3(256), 0, s, buf,
+                           payload_off, payload_len, pair_user, pair_msg, pair_dst, 0u,
+                           seq_state, n_pairs_ptr, capacity, units_per_pair, egress);
     else
-        hipLaunchKernelGGL((k3_fanout_flat2_t<0>), dim3(grid), dim3(256), 0, s, buf,
-                           payload_off, payload_len, pair_user, pair_msg, pair_dst, seq_base,
-                           n_pairs_ptr, capacity, units_per_pair, egress);
-}
-
-}  // extern "C"
-
-// ---------------------------------------------------------------------------
-// hipGraph-capturable variants: the per-tick message sequence base lives in
-// a DEVICE counter (seq_state[0]) so the whole tick can be captured once and
-// replayed (a host-passed seq base would be frozen into the graph).
-// k_seq_advance bumps the counter at the end of the tick.
-// ---------------------------------------------------------------------------
-template <int NT>
-__global__ void __launch_bounds__(256) k3_fanout_flat3_t(
-    const uint8_t* __restrict__ buf,
-    const int64_t* __restrict__ payload_off,
-    const int32_t* __restrict__ payload_len,
-    const int32_t* __restrict__ pair_user,
-    const int32_t* __restrict__ pair_msg,
-    const int64_t* __restrict__ pair_dst,
-    const uint32_t* __restrict__ seq_state,   // [0] = seq base (device counter)
-    const int32_t* __restrict__ n_pairs_ptr,
-    int32_t capacity,
-    int32_t units_per_pair,
-    uint8_t* __restrict__ egress)
-{
-    typedef unsigned int v4u __attribute__((ext_vector_type(4)));
-    const uint32_t seq_base = seq_state[0];
-    int np = *n_pairs_ptr;
-    if (np > capacity) np = capacity;
-    const int64_t n_units = (int64_t)np * units_per_pair;
-    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < n_units; f += stride) {
-        const int p = (int)(f / units_per_pair);
-        const int unit = (int)(f - (int64_t)p * units_per_pair);
-        const int u = pair_user[p];
-        if (u < 0) continue;
-        const int mi = pair_msg[p];
-        uint8_t* dst = egress + pair_dst[p] + (size_t)unit * 16;
-        if (unit == 0) {
-            uint32_t hdr[4] = {(uint32_t)payload_len[mi], seq_base + (uint32_t)mi, 0, 0};
-            v4u h; memcpy(&h, hdr, 16);
-            if (NT) __builtin_nontemporal_store(h, (v4u*)dst);
-            else memcpy(dst, hdr, 16);
-            continue;
-        }
-        const uint8_t* src = buf + payload_off[mi] + (size_t)(unit - 1) * 16;
-        const int32_t len = payload_len[mi];
-        const int32_t coff = (unit - 1) * 16;
-        if (coff + 16 <= len && (((uintptr_t)src) & 15) == 0) {
-            v4u v = *(const v4u*)src;
-            if (NT) __builtin_nontemporal_store(v, (v4u*)dst);
-            else *(v4u*)dst = v;
-        } else {
-            for (int b = 0; b < 16 && coff + b < len; ++b) dst[b] = src[b];
-        }
-    }
+        hipLaunchKernelGGL((k3_fanout_flat_t<0, true>), dim3(grid), dim3(256), 0, s, buf,
+                           payload_off, payload_len, pair_user, pair_msg, pair_dst, 0u,
+                           seq_state, n_pairs_ptr, capacity, units_per_pair, egress);
 }
 
 extern "C" __global__ void k_seq_advance(uint32_t* seq_state, int32_t m) {
@@ -899,22 +681,6 @@ extern "C" __global__ void k_seq_advance(uint32_t* seq_state, int32_t m) {
 
 extern "C" {
 
-void launch_k3_fanout_flat3(const uint8_t* buf, const int64_t* payload_off,
-                            const int32_t* payload_len, const int32_t* pair_user,
-                            const int32_t* pair_msg, const int64_t* pair_dst,
-                            const uint32_t* seq_state, const int32_t* n_pairs_ptr,
-                            int32_t capacity, int32_t units_per_pair, uint8_t* egress, int nt,
-                            int grid, hipStream_t s) {
-    if (grid <= 0) grid = 8192;
-    if (nt)
-        hipLaunchKernelGGL((k3_fanout_flat3_t<1>), dim3(grid), dim3(256), 0, s, buf,
-                           payload_off, payload_len, pair_user, pair_msg, pair_dst, seq_state,
-                           n_pairs_ptr, capacity, units_per_pair, egress);
-    else
-        hipLaunchKernelGGL((k3_fanout_flat3_t<0>), dim3(grid), dim3(256), 0, s, buf,
-                           payload_off, payload_len, pair_user, pair_msg, pair_dst, seq_state,
-                           n_pairs_ptr, capacity, units_per_pair, egress);
-}
 
 void launch_k_seq_advance(uint32_t* seq_state, int32_t m, hipStream_t s) {
     hipLaunchKernelGGL(k_seq_advance, dim3(1), dim3(1), 0, s, seq_state, m);
