@@ -46,13 +46,18 @@ from pushcdn_amd.proto import message as msglib
 
 
 def build_mixed_batch(n_msgs, payload_bytes, n_topics, n_users, seed):
-    """Alternating Direct/Broadcast batch (config 4). Per-position wire sizes
-    are deterministic (fixed-width recipient keys), so the offsets table is
-    identical on every rank and one device offsets tensor serves every
-    mesh-exchanged batch. Non-uniform across positions -> wave-per-pair K3."""
+    """Alternating Direct/Broadcast batch (config 4). BOTH message types are
+    padded to one common wire size (Direct carries a recipient key, so it is
+    a few words larger than Broadcast), which makes the whole batch uniform:
+    identical offsets on every rank AND the flat ~100%-lane-utilization K3
+    path instead of wave-per-pair. Trailing zero padding is inert to the
+    capnp parser (it reads only the declared segment)."""
     import random
 
     rng = random.Random(seed)
+    probe_b = msglib.serialize(msglib.Broadcast([0], bytes(payload_bytes)))
+    probe_d = msglib.serialize(msglib.Direct(b"user-00000000", bytes(payload_bytes)))
+    wire_len = (max(len(probe_b), len(probe_d)) + 15) & ~15
     buf = bytearray()
     offsets = [0]
     for i in range(n_msgs):
@@ -62,10 +67,10 @@ def build_mixed_batch(n_msgs, payload_bytes, n_topics, n_users, seed):
         else:
             msg = msglib.Direct(f"user-{rng.randrange(n_users):08d}".encode(), payload)
         raw = msglib.serialize(msg)
-        padded = (len(raw) + 15) & ~15
-        buf += raw + b"\x00" * (padded - len(raw))
+        assert len(raw) <= wire_len
+        buf += raw + b"\x00" * (wire_len - len(raw))
         offsets.append(len(buf))
-    return bytes(buf), offsets, None
+    return bytes(buf), offsets, wire_len
 
 
 def build_batch(n_msgs: int, payload_bytes: int, n_topics: int, seed: int):
