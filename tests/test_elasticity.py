@@ -116,3 +116,36 @@ def test_user_evicted_on_dead_connection(tmp_path):
         await stop_stack([broker], marshal, client)
 
     run(go())
+
+
+def test_concurrent_sends_single_reconnect(tmp_path):
+    """Concurrent sends during a dead connection trigger exactly ONE
+    reconnect (the connecting-guard semaphore, reference lib.rs:204-258)."""
+    async def go():
+        db = new_db(tmp_path)
+        broker = make_broker(db)
+        await broker.start()
+        await broker.discovery.perform_heartbeat(0, 60)
+        marshal, endpoint = make_marshal(db)
+        await marshal.start()
+        client = make_client(endpoint, seed=41, topics=[0])
+        await client.ensure_initialized()
+
+        connects = 0
+        orig = client._connect_once
+
+        async def counting_connect():
+            nonlocal connects
+            connects += 1
+            return await orig()
+
+        client._connect_once = counting_connect
+        client._disconnect_on_error()  # force a dead connection
+        await asyncio.gather(*(
+            client.send_broadcast_message([0], f"c{i}".encode()) for i in range(8)
+        ))
+        assert connects == 1, f"expected one reconnect, got {connects}"
+        assert len(broker.connections.users) == 1
+        await stop_stack([broker], marshal, client)
+
+    run(go())

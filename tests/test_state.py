@@ -219,3 +219,25 @@ def test_global_permits_feature(tmp_path):
         assert await d.validate_permit(id_(2), permit) is None
 
     asyncio.run(go())
+
+
+def test_versioned_tombstone_purged_after_diff():
+    """Tombstones ship once and are purged (reference versioned_map purge)."""
+    vm = VersionedMap("a")
+    vm.insert(b"k", "v")
+    vm.diff()
+    vm.remove(b"k")
+    d = vm.diff()
+    assert d[b"k"].value is None
+    # purged: a full sync no longer carries the tombstone
+    assert b"k" not in vm.get_full()
+    assert len(vm) == 0
+
+
+def test_relational_map_reassociate_after_remove():
+    rm = RelationalMap()
+    rm.associate_key_with_values(b"u", [1])
+    rm.remove_key(b"u")
+    rm.associate_key_with_values(b"u", [2])
+    assert rm.get_values_by_key(b"u") == {2}
+    assert rm.get_keys_by_value(1) == set()
