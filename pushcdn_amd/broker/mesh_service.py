@@ -68,6 +68,7 @@ class MeshBroker(Broker):
         super().__init__(config)
         self.batch_capacity = batch_capacity
         self.mesh: Optional[RcclMesh] = None
+        self._carry: List[Bytes] = []  # messages that didn't fit last tick
 
     # the framed broker mesh is replaced by the collective: no dialing, no
     # framed sync blasts; heartbeats still publish load for the marshal
@@ -99,9 +100,20 @@ class MeshBroker(Broker):
             else None
         )
         while True:
-            batch: List[Bytes] = []
+            # collect up to a capacity-bounded batch; an oversize tick must
+            # NEVER raise here — that would stall every peer's collective
+            batch: List[Bytes] = self._carry
+            self._carry = []
+            used = sum((len(r.data) + 15) & ~15 for r in batch)
+            budget = self.batch_capacity - 16 * 4096  # header headroom
             while not self._gpu_queue.empty() and len(batch) < 4096:
-                batch.append(self._gpu_queue.get_nowait()[0])
+                raw = self._gpu_queue.get_nowait()[0]
+                padded = (len(raw.data) + 15) & ~15
+                if used + padded > budget:
+                    self._carry.append(raw)  # next tick
+                    break
+                batch.append(raw)
+                used += padded
             msgs = [raw.data for raw in batch]
             host_buf, n_local, _used = pack_mesh_batch(msgs, self.batch_capacity)
             if dev_buf is not None:
