@@ -202,22 +202,37 @@ def main() -> None:
     from pushcdn_amd.parallel.mesh import RcclMesh
 
     mesh = RcclMesh(torch.device(device), batch_capacity=cap)
-    dev_buf = None if use_cpu else torch.zeros(cap, dtype=torch.uint8, device=device)
     use_graph = (not use_cpu) and args.graph and args.mode == "broadcast"
+
+    # Double-buffered ingest: the NEXT step's H2D copy runs on a side stream
+    # while this step's kernels execute (the socket-read/compute overlap a
+    # real broker gets from its per-connection reader tasks).
+    if not use_cpu:
+        dev_bufs = [torch.zeros(cap, dtype=torch.uint8, device=device) for _ in range(2)]
+        copy_stream = torch.cuda.Stream(device=device)
+        copy_done = [torch.cuda.Event(), torch.cuda.Event()]
+        tick_done = [torch.cuda.Event(), torch.cuda.Event()]
+        with torch.cuda.stream(copy_stream):
+            dev_bufs[0].copy_(pinned[0], non_blocking=True)
+            copy_done[0].record(copy_stream)
 
     def step(i: int) -> None:
         v = i % n_variants
-        src = pinned[v]
         if use_cpu:
-            buf = src
+            buf = pinned[v]
         else:
-            dev_buf.copy_(src, non_blocking=True)  # fixed-address ingest buffer
-            buf = dev_buf
+            cur, nxt = i % 2, (i + 1) % 2
+            torch.cuda.current_stream().wait_event(copy_done[cur])
+            with torch.cuda.stream(copy_stream):
+                copy_stream.wait_event(tick_done[nxt])  # buffer free?
+                dev_bufs[nxt].copy_(pinned[(i + 1) % n_variants], non_blocking=True)
+                copy_done[nxt].record(copy_stream)
+            buf = dev_bufs[cur]
         # broker->broker mesh: all-gather this tick's batches over xGMI
         for r, view, n_msgs, nbytes in mesh.exchange(buf, args.batch, len(host_batches[v][0])):
             if use_graph:
                 # hipGraph-captured tick (captured per fixed buffer; the
-                # gathered views and dev_buf are stable addresses)
+                # gathered views and ingest buffers are stable addresses)
                 eng.tick_graphed(view, dev_offsets, wire_len)
             else:
                 eng.tick(
@@ -227,6 +242,8 @@ def main() -> None:
                     host_offsets=None if not use_cpu else host_batches[v][1],
                     uniform_wire_len=wire_len,
                 )
+        if not use_cpu:
+            tick_done[i % 2].record(torch.cuda.current_stream())
         eng.drain_cursors()
 
     def barrier_sync() -> None:
