@@ -114,6 +114,8 @@ def main() -> None:
     p.add_argument("--population", choices=["all", "modulo"], default="all",
                    help="all: every subscriber on every benched topic (dense fan-out); "
                         "modulo: user u subscribes only to topic u%%topics (sparse)")
+    p.add_argument("--no-overlap", action="store_true",
+                   help="disable the double-buffered ingest overlap (A/B)")
     p.add_argument("--graph", action="store_true",
                    help="hipGraph-capture the tick (measured ~2%% slower than "
                         "eager at this kernel count; kept for A/B)")
@@ -222,11 +224,14 @@ def main() -> None:
             buf = pinned[v]
         else:
             cur, nxt = i % 2, (i + 1) % 2
-            torch.cuda.current_stream().wait_event(copy_done[cur])
-            with torch.cuda.stream(copy_stream):
-                copy_stream.wait_event(tick_done[nxt])  # buffer free?
-                dev_bufs[nxt].copy_(pinned[(i + 1) % n_variants], non_blocking=True)
-                copy_done[nxt].record(copy_stream)
+            if args.no_overlap:
+                dev_bufs[cur].copy_(pinned[v], non_blocking=True)
+            else:
+                torch.cuda.current_stream().wait_event(copy_done[cur])
+                with torch.cuda.stream(copy_stream):
+                    copy_stream.wait_event(tick_done[nxt])  # buffer free?
+                    dev_bufs[nxt].copy_(pinned[(i + 1) % n_variants], non_blocking=True)
+                    copy_done[nxt].record(copy_stream)
             buf = dev_bufs[cur]
         # broker->broker mesh: all-gather this tick's batches over xGMI
         for r, view, n_msgs, nbytes in mesh.exchange(buf, args.batch, len(host_batches[v][0])):
