@@ -12,6 +12,7 @@
 
 #include "bls/bls.h"
 #include "wire/message.h"
+#include "state/versioned_map.h"
 
 namespace py = pybind11;
 using namespace bn254;
@@ -278,6 +279,57 @@ static py::object w_deserialize(const py::bytes& raw) {
     return d;
 }
 
+// ---------------------------------------------------------------------------
+// native CRDT bindings (pushcdn_amd/broker/versioned_map.py delegates here)
+// ---------------------------------------------------------------------------
+class PyVersionedMap {
+  public:
+    explicit PyVersionedMap(const std::string& cid) : vm_(cid) {}
+    void insert(const py::bytes& k, const py::bytes& v) {
+        vm_.insert((std::string)k, (std::string)v);
+    }
+    void remove(const py::bytes& k) { vm_.remove((std::string)k); }
+    py::object get(const py::bytes& k) {
+        auto v = vm_.get((std::string)k);
+        if (!v) return py::none();
+        return py::bytes(*v);
+    }
+    size_t size() const { return vm_.size(); }
+    py::bytes diff() {
+        auto d = vm_.diff();
+        auto raw = state::VersionedMap::serialize_delta(d);
+        return py::bytes((const char*)raw.data(), raw.size());
+    }
+    py::bytes get_full() const {
+        auto raw = state::VersionedMap::serialize_delta(vm_.get_full());
+        return py::bytes((const char*)raw.data(), raw.size());
+    }
+    // merge a serialized delta; returns [(key, old|None, new|None), ...]
+    py::list merge(const py::bytes& delta) {
+        auto v = to_vec(delta);
+        std::map<std::string, state::Versioned> d;
+        if (!state::VersionedMap::deserialize_delta(v.data(), v.size(), &d))
+            throw std::invalid_argument("malformed delta");
+        py::list out;
+        for (const auto& c : vm_.merge(d)) {
+            out.append(py::make_tuple(
+                py::bytes(c.key),
+                c.old_value ? py::object(py::bytes(*c.old_value)) : py::object(py::none()),
+                c.new_value ? py::object(py::bytes(*c.new_value)) : py::object(py::none())));
+        }
+        return out;
+    }
+    py::list items() const {
+        py::list out;
+        for (const auto& [k, e] : vm_.get_full())
+            if (e.value) out.append(py::make_tuple(py::bytes(k), py::bytes(*e.value)));
+        return out;
+    }
+
+  private:
+    state::VersionedMap vm_;
+};
+
 PYBIND11_MODULE(pushcdn_core, m) {
     m.doc() = "pushcdn host core: BLS-over-BN254";
     m.def("keygen", &keygen, "deterministic BLS keypair from a u64 seed -> (sk, vk)");
@@ -303,4 +355,14 @@ PYBIND11_MODULE(pushcdn_core, m) {
     m.def("wire_serialize_topics", &w_ser_topics);
     m.def("wire_serialize_sync", &w_ser_sync);
     m.def("wire_deserialize", &w_deserialize);
+    py::class_<PyVersionedMap>(m, "VersionedMap")
+        .def(py::init<const std::string&>())
+        .def("insert", &PyVersionedMap::insert)
+        .def("remove", &PyVersionedMap::remove)
+        .def("get", &PyVersionedMap::get)
+        .def("__len__", &PyVersionedMap::size)
+        .def("diff", &PyVersionedMap::diff)
+        .def("get_full", &PyVersionedMap::get_full)
+        .def("merge", &PyVersionedMap::merge)
+        .def("items", &PyVersionedMap::items);
 }

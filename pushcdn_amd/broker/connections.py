@@ -26,6 +26,48 @@ Topic = int
 UserPubKey = bytes
 
 
+class SyncMap:
+    """bytes->bytes replicated map over the NATIVE CRDT (pushcdn_core
+    VersionedMap, csrc/state/versioned_map.h) with a pure-Python fallback.
+    Deltas travel as the documented length-prefixed encoding — the two
+    implementations are wire-compatible (cross-checked in test_state.py)."""
+
+    def __init__(self, conflict_id: str) -> None:
+        self._native = False
+        try:
+            from ..ops.build import build_core
+
+            self._m = build_core().VersionedMap(conflict_id)
+            self._native = True
+        except Exception:
+            self._m = VersionedMap(conflict_id)
+
+    def insert(self, key: bytes, value: bytes) -> None:
+        self._m.insert(key, value)
+
+    def remove(self, key: bytes) -> None:
+        self._m.remove(key)
+
+    def get(self, key: bytes) -> Optional[bytes]:
+        return self._m.get(key)
+
+    def diff_bytes(self) -> bytes:
+        if self._native:
+            return self._m.diff()
+        return serialize_delta(self._m.diff(), lambda k: k, lambda v: v)
+
+    def full_bytes(self) -> bytes:
+        if self._native:
+            return self._m.get_full()
+        return serialize_delta(self._m.get_full(), lambda k: k, lambda v: v)
+
+    def merge_bytes(self, raw: bytes) -> List[Tuple[bytes, Optional[bytes], Optional[bytes]]]:
+        if self._native:
+            return list(self._m.merge(raw))
+        delta = deserialize_delta(raw, lambda k: k, lambda v: v)
+        return self._m.merge(delta)
+
+
 class Connections:
     def __init__(self, identity: BrokerIdentifier) -> None:
         self.identity = identity
@@ -33,11 +75,11 @@ class Connections:
         # their Connection + task handles here)
         self.users: Dict[UserPubKey, object] = {}
         self.brokers: Dict[BrokerIdentifier, object] = {}
-        # routing state
-        self.direct_map: VersionedMap[UserPubKey, str, str] = VersionedMap(str(identity))
+        # routing state (CRDTs are native C++ when pushcdn_core is built)
+        self.direct_map = SyncMap(str(identity))
         self.user_topics: RelationalMap[UserPubKey, Topic] = RelationalMap()
         self.broker_topics: RelationalMap[str, Topic] = RelationalMap()
-        self.topic_sync_map: VersionedMap[Topic, bool, str] = VersionedMap(str(identity))
+        self.topic_sync_map = SyncMap(str(identity))
         self._previous_local_topics: Set[Topic] = set()
 
     # ------------------------------ users ------------------------------
@@ -48,7 +90,7 @@ class Connections:
         old = self.users.pop(pubkey, None)
         self.users[pubkey] = handle
         self.user_topics.associate_key_with_values(pubkey, topics)
-        self.direct_map.insert(pubkey, str(self.identity))
+        self.direct_map.insert(pubkey, str(self.identity).encode())
         NUM_USERS_CONNECTED.set(len(self.users))
         return old
 
@@ -57,7 +99,7 @@ class Connections:
         if handle is not None:
             self.user_topics.remove_key(pubkey)
             # only remove from the direct map if we still own the user
-            if self.direct_map.get(pubkey) == str(self.identity):
+            if self.direct_map.get(pubkey) == str(self.identity).encode():
                 self.direct_map.remove(pubkey)
             NUM_USERS_CONNECTED.set(len(self.users))
         return handle
@@ -112,24 +154,24 @@ class Connections:
 
     def get_broker_identifier_of_user(self, pubkey: UserPubKey) -> Optional[BrokerIdentifier]:
         owner = self.direct_map.get(pubkey)
-        return BrokerIdentifier.parse(owner) if owner else None
+        return BrokerIdentifier.parse(owner.decode()) if owner else None
 
     # ------------------------------ sync ------------------------------
 
     def get_partial_user_sync(self) -> bytes:
-        return serialize_delta(self.direct_map.diff(), bytes, lambda v: v.encode())
+        return self.direct_map.diff_bytes()
 
     def get_full_user_sync(self) -> bytes:
-        return serialize_delta(self.direct_map.get_full(), bytes, lambda v: v.encode())
+        return self.direct_map.full_bytes()
 
     def apply_user_sync(self, data: bytes) -> List[UserPubKey]:
         """Merge a remote user delta; returns local users to kick (now owned
         elsewhere — reference connections/mod.rs:154-162)."""
-        delta = deserialize_delta(data, bytes, lambda b: b.decode())
-        changed = self.direct_map.merge(delta)
+        changed = self.direct_map.merge_bytes(data)
+        me = str(self.identity).encode()
         to_kick = []
         for key, _old, new in changed:
-            if new is not None and new != str(self.identity) and key in self.users:
+            if new is not None and new != me and key in self.users:
                 to_kick.append(key)
         return to_kick
 
@@ -138,26 +180,24 @@ class Connections:
         (reference connections/mod.rs:205-237)."""
         current = self.user_topics.get_values()
         for t in current - self._previous_local_topics:
-            self.topic_sync_map.insert(t, True)
+            self.topic_sync_map.insert(bytes([t]), b"\x01")
         for t in self._previous_local_topics - current:
-            self.topic_sync_map.remove(t)
+            self.topic_sync_map.remove(bytes([t]))
         self._previous_local_topics = current
 
     def get_partial_topic_sync(self) -> bytes:
         self._local_topic_updates()
-        return serialize_delta(
-            self.topic_sync_map.diff(), lambda t: bytes([t]), lambda v: b"\x01"
-        )
+        return self.topic_sync_map.diff_bytes()
 
     def get_full_topic_sync(self) -> bytes:
         self._local_topic_updates()
-        return serialize_delta(
-            self.topic_sync_map.get_full(), lambda t: bytes([t]), lambda v: b"\x01"
-        )
+        return self.topic_sync_map.full_bytes()
 
     def apply_topic_sync(self, broker: BrokerIdentifier, data: bytes) -> None:
         """Apply a remote broker's topic interests: subscribe/unsubscribe the
-        broker per changed topic (reference connections/mod.rs:165-191)."""
+        broker per changed topic (reference connections/mod.rs:165-191).
+        The delta is applied directly (not merged into our own map: it
+        describes the PEER's interests, keyed per peer in broker_topics)."""
         delta = deserialize_delta(data, lambda b: b[0], lambda b: True)
         for topic, e in delta.items():
             if e.value:

@@ -241,3 +241,30 @@ def test_relational_map_reassociate_after_remove():
     rm.associate_key_with_values(b"u", [2])
     assert rm.get_values_by_key(b"u") == {2}
     assert rm.get_keys_by_value(1) == set()
+
+
+def test_native_and_python_crdt_wire_compatible():
+    """The native C++ VersionedMap's delta bytes must be parseable by the
+    Python implementation and vice versa (same documented encoding)."""
+    from pushcdn_amd.ops.build import build_core
+
+    core = build_core()
+    native = core.VersionedMap("nat")
+    native.insert(b"alice", b"broker-1")
+    native.insert(b"bob", b"broker-2")
+    native.remove(b"bob")
+    raw = native.get_full()
+
+    # python side parses the native delta
+    delta = deserialize_delta(raw, lambda k: k, lambda v: v)
+    assert delta[b"alice"].value == b"broker-1"
+    assert delta[b"bob"].value is None  # tombstone
+
+    # python-serialized delta merges into a native map
+    pyvm = VersionedMap("py")
+    pyvm.insert(b"carol", b"broker-3")
+    pyraw = serialize_delta(pyvm.get_full(), lambda k: k, lambda v: v)
+    native2 = core.VersionedMap("nat2")
+    changed = native2.merge(pyraw)
+    assert changed == [(b"carol", None, b"broker-3")]
+    assert native2.get(b"carol") == b"broker-3"
