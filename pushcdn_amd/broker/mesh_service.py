@@ -63,10 +63,16 @@ def unpack_mesh_offsets(view: torch.Tensor, n_messages: int) -> torch.Tensor:
 
 
 class MeshBroker(Broker):
-    def __init__(self, config: BrokerConfig, batch_capacity: int = 1 << 22) -> None:
+    def __init__(self, config: BrokerConfig, batch_capacity: int = 1 << 22,
+                 interest_routed: bool = False) -> None:
         assert config.data_plane == "gpu", "MeshBroker is the GPU data-plane broker"
         super().__init__(config)
         self.batch_capacity = batch_capacity
+        # interest_routed: ship batches only to peers whose subscribers
+        # intersect the batch's topics (grouped P2P on the xGMI links)
+        # instead of the all-gather. Direct traffic still ships everywhere
+        # (ownership is not synced in mesh mode).
+        self.interest_routed = interest_routed
         self.mesh: Optional[RcclMesh] = None
         self._carry: List[Bytes] = []  # messages that didn't fit last tick
 
@@ -115,13 +121,36 @@ class MeshBroker(Broker):
                 batch.append(raw)
                 used += padded
             msgs = [raw.data for raw in batch]
-            host_buf, n_local, _used = pack_mesh_batch(msgs, self.batch_capacity)
+            host_buf, n_local, used_bytes = pack_mesh_batch(msgs, self.batch_capacity)
             if dev_buf is not None:
                 dev_buf.copy_(host_buf, non_blocking=True)
                 send_buf = dev_buf
             else:
                 send_buf = host_buf
-            for rank, view, n_msgs, _nbytes in self.mesh.exchange(send_buf, n_local, 0):
+            if self.interest_routed:
+                from ..proto import message as msglib
+
+                batch_topics = 0
+                has_direct = False
+                for raw in msgs:
+                    try:
+                        r = msglib.parse_offsets(raw)
+                    except Exception:
+                        continue
+                    if r["disc"] == 4:
+                        for t in raw[r["topics_off"] : r["topics_off"] + r["topics_cnt"]]:
+                            batch_topics |= 1 << t
+                    elif r["disc"] == 3:
+                        has_direct = True
+                interests = 0
+                for t in self.connections.user_topics.get_values():
+                    interests |= 1 << (t & 0xFF)
+                exchanged = self.mesh.exchange_interest(
+                    send_buf, n_local, used_bytes, batch_topics, interests, has_direct
+                )
+            else:
+                exchanged = self.mesh.exchange(send_buf, n_local, 0)
+            for rank, view, n_msgs, _nbytes in exchanged:
                 if n_msgs == 0:
                     continue
                 offsets = unpack_mesh_offsets(view, n_msgs)

@@ -119,6 +119,75 @@ class RcclMesh:
                 ))
         return out
 
+    def exchange_interest(self, batch: torch.Tensor, n_messages: int, batch_bytes: int,
+                          batch_topics: int, interests: int, has_direct: bool) -> list:
+        """Interest-routed exchange in two phases:
+          1. all-gather tiny metadata: {n, bytes, batch-topic bitmap (256b),
+             interest bitmap (256b), has_direct} per rank
+          2. every rank computes the SAME sender->receiver matrix from the
+             metadata (sender s ships to r iff s's batch topics intersect
+             r's interests, or s has direct traffic), then posts matching
+             grouped isend/irecv of ONLY the used bytes.
+        This is the reference's per-interested-broker fan-out
+        (handler.rs:262-265) on xGMI point-to-point links."""
+        if not self.enabled:
+            return [(0, batch, n_messages, batch_bytes)]
+        assert batch.numel() == self.capacity
+
+        def pack_bits(v: int) -> list:
+            out = []
+            for i in range(4):
+                w = (v >> (64 * i)) & ((1 << 64) - 1)
+                out.append(w - (1 << 64) if w >= (1 << 63) else w)
+            return out
+
+        meta_local = torch.tensor(
+            [n_messages, batch_bytes] + pack_bits(batch_topics) + pack_bits(interests)
+            + [1 if has_direct else 0],
+            dtype=torch.int64, device=self.device)
+        stride = meta_local.numel()
+        meta = torch.zeros(self.world_size * stride, dtype=torch.int64, device=self.device)
+        self.dist.all_gather_into_tensor(meta, meta_local)
+        mh = meta.to("cpu").tolist()
+
+        def unpack_bits(words) -> int:
+            v = 0
+            for i, w in enumerate(words):
+                v |= (w & ((1 << 64) - 1)) << (64 * i)
+            return v
+
+        n_of = lambda r: int(mh[r * stride])
+        bytes_of = lambda r: int(mh[r * stride + 1])
+        topics_of = lambda r: unpack_bits(mh[r * stride + 2 : r * stride + 6])
+        interest_of = lambda r: unpack_bits(mh[r * stride + 6 : r * stride + 10])
+        direct_of = lambda r: bool(mh[r * stride + 10])
+
+        def ships(s: int, r: int) -> bool:
+            if s == r or n_of(s) == 0:
+                return False
+            return bool(topics_of(s) & interest_of(r)) or direct_of(s)
+
+        ops = []
+        for r in range(self.world_size):
+            if r == self.rank:
+                continue
+            if ships(self.rank, r):
+                ops.append(self.dist.P2POp(self.dist.isend, batch[: bytes_of(self.rank)], r))
+            if ships(r, self.rank):
+                ops.append(self.dist.P2POp(
+                    self.dist.irecv,
+                    self._gathered[r * self.capacity : r * self.capacity + bytes_of(r)], r))
+        if ops:
+            for req in self.dist.batch_isend_irecv(ops):
+                req.wait()
+        out = [(self.rank, batch, n_messages, batch_bytes)]
+        for r in range(self.world_size):
+            if r != self.rank and ships(r, self.rank):
+                out.append((r,
+                            self._gathered[r * self.capacity : (r + 1) * self.capacity],
+                            n_of(r), bytes_of(r)))
+        return out
+
     def barrier(self) -> None:
         if self.enabled:
             self.dist.barrier()

@@ -98,3 +98,26 @@ def test_mesh_broker_service_gloo(tmp_path):
     assert out.returncode == 0, (out.stdout[-2000:], out.stderr[-3000:])
     assert "rank 0 mesh-service OK" in out.stdout
     assert "rank 1 mesh-service OK" in out.stdout
+
+
+def test_mesh_broker_interest_routed_gloo(tmp_path):
+    """Same end-to-end scenario, but with interest-routed grouped P2P
+    instead of the all-gather."""
+    script = tmp_path / "mesh_worker_p2p.py"
+    script.write_text(WORKER.replace(
+        "broker = MeshBroker(cfg, batch_capacity=1 << 14)",
+        "broker = MeshBroker(cfg, batch_capacity=1 << 14, interest_routed=True)"))
+    env = dict(os.environ)
+    env["PUSHCDN_REPO"] = str(REPO)
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node=2",
+            "--master-addr", "127.0.0.1", "--master-port", "29527",
+            str(script),
+        ],
+        capture_output=True, text=True, timeout=300, env=env,
+    )
+    assert out.returncode == 0, (out.stdout[-2000:], out.stderr[-3000:])
+    assert "rank 0 mesh-service OK" in out.stdout
+    assert "rank 1 mesh-service OK" in out.stdout
