@@ -32,7 +32,7 @@ async def main(seconds: int) -> None:
         heartbeat_interval_s=1.0, data_plane="gpu", gpu_device="cuda:0",
         gpu_max_users=64, gpu_ring_bytes=1 << 20, gpu_tick_interval_s=0.002,
     )
-    broker = MeshBroker(cfg, batch_capacity=1 << 20)
+    broker = MeshBroker(cfg, batch_capacity=1 << 20, interest_routed=False)
     await broker.start()
     await broker.discovery.perform_heartbeat(0, 600)
     marshal = Marshal(MarshalConfig(bind_endpoint="soak-marshal",
