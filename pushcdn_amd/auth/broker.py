@@ -3,7 +3,7 @@
 from __future__ import annotations
 
 import time
-from typing import Optional, Sequence, Tuple
+from typing import Optional, Tuple
 
 from ..crypto import bls
 from ..discovery import BrokerIdentifier, DiscoveryClient

@@ -14,7 +14,6 @@ stays torch-free so the control plane is testable anywhere.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
 from typing import Callable, Dict, List, Optional, Sequence, Set, Tuple
 
 from ..discovery import BrokerIdentifier

@@ -8,7 +8,7 @@ values are topics.  Both directions are kept consistent on every mutation.
 
 from __future__ import annotations
 
-from typing import Dict, Generic, Iterable, List, Set, TypeVar
+from typing import Dict, Generic, Iterable, Set, TypeVar
 
 K = TypeVar("K")
 V = TypeVar("V")

@@ -18,7 +18,7 @@ import asyncio
 import random
 import socket
 from dataclasses import dataclass, field
-from typing import Callable, Dict, List, Optional, Sequence, Tuple
+from typing import Callable, Dict, List, Optional, Sequence
 
 from ..auth import BrokerAuth
 from ..auth.broker import BrokerAuth as _BA
@@ -28,9 +28,8 @@ from ..proto import message as m
 from ..proto.errors import ConnectionError_, TopicError
 from ..proto.limiter import Bytes, Limiter
 from ..proto.topic import TopicSpace, ALL_TOPICS
-from ..proto.transports.base import Connection, Protocol
+from ..proto.transports.base import Connection
 from ..utils.log import get_logger, ident
-from ..utils.mnemonic import mnemonic
 from .connections import Connections
 
 log = get_logger("broker")

@@ -16,15 +16,13 @@ import asyncio
 import tempfile
 import uuid
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import List, Sequence
 
 from ..crypto import bls
 from ..discovery import BrokerIdentifier
 from ..proto import message as m
-from ..proto.limiter import Limiter
 from ..proto.transports.base import Connection
 from ..proto.transports.memory import Memory, gen_testing_connection_pair
-from .connections import Connections
 from .service import Broker, BrokerConfig, BrokerHandle, UserHandle
 from .versioned_map import Versioned, serialize_delta
 

@@ -16,7 +16,7 @@ host path; ``min_batch`` falls back automatically.
 from __future__ import annotations
 
 import asyncio
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 from . import bls
 

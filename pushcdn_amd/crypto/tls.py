@@ -10,11 +10,9 @@ scripts/gen-ca.bash does the same); the ssl stdlib handles the handshake.
 
 from __future__ import annotations
 
-import os
 import ssl
 import subprocess
 import tempfile
-from dataclasses import dataclass
 from pathlib import Path
 from typing import Optional, Tuple
 

@@ -12,7 +12,7 @@ Semantics match the reference broker: topic match = union of subscriber sets
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 

@@ -19,7 +19,7 @@ driver of this process group (torchrun / the service supervisor) owns that —
 from __future__ import annotations
 
 import os
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 

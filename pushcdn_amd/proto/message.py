@@ -25,7 +25,7 @@ Python code.
 from __future__ import annotations
 
 import struct
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Union
 
 from .errors import SerializeError, DeserializeError

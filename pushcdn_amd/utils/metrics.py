@@ -14,7 +14,7 @@ from __future__ import annotations
 
 import threading
 import time
-from typing import Dict, List, Optional, Tuple
+from typing import List, Tuple
 
 try:
     from prometheus_client import (
