@@ -4,18 +4,33 @@
 // (cdn-broker/src/tasks/user/handler.rs:95-163 + broker/handler.rs:197-272)
 // as batched CDNA4 kernels over HBM-resident tables:
 //
-//   K4  parse_batch    — on-device Cap'n Proto validation/field extraction
+//   K4  k4_parse_batch — on-device Cap'n Proto validation/field extraction
 //                        (reference cdn-proto/src/message.rs:212-312)
-//   K2a topic_mask     — per-message OR of subscription-bitmap topic rows
-//                        (reference connections/mod.rs:94-124)
-//   K2b assign_emit    — per-user ordered scan: assigns egress-ring offsets
-//                        and emits the (msg,user,dst) delivery pair list
-//                        (the per-connection FIFO the reference gets from its
-//                        per-conn channel actors, protocols/mod.rs:139-217)
-//   K3  fanout_copy    — N-way payload fan-out into per-user egress rings
-//                        (reference user/sender.rs:16-33 Arc-clone push)
-//   K5  direct_lookup  — open-addressing probe user-hash -> owner
+//   K2a k2a_topic_mask[_t] — per-message OR of subscription-bitmap topic
+//                        rows (reference connections/mod.rs:94-124);
+//                        the _t variant writes the mask TRANSPOSED [W][M]
+//                        so K2b scans contiguous memory
+//   K2b k2b_fused_t    — the production emit: per-user ordered scan with
+//                        wave-aggregated slot claims (one atomic per wave64)
+//                        + closed-form ring math for uniform records; the
+//                        per-connection FIFO the reference gets from its
+//                        per-conn channel actors (protocols/mod.rs:139-217).
+//                        (k2b_count/k2b_emit are the straightforward
+//                        two-pass reference pair, kept for golden tests.)
+//   K3  k3_fanout_flat_t — the production fan-out for uniform records
+//                        (unit-per-lane flat index, ~100% lane utilization,
+//                        non-temporal 16 B stores; seq from value or device
+//                        counter for hipGraph capture); k3_fanout_wave_t is
+//                        the general mixed-size path (one wave per pair,
+//                        best for records >= 4 KiB); k3_fanout is the plain
+//                        reference variant for golden tests. All mirror the
+//                        reference's raw-bytes Arc-clone push
+//                        (user/sender.rs:16-33).
+//   K5  k5_direct_lookup — open-addressing probe user-hash -> owner
 //                        (reference connections/mod.rs:69-71 DirectMap get)
+//   K5b k5b_emit_direct — on-device direct-delivery pair emission (appends
+//                        to the same pair list; no host sync)
+//   K2c k2c_apply_subs — ordered subscription updates
 //
 // Design notes (per /opt/skills/guides/cdna_hip_programming.md):
 //  - wave = 64; all copies are uint4 (16 B/lane) vectorized
