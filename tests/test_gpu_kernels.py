@@ -379,3 +379,62 @@ def test_fused_t_capacity_drop(ops):
     assert int(drops.cpu()[0]) == 64 - cap
     assert int(n_pairs.cpu()[0]) >= cap
     assert bool((pu[:cap].cpu() >= 0).all())  # every in-capacity slot written
+
+
+def _random_message(rng: random.Random) -> "m.Message":
+    kind = rng.randrange(9)
+    blob = lambda n: rng.randbytes(rng.randrange(n))
+    tl = lambda: [rng.randrange(256) for _ in range(rng.randrange(16))]
+    u64 = lambda: rng.getrandbits(64)
+    return [
+        lambda: m.AuthenticateWithKey(blob(128), u64(), blob(128)),
+        lambda: m.AuthenticateWithPermit(u64()),
+        lambda: m.AuthenticateResponse(u64(), "ctx-%d" % rng.getrandbits(16)),
+        lambda: m.Direct(blob(64), blob(512)),
+        lambda: m.Broadcast(tl(), blob(512)),
+        lambda: m.Subscribe(tl()),
+        lambda: m.Unsubscribe(tl()),
+        lambda: m.UserSync(blob(512)),
+        lambda: m.TopicSync(blob(512)),
+    ][kind]()
+
+
+def test_parse_batch_fuzz_corrupted(ops):
+    """Seeded fuzz: K4 must agree with the host structural parser on EVERY
+    record of a batch where ~60% of frames are bit-flipped, truncated or
+    garbage-spliced — same accept/reject decision, same extracted fields,
+    and no device fault (memory-safety of the on-device decoder)."""
+    rng = random.Random(0xC0FFEE)
+    frames = []
+    for i in range(768):
+        raw = bytearray(m.serialize(_random_message(rng)))
+        if rng.random() < 0.6 and raw:
+            c = rng.randrange(3)
+            if c == 0:
+                for _ in range(rng.randint(1, 6)):
+                    raw[rng.randrange(len(raw))] ^= 1 << rng.randrange(8)
+            elif c == 1:
+                raw = raw[: rng.randrange(len(raw) + 1)]
+            elif len(raw) >= 8:
+                at = rng.randrange(len(raw) - 7)
+                raw[at : at + 8] = rng.randbytes(8)
+        frames.append(bytes(raw))
+    buf, offsets = b"", [0]
+    for f in frames:
+        buf += f
+        offsets.append(len(buf))
+    pr = ref.parse_batch(buf, offsets)
+    dbuf, doff = to_dev(buf, offsets)
+    disc, poff, plen, toff, tcnt, rhash, ts = ops.parse_batch(dbuf, doff)
+    torch.cuda.synchronize()
+    n_valid = int((pr.disc >= 0).sum())
+    assert 0 < n_valid < len(frames)  # fuzz actually exercised both paths
+    for name, dev, host in [
+        ("disc", disc, pr.disc), ("payload_off", poff, pr.payload_off),
+        ("payload_len", plen, pr.payload_len), ("topics_off", toff, pr.topics_off),
+        ("topics_cnt", tcnt, pr.topics_cnt), ("recip_hash", rhash, pr.recip_hash),
+        ("timestamp", ts, pr.timestamp),
+    ]:
+        d = dev.cpu()
+        bad = (d != host).nonzero().flatten().tolist()
+        assert not bad, f"{name} mismatch at records {bad[:8]} (host disc={[int(pr.disc[i]) for i in bad[:8]]})"
