@@ -136,10 +136,16 @@ extern "C" __global__ void k4_parse_batch(
 {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= M) return;
+    // defaults; field values are buffered in locals and committed only
+    // once the whole frame validates, so rejected records are all-zero
+    // (bit-identical to the host mirror, proto/message.py parse_offsets)
     out.disc[i] = -1;
     out.payload_off[i] = 0; out.payload_len[i] = 0;
     out.topics_off[i] = 0;  out.topics_cnt[i] = 0;
     out.recip_hash[i] = 0;  out.timestamp[i] = 0;
+    int64_t o_poff = 0, o_toff = 0, o_ts = 0;
+    uint64_t o_rhash = 0;
+    int32_t o_plen = 0, o_tcnt = 0;
 
     int64_t beg = offsets[i], end = offsets[i + 1];
     if (end - beg < 8 + 8) return;  // header + root ptr
@@ -165,32 +171,32 @@ extern "C" __global__ void k4_parse_batch(
         int64_t it; int32_t idw, ipw;
         if (!read_struct_ptr(seg, nwords, up, &it, &idw, &ipw)) return;
         if (idw < 1 || ipw < 2) return;
-        out.timestamp[i] = ld_u64(seg + it * 8);
+        o_ts = (int64_t)ld_u64(seg + it * 8);
         int64_t ko; int32_t kl;
         if (!read_byte_list(seg, nwords, it + idw, &ko, &kl)) return;
-        out.payload_off[i] = kl ? seg_base + ko : 0;  // public key bytes
-        out.payload_len[i] = kl;
+        o_poff = kl ? seg_base + ko : 0;  // public key bytes
+        o_plen = kl;
         int64_t so; int32_t sl;
         if (!read_byte_list(seg, nwords, it + idw + 1, &so, &sl)) return;
-        out.topics_off[i] = sl ? seg_base + so : 0;   // signature bytes (reused slot)
-        out.topics_cnt[i] = sl;
+        o_toff = sl ? seg_base + so : 0;   // signature bytes (reused slot)
+        o_tcnt = sl;
         break;
     }
     case 1: {  // AuthenticateWithPermit
         int64_t it; int32_t idw, ipw;
         if (!read_struct_ptr(seg, nwords, up, &it, &idw, &ipw)) return;
         if (idw < 1) return;
-        out.timestamp[i] = ld_u64(seg + it * 8);  // permit
+        o_ts = (int64_t)ld_u64(seg + it * 8);  // permit
         break;
     }
     case 2: {  // AuthenticateResponse
         int64_t it; int32_t idw, ipw;
         if (!read_struct_ptr(seg, nwords, up, &it, &idw, &ipw)) return;
         if (idw < 1 || ipw < 1) return;
-        out.timestamp[i] = ld_u64(seg + it * 8);  // permit
+        o_ts = (int64_t)ld_u64(seg + it * 8);  // permit
         int64_t co; int32_t cl;
         if (!read_byte_list(seg, nwords, it + idw, &co, &cl)) return;
-        out.payload_off[i] = cl ? seg_base + co : 0; out.payload_len[i] = cl;
+        o_poff = cl ? seg_base + co : 0; o_plen = cl;
         break;
     }
     case 3: {  // Direct
@@ -199,10 +205,10 @@ extern "C" __global__ void k4_parse_batch(
         if (ipw < 2) return;
         int64_t ro; int32_t rl;
         if (!read_byte_list(seg, nwords, it + idw, &ro, &rl)) return;
-        out.recip_hash[i] = fnv1a64(seg + ro, rl);
+        o_rhash = fnv1a64(seg + ro, rl);
         int64_t mo; int32_t ml;
         if (!read_byte_list(seg, nwords, it + idw + 1, &mo, &ml)) return;
-        out.payload_off[i] = ml ? seg_base + mo : 0; out.payload_len[i] = ml;
+        o_poff = ml ? seg_base + mo : 0; o_plen = ml;
         break;
     }
     case 4: {  // Broadcast
@@ -211,27 +217,30 @@ extern "C" __global__ void k4_parse_batch(
         if (ipw < 2) return;
         int64_t to; int32_t tc;
         if (!read_byte_list(seg, nwords, it + idw, &to, &tc)) return;
-        out.topics_off[i] = tc ? seg_base + to : 0; out.topics_cnt[i] = tc;
+        o_toff = tc ? seg_base + to : 0; o_tcnt = tc;
         int64_t mo; int32_t ml;
         if (!read_byte_list(seg, nwords, it + idw + 1, &mo, &ml)) return;
-        out.payload_off[i] = ml ? seg_base + mo : 0; out.payload_len[i] = ml;
+        o_poff = ml ? seg_base + mo : 0; o_plen = ml;
         break;
     }
     case 5: case 6: {  // Subscribe / Unsubscribe
         int64_t to; int32_t tc;
         if (!read_byte_list(seg, nwords, up, &to, &tc)) return;
-        out.topics_off[i] = tc ? seg_base + to : 0; out.topics_cnt[i] = tc;
+        o_toff = tc ? seg_base + to : 0; o_tcnt = tc;
         break;
     }
     case 7: case 8: {  // UserSync / TopicSync
         int64_t dof; int32_t dl;
         if (!read_byte_list(seg, nwords, up, &dof, &dl)) return;
-        out.payload_off[i] = dl ? seg_base + dof : 0; out.payload_len[i] = dl;
+        o_poff = dl ? seg_base + dof : 0; o_plen = dl;
         break;
     }
     default:
         return;
     }
+    out.payload_off[i] = o_poff; out.payload_len[i] = o_plen;
+    out.topics_off[i] = o_toff;  out.topics_cnt[i] = o_tcnt;
+    out.recip_hash[i] = (int64_t)o_rhash; out.timestamp[i] = o_ts;
     out.disc[i] = (int32_t)disc;
 }
 
