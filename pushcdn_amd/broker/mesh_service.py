@@ -69,9 +69,9 @@ class MeshBroker(Broker):
         super().__init__(config)
         self.batch_capacity = batch_capacity
         # interest_routed: ship batches only to peers whose subscribers
-        # intersect the batch's topics (grouped P2P on the xGMI links)
-        # instead of the all-gather. Direct traffic still ships everywhere
-        # (ownership is not synced in mesh mode).
+        # intersect the batch's topics, or whose owned-user digest
+        # intersects the batch's direct-recipient digest (grouped P2P on
+        # the xGMI links), instead of the all-gather.
         self.interest_routed = interest_routed
         self.mesh: Optional[RcclMesh] = None
         self._carry: List[Bytes] = []  # messages that didn't fit last tick
@@ -129,9 +129,10 @@ class MeshBroker(Broker):
                 send_buf = host_buf
             if self.interest_routed:
                 from ..proto import message as msglib
+                from ..utils.keyhash import fnv1a64
 
                 batch_topics = 0
-                has_direct = False
+                direct_bits = 0  # 64b digest of this batch's direct recipients
                 for raw in msgs:
                     try:
                         r = msglib.parse_offsets(raw)
@@ -141,12 +142,18 @@ class MeshBroker(Broker):
                         for t in raw[r["topics_off"] : r["topics_off"] + r["topics_cnt"]]:
                             batch_topics |= 1 << t
                     elif r["disc"] == 3:
-                        has_direct = True
+                        direct_bits |= 1 << (fnv1a64(r["recipient"]) & 63)
                 interests = 0
                 for t in self.connections.user_topics.get_values():
                     interests |= 1 << (t & 0xFF)
+                # 64b digest of the direct users owned (connected) here —
+                # the mesh-plane analog of the reference's DirectMap
+                owned_bits = 0
+                for pubkey in self._gpu_user_by_slot.values():
+                    owned_bits |= 1 << (fnv1a64(pubkey) & 63)
                 exchanged = self.mesh.exchange_interest(
-                    send_buf, n_local, used_bytes, batch_topics, interests, has_direct
+                    send_buf, n_local, used_bytes, batch_topics, interests,
+                    direct_bits=direct_bits, owned_bits=owned_bits,
                 )
             else:
                 exchanged = self.mesh.exchange(send_buf, n_local, 0)

@@ -120,30 +120,38 @@ class RcclMesh:
         return out
 
     def exchange_interest(self, batch: torch.Tensor, n_messages: int, batch_bytes: int,
-                          batch_topics: int, interests: int, has_direct: bool) -> list:
+                          batch_topics: int, interests: int,
+                          direct_bits: int = 0, owned_bits: int = 0) -> list:
         """Interest-routed exchange in two phases:
           1. all-gather tiny metadata: {n, bytes, batch-topic bitmap (256b),
-             interest bitmap (256b), has_direct} per rank
+             interest bitmap (256b), direct-recipient digest (64b),
+             owned-user digest (64b)} per rank
           2. every rank computes the SAME sender->receiver matrix from the
              metadata (sender s ships to r iff s's batch topics intersect
-             r's interests, or s has direct traffic), then posts matching
-             grouped isend/irecv of ONLY the used bytes.
-        This is the reference's per-interested-broker fan-out
-        (handler.rs:262-265) on xGMI point-to-point links."""
+             r's interests, or s's direct-recipient digest intersects r's
+             owned-user digest), then posts matching grouped isend/irecv of
+             ONLY the used bytes.
+        The digests are 64-bit blooms over fnv1a64(user pubkey) & 63: a
+        digest collision only costs a spurious ship (the local K5 lookup
+        still drops non-owned directs); a recipient whose owner is connected
+        ALWAYS reaches that owner.  This is the reference's
+        per-interested-broker fan-out (handler.rs:262-265) plus its
+        DirectMap ownership routing (handler.rs:197-237) on xGMI
+        point-to-point links."""
         if not self.enabled:
             return [(0, batch, n_messages, batch_bytes)]
         assert batch.numel() == self.capacity
 
+        def _i64(w: int) -> int:
+            w &= (1 << 64) - 1
+            return w - (1 << 64) if w >= (1 << 63) else w
+
         def pack_bits(v: int) -> list:
-            out = []
-            for i in range(4):
-                w = (v >> (64 * i)) & ((1 << 64) - 1)
-                out.append(w - (1 << 64) if w >= (1 << 63) else w)
-            return out
+            return [_i64(v >> (64 * i)) for i in range(4)]
 
         meta_local = torch.tensor(
             [n_messages, batch_bytes] + pack_bits(batch_topics) + pack_bits(interests)
-            + [1 if has_direct else 0],
+            + [_i64(direct_bits), _i64(owned_bits)],
             dtype=torch.int64, device=self.device)
         stride = meta_local.numel()
         meta = torch.zeros(self.world_size * stride, dtype=torch.int64, device=self.device)
@@ -160,12 +168,13 @@ class RcclMesh:
         bytes_of = lambda r: int(mh[r * stride + 1])
         topics_of = lambda r: unpack_bits(mh[r * stride + 2 : r * stride + 6])
         interest_of = lambda r: unpack_bits(mh[r * stride + 6 : r * stride + 10])
-        direct_of = lambda r: bool(mh[r * stride + 10])
+        direct_of = lambda r: mh[r * stride + 10] & ((1 << 64) - 1)
+        owned_of = lambda r: mh[r * stride + 11] & ((1 << 64) - 1)
 
         def ships(s: int, r: int) -> bool:
             if s == r or n_of(s) == 0:
                 return False
-            return bool(topics_of(s) & interest_of(r)) or direct_of(s)
+            return bool(topics_of(s) & interest_of(r)) or bool(direct_of(s) & owned_of(r))
 
         ops = []
         for r in range(self.world_size):
