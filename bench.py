@@ -138,13 +138,13 @@ def main() -> None:
         torch.cuda.set_device(local_rank)
         device = f"cuda:{local_rank}"
 
-    from pushcdn_amd.broker.gpu_engine import GpuBrokerEngine
+    from pushcdn_amd.broker.gpu_engine import GpuBrokerEngine, ring_rec
 
     n_local_users = (args.subscribers + world_size - 1) // world_size
     ring_bytes = 1 << 9
     # Ring must hold one step's worth of WIRE messages (payload + ~64 B capnp
-    # envelope), each as a 16 B record header + 16-aligned body.
-    wire_est = 16 + ((args.payload + 64 + 15) & ~15)
+    # envelope), each as a 64-aligned ring record (see gpu_engine.ring_rec).
+    wire_est = ring_rec((args.payload + 64 + 15) & ~15)
     need = world_size * args.batch * wire_est
     while ring_bytes < need * 2:
         ring_bytes <<= 1

@@ -50,6 +50,16 @@
 #define CDN_CHECK 1
 #endif
 
+// Egress ring record stride: 16 B header + payload, padded so every record
+// START (and the ring base) is 64-byte aligned — one HBM line. K3 writes the
+// FULL stride (zero pad) so no cache line is ever partially written:
+// non-temporal stores then never read-modify-write. Host mirrors:
+// pushcdn_amd/broker/gpu_engine.py ring_rec / ops/reference.py.
+#define RING_ALIGN 64ull
+__host__ __device__ inline uint64_t ring_rec(int32_t len) {
+    return ((uint64_t)len + 16 + (RING_ALIGN - 1)) & ~(RING_ALIGN - 1);
+}
+
 // ---------------------------------------------------------------------------
 // 64-bit FNV-1a — the routing hash for user public keys. Host mirror in
 // csrc/common/hash.h / pushcdn_amd/utils/keyhash.py must match bit-for-bit.
@@ -330,9 +340,8 @@ extern "C" __global__ void k2b_emit(
     for (int m = 0; m < M; ++m) {
         if (!(mask[(int64_t)m * W + w] & bit)) continue;
         int32_t len = payload_len[m];
-        // 16-byte record header + payload padded to 16 so every payload copy
-        // in K3 is uint4-aligned (ring_bytes must be a multiple of 16).
-        uint64_t rec = 16 + (uint64_t)((len + 15) & ~15);
+        // 64-aligned record stride (see ring_rec)
+        uint64_t rec = ring_rec(len);
         if (wpos + rec > (uint64_t)ring_bytes) {
             pair_user[slot] = -1; pair_msg[slot] = m; pair_dst[slot] = 0;
             slot++; dropped++; continue;
@@ -662,7 +671,14 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
             if (NT) __builtin_nontemporal_store(v, (v4u*)dst);
             else *(v4u*)dst = v;
         } else {
-            for (int b = 0; b < 16 && coff + b < len; ++b) dst[b] = src[b];
+            // tail / pad unit: zero-fill to a full 16 B store so the whole
+            // record stride is written (full-line NT writes, no RMW)
+            uint8_t tmp[16];
+            #pragma unroll
+            for (int b = 0; b < 16; ++b) tmp[b] = (coff + b < len) ? src[b] : 0;
+            v4u v; memcpy(&v, tmp, 16);
+            if (NT) __builtin_nontemporal_store(v, (v4u*)dst);
+            else *(v4u*)dst = v;
         }
     }
 }
@@ -811,7 +827,7 @@ extern "C" __global__ void k2b_fused_t(
     for (int m = 0; m < M; ++m) {
         if (!(col[m] & bit)) continue;
         int32_t len = payload_len[m];
-        uint64_t rec = 16 + (uint64_t)((len + 15) & ~15);
+        uint64_t rec = ring_rec(len);
         bool fits_ring = (wpos + rec <= (uint64_t)ring_bytes);
         bool fits_cap = (slot < capacity);
         if (!fits_ring || !fits_cap) {
@@ -884,7 +900,7 @@ extern "C" __global__ void k5b_emit_direct(
     int u = owner[i];
     if (u < 0) return;
     int32_t len = payload_len[i];
-    uint64_t rec = 16 + (uint64_t)((len + 15) & ~15);
+    uint64_t rec = ring_rec(len);
     uint64_t old = atomicAdd((unsigned long long*)&ring_wpos[u], (unsigned long long)rec);
     int slot = atomicAdd(n_pairs, 1);
     if (old + rec > (uint64_t)ring_bytes || slot >= capacity) {

@@ -32,6 +32,17 @@ from typing import Dict, List, Optional, Tuple
 
 import torch
 
+RING_ALIGN = 64
+
+
+def ring_rec(length: int) -> int:
+    """Egress ring record stride: 16 B header + payload, padded so every
+    record start is 64 B (one HBM line) aligned.  Host mirror of
+    csrc/hip/dataplane.hip ring_rec — K3 writes the full stride with zero
+    pad so non-temporal stores never partially write a line."""
+    return (length + 16 + (RING_ALIGN - 1)) & ~(RING_ALIGN - 1)
+
+
 from ..utils.keyhash import fnv1a64
 
 
@@ -203,7 +214,7 @@ class GpuBrokerEngine:
         # count stays on device
         self._n_pairs.zero_()
         uniform = self.fanout_wire and uniform_wire_len is not None
-        rec = (16 + ((uniform_wire_len + 15) & ~15)) if uniform else 0
+        rec = ring_rec(uniform_wire_len) if uniform else 0
         ops.assign_emit_fused_t(
             mask_t, payload_len, self.ring_wpos, self.ring_bytes, self.n_users,
             self._pair_user, self._pair_msg, self._pair_dst, self._drops, self._n_pairs,
@@ -266,7 +277,7 @@ class GpuBrokerEngine:
         ~8 kernel launches). Requires fanout_wire + uniform records +
         direct_enabled=False (the broadcast-bench shape)."""
         assert self.fanout_wire and not self.direct_enabled
-        units = 1 + ((uniform_wire_len + 15) & ~15) // 16
+        units = ring_rec(uniform_wire_len) // 16
         key = (buf.data_ptr(), offsets.data_ptr(), units)
         g = self._graphs.get(key)
         if g is None:
@@ -313,7 +324,7 @@ class GpuBrokerEngine:
             if int(pr.disc[i]) == 3 and int(owner[i]) >= 0:
                 u = int(owner[i])
                 length = int(pr.payload_len[i])
-                rec = 16 + ((length + 15) & ~15)
+                rec = ring_rec(length)
                 wpos = int(self.ring_wpos[u])
                 if wpos + rec <= self.ring_bytes:
                     dst = u * self.ring_bytes + wpos
@@ -350,5 +361,5 @@ def parse_ring_records(ring: bytes, wpos: int) -> List[Tuple[int, bytes]]:
         seq = int.from_bytes(ring[pos + 4 : pos + 8], "little")
         payload = ring[pos + 16 : pos + 16 + length]
         out.append((seq, payload))
-        pos += 16 + ((length + 15) & ~15)
+        pos += ring_rec(length)
     return out

@@ -17,8 +17,7 @@ from typing import List, Tuple
 import torch
 
 from ..utils.keyhash import fnv1a64
-
-RECORD_ALIGN = 16
+from ..broker.gpu_engine import ring_rec
 
 
 @dataclass
@@ -114,7 +113,7 @@ def assign_emit(
             if not (int(mask[m, w]) & bit):
                 continue
             length = int(payload_len[m])
-            rec = 16 + ((length + 15) & ~15)
+            rec = ring_rec(length)
             if wpos + rec > ring_bytes:
                 drops += 1
                 pair_user.append(-1)

@@ -370,8 +370,11 @@ def test_fused_t_capacity_drop(ops):
     n_pairs = torch.zeros(1, dtype=torch.int32, device="cuda")
     woff = doff[:-1].contiguous()
     wlen = (doff[1:] - doff[:-1]).to(torch.int32).contiguous()
-    rec = 16 + padded
-    ops.assign_emit_fused_t(mask_t, wlen, wpos, 1 << 12, n_users, pu, pm, pd, drops,
+    from pushcdn_amd.broker.gpu_engine import ring_rec
+
+    rec = ring_rec(padded)
+    # ring sized so all 64 deliveries FIT — only pair-capacity drops counted
+    ops.assign_emit_fused_t(mask_t, wlen, wpos, 64 * rec, n_users, pu, pm, pd, drops,
                             n_pairs, rec)
     torch.cuda.synchronize()
     # wave-aggregated claims may overshoot the clamp in the counter itself;
