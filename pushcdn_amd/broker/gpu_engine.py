@@ -32,14 +32,14 @@ from typing import Dict, List, Optional, Tuple
 
 import torch
 
-RING_ALIGN = 64
+RING_ALIGN = 16
 
 
 def ring_rec(length: int) -> int:
-    """Egress ring record stride: 16 B header + payload, padded so every
-    record start is 64 B (one HBM line) aligned.  Host mirror of
-    csrc/hip/dataplane.hip ring_rec — K3 writes the full stride with zero
-    pad so non-temporal stores never partially write a line."""
+    """Egress ring record stride: 16 B header + payload padded to 16 B
+    (uint4-aligned record starts).  Host mirror of csrc/hip/dataplane.hip
+    ring_rec; K3 zero-pads the tail unit so each record is written with
+    full vector stores."""
     return (length + 16 + (RING_ALIGN - 1)) & ~(RING_ALIGN - 1)
 
 
