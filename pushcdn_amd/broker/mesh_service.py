@@ -75,6 +75,15 @@ class MeshBroker(Broker):
         self.interest_routed = interest_routed
         self.mesh: Optional[RcclMesh] = None
         self._carry: List[Bytes] = []  # messages that didn't fit last tick
+        # The collective exchange + kernel ticks run on ONE dedicated thread
+        # so the asyncio loop keeps serving user IO / heartbeats while this
+        # rank waits for its peers to align on the collective.  A single
+        # worker also guarantees the communicator is never entered
+        # concurrently (NCCL/gloo requirement).
+        import concurrent.futures
+
+        self._mesh_executor = concurrent.futures.ThreadPoolExecutor(
+            max_workers=1, thread_name_prefix="mesh-tick")
 
     # the framed broker mesh is replaced by the collective: no dialing, no
     # framed sync blasts; heartbeats still publish load for the marshal
@@ -92,6 +101,54 @@ class MeshBroker(Broker):
 
     async def _send_partial_syncs(self) -> None:
         pass
+
+    def _blocking_mesh_tick(self, msgs: List[bytes], dev_buf: Optional[torch.Tensor],
+                            interests: int, owned_bits: int) -> torch.Tensor:
+        """One mesh tick's blocking half (runs on the dedicated mesh thread):
+        pack, H2D, collective exchange, kernel tick per received batch,
+        cursor drain.  Returns the drained ring cursors."""
+        host_buf, n_local, used_bytes = pack_mesh_batch(msgs, self.batch_capacity)
+        if dev_buf is not None:
+            dev_buf.copy_(host_buf, non_blocking=True)
+            send_buf = dev_buf
+        else:
+            send_buf = host_buf
+        if self.interest_routed:
+            from ..proto import message as msglib
+            from ..utils.keyhash import fnv1a64
+
+            batch_topics = 0
+            direct_bits = 0  # 64b digest of this batch's direct recipients
+            for raw in msgs:
+                try:
+                    r = msglib.parse_offsets(raw)
+                except Exception:
+                    continue
+                if r["disc"] == 4:
+                    for t in raw[r["topics_off"] : r["topics_off"] + r["topics_cnt"]]:
+                        batch_topics |= 1 << t
+                elif r["disc"] == 3:
+                    direct_bits |= 1 << (fnv1a64(r["recipient"]) & 63)
+            exchanged = self.mesh.exchange_interest(
+                send_buf, n_local, used_bytes, batch_topics, interests,
+                direct_bits=direct_bits, owned_bits=owned_bits,
+            )
+        else:
+            exchanged = self.mesh.exchange(send_buf, n_local, 0)
+        for rank, view, n_msgs, _nbytes in exchanged:
+            if n_msgs == 0:
+                continue
+            offsets = unpack_mesh_offsets(view, n_msgs)
+            if self._engine.use_gpu_ops:
+                self._engine.tick(view, offsets)
+            else:
+                host_bytes = bytes(view.numpy().tobytes())
+                self._engine.tick(
+                    view, offsets,
+                    host_batch=host_bytes,
+                    host_offsets=[int(x) for x in offsets],
+                )
+        return self._engine.drain_cursors()
 
     async def _gpu_tick_task(self) -> None:
         """Fixed-cadence mesh tick: pack queued local messages (possibly
@@ -121,56 +178,21 @@ class MeshBroker(Broker):
                 batch.append(raw)
                 used += padded
             msgs = [raw.data for raw in batch]
-            host_buf, n_local, used_bytes = pack_mesh_batch(msgs, self.batch_capacity)
-            if dev_buf is not None:
-                dev_buf.copy_(host_buf, non_blocking=True)
-                send_buf = dev_buf
-            else:
-                send_buf = host_buf
+            # digest inputs come from loop-owned state (connections maps),
+            # so compute them HERE; the exchange + ticks then run off-loop
+            interests = owned_bits = 0
             if self.interest_routed:
-                from ..proto import message as msglib
                 from ..utils.keyhash import fnv1a64
 
-                batch_topics = 0
-                direct_bits = 0  # 64b digest of this batch's direct recipients
-                for raw in msgs:
-                    try:
-                        r = msglib.parse_offsets(raw)
-                    except Exception:
-                        continue
-                    if r["disc"] == 4:
-                        for t in raw[r["topics_off"] : r["topics_off"] + r["topics_cnt"]]:
-                            batch_topics |= 1 << t
-                    elif r["disc"] == 3:
-                        direct_bits |= 1 << (fnv1a64(r["recipient"]) & 63)
-                interests = 0
                 for t in self.connections.user_topics.get_values():
                     interests |= 1 << (t & 0xFF)
                 # 64b digest of the direct users owned (connected) here —
                 # the mesh-plane analog of the reference's DirectMap
-                owned_bits = 0
                 for pubkey in self._gpu_user_by_slot.values():
                     owned_bits |= 1 << (fnv1a64(pubkey) & 63)
-                exchanged = self.mesh.exchange_interest(
-                    send_buf, n_local, used_bytes, batch_topics, interests,
-                    direct_bits=direct_bits, owned_bits=owned_bits,
-                )
-            else:
-                exchanged = self.mesh.exchange(send_buf, n_local, 0)
-            for rank, view, n_msgs, _nbytes in exchanged:
-                if n_msgs == 0:
-                    continue
-                offsets = unpack_mesh_offsets(view, n_msgs)
-                if self._engine.use_gpu_ops:
-                    self._engine.tick(view, offsets)
-                else:
-                    host_bytes = bytes(view.numpy().tobytes())
-                    self._engine.tick(
-                        view, offsets,
-                        host_batch=host_bytes,
-                        host_offsets=[int(x) for x in offsets],
-                    )
-            wpos = self._engine.drain_cursors()
+            wpos = await asyncio.get_running_loop().run_in_executor(
+                self._mesh_executor, self._blocking_mesh_tick,
+                msgs, dev_buf, interests, owned_bits)
             for slot, pubkey in list(self._gpu_user_by_slot.items()):
                 nbytes = int(wpos[slot])
                 if nbytes == 0:
