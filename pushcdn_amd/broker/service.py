@@ -257,11 +257,14 @@ class Broker:
             connection.close()
             return
         pubkey, raw_topics = result
+        # the INITIAL subscribe tolerates an all-invalid list — the user
+        # just connects with no topics (reference user/handler.rs:47
+        # discards the prune error); later Subscribe/Unsubscribe frames
+        # disconnect on it (handler.rs:140-156)
         try:
             topics = self.config.topic_space.prune(raw_topics)
         except TopicError:
-            connection.close()
-            return
+            topics = []
         handle = UserHandle(connection=connection)
         old = self.connections.add_user(pubkey, handle, topics)
         log.info("user connected: %s topics=%s%s", ident(pubkey), topics,
@@ -326,7 +329,12 @@ class Broker:
                         self._engine.subscribe(handle.gpu_index, topics)
                     raw.drop()
                 elif isinstance(msg, m.Unsubscribe):
-                    topics = [t & 0xFF for t in msg.topics]
+                    # pruned like Subscribe: an all-invalid unsubscribe
+                    # disconnects (reference user/handler.rs:150-156)
+                    try:
+                        topics = self.config.topic_space.prune(msg.topics)
+                    except TopicError:
+                        break
                     self.connections.unsubscribe_user(pubkey, topics)
                     if self._engine is not None and handle.gpu_index is not None:
                         self._engine.unsubscribe(handle.gpu_index, topics)

@@ -64,6 +64,7 @@ class TestDefinition:
 
     connected_users: List[TestUser] = field(default_factory=list)
     connected_brokers: List[TestBroker] = field(default_factory=list)
+    topic_space: object = None  # TopicSpace override (default ALL_TOPICS)
 
     async def into_run(self) -> TestRun:
         n = uuid.uuid4().hex[:8]
@@ -81,6 +82,8 @@ class TestDefinition:
             sync_interval_s=3600,
             whitelist_interval_s=3600,
         )
+        if self.topic_space is not None:
+            cfg.topic_space = self.topic_space
         broker = Broker(cfg)
         await broker.start()
 

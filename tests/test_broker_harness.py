@@ -168,3 +168,33 @@ def test_message_hooks_skip_and_disconnect():
         await tr.close()
 
     run(go())
+
+
+def test_invalid_topic_subscribe_disconnects():
+    """Subscribing (or unsubscribing) with NO valid topics kills the
+    connection; a partially valid list is pruned and kept (reference
+    tests/src/tests/subscribe.rs:123-199, user/handler.rs:140-156)."""
+    from pushcdn_amd.proto.topic import TEST_TOPIC_SPACE
+
+    async def go():
+        run_ = await TestDefinition(
+            connected_users=[TestUser(topics=[0]), TestUser(topics=[0])],
+            topic_space=TEST_TOPIC_SPACE,  # valid: {0, 1}
+        ).into_run()
+        broker = run_.broker
+        # partially valid: pruned to [1], connection stays
+        await run_.users[0].send_message(m.Subscribe(topics=[1, 7]))
+        await asyncio.sleep(0.1)
+        assert at_index(0) in broker.connections.users
+        assert broker.connections.user_topics.get_values_by_key(at_index(0)) == {0, 1}
+        # all-invalid subscribe: disconnected
+        await run_.users[0].send_message(m.Subscribe(topics=[7, 9]))
+        await asyncio.sleep(0.2)
+        assert at_index(0) not in broker.connections.users
+        # all-invalid unsubscribe: disconnected too
+        await run_.users[1].send_message(m.Unsubscribe(topics=[250]))
+        await asyncio.sleep(0.2)
+        assert at_index(1) not in broker.connections.users
+        await run_.close()
+
+    asyncio.run(asyncio.wait_for(go(), timeout=30))
