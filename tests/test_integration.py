@@ -227,6 +227,56 @@ def test_two_broker_mesh_broadcast_and_direct(tmp_path):
     run(go())
 
 
+def test_double_connect_across_brokers_kicks_old(tmp_path):
+    """Same keypair connects to broker 1, then (marshal steered by load
+    reports) to broker 2: the UserSync CRDT moves ownership and broker 1
+    kicks its stale session (reference double_connect.rs:76-141)."""
+
+    async def go():
+        db = new_db(tmp_path)
+        kp = bls.KeyPair.from_seed(1000)
+        b1 = make_broker(db, keypair=kp, tag="dc-one")
+        b2 = make_broker(db, keypair=kp, tag="dc-two")
+        await b1.start()
+        await b2.start()
+        await b1.discovery.perform_heartbeat(0, 60)
+        await b2.discovery.perform_heartbeat(0, 60)
+        await asyncio.sleep(0.8)  # mesh forms
+        marshal, endpoint = make_marshal(db)
+        await marshal.start()
+
+        # steer the first session to b1
+        await b1.discovery.perform_heartbeat(0, 60)
+        await b2.discovery.perform_heartbeat(10, 60)
+        first = make_client(endpoint, seed=77, topics=[0])
+        await first.ensure_initialized()
+        assert len(b1.connections.users) == 1 and len(b2.connections.users) == 0
+
+        # steer the second session (same key!) to b2
+        await b1.discovery.perform_heartbeat(10, 60)
+        await b2.discovery.perform_heartbeat(0, 60)
+        second = make_client(endpoint, seed=77, topics=[0])
+        await second.ensure_initialized()
+        assert len(b2.connections.users) == 1
+
+        # b1 learns via UserSync that b2 owns the key now and kicks its
+        # stale session (sync_interval_s=0.2 in make_broker)
+        for _ in range(40):
+            if len(b1.connections.users) == 0:
+                break
+            await asyncio.sleep(0.1)
+        assert len(b1.connections.users) == 0
+
+        # and the surviving session is fully functional
+        await second.send_direct_message(second.public_key, b"to-survivor")
+        msg = await asyncio.wait_for(second.receive_message(), timeout=5)
+        assert isinstance(msg, m.Direct) and msg.message == b"to-survivor"
+
+        await stop_stack([b1, b2], marshal, first, second)
+
+    run(go())
+
+
 def test_cli_parser_smoke():
     """CLI surface exists with the reference's flag set (no daemon start)."""
     from pushcdn_amd import cli
