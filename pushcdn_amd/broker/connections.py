@@ -14,7 +14,7 @@ stays torch-free so the control plane is testable anywhere.
 
 from __future__ import annotations
 
-from typing import Callable, Dict, List, Optional, Sequence, Set, Tuple
+from typing import Dict, List, Optional, Sequence, Set, Tuple
 
 from ..discovery import BrokerIdentifier
 from ..utils.metrics import NUM_BROKERS_CONNECTED, NUM_USERS_CONNECTED

@@ -26,7 +26,6 @@ the engine's semantics are testable without a GPU.
 
 from __future__ import annotations
 
-import time
 from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 

@@ -16,8 +16,6 @@ from __future__ import annotations
 import argparse
 import asyncio
 import random
-import sys
-import time
 
 
 def _broker_args(p: argparse.ArgumentParser) -> None:

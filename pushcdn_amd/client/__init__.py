@@ -18,7 +18,7 @@ from ..auth.user import UserAuth
 from ..crypto import bls
 from ..proto import message as m
 from ..proto.errors import ConnectionError_
-from ..proto.limiter import Bytes, Limiter
+from ..proto.limiter import Limiter
 from ..proto.transports.base import Connection
 
 RECONNECT_ATTEMPT_TIMEOUT_S = 10.0

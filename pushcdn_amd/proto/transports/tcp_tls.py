@@ -15,7 +15,7 @@ from ...crypto import tls as tlslib
 from ..errors import ConnectionError_
 from ..limiter import Limiter
 from .base import Connection, Listener, Protocol, UnfinalizedConnection
-from .tcp import TcpListener, _set_nodelay, parse_endpoint
+from .tcp import _set_nodelay, parse_endpoint
 
 
 class TcpTlsUnfinalized(UnfinalizedConnection):

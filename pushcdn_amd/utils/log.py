@@ -13,7 +13,6 @@ import logging
 import os
 import sys
 import time
-from typing import Any
 
 from .mnemonic import mnemonic
 

@@ -1,7 +1,6 @@
 #!/usr/bin/env python3
 """Grid + batch sweeps for the broadcast tick (within-process A/B)."""
 
-import json
 import random
 import sys
 import time

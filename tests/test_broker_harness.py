@@ -8,7 +8,6 @@
 
 import asyncio
 
-import pytest
 
 from pushcdn_amd.broker.testing import (
     TestBroker,

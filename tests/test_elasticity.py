@@ -3,9 +3,7 @@ subscription replay after its broker dies; broker mesh self-healing after a
 peer kill/rejoin; eviction on send failure."""
 
 import asyncio
-import uuid
 
-import pytest
 
 from tests.test_integration import (
     make_broker,

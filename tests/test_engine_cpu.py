@@ -2,7 +2,6 @@
 broadcast fan-out to subscribers only, per-user FIFO order, direct routing,
 ring records, drops on ring-full."""
 
-import torch
 
 from pushcdn_amd.broker.gpu_engine import GpuBrokerEngine, parse_ring_records
 from pushcdn_amd.proto import message as m

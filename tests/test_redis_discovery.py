@@ -8,7 +8,6 @@ one-time permits, whitelist (empty set = allow all)."""
 import asyncio
 import time
 
-import pytest
 
 from pushcdn_amd.discovery import BrokerIdentifier
 from pushcdn_amd.discovery.redis import RedisDiscovery

@@ -83,8 +83,6 @@ def test_marshal_gpu_batch_verify(tmp_path):
     import time as _time
 
     from pushcdn_amd.marshal import Marshal, MarshalConfig
-    from pushcdn_amd.auth.marshal import MarshalAuth
-    from pushcdn_amd.crypto.gpu_verify import GpuBatchVerifier
 
     async def go():
         db = new_db(tmp_path)

@@ -16,7 +16,7 @@ from pushcdn_amd.broker.versioned_map import (
 )
 from pushcdn_amd.discovery import BrokerIdentifier
 from pushcdn_amd.discovery.embedded import EmbeddedDiscovery
-from pushcdn_amd.proto.topic import TEST_TOPIC_SPACE, TopicSpace
+from pushcdn_amd.proto.topic import TEST_TOPIC_SPACE
 from pushcdn_amd.proto.errors import TopicError
 
 
