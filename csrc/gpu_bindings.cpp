@@ -54,6 +54,10 @@ void launch_k2a_topic_mask_t(const uint64_t*, const uint8_t*, const int64_t*, co
 void launch_k2b_fused_t(const uint64_t*, const int32_t*, int32_t, int32_t, int32_t, int64_t,
                         int32_t, int32_t, uint64_t*, int32_t*, int32_t*, int32_t*, int64_t*,
                         uint32_t*, hipStream_t);
+void launch_k2b_blocks_t(const uint64_t*, int32_t, int32_t, int32_t, int64_t, int32_t,
+                         int32_t, uint64_t*, int32_t*, int32_t*, int32_t*, int32_t*,
+                         int32_t*, int64_t*, int32_t*, int32_t*, int64_t*, uint32_t*,
+                         hipStream_t);
 }
 
 #define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
@@ -284,6 +288,34 @@ void assign_emit_fused_t(torch::Tensor mask_t, torch::Tensor payload_len,
                        (uint32_t*)drops.data_ptr<int32_t>(), cur_stream());
 }
 
+void assign_emit_blocks_t(torch::Tensor mask_t, torch::Tensor ring_wpos,
+                          int64_t ring_bytes, int64_t n_users,
+                          torch::Tensor bcount, torch::Tensor pprefix, torch::Tensor ubase,
+                          torch::Tensor ufit, torch::Tensor udst,
+                          torch::Tensor pair_user, torch::Tensor pair_msg,
+                          torch::Tensor pair_dst, torch::Tensor drops, torch::Tensor n_pairs,
+                          int64_t uniform_rec) {
+    CHECK_DEV(mask_t); CHECK_CONTIG(mask_t);
+    int32_t W = (int32_t)mask_t.size(0);
+    int32_t M = (int32_t)mask_t.size(1);
+    TORCH_CHECK(ring_bytes % 16 == 0 && uniform_rec > 0);
+    int64_t NB = (M + 31) / 32;
+    TORCH_CHECK(bcount.numel() >= NB * W * 64 && pprefix.numel() >= NB * W * 64,
+                "k2b block scratch too small");
+    TORCH_CHECK(ubase.numel() >= W * 64 && ufit.numel() >= W * 64 && udst.numel() >= W * 64);
+    int32_t capacity = (int32_t)pair_user.size(0);
+    launch_k2b_blocks_t((const uint64_t*)mask_t.data_ptr<int64_t>(), M, W, (int32_t)n_users,
+                        ring_bytes, capacity, (int32_t)uniform_rec,
+                        (uint64_t*)ring_wpos.data_ptr<int64_t>(),
+                        n_pairs.data_ptr<int32_t>(),
+                        bcount.data_ptr<int32_t>(), pprefix.data_ptr<int32_t>(),
+                        ubase.data_ptr<int32_t>(), ufit.data_ptr<int32_t>(),
+                        udst.data_ptr<int64_t>(),
+                        pair_user.data_ptr<int32_t>(), pair_msg.data_ptr<int32_t>(),
+                        pair_dst.data_ptr<int64_t>(),
+                        (uint32_t*)drops.data_ptr<int32_t>(), cur_stream());
+}
+
 void emit_direct(torch::Tensor disc, torch::Tensor owner, torch::Tensor payload_off,
                  torch::Tensor payload_len, int64_t ring_bytes, torch::Tensor ring_wpos,
                  torch::Tensor n_pairs, torch::Tensor pair_user, torch::Tensor pair_msg,
@@ -316,5 +348,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("emit_direct", &emit_direct,
           "K5b: on-device direct-delivery pair emission (no host sync)");
     m.def("assign_emit_fused_t", &assign_emit_fused_t,
-          "K2b fused on transposed mask (sharded counters, uniform-rec fast path)");
+          "K2b fused on transposed mask (wave-aggregated claims, uniform-rec fast path)");
+    m.def("assign_emit_blocks_t", &assign_emit_blocks_t,
+          "K2b block-parallel (P1 count / P2 bases / P3 emit) for uniform records");
 }

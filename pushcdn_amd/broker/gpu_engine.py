@@ -197,6 +197,44 @@ class GpuBrokerEngine:
         assert host_batch is not None and host_offsets is not None
         return self._tick_cpu(host_batch, host_offsets)
 
+    def _k2b_block_scratch(self, M: int):
+        """Lazily (re)allocate the block-K2b scratch for batches up to M
+        messages: [NB][W*64] counts + prefixes, per-user base/fit/dst."""
+        NB = (M + 31) // 32
+        W64 = ((self.n_users + 63) // 64) * 64
+        cur = getattr(self, "_k2b_scratch", None)
+        if cur is not None and cur[0] >= NB:
+            return cur[1]
+        o32 = dict(dtype=torch.int32, device=self.device)
+        bufs = (
+            torch.empty(NB * W64, **o32),              # bcount
+            torch.empty(NB * W64, **o32),              # pprefix
+            torch.empty(W64, **o32),                   # ubase
+            torch.empty(W64, **o32),                   # ufit
+            torch.empty(W64, dtype=torch.int64, device=self.device),  # udst
+        )
+        self._k2b_scratch = (NB, bufs)
+        return bufs
+
+    def _assign_emit(self, ops, mask_t, payload_len, rec: int, M: int) -> None:
+        """Dispatch K2b: block-parallel pipeline for uniform records (fills
+        the chip at any population — the one-lane-per-user fused kernel
+        runs only ~W waves), fused kernel otherwise."""
+        if rec and self.use_gpu_ops:
+            bcount, pprefix, ubase, ufit, udst = self._k2b_block_scratch(M)
+            ops.assign_emit_blocks_t(
+                mask_t, self.ring_wpos, self.ring_bytes, self.n_users,
+                bcount, pprefix, ubase, ufit, udst,
+                self._pair_user, self._pair_msg, self._pair_dst,
+                self._drops, self._n_pairs, rec,
+            )
+        else:
+            ops.assign_emit_fused_t(
+                mask_t, payload_len, self.ring_wpos, self.ring_bytes, self.n_users,
+                self._pair_user, self._pair_msg, self._pair_dst, self._drops,
+                self._n_pairs, rec,
+            )
+
     def _tick_gpu(self, buf: torch.Tensor, offsets: torch.Tensor,
                   uniform_wire_len: Optional[int] = None) -> TickStats:
         ops = self._ops
@@ -214,11 +252,7 @@ class GpuBrokerEngine:
         self._n_pairs.zero_()
         uniform = self.fanout_wire and uniform_wire_len is not None
         rec = ring_rec(uniform_wire_len) if uniform else 0
-        ops.assign_emit_fused_t(
-            mask_t, payload_len, self.ring_wpos, self.ring_bytes, self.n_users,
-            self._pair_user, self._pair_msg, self._pair_dst, self._drops, self._n_pairs,
-            rec,
-        )
+        self._assign_emit(ops, mask_t, payload_len, rec, M)
         if self.direct_enabled:
             # K5 lookup + K5b on-device delivery-pair emission: direct pairs
             # append to the same pair list, all consumed by the single
@@ -259,11 +293,7 @@ class GpuBrokerEngine:
         payload_len = (offsets[1:] - offsets[:-1]).to(torch.int32).contiguous()
         self._n_pairs.zero_()
         rec = units * 16
-        ops.assign_emit_fused_t(
-            mask_t, payload_len, self.ring_wpos, self.ring_bytes, self.n_users,
-            self._pair_user, self._pair_msg, self._pair_dst, self._drops, self._n_pairs,
-            rec,
-        )
+        self._assign_emit(ops, mask_t, payload_len, rec, M)
         ops.fanout_flat3(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
                          self._pair_dst, self._seq_dev, self._n_pairs, units, self.egress,
                          1 if self.nt_fanout else 0, 0)

@@ -441,3 +441,65 @@ def test_parse_batch_fuzz_corrupted(ops):
         d = dev.cpu()
         bad = (d != host).nonzero().flatten().tolist()
         assert not bad, f"{name} mismatch at records {bad[:8]} (host disc={[int(pr.disc[i]) for i in bad[:8]]})"
+
+
+def _run_k2b(ops, variant, mask_t, n_users, ring_bytes, cap, rec, wpos0):
+    wpos = wpos0.clone()
+    pu = torch.full((cap,), -7, dtype=torch.int32, device="cuda")
+    pm = torch.full((cap,), -7, dtype=torch.int32, device="cuda")
+    pd = torch.full((cap,), -7, dtype=torch.int64, device="cuda")
+    drops = torch.zeros(1, dtype=torch.int32, device="cuda")
+    n_pairs = torch.zeros(1, dtype=torch.int32, device="cuda")
+    M = mask_t.shape[1]
+    if variant == "fused":
+        wlen = torch.full((M,), rec - 16, dtype=torch.int32, device="cuda")
+        ops.assign_emit_fused_t(mask_t, wlen, wpos, ring_bytes, n_users,
+                                pu, pm, pd, drops, n_pairs, rec)
+    else:
+        W64 = mask_t.shape[0] * 64
+        NB = (M + 31) // 32
+        o32 = dict(dtype=torch.int32, device="cuda")
+        ops.assign_emit_blocks_t(
+            mask_t, wpos, ring_bytes, n_users,
+            torch.empty(NB * W64, **o32), torch.empty(NB * W64, **o32),
+            torch.empty(W64, **o32), torch.empty(W64, **o32),
+            torch.empty(W64, dtype=torch.int64, device="cuda"),
+            pu, pm, pd, drops, n_pairs, rec)
+    torch.cuda.synchronize()
+    return wpos.cpu(), pu.cpu(), pm.cpu(), pd.cpu(), int(drops.cpu()[0]), int(n_pairs.cpu()[0])
+
+
+def _pairs_as_set(pu, pm, pd, n):
+    n = min(n, pu.shape[0])
+    return sorted((int(pu[i]), int(pm[i]), int(pd[i])) for i in range(n))
+
+
+def test_k2b_blocks_matches_fused(ops):
+    """The block-parallel K2b (P1/P2/P3) must produce the same delivery set,
+    ring cursors and drop counts as the one-lane-per-user fused kernel —
+    including ring-full and pair-capacity-clamped regimes."""
+    rng = random.Random(99)
+    n_users, M = 500, 100   # W=8, NB=4: exercises partial blocks + words
+    W = (n_users + 63) // 64
+    mask = torch.zeros((W, M), dtype=torch.int64)
+    for u in range(n_users):
+        for m in range(M):
+            if rng.random() < 0.3:
+                v = int(mask[u >> 6, m]) | (1 << (u & 63))
+                mask[u >> 6, m] = v - (1 << 64) if v >= (1 << 63) else v
+    mask_t = mask.to("cuda")
+    rec = 1040  # 16B header + 1024
+    for ring_bytes, cap, wstart in [
+        (1 << 20, 1 << 18, 0),            # everything fits
+        (rec * 10, 1 << 18, 0),           # ring-full drops (10 records/user)
+        (1 << 20, 5000, 0),               # pair-capacity clamp
+        (rec * 16, 1 << 18, rec * 12),    # pre-advanced cursors (fit=4)
+    ]:
+        wpos0 = torch.full((n_users,), wstart, dtype=torch.int64, device="cuda")
+        wf, puf, pmf, pdf, df, nf = _run_k2b(ops, "fused", mask_t, n_users,
+                                             ring_bytes, cap, rec, wpos0)
+        wb, pub, pmb, pdb, db, nb = _run_k2b(ops, "blocks", mask_t, n_users,
+                                             ring_bytes, cap, rec, wpos0)
+        assert torch.equal(wf, wb), (ring_bytes, cap)
+        assert df == db and nf == nb, (df, db, nf, nb)
+        assert _pairs_as_set(puf, pmf, pdf, nf) == _pairs_as_set(pub, pmb, pdb, nb)
