@@ -492,7 +492,6 @@ def test_k2b_blocks_matches_fused(ops):
     for ring_bytes, cap, wstart in [
         (1 << 20, 1 << 18, 0),            # everything fits
         (rec * 10, 1 << 18, 0),           # ring-full drops (10 records/user)
-        (1 << 20, 5000, 0),               # pair-capacity clamp
         (rec * 16, 1 << 18, rec * 12),    # pre-advanced cursors (fit=4)
     ]:
         wpos0 = torch.full((n_users,), wstart, dtype=torch.int64, device="cuda")
@@ -503,3 +502,17 @@ def test_k2b_blocks_matches_fused(ops):
         assert torch.equal(wf, wb), (ring_bytes, cap)
         assert df == db and nf == nb, (df, db, nf, nb)
         assert _pairs_as_set(puf, pmf, pdf, nf) == _pairs_as_set(pub, pmb, pdb, nb)
+
+    # pair-capacity clamp: WHICH users land under the capacity boundary is
+    # atomic-claim-order dependent (true of the fused kernel run-to-run
+    # too), so compare aggregate invariants, not per-user state
+    cap = 5000
+    wpos0 = torch.zeros(n_users, dtype=torch.int64, device="cuda")
+    wf, puf, pmf, pdf, df, nf = _run_k2b(ops, "fused", mask_t, n_users,
+                                         1 << 20, cap, rec, wpos0)
+    wb, pub, pmb, pdb, db, nb = _run_k2b(ops, "blocks", mask_t, n_users,
+                                         1 << 20, cap, rec, wpos0)
+    assert nf == nb and df == db                      # totals deterministic
+    assert int(wf.sum()) == int(wb.sum()) == cap * rec  # every slot emitted
+    for pu in (puf, pub):
+        assert bool((pu[:cap] >= 0).all())            # all in-capacity slots real
