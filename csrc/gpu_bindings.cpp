@@ -18,17 +18,17 @@ struct ParseOut {
     uint64_t* timestamp;
 };
 
+struct alignas(16) PairRec { int32_t user; int32_t msg; int64_t dst; };
+
 extern "C" {
 void launch_k4_parse(const uint8_t*, const int64_t*, int32_t, ParseOut, hipStream_t);
 void launch_k2a_topic_mask(const uint64_t*, const uint8_t*, const int64_t*, const int32_t*,
                            const int32_t*, uint64_t*, int32_t, int32_t, hipStream_t);
 void launch_k2b_count(const uint64_t*, int32_t, int32_t, int32_t, int32_t*, hipStream_t);
 void launch_k2b_emit(const uint64_t*, const int64_t*, const int32_t*, const int32_t*, int32_t,
-                     int32_t, int32_t, int64_t, uint64_t*, int32_t*, int32_t*, int64_t*,
-                     uint32_t*, hipStream_t);
-void launch_k3_fanout(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
-                      const int32_t*, const int64_t*, const uint32_t*, int32_t, uint8_t*,
-                      hipStream_t);
+                     int32_t, int32_t, int64_t, uint64_t*, PairRec*, uint32_t*, hipStream_t);
+void launch_k3_fanout(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
+                      const uint32_t*, int32_t, uint8_t*, hipStream_t);
 void launch_k5_direct_lookup(const uint64_t*, const int32_t*, int64_t, const uint64_t*, int32_t,
                              int32_t*, hipStream_t);
 void launch_k2c_apply_subs(uint64_t*, const uint8_t*, const int64_t*, const int32_t*,
@@ -36,32 +36,36 @@ void launch_k2c_apply_subs(uint64_t*, const uint8_t*, const int64_t*, const int3
 void launch_k1_bls_verify(const uint8_t*, const uint8_t*, uint8_t*, const int64_t*, int32_t,
                           int32_t*, hipStream_t);
 void launch_k1_hash_to_g1(uint8_t*, const int64_t*, int32_t, uint8_t*, hipStream_t);
-void launch_k3_fanout_wave(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
-                           const int32_t*, const int64_t*, const uint32_t*, const int32_t*,
-                           uint8_t*, int, int, hipStream_t);
-void launch_k3_fanout_flat2(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
-                            const int32_t*, const int64_t*, uint32_t, const int32_t*, int32_t,
-                            int32_t, uint8_t*, int, int, hipStream_t);
-void launch_k3_fanout_flat3(const uint8_t*, const int64_t*, const int32_t*, const int32_t*,
-                            const int32_t*, const int64_t*, const uint32_t*, const int32_t*,
-                            int32_t, int32_t, uint8_t*, int, int, hipStream_t);
+void launch_k3_fanout_wave(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
+                           const uint32_t*, const int32_t*, uint8_t*, int, int, hipStream_t);
+void launch_k3_fanout_flat2(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
+                            uint32_t, const int32_t*, int32_t, int32_t, uint8_t*, int, int,
+                            hipStream_t);
+void launch_k3_fanout_flat3(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
+                            const uint32_t*, const int32_t*, int32_t, int32_t, uint8_t*, int,
+                            int, hipStream_t);
 void launch_k_seq_advance(uint32_t*, int32_t, hipStream_t);
 void launch_k5b_emit_direct(const int32_t*, const int32_t*, const int64_t*, const int32_t*,
-                            int32_t, int64_t, int32_t, uint64_t*, int32_t*, int32_t*, int32_t*,
-                            int64_t*, uint32_t*, hipStream_t);
+                            int32_t, int64_t, int32_t, uint64_t*, int32_t*, PairRec*,
+                            uint32_t*, hipStream_t);
 void launch_k2a_topic_mask_t(const uint64_t*, const uint8_t*, const int64_t*, const int32_t*,
                              const int32_t*, uint64_t*, int32_t, int32_t, hipStream_t);
 void launch_k2b_fused_t(const uint64_t*, const int32_t*, int32_t, int32_t, int32_t, int64_t,
-                        int32_t, int32_t, uint64_t*, int32_t*, int32_t*, int32_t*, int64_t*,
-                        uint32_t*, hipStream_t);
+                        int32_t, int32_t, uint64_t*, int32_t*, PairRec*, uint32_t*,
+                        hipStream_t);
 void launch_k2b_blocks_t(const uint64_t*, int32_t, int32_t, int32_t, int64_t, int32_t,
                          int32_t, uint64_t*, int32_t*, int32_t*, int32_t*, int32_t*,
-                         int32_t*, int64_t*, int32_t*, int32_t*, int64_t*, uint32_t*,
-                         hipStream_t);
+                         int32_t*, int64_t*, PairRec*, uint32_t*, hipStream_t);
 }
 
 #define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+static inline PairRec* pair_ptr(torch::Tensor& pairs) {
+    TORCH_CHECK(pairs.dim() == 2 && pairs.size(1) == 4 && pairs.dtype() == torch::kInt32
+                && pairs.is_contiguous(), "pairs must be a contiguous int32 [cap, 4] tensor");
+    return (PairRec*)pairs.data_ptr<int32_t>();
+}
 
 static inline hipStream_t cur_stream() {
     return at::hip::getCurrentHIPStream().stream();
@@ -124,17 +128,16 @@ std::vector<torch::Tensor> assign_emit(torch::Tensor mask, torch::Tensor payload
     auto cum = counts.cumsum(0).to(torch::kInt32);
     auto pair_base = (cum - counts).contiguous();  // exclusive scan
     int64_t total = cum[n_users - 1].item<int32_t>();  // one small D2H sync
-    auto pair_user = torch::empty({total}, o32);
-    auto pair_msg = torch::empty({total}, o32);
-    auto pair_dst = torch::empty({total},
-                                 torch::TensorOptions().dtype(torch::kInt64).device(mask.device()));
+    auto pairs = torch::empty({total, 4}, o32);
     auto drops = torch::zeros({1}, o32);
     launch_k2b_emit((const uint64_t*)mask.data_ptr<int64_t>(), payload_off.data_ptr<int64_t>(),
                     payload_len.data_ptr<int32_t>(), pair_base.data_ptr<int32_t>(), M, W,
                     (int32_t)n_users, ring_bytes, (uint64_t*)ring_wpos.data_ptr<int64_t>(),
-                    pair_user.data_ptr<int32_t>(), pair_msg.data_ptr<int32_t>(),
-                    pair_dst.data_ptr<int64_t>(), (uint32_t*)drops.data_ptr<int32_t>(),
-                    cur_stream());
+                    pair_ptr(pairs), (uint32_t*)drops.data_ptr<int32_t>(), cur_stream());
+    // split the AoS records back into the golden-test API's three tensors
+    auto pair_user = pairs.select(1, 0).contiguous();
+    auto pair_msg = pairs.select(1, 1).contiguous();
+    auto pair_dst = pairs.slice(1, 2, 4).contiguous().view(torch::kInt64).squeeze(1);
     return {pair_user, pair_msg, pair_dst, drops};
 }
 
@@ -144,9 +147,13 @@ void fanout(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_
     CHECK_DEV(egress); CHECK_CONTIG(egress);
     int32_t n_pairs = (int32_t)pair_user.size(0);
     if (n_pairs == 0) return;
+    auto pairs = torch::empty({n_pairs, 4},
+                              torch::TensorOptions().dtype(torch::kInt32).device(buf.device()));
+    pairs.select(1, 0).copy_(pair_user);
+    pairs.select(1, 1).copy_(pair_msg);
+    pairs.slice(1, 2, 4).view(torch::kInt64).squeeze(1).copy_(pair_dst);
     launch_k3_fanout(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
-                     payload_len.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
-                     pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
+                     payload_len.data_ptr<int32_t>(), pair_ptr(pairs),
                      (const uint32_t*)msg_seq.data_ptr<int32_t>(), n_pairs,
                      egress.data_ptr<uint8_t>(), cur_stream());
 }
@@ -208,41 +215,38 @@ torch::Tensor hash_to_g1_batch(torch::Tensor msgs, torch::Tensor moff) {
 }
 
 void fanout_wave(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_len,
-                 torch::Tensor pair_user, torch::Tensor pair_msg, torch::Tensor pair_dst,
+                 torch::Tensor pairs,
                  torch::Tensor msg_seq, torch::Tensor n_pairs, torch::Tensor egress,
                  int64_t nt, int64_t grid) {
     CHECK_DEV(egress); CHECK_CONTIG(egress);
     launch_k3_fanout_wave(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
-                          payload_len.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
-                          pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
+                          payload_len.data_ptr<int32_t>(), pair_ptr(pairs),
                           (const uint32_t*)msg_seq.data_ptr<int32_t>(),
                           n_pairs.data_ptr<int32_t>(), egress.data_ptr<uint8_t>(), (int)nt,
                           (int)grid, cur_stream());
 }
 
 void fanout_flat2(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_len,
-                  torch::Tensor pair_user, torch::Tensor pair_msg, torch::Tensor pair_dst,
+                  torch::Tensor pairs,
                   int64_t seq_base, torch::Tensor n_pairs, int64_t units_per_pair,
                   torch::Tensor egress, int64_t nt, int64_t grid) {
     CHECK_DEV(egress); CHECK_CONTIG(egress);
-    int32_t capacity = (int32_t)pair_user.size(0);
+    int32_t capacity = (int32_t)pairs.size(0);
     launch_k3_fanout_flat2(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
-                           payload_len.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
-                           pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
+                           payload_len.data_ptr<int32_t>(), pair_ptr(pairs),
                            (uint32_t)seq_base, n_pairs.data_ptr<int32_t>(), capacity,
                            (int32_t)units_per_pair, egress.data_ptr<uint8_t>(), (int)nt,
                            (int)grid, cur_stream());
 }
 
 void fanout_flat3(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_len,
-                  torch::Tensor pair_user, torch::Tensor pair_msg, torch::Tensor pair_dst,
+                  torch::Tensor pairs,
                   torch::Tensor seq_state, torch::Tensor n_pairs, int64_t units_per_pair,
                   torch::Tensor egress, int64_t nt, int64_t grid) {
     CHECK_DEV(egress); CHECK_CONTIG(egress);
-    int32_t capacity = (int32_t)pair_user.size(0);
+    int32_t capacity = (int32_t)pairs.size(0);
     launch_k3_fanout_flat3(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
-                           payload_len.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
-                           pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
+                           payload_len.data_ptr<int32_t>(), pair_ptr(pairs),
                            (const uint32_t*)seq_state.data_ptr<int32_t>(),
                            n_pairs.data_ptr<int32_t>(), capacity, (int32_t)units_per_pair,
                            egress.data_ptr<uint8_t>(), (int)nt, (int)grid, cur_stream());
@@ -271,20 +275,18 @@ torch::Tensor topic_mask_t(torch::Tensor sub_bitmap, torch::Tensor buf,
 
 void assign_emit_fused_t(torch::Tensor mask_t, torch::Tensor payload_len,
                          torch::Tensor ring_wpos, int64_t ring_bytes, int64_t n_users,
-                         torch::Tensor pair_user, torch::Tensor pair_msg,
-                         torch::Tensor pair_dst, torch::Tensor drops, torch::Tensor n_pairs,
+                         torch::Tensor pairs, torch::Tensor drops, torch::Tensor n_pairs,
                          int64_t uniform_rec) {
     CHECK_DEV(mask_t); CHECK_CONTIG(mask_t);
     int32_t W = (int32_t)mask_t.size(0);
     int32_t M = (int32_t)mask_t.size(1);
     TORCH_CHECK(ring_bytes % 16 == 0);
-    int32_t capacity = (int32_t)pair_user.size(0);
+    int32_t capacity = (int32_t)pairs.size(0);
     launch_k2b_fused_t((const uint64_t*)mask_t.data_ptr<int64_t>(),
                        payload_len.data_ptr<int32_t>(), M, W, (int32_t)n_users, ring_bytes,
                        capacity, (int32_t)uniform_rec,
                        (uint64_t*)ring_wpos.data_ptr<int64_t>(),
-                       n_pairs.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
-                       pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
+                       n_pairs.data_ptr<int32_t>(), pair_ptr(pairs),
                        (uint32_t*)drops.data_ptr<int32_t>(), cur_stream());
 }
 
@@ -292,8 +294,7 @@ void assign_emit_blocks_t(torch::Tensor mask_t, torch::Tensor ring_wpos,
                           int64_t ring_bytes, int64_t n_users,
                           torch::Tensor bcount, torch::Tensor pprefix, torch::Tensor ubase,
                           torch::Tensor ufit, torch::Tensor udst,
-                          torch::Tensor pair_user, torch::Tensor pair_msg,
-                          torch::Tensor pair_dst, torch::Tensor drops, torch::Tensor n_pairs,
+                          torch::Tensor pairs, torch::Tensor drops, torch::Tensor n_pairs,
                           int64_t uniform_rec) {
     CHECK_DEV(mask_t); CHECK_CONTIG(mask_t);
     int32_t W = (int32_t)mask_t.size(0);
@@ -303,7 +304,7 @@ void assign_emit_blocks_t(torch::Tensor mask_t, torch::Tensor ring_wpos,
     TORCH_CHECK(bcount.numel() >= NB * W * 64 && pprefix.numel() >= NB * W * 64,
                 "k2b block scratch too small");
     TORCH_CHECK(ubase.numel() >= W * 64 && ufit.numel() >= W * 64 && udst.numel() >= W * 64);
-    int32_t capacity = (int32_t)pair_user.size(0);
+    int32_t capacity = (int32_t)pairs.size(0);
     launch_k2b_blocks_t((const uint64_t*)mask_t.data_ptr<int64_t>(), M, W, (int32_t)n_users,
                         ring_bytes, capacity, (int32_t)uniform_rec,
                         (uint64_t*)ring_wpos.data_ptr<int64_t>(),
@@ -311,23 +312,20 @@ void assign_emit_blocks_t(torch::Tensor mask_t, torch::Tensor ring_wpos,
                         bcount.data_ptr<int32_t>(), pprefix.data_ptr<int32_t>(),
                         ubase.data_ptr<int32_t>(), ufit.data_ptr<int32_t>(),
                         udst.data_ptr<int64_t>(),
-                        pair_user.data_ptr<int32_t>(), pair_msg.data_ptr<int32_t>(),
-                        pair_dst.data_ptr<int64_t>(),
+                        pair_ptr(pairs),
                         (uint32_t*)drops.data_ptr<int32_t>(), cur_stream());
 }
 
 void emit_direct(torch::Tensor disc, torch::Tensor owner, torch::Tensor payload_off,
                  torch::Tensor payload_len, int64_t ring_bytes, torch::Tensor ring_wpos,
-                 torch::Tensor n_pairs, torch::Tensor pair_user, torch::Tensor pair_msg,
-                 torch::Tensor pair_dst, torch::Tensor drops) {
+                 torch::Tensor n_pairs, torch::Tensor pairs, torch::Tensor drops) {
     int32_t M = (int32_t)disc.size(0);
     if (M == 0) return;
-    int32_t capacity = (int32_t)pair_user.size(0);
+    int32_t capacity = (int32_t)pairs.size(0);
     launch_k5b_emit_direct(disc.data_ptr<int32_t>(), owner.data_ptr<int32_t>(),
                            payload_off.data_ptr<int64_t>(), payload_len.data_ptr<int32_t>(),
                            M, ring_bytes, capacity, (uint64_t*)ring_wpos.data_ptr<int64_t>(),
-                           n_pairs.data_ptr<int32_t>(), pair_user.data_ptr<int32_t>(),
-                           pair_msg.data_ptr<int32_t>(), pair_dst.data_ptr<int64_t>(),
+                           n_pairs.data_ptr<int32_t>(), pair_ptr(pairs),
                            (uint32_t*)drops.data_ptr<int32_t>(), cur_stream());
 }
 

@@ -62,6 +62,19 @@ __host__ __device__ inline uint64_t ring_rec(int32_t len) {
     return ((uint64_t)len + 16 + (RING_ALIGN - 1)) & ~(RING_ALIGN - 1);
 }
 
+// A delivery pair as ONE 16-byte record {user, msg, ring dst}: emitters do a
+// single dword4 store per pair and K3 a single dword4 load per unit.  The
+// SoA layout (three arrays) cost ~3x scattered sub-line stores per pair —
+// measured 100 us/tick in k2b_p3_emit at 2.56M pairs before this change.
+struct alignas(16) PairRec { int32_t user; int32_t msg; int64_t dst; };
+static_assert(sizeof(PairRec) == 16, "PairRec must be one dword4");
+typedef unsigned int cdn_v4u __attribute__((ext_vector_type(4)));
+__device__ inline void store_pair(PairRec* p, int32_t u, int32_t m, int64_t d) {
+    PairRec r{u, m, d};
+    cdn_v4u v; memcpy(&v, &r, 16);
+    *(cdn_v4u*)p = v;
+}
+
 // ---------------------------------------------------------------------------
 // 64-bit FNV-1a — the routing hash for user public keys. Host mirror in
 // csrc/common/hash.h / pushcdn_amd/utils/keyhash.py must match bit-for-bit.
@@ -327,9 +340,7 @@ extern "C" __global__ void k2b_emit(
     int32_t M, int32_t W, int32_t n_users,
     int64_t ring_bytes,
     uint64_t* __restrict__ ring_wpos,         // [n_users] persistent write cursor
-    int32_t* __restrict__ pair_user,          // [total]
-    int32_t* __restrict__ pair_msg,           // [total]
-    int64_t* __restrict__ pair_dst,           // [total] byte offset in egress tensor
+    PairRec* __restrict__ pairs,              // [total] {user,msg,dst}
     uint32_t* __restrict__ drops)             // [1] dropped deliveries (ring full)
 {
     int u = blockIdx.x * blockDim.x + threadIdx.x;
@@ -345,12 +356,10 @@ extern "C" __global__ void k2b_emit(
         // 64-aligned record stride (see ring_rec)
         uint64_t rec = ring_rec(len);
         if (wpos + rec > (uint64_t)ring_bytes) {
-            pair_user[slot] = -1; pair_msg[slot] = m; pair_dst[slot] = 0;
+            store_pair(pairs + slot, -1, m, 0);
             slot++; dropped++; continue;
         }
-        pair_user[slot] = u;
-        pair_msg[slot] = m;
-        pair_dst[slot] = (int64_t)u * ring_bytes + wpos;
+        store_pair(pairs + slot, u, m, (int64_t)u * ring_bytes + wpos);
         slot++;
         wpos += rec;
     }
@@ -368,19 +377,18 @@ extern "C" __global__ void k3_fanout(
     const uint8_t* __restrict__ buf,
     const int64_t* __restrict__ payload_off,
     const int32_t* __restrict__ payload_len,
-    const int32_t* __restrict__ pair_user,
-    const int32_t* __restrict__ pair_msg,
-    const int64_t* __restrict__ pair_dst,
+    const PairRec* __restrict__ pairs,
     const uint32_t* __restrict__ msg_seq,   // [M] global sequence numbers
     int32_t n_pairs,
     uint8_t* __restrict__ egress)
 {
     for (int p = blockIdx.x; p < n_pairs; p += gridDim.x) {
-        if (pair_user[p] < 0) continue;  // dropped (ring full)
-        int m = pair_msg[p];
+        const PairRec pr = pairs[p];
+        if (pr.user < 0) continue;  // dropped (ring full)
+        int m = pr.msg;
         int32_t len = payload_len[m];
         const uint8_t* src = buf + payload_off[m];
-        uint8_t* dst = egress + pair_dst[p];
+        uint8_t* dst = egress + pr.dst;
         if (threadIdx.x == 0) {
             uint32_t hdr[4] = {(uint32_t)len, msg_seq[m], 0, 0};
             memcpy(dst, hdr, 16);
@@ -498,22 +506,21 @@ void launch_k2b_count(const uint64_t* mask, int32_t M, int32_t W, int32_t n_user
 
 void launch_k2b_emit(const uint64_t* mask, const int64_t* payload_off,
                      const int32_t* payload_len, const int32_t* pair_base, int32_t M, int32_t W,
-                     int32_t n_users, int64_t ring_bytes, uint64_t* ring_wpos, int32_t* pair_user,
-                     int32_t* pair_msg, int64_t* pair_dst, uint32_t* drops, hipStream_t s) {
+                     int32_t n_users, int64_t ring_bytes, uint64_t* ring_wpos, PairRec* pairs,
+                     uint32_t* drops, hipStream_t s) {
     int threads = 256, blocks = (n_users + threads - 1) / threads;
     hipLaunchKernelGGL(k2b_emit, dim3(blocks), dim3(threads), 0, s, mask, payload_off,
-                       payload_len, pair_base, M, W, n_users, ring_bytes, ring_wpos, pair_user,
-                       pair_msg, pair_dst, drops);
+                       payload_len, pair_base, M, W, n_users, ring_bytes, ring_wpos, pairs,
+                       drops);
 }
 
 void launch_k3_fanout(const uint8_t* buf, const int64_t* payload_off,
-                      const int32_t* payload_len, const int32_t* pair_user,
-                      const int32_t* pair_msg, const int64_t* pair_dst, const uint32_t* msg_seq,
+                      const int32_t* payload_len, const PairRec* pairs, const uint32_t* msg_seq,
                       int32_t n_pairs, uint8_t* egress, hipStream_t s) {
     int blocks = n_pairs < 16384 ? n_pairs : 16384;
     if (blocks == 0) return;
     hipLaunchKernelGGL(k3_fanout, dim3(blocks), dim3(128), 0, s, buf, payload_off, payload_len,
-                       pair_user, pair_msg, pair_dst, msg_seq, n_pairs, egress);
+                       pairs, msg_seq, n_pairs, egress);
 }
 
 void launch_k5_direct_lookup(const uint64_t* table_keys, const int32_t* table_vals,
@@ -548,9 +555,7 @@ __global__ void __launch_bounds__(256) k3_fanout_wave_t(
     const uint8_t* __restrict__ buf,
     const int64_t* __restrict__ payload_off,
     const int32_t* __restrict__ payload_len,
-    const int32_t* __restrict__ pair_user,
-    const int32_t* __restrict__ pair_msg,
-    const int64_t* __restrict__ pair_dst,
+    const PairRec* __restrict__ pairs,
     const uint32_t* __restrict__ msg_seq,
     const int32_t* __restrict__ n_pairs_ptr,
     uint8_t* __restrict__ egress)
@@ -560,12 +565,12 @@ __global__ void __launch_bounds__(256) k3_fanout_wave_t(
     const int wave_in_wg = threadIdx.x >> 6;
     const int waves_total = gridDim.x * 4;
     for (int p = blockIdx.x * 4 + wave_in_wg; p < n_pairs; p += waves_total) {
-        const int u = pair_user[p];
-        if (u < 0) continue;
-        const int mi = pair_msg[p];
+        const PairRec pr = pairs[p];
+        if (pr.user < 0) continue;
+        const int mi = pr.msg;
         const int32_t len = payload_len[mi];
         const uint8_t* src = buf + payload_off[mi];
-        uint8_t* dst = egress + pair_dst[p];
+        uint8_t* dst = egress + pr.dst;
         if (lane == 0) {
             uint32_t hdr[4] = {(uint32_t)len, msg_seq[mi], 0, 0};
             if (NT) {
@@ -602,18 +607,17 @@ __global__ void __launch_bounds__(256) k3_fanout_wave_t(
 extern "C" {
 
 void launch_k3_fanout_wave(const uint8_t* buf, const int64_t* payload_off,
-                           const int32_t* payload_len, const int32_t* pair_user,
-                           const int32_t* pair_msg, const int64_t* pair_dst,
+                           const int32_t* payload_len, const PairRec* pairs,
                            const uint32_t* msg_seq, const int32_t* n_pairs_ptr,
                            uint8_t* egress, int nt, int grid, hipStream_t s) {
     if (grid <= 0) grid = 4096;  // 4096 WGs x 4 waves = 16384 concurrent pairs
     if (nt)
         hipLaunchKernelGGL((k3_fanout_wave_t<1>), dim3(grid), dim3(256), 0, s, buf, payload_off,
-                           payload_len, pair_user, pair_msg, pair_dst, msg_seq, n_pairs_ptr,
+                           payload_len, pairs, msg_seq, n_pairs_ptr,
                            egress);
     else
         hipLaunchKernelGGL((k3_fanout_wave_t<0>), dim3(grid), dim3(256), 0, s, buf, payload_off,
-                           payload_len, pair_user, pair_msg, pair_dst, msg_seq, n_pairs_ptr,
+                           payload_len, pairs, msg_seq, n_pairs_ptr,
                            egress);
 }
 
@@ -635,9 +639,7 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
     const uint8_t* __restrict__ buf,
     const int64_t* __restrict__ payload_off,
     const int32_t* __restrict__ payload_len,
-    const int32_t* __restrict__ pair_user,
-    const int32_t* __restrict__ pair_msg,
-    const int64_t* __restrict__ pair_dst,
+    const PairRec* __restrict__ pairs,
     uint32_t seq_base_val,
     const uint32_t* __restrict__ seq_state,
     const int32_t* __restrict__ n_pairs_ptr,
@@ -654,10 +656,10 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
     for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < n_units; f += stride) {
         const int p = (int)(f / units_per_pair);
         const int unit = (int)(f - (int64_t)p * units_per_pair);
-        const int u = pair_user[p];
-        if (u < 0) continue;
-        const int mi = pair_msg[p];
-        uint8_t* dst = egress + pair_dst[p] + (size_t)unit * 16;
+        const PairRec pr = pairs[p];   // one dword4 load
+        if (pr.user < 0) continue;
+        const int mi = pr.msg;
+        uint8_t* dst = egress + pr.dst + (size_t)unit * 16;
         if (unit == 0) {
             uint32_t hdr[4] = {(uint32_t)payload_len[mi], seq_base + (uint32_t)mi, 0, 0};
             v4u h; memcpy(&h, hdr, 16);
@@ -692,33 +694,33 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
 
 extern "C" void launch_k3_fanout_flat2(
     const uint8_t* buf, const int64_t* payload_off, const int32_t* payload_len,
-    const int32_t* pair_user, const int32_t* pair_msg, const int64_t* pair_dst,
+    const PairRec* pairs,
     uint32_t seq_base, const int32_t* n_pairs_ptr, int32_t capacity, int32_t units_per_pair,
     uint8_t* egress, int nt, int grid, hipStream_t s) {
     if (grid <= 0) grid = 16384;  // swept: 16384 > 8192 > 4096 (~0.5% each)
     if (nt)
         hipLaunchKernelGGL((k3_fanout_flat_t<1, false>), dim3(grid), dim3(256), 0, s, buf,
-                           payload_off, payload_len, pair_user, pair_msg, pair_dst, seq_base,
+                           payload_off, payload_len, pairs, seq_base,
                            nullptr, n_pairs_ptr, capacity, units_per_pair, egress);
     else
         hipLaunchKernelGGL((k3_fanout_flat_t<0, false>), dim3(grid), dim3(256), 0, s, buf,
-                           payload_off, payload_len, pair_user, pair_msg, pair_dst, seq_base,
+                           payload_off, payload_len, pairs, seq_base,
                            nullptr, n_pairs_ptr, capacity, units_per_pair, egress);
 }
 
 extern "C" void launch_k3_fanout_flat3(
     const uint8_t* buf, const int64_t* payload_off, const int32_t* payload_len,
-    const int32_t* pair_user, const int32_t* pair_msg, const int64_t* pair_dst,
+    const PairRec* pairs,
     const uint32_t* seq_state, const int32_t* n_pairs_ptr, int32_t capacity,
     int32_t units_per_pair, uint8_t* egress, int nt, int grid, hipStream_t s) {
     if (grid <= 0) grid = 16384;
     if (nt)
         hipLaunchKernelGGL((k3_fanout_flat_t<1, true>), dim3(grid), dim3(256), 0, s, buf,
-                           payload_off, payload_len, pair_user, pair_msg, pair_dst, 0u,
+                           payload_off, payload_len, pairs, 0u,
                            seq_state, n_pairs_ptr, capacity, units_per_pair, egress);
     else
         hipLaunchKernelGGL((k3_fanout_flat_t<0, true>), dim3(grid), dim3(256), 0, s, buf,
-                           payload_off, payload_len, pair_user, pair_msg, pair_dst, 0u,
+                           payload_off, payload_len, pairs, 0u,
                            seq_state, n_pairs_ptr, capacity, units_per_pair, egress);
 }
 
@@ -775,9 +777,7 @@ extern "C" __global__ void k2b_fused_t(
     int64_t ring_bytes, int32_t capacity, int32_t uniform_rec,
     uint64_t* __restrict__ ring_wpos,
     int32_t* __restrict__ n_pairs,            // [1] global pair counter (zeroed)
-    int32_t* __restrict__ pair_user,
-    int32_t* __restrict__ pair_msg,
-    int64_t* __restrict__ pair_dst,
+    PairRec* __restrict__ pairs,
     uint32_t* __restrict__ drops)
 {
     const int u = blockIdx.x * blockDim.x + threadIdx.x;
@@ -814,13 +814,11 @@ extern "C" __global__ void k2b_fused_t(
             if (!(col[m] & bit)) continue;
             const bool ok = (emitted < fit) && (slot < capacity);
             if (!ok) {
-                if (slot < capacity) { pair_user[slot] = -1; pair_msg[slot] = m; pair_dst[slot] = 0; slot++; }
+                if (slot < capacity) { store_pair(pairs + slot, -1, m, 0); slot++; }
                 dropped++;
                 continue;
             }
-            pair_user[slot] = u;
-            pair_msg[slot] = m;
-            pair_dst[slot] = dst_base + (int64_t)emitted * uniform_rec;
+            store_pair(pairs + slot, u, m, dst_base + (int64_t)emitted * uniform_rec);
             slot++;
             emitted++;
         }
@@ -838,13 +836,11 @@ extern "C" __global__ void k2b_fused_t(
         bool fits_ring = (wpos + rec <= (uint64_t)ring_bytes);
         bool fits_cap = (slot < capacity);
         if (!fits_ring || !fits_cap) {
-            if (fits_cap) { pair_user[slot] = -1; pair_msg[slot] = m; pair_dst[slot] = 0; slot++; }
+            if (fits_cap) { store_pair(pairs + slot, -1, m, 0); slot++; }
             dropped++;
             continue;
         }
-        pair_user[slot] = u;
-        pair_msg[slot] = m;
-        pair_dst[slot] = (int64_t)u * ring_bytes + wpos;
+        store_pair(pairs + slot, u, m, (int64_t)u * ring_bytes + wpos);
         slot++;
         wpos += rec;
     }
@@ -955,9 +951,7 @@ extern "C" __global__ void k2b_p3_emit(
     const int64_t* __restrict__ udst,
     int32_t M, int32_t W, int32_t NB, int32_t n_users,
     int32_t capacity, int32_t uniform_rec,
-    int32_t* __restrict__ pair_user,
-    int32_t* __restrict__ pair_msg,
-    int64_t* __restrict__ pair_dst)
+    PairRec* __restrict__ pairs)
 {
     const int wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     if (wave >= W * NB) return;
@@ -975,13 +969,10 @@ extern "C" __global__ void k2b_p3_emit(
         if (!(col[m] & bit)) continue;
         const int slot = base + p;
         if (slot < capacity) {
-            if (p < fit) {
-                pair_user[slot] = u;
-                pair_msg[slot] = m;
-                pair_dst[slot] = dstb + (int64_t)p * uniform_rec;
-            } else {  // ring full: placeholder, counted in P2
-                pair_user[slot] = -1; pair_msg[slot] = m; pair_dst[slot] = 0;
-            }
+            if (p < fit)
+                store_pair(pairs + slot, u, m, dstb + (int64_t)p * uniform_rec);
+            else  // ring full: placeholder, counted in P2
+                store_pair(pairs + slot, -1, m, 0);
         }
         p++;
     }
@@ -992,7 +983,7 @@ extern "C" void launch_k2b_blocks_t(
     int64_t ring_bytes, int32_t capacity, int32_t uniform_rec,
     uint64_t* ring_wpos, int32_t* n_pairs,
     int32_t* bcount, int32_t* pprefix, int32_t* ubase, int32_t* ufit, int64_t* udst,
-    int32_t* pair_user, int32_t* pair_msg, int64_t* pair_dst,
+    PairRec* pairs,
     uint32_t* drops, hipStream_t s) {
     const int NB = (M + K2B_BLK - 1) / K2B_BLK;
     const int waves = W * NB;
@@ -1006,18 +997,18 @@ extern "C" void launch_k2b_blocks_t(
                        ring_wpos, n_pairs, pprefix, ubase, ufit, udst, drops);
     hipLaunchKernelGGL(k2b_p3_emit, dim3(blocks_wb), dim3(threads), 0, s,
                        mask_t, pprefix, ubase, ufit, udst, M, W, NB, n_users,
-                       capacity, uniform_rec, pair_user, pair_msg, pair_dst);
+                       capacity, uniform_rec, pairs);
 }
 
 void launch_k2b_fused_t(const uint64_t* mask_t, const int32_t* payload_len, int32_t M,
                         int32_t W, int32_t n_users, int64_t ring_bytes, int32_t capacity,
                         int32_t uniform_rec, uint64_t* ring_wpos, int32_t* n_pairs,
-                        int32_t* pair_user, int32_t* pair_msg, int64_t* pair_dst,
+                        PairRec* pairs,
                         uint32_t* drops, hipStream_t s) {
     int threads = 256, blocks = (n_users + threads - 1) / threads;
     hipLaunchKernelGGL(k2b_fused_t, dim3(blocks), dim3(threads), 0, s, mask_t, payload_len, M,
                        W, n_users, ring_bytes, capacity, uniform_rec, ring_wpos, n_pairs,
-                       pair_user, pair_msg, pair_dst, drops);
+                       pairs, drops);
 }
 
 }  // extern "C"
@@ -1040,9 +1031,7 @@ extern "C" __global__ void k5b_emit_direct(
     int64_t ring_bytes, int32_t capacity,
     uint64_t* __restrict__ ring_wpos,
     int32_t* __restrict__ n_pairs,
-    int32_t* __restrict__ pair_user,
-    int32_t* __restrict__ pair_msg,
-    int64_t* __restrict__ pair_dst,
+    PairRec* __restrict__ pairs,
     uint32_t* __restrict__ drops)
 {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -1060,22 +1049,19 @@ extern "C" __global__ void k5b_emit_direct(
         if (old + rec > (uint64_t)ring_bytes)
             atomicAdd((unsigned long long*)&ring_wpos[u],
                       (unsigned long long)(-(long long)rec));
-        if (slot < capacity) { pair_user[slot] = -1; pair_msg[slot] = i; pair_dst[slot] = 0; }
+        if (slot < capacity) store_pair(pairs + slot, -1, i, 0);
         atomicAdd(drops, 1u);
         return;
     }
-    pair_user[slot] = u;
-    pair_msg[slot] = i;
-    pair_dst[slot] = (int64_t)u * ring_bytes + (int64_t)old;
+    store_pair(pairs + slot, u, i, (int64_t)u * ring_bytes + (int64_t)old);
 }
 
 extern "C" void launch_k5b_emit_direct(
     const int32_t* disc, const int32_t* owner, const int64_t* payload_off,
     const int32_t* payload_len, int32_t M, int64_t ring_bytes, int32_t capacity,
-    uint64_t* ring_wpos, int32_t* n_pairs, int32_t* pair_user, int32_t* pair_msg,
-    int64_t* pair_dst, uint32_t* drops, hipStream_t s) {
+    uint64_t* ring_wpos, int32_t* n_pairs, PairRec* pairs, uint32_t* drops, hipStream_t s) {
     int threads = 256, blocks = (M + threads - 1) / threads;
     hipLaunchKernelGGL(k5b_emit_direct, dim3(blocks), dim3(threads), 0, s, disc, owner,
                        payload_off, payload_len, M, ring_bytes, capacity, ring_wpos, n_pairs,
-                       pair_user, pair_msg, pair_dst, drops);
+                       pairs, drops);
 }

@@ -106,9 +106,10 @@ class GpuBrokerEngine:
         self.total = TickStats()
         if self.use_gpu_ops:
             o32 = dict(dtype=torch.int32, device=dev)
-            self._pair_user = torch.empty(pair_capacity, **o32)
-            self._pair_msg = torch.empty(pair_capacity, **o32)
-            self._pair_dst = torch.empty(pair_capacity, dtype=torch.int64, device=dev)
+            # delivery pairs as 16 B AoS records {i32 user, i32 msg, i64 dst}
+            # — one dword4 store per pair in the emitters, one dword4 load
+            # per unit in K3 (SoA cost 3x scattered sub-line stores)
+            self._pairs = torch.empty((pair_capacity, 4), **o32)
             self._drops = torch.zeros(1, **o32)
             self._n_pairs = torch.zeros(1, **o32)
             self._seq_dev = torch.zeros(1, **o32)  # device seq counter (graph path)
@@ -225,14 +226,12 @@ class GpuBrokerEngine:
             ops.assign_emit_blocks_t(
                 mask_t, self.ring_wpos, self.ring_bytes, self.n_users,
                 bcount, pprefix, ubase, ufit, udst,
-                self._pair_user, self._pair_msg, self._pair_dst,
-                self._drops, self._n_pairs, rec,
+                self._pairs, self._drops, self._n_pairs, rec,
             )
         else:
             ops.assign_emit_fused_t(
                 mask_t, payload_len, self.ring_wpos, self.ring_bytes, self.n_users,
-                self._pair_user, self._pair_msg, self._pair_dst, self._drops,
-                self._n_pairs, rec,
+                self._pairs, self._drops, self._n_pairs, rec,
             )
 
     def _tick_gpu(self, buf: torch.Tensor, offsets: torch.Tensor,
@@ -261,8 +260,7 @@ class GpuBrokerEngine:
             # same padded wire length as broadcasts.)
             owner = ops.direct_lookup(self.direct_keys, self.direct_vals, recip_hash)
             ops.emit_direct(disc, owner, payload_off, payload_len, self.ring_bytes,
-                            self.ring_wpos, self._n_pairs, self._pair_user,
-                            self._pair_msg, self._pair_dst, self._drops)
+                            self.ring_wpos, self._n_pairs, self._pairs, self._drops)
         seq_base = self.seq
         self.seq += M
         nt = 1 if self.nt_fanout else 0
@@ -272,13 +270,12 @@ class GpuBrokerEngine:
         # costs ~3% (measured on the 64 KiB mixed bench) — use wave there.
         if uniform and rec <= 4096:
             units = rec // 16
-            ops.fanout_flat2(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
-                             self._pair_dst, seq_base, self._n_pairs, units, self.egress,
-                             nt, 0)
+            ops.fanout_flat2(buf, payload_off, payload_len, self._pairs, seq_base,
+                             self._n_pairs, units, self.egress, nt, 0)
         else:
             seq = torch.arange(seq_base, seq_base + M, dtype=torch.int32, device=self.device)
-            ops.fanout_wave(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
-                            self._pair_dst, seq, self._n_pairs, self.egress, nt, 0)
+            ops.fanout_wave(buf, payload_off, payload_len, self._pairs, seq,
+                            self._n_pairs, self.egress, nt, 0)
         return TickStats(n_messages=M)
 
     def _graph_tick_body(self, buf: torch.Tensor, offsets: torch.Tensor, units: int) -> None:
@@ -294,8 +291,8 @@ class GpuBrokerEngine:
         self._n_pairs.zero_()
         rec = units * 16
         self._assign_emit(ops, mask_t, payload_len, rec, M)
-        ops.fanout_flat3(buf, payload_off, payload_len, self._pair_user, self._pair_msg,
-                         self._pair_dst, self._seq_dev, self._n_pairs, units, self.egress,
+        ops.fanout_flat3(buf, payload_off, payload_len, self._pairs, self._seq_dev,
+                         self._n_pairs, units, self.egress,
                          1 if self.nt_fanout else 0, 0)
         ops.seq_advance(self._seq_dev, M)
 

@@ -363,9 +363,7 @@ def test_fused_t_capacity_drop(ops):
     mask_t = ops.topic_mask_t(sub.to("cuda"), dbuf, toff, tcnt, disc)
     wpos = torch.zeros(n_users, dtype=torch.int64, device="cuda")
     cap = 40  # < 64 deliveries
-    pu = torch.empty(cap, dtype=torch.int32, device="cuda")
-    pm = torch.empty(cap, dtype=torch.int32, device="cuda")
-    pd = torch.empty(cap, dtype=torch.int64, device="cuda")
+    pairs = torch.empty((cap, 4), dtype=torch.int32, device="cuda")
     drops = torch.zeros(1, dtype=torch.int32, device="cuda")
     n_pairs = torch.zeros(1, dtype=torch.int32, device="cuda")
     woff = doff[:-1].contiguous()
@@ -374,14 +372,14 @@ def test_fused_t_capacity_drop(ops):
 
     rec = ring_rec(padded)
     # ring sized so all 64 deliveries FIT — only pair-capacity drops counted
-    ops.assign_emit_fused_t(mask_t, wlen, wpos, 64 * rec, n_users, pu, pm, pd, drops,
+    ops.assign_emit_fused_t(mask_t, wlen, wpos, 64 * rec, n_users, pairs, drops,
                             n_pairs, rec)
     torch.cuda.synchronize()
     # wave-aggregated claims may overshoot the clamp in the counter itself;
     # what matters: exactly `cap` pairs were written and the rest counted
     assert int(drops.cpu()[0]) == 64 - cap
     assert int(n_pairs.cpu()[0]) >= cap
-    assert bool((pu[:cap].cpu() >= 0).all())  # every in-capacity slot written
+    assert bool((pairs[:cap, 0].cpu() >= 0).all())  # every in-capacity slot written
 
 
 def _random_message(rng: random.Random) -> "m.Message":
@@ -445,16 +443,14 @@ def test_parse_batch_fuzz_corrupted(ops):
 
 def _run_k2b(ops, variant, mask_t, n_users, ring_bytes, cap, rec, wpos0):
     wpos = wpos0.clone()
-    pu = torch.full((cap,), -7, dtype=torch.int32, device="cuda")
-    pm = torch.full((cap,), -7, dtype=torch.int32, device="cuda")
-    pd = torch.full((cap,), -7, dtype=torch.int64, device="cuda")
+    pairs = torch.full((cap, 4), -7, dtype=torch.int32, device="cuda")
     drops = torch.zeros(1, dtype=torch.int32, device="cuda")
     n_pairs = torch.zeros(1, dtype=torch.int32, device="cuda")
     M = mask_t.shape[1]
     if variant == "fused":
         wlen = torch.full((M,), rec - 16, dtype=torch.int32, device="cuda")
         ops.assign_emit_fused_t(mask_t, wlen, wpos, ring_bytes, n_users,
-                                pu, pm, pd, drops, n_pairs, rec)
+                                pairs, drops, n_pairs, rec)
     else:
         W64 = mask_t.shape[0] * 64
         NB = (M + 31) // 32
@@ -464,14 +460,15 @@ def _run_k2b(ops, variant, mask_t, n_users, ring_bytes, cap, rec, wpos0):
             torch.empty(NB * W64, **o32), torch.empty(NB * W64, **o32),
             torch.empty(W64, **o32), torch.empty(W64, **o32),
             torch.empty(W64, dtype=torch.int64, device="cuda"),
-            pu, pm, pd, drops, n_pairs, rec)
+            pairs, drops, n_pairs, rec)
     torch.cuda.synchronize()
-    return wpos.cpu(), pu.cpu(), pm.cpu(), pd.cpu(), int(drops.cpu()[0]), int(n_pairs.cpu()[0])
+    return wpos.cpu(), pairs.cpu(), int(drops.cpu()[0]), int(n_pairs.cpu()[0])
 
 
-def _pairs_as_set(pu, pm, pd, n):
-    n = min(n, pu.shape[0])
-    return sorted((int(pu[i]), int(pm[i]), int(pd[i])) for i in range(n))
+def _pairs_as_set(pairs, n):
+    n = min(n, pairs.shape[0])
+    pd = pairs[:, 2:4].contiguous().view(torch.int64).flatten()
+    return sorted((int(pairs[i, 0]), int(pairs[i, 1]), int(pd[i])) for i in range(n))
 
 
 def test_k2b_blocks_matches_fused(ops):
@@ -495,24 +492,24 @@ def test_k2b_blocks_matches_fused(ops):
         (rec * 16, 1 << 18, rec * 12),    # pre-advanced cursors (fit=4)
     ]:
         wpos0 = torch.full((n_users,), wstart, dtype=torch.int64, device="cuda")
-        wf, puf, pmf, pdf, df, nf = _run_k2b(ops, "fused", mask_t, n_users,
-                                             ring_bytes, cap, rec, wpos0)
-        wb, pub, pmb, pdb, db, nb = _run_k2b(ops, "blocks", mask_t, n_users,
-                                             ring_bytes, cap, rec, wpos0)
+        wf, prf, df, nf = _run_k2b(ops, "fused", mask_t, n_users,
+                                   ring_bytes, cap, rec, wpos0)
+        wb, prb, db, nb = _run_k2b(ops, "blocks", mask_t, n_users,
+                                   ring_bytes, cap, rec, wpos0)
         assert torch.equal(wf, wb), (ring_bytes, cap)
         assert df == db and nf == nb, (df, db, nf, nb)
-        assert _pairs_as_set(puf, pmf, pdf, nf) == _pairs_as_set(pub, pmb, pdb, nb)
+        assert _pairs_as_set(prf, nf) == _pairs_as_set(prb, nb)
 
     # pair-capacity clamp: WHICH users land under the capacity boundary is
     # atomic-claim-order dependent (true of the fused kernel run-to-run
     # too), so compare aggregate invariants, not per-user state
     cap = 5000
     wpos0 = torch.zeros(n_users, dtype=torch.int64, device="cuda")
-    wf, puf, pmf, pdf, df, nf = _run_k2b(ops, "fused", mask_t, n_users,
-                                         1 << 20, cap, rec, wpos0)
-    wb, pub, pmb, pdb, db, nb = _run_k2b(ops, "blocks", mask_t, n_users,
-                                         1 << 20, cap, rec, wpos0)
+    wf, prf, df, nf = _run_k2b(ops, "fused", mask_t, n_users,
+                               1 << 20, cap, rec, wpos0)
+    wb, prb, db, nb = _run_k2b(ops, "blocks", mask_t, n_users,
+                               1 << 20, cap, rec, wpos0)
     assert nf == nb and df == db                      # totals deterministic
     assert int(wf.sum()) == int(wb.sum()) == cap * rec  # every slot emitted
-    for pu in (puf, pub):
-        assert bool((pu[:cap] >= 0).all())            # all in-capacity slots real
+    for pr in (prf, prb):
+        assert bool((pr[:cap, 0] >= 0).all())         # all in-capacity slots real
