@@ -1052,10 +1052,14 @@ extern "C" __global__ void k2b_p3_emit_m(
 
 }  // extern "C"  (template below needs C++ linkage)
 
-// K3m: one wave stages a message's wire bytes in registers (<= 4 KiB
-// record = 4 x 16 B per lane) and streams them to a segment of the
-// message's recipients.  Reads: units x segments per message; writes: the
-// same non-temporal full-record stores as the flat kernel.
+// K3m: each workgroup stages one message's full record (16 B header +
+// wire bytes, <= 4 KiB) in LDS once, then streams the flat unit stream of
+// its recipient SEGMENT at ~full lane utilization — the flat kernel
+// re-reads every unit from L2 once per recipient (2.9 GB/tick at the
+// headline config), capping it ~20% under the 5.5 TB/s write-only
+// roofline measured for the same store pattern (scripts/k3_roofline.hip).
+// (A register-staged variant was tried first and measured SLOWER: with
+// 70 units/record its second register pass idles 58 of 64 lanes.)
 template <int NT, bool SEQ_FROM_PTR>
 __global__ void __launch_bounds__(256) k3_fanout_m(
     const uint8_t* __restrict__ buf,
@@ -1068,44 +1072,46 @@ __global__ void __launch_bounds__(256) k3_fanout_m(
     int32_t units_per_pair, int32_t wire_len,
     uint8_t* __restrict__ egress)
 {
-    const int wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-    if (wave >= M * groups) return;
-    const int m = wave / groups, g = wave - (wave / groups) * groups;
-    const int lane = threadIdx.x & 63;
+    __shared__ cdn_v4u smsg[256];  // up to 4 KiB record (header + wire)
+    const int wg = blockIdx.x;     // one workgroup per (message, segment)
+    if (wg >= M * groups) return;
+    const int m = wg / groups, g = wg - (wg / groups) * groups;
+    const int tid = threadIdx.x;
     const uint32_t seq_base = SEQ_FROM_PTR ? seq_state[0] : seq_base_val;
 
-    // stage the record in registers: unit 0 is the 16 B header, units
-    // 1..units-1 the wire bytes (uniform, 16-aligned)
-    cdn_v4u regs[4];
-    const int passes = (units_per_pair + 63) >> 6;
     const uint8_t* src = buf + payload_off[m];
-    for (int r = 0; r < passes; ++r) {
-        const int unit = (r << 6) + lane;
-        if (unit == 0) {
+    for (int u = tid; u < units_per_pair; u += (int)blockDim.x) {
+        if (u == 0) {
             uint32_t hdr[4] = {(uint32_t)wire_len, seq_base + (uint32_t)m, 0, 0};
-            memcpy(&regs[0], hdr, 16);
-        } else if (unit < units_per_pair) {
-            regs[r] = *(const cdn_v4u*)(src + ((size_t)unit - 1) * 16);
+            memcpy(&smsg[0], hdr, 16);
+        } else {
+            smsg[u] = *(const cdn_v4u*)(src + ((size_t)u - 1) * 16);
         }
     }
+    __syncthreads();
 
     int lo = mbase[m], hi = mbase[m + 1];
     if (hi > capacity) hi = capacity;
     const int cnt = hi - lo;
     if (cnt <= 0) return;
     const int seg = (cnt + groups - 1) / groups;
-    const int i0 = lo + g * seg;
-    const int i1 = min(hi, i0 + seg);
-    for (int i = i0; i < i1; ++i) {
-        const int64_t dst0 = mdst[i];
-        if (dst0 < 0) continue;  // ring-full drop
-        uint8_t* dst = egress + dst0;
-        for (int r = 0; r < passes; ++r) {
-            const int unit = (r << 6) + lane;
-            if (unit >= units_per_pair) break;
-            if (NT) __builtin_nontemporal_store(regs[r], (cdn_v4u*)(dst + (size_t)unit * 16));
-            else *(cdn_v4u*)(dst + (size_t)unit * 16) = regs[r];
+    const int i0 = g * seg;
+    const int i1 = min(cnt, i0 + seg);
+    if (i0 >= i1) return;
+    // flat unit stream over the segment: f -> (recipient i, unit u),
+    // strength-reduced (no division in the loop)
+    const int64_t n_units = (int64_t)(i1 - i0) * units_per_pair;
+    int i = i0 + tid / units_per_pair;
+    int u = tid - (tid / units_per_pair) * units_per_pair;
+    for (int64_t f = tid; f < n_units; f += (int)blockDim.x) {
+        const int64_t dst0 = mdst[lo + i];
+        if (dst0 >= 0) {
+            uint8_t* dst = egress + dst0 + (size_t)u * 16;
+            if (NT) __builtin_nontemporal_store(smsg[u], (cdn_v4u*)dst);
+            else *(cdn_v4u*)dst = smsg[u];
         }
+        u += (int)blockDim.x;
+        while (u >= units_per_pair) { u -= units_per_pair; ++i; }
     }
 }
 
@@ -1131,11 +1137,10 @@ extern "C" void launch_k3_fanout_m(
     const int64_t* mdst, uint32_t seq_base, const uint32_t* seq_state, int32_t M,
     int32_t capacity, int32_t units_per_pair, int32_t wire_len, uint8_t* egress,
     int nt, hipStream_t s) {
-    int groups = 8192 / (M > 0 ? M : 1);
+    int groups = 2048 / (M > 0 ? M : 1);  // >=2048 WGs fills 8 XCDs
     if (groups < 1) groups = 1;
     if (groups > 256) groups = 256;
-    const int waves = M * groups;
-    const int blocks = (waves * 64 + 255) / 256;
+    const int blocks = M * groups;
     if (seq_state) {
         if (nt) hipLaunchKernelGGL((k3_fanout_m<1, true>), dim3(blocks), dim3(256), 0, s,
                     buf, payload_off, mbase, mdst, 0u, seq_state, M, groups, capacity,
