@@ -55,13 +55,7 @@ void launch_k2b_fused_t(const uint64_t*, const int32_t*, int32_t, int32_t, int32
                         hipStream_t);
 void launch_k2b_blocks_t(const uint64_t*, int32_t, int32_t, int32_t, int64_t, int32_t,
                          int32_t, uint64_t*, int32_t*, int32_t*, int32_t*, int32_t*,
-                         int32_t*, int64_t*, PairRec*, uint32_t*, int, hipStream_t);
-void launch_k2b_m_index(const uint64_t*, const int32_t*, const int32_t*, const int64_t*,
-                        int32_t, int32_t, int32_t, int32_t, int32_t, int32_t*, int32_t*,
-                        int32_t*, int64_t*, hipStream_t);
-void launch_k3_fanout_m(const uint8_t*, const int64_t*, const int32_t*, const int64_t*,
-                        uint32_t, const uint32_t*, int32_t, int32_t, int32_t, int32_t,
-                        uint8_t*, int, hipStream_t);
+                         int32_t*, int64_t*, PairRec*, uint32_t*, hipStream_t);
 }
 
 #define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
@@ -301,7 +295,7 @@ void assign_emit_blocks_t(torch::Tensor mask_t, torch::Tensor ring_wpos,
                           torch::Tensor bcount, torch::Tensor pprefix, torch::Tensor ubase,
                           torch::Tensor ufit, torch::Tensor udst,
                           torch::Tensor pairs, torch::Tensor drops, torch::Tensor n_pairs,
-                          int64_t uniform_rec, int64_t emit_pairs) {
+                          int64_t uniform_rec) {
     CHECK_DEV(mask_t); CHECK_CONTIG(mask_t);
     int32_t W = (int32_t)mask_t.size(0);
     int32_t M = (int32_t)mask_t.size(1);
@@ -319,43 +313,7 @@ void assign_emit_blocks_t(torch::Tensor mask_t, torch::Tensor ring_wpos,
                         ubase.data_ptr<int32_t>(), ufit.data_ptr<int32_t>(),
                         udst.data_ptr<int64_t>(),
                         pair_ptr(pairs),
-                        (uint32_t*)drops.data_ptr<int32_t>(), (int)emit_pairs,
-                        cur_stream());
-}
-
-void build_m_index(torch::Tensor mask_t, torch::Tensor pprefix, torch::Tensor ufit,
-                   torch::Tensor udst, int64_t n_users, int64_t uniform_rec,
-                   torch::Tensor cpx, torch::Tensor tcount, torch::Tensor mbase,
-                   torch::Tensor mdst) {
-    CHECK_DEV(mask_t); CHECK_CONTIG(mask_t);
-    int32_t W = (int32_t)mask_t.size(0);
-    int32_t M = (int32_t)mask_t.size(1);
-    TORCH_CHECK(cpx.numel() >= (int64_t)W * M && tcount.numel() >= M
-                && mbase.numel() >= M + 1, "m-index scratch too small");
-    int32_t capacity = (int32_t)mdst.size(0);
-    launch_k2b_m_index((const uint64_t*)mask_t.data_ptr<int64_t>(),
-                       pprefix.data_ptr<int32_t>(), ufit.data_ptr<int32_t>(),
-                       udst.data_ptr<int64_t>(), M, W, (int32_t)n_users, capacity,
-                       (int32_t)uniform_rec, cpx.data_ptr<int32_t>(),
-                       tcount.data_ptr<int32_t>(), mbase.data_ptr<int32_t>(),
-                       mdst.data_ptr<int64_t>(), cur_stream());
-}
-
-void fanout_m(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor mbase,
-              torch::Tensor mdst, int64_t seq_base, torch::Tensor seq_state,
-              int64_t units_per_pair, int64_t wire_len, torch::Tensor egress,
-              int64_t nt, int64_t use_seq_state) {
-    CHECK_DEV(egress); CHECK_CONTIG(egress);
-    int32_t M = (int32_t)payload_off.size(0);
-    TORCH_CHECK(units_per_pair <= 256, "K3m stages at most a 4 KiB record in registers");
-    int32_t capacity = (int32_t)mdst.size(0);
-    launch_k3_fanout_m(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
-                       mbase.data_ptr<int32_t>(), mdst.data_ptr<int64_t>(),
-                       (uint32_t)seq_base,
-                       use_seq_state ? (const uint32_t*)seq_state.data_ptr<int32_t>()
-                                     : nullptr,
-                       M, capacity, (int32_t)units_per_pair, (int32_t)wire_len,
-                       egress.data_ptr<uint8_t>(), (int)nt, cur_stream());
+                        (uint32_t*)drops.data_ptr<int32_t>(), cur_stream());
 }
 
 void emit_direct(torch::Tensor disc, torch::Tensor owner, torch::Tensor payload_off,
@@ -391,8 +349,4 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "K2b fused on transposed mask (wave-aggregated claims, uniform-rec fast path)");
     m.def("assign_emit_blocks_t", &assign_emit_blocks_t,
           "K2b block-parallel (P1 count / P2 bases / P3 emit) for uniform records");
-    m.def("build_m_index", &build_m_index,
-          "K2b message-major recipient index (col prefixes + mbase + mdst)");
-    m.def("fanout_m", &fanout_m,
-          "K3m: register-staged message-major fan-out (read once per wave)");
 }

@@ -978,193 +978,13 @@ extern "C" __global__ void k2b_p3_emit(
     }
 }
 
-// ---------------------------------------------------------------------------
-// Message-major fan-out path (broadcast-only, uniform records): the flat K3
-// re-reads every payload unit from L2 once PER RECIPIENT (2.9 GB/tick at
-// the headline config) which caps it at ~4.5 TB/s against a measured
-// 5.5 TB/s write-only roofline for the same store pattern
-// (scripts/k3_roofline.hip).  Here K2b additionally emits a message-major
-// recipient list (8 B ring-dst per delivery, computed from column prefix
-// sums over the mask), and K3m loads each message ONCE per wave into
-// registers and streams it to a segment of its recipients — the read side
-// drops from O(deliveries) to O(messages x segments).
-// ---------------------------------------------------------------------------
-extern "C" __global__ void k2b_m1_colprefix(
-    const uint64_t* __restrict__ mask_t,   // [W][M]
-    int32_t M, int32_t W,
-    int32_t* __restrict__ cpx,             // [W][M] out: prefix over w
-    int32_t* __restrict__ tcount)          // [M] out: recipients per message
-{
-    int m = blockIdx.x * blockDim.x + threadIdx.x;
-    if (m >= M) return;
-    int run = 0;
-    for (int w = 0; w < W; ++w) {
-        cpx[(int64_t)w * M + m] = run;
-        run += __popcll(mask_t[(int64_t)w * M + m]);
-    }
-    tcount[m] = run;
-}
-
-extern "C" __global__ void k2b_m2_mbase(
-    const int32_t* __restrict__ tcount, int32_t M,
-    int32_t* __restrict__ mbase)           // [M+1] out: exclusive scan
-{
-    if (threadIdx.x != 0 || blockIdx.x != 0) return;
-    int run = 0;
-    for (int m = 0; m < M; ++m) { mbase[m] = run; run += tcount[m]; }
-    mbase[M] = run;
-}
-
-extern "C" __global__ void k2b_p3_emit_m(
-    const uint64_t* __restrict__ mask_t,   // [W][M]
-    const int32_t* __restrict__ pprefix,   // [NB][W*64] per-user p offsets
-    const int32_t* __restrict__ ufit,      // [W*64] ring fit (records)
-    const int64_t* __restrict__ udst,      // [W*64] ring dst base
-    const int32_t* __restrict__ cpx,       // [W][M]
-    const int32_t* __restrict__ mbase,     // [M+1]
-    int32_t M, int32_t W, int32_t NB, int32_t n_users,
-    int32_t capacity, int32_t uniform_rec,
-    int64_t* __restrict__ mdst)            // [capacity] out: dst per delivery,
-                                           //   message-major; -1 = dropped
-{
-    const int wave = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-    if (wave >= W * NB) return;
-    const int w = wave / NB, b = wave - (wave / NB) * NB;
-    const int lane = threadIdx.x & 63;
-    const int u = w * 64 + lane;
-    if (u >= n_users) return;
-    const uint64_t bit = 1ull << lane;
-    const uint64_t lower = bit - 1;
-    const uint64_t* col = mask_t + (int64_t)w * M;
-    int p = pprefix[(int64_t)b * (W * 64) + u];
-    const int fit = ufit[u];
-    const int64_t dstb = udst[u];
-    const int m0 = b * K2B_BLK, m1 = min(M, m0 + K2B_BLK);
-    for (int m = m0; m < m1; ++m) {
-        const uint64_t word = col[m];
-        if (!(word & bit)) continue;
-        const int mslot = mbase[m] + cpx[(int64_t)w * M + m] + __popcll(word & lower);
-        if (mslot < capacity)
-            mdst[mslot] = (p < fit) ? dstb + (int64_t)p * uniform_rec : -1;
-        p++;
-    }
-}
-
-}  // extern "C"  (template below needs C++ linkage)
-
-// K3m: each workgroup stages one message's full record (16 B header +
-// wire bytes, <= 4 KiB) in LDS once, then streams the flat unit stream of
-// its recipient SEGMENT at ~full lane utilization — the flat kernel
-// re-reads every unit from L2 once per recipient (2.9 GB/tick at the
-// headline config), capping it ~20% under the 5.5 TB/s write-only
-// roofline measured for the same store pattern (scripts/k3_roofline.hip).
-// (A register-staged variant was tried first and measured SLOWER: with
-// 70 units/record its second register pass idles 58 of 64 lanes.)
-template <int NT, bool SEQ_FROM_PTR>
-__global__ void __launch_bounds__(256) k3_fanout_m(
-    const uint8_t* __restrict__ buf,
-    const int64_t* __restrict__ payload_off,  // [M] wire offsets
-    const int32_t* __restrict__ mbase,        // [M+1]
-    const int64_t* __restrict__ mdst,         // [capacity]
-    uint32_t seq_base_val,
-    const uint32_t* __restrict__ seq_state,
-    int32_t M, int32_t groups, int32_t capacity,
-    int32_t units_per_pair, int32_t wire_len,
-    uint8_t* __restrict__ egress)
-{
-    __shared__ cdn_v4u smsg[256];  // up to 4 KiB record (header + wire)
-    const int wg = blockIdx.x;     // one workgroup per (message, segment)
-    if (wg >= M * groups) return;
-    const int m = wg / groups, g = wg - (wg / groups) * groups;
-    const int tid = threadIdx.x;
-    const uint32_t seq_base = SEQ_FROM_PTR ? seq_state[0] : seq_base_val;
-
-    const uint8_t* src = buf + payload_off[m];
-    for (int u = tid; u < units_per_pair; u += (int)blockDim.x) {
-        if (u == 0) {
-            uint32_t hdr[4] = {(uint32_t)wire_len, seq_base + (uint32_t)m, 0, 0};
-            memcpy(&smsg[0], hdr, 16);
-        } else {
-            smsg[u] = *(const cdn_v4u*)(src + ((size_t)u - 1) * 16);
-        }
-    }
-    __syncthreads();
-
-    int lo = mbase[m], hi = mbase[m + 1];
-    if (hi > capacity) hi = capacity;
-    const int cnt = hi - lo;
-    if (cnt <= 0) return;
-    const int seg = (cnt + groups - 1) / groups;
-    const int i0 = g * seg;
-    const int i1 = min(cnt, i0 + seg);
-    if (i0 >= i1) return;
-    // flat unit stream over the segment: f -> (recipient i, unit u),
-    // strength-reduced (no division in the loop)
-    const int64_t n_units = (int64_t)(i1 - i0) * units_per_pair;
-    int i = i0 + tid / units_per_pair;
-    int u = tid - (tid / units_per_pair) * units_per_pair;
-    for (int64_t f = tid; f < n_units; f += (int)blockDim.x) {
-        const int64_t dst0 = mdst[lo + i];
-        if (dst0 >= 0) {
-            uint8_t* dst = egress + dst0 + (size_t)u * 16;
-            if (NT) __builtin_nontemporal_store(smsg[u], (cdn_v4u*)dst);
-            else *(cdn_v4u*)dst = smsg[u];
-        }
-        u += (int)blockDim.x;
-        while (u >= units_per_pair) { u -= units_per_pair; ++i; }
-    }
-}
-
-extern "C" {
-
-void launch_k2b_m_index(
-    const uint64_t* mask_t, const int32_t* pprefix, const int32_t* ufit,
-    const int64_t* udst, int32_t M, int32_t W, int32_t n_users, int32_t capacity,
-    int32_t uniform_rec, int32_t* cpx, int32_t* tcount, int32_t* mbase,
-    int64_t* mdst, hipStream_t s) {
-    const int NB = (M + K2B_BLK - 1) / K2B_BLK;
-    const int threads = 256;
-    hipLaunchKernelGGL(k2b_m1_colprefix, dim3((M + threads - 1) / threads), dim3(threads),
-                       0, s, mask_t, M, W, cpx, tcount);
-    hipLaunchKernelGGL(k2b_m2_mbase, dim3(1), dim3(64), 0, s, tcount, M, mbase);
-    hipLaunchKernelGGL(k2b_p3_emit_m, dim3((W * NB * 64 + threads - 1) / threads),
-                       dim3(threads), 0, s, mask_t, pprefix, ufit, udst, cpx, mbase,
-                       M, W, NB, n_users, capacity, uniform_rec, mdst);
-}
-
-extern "C" void launch_k3_fanout_m(
-    const uint8_t* buf, const int64_t* payload_off, const int32_t* mbase,
-    const int64_t* mdst, uint32_t seq_base, const uint32_t* seq_state, int32_t M,
-    int32_t capacity, int32_t units_per_pair, int32_t wire_len, uint8_t* egress,
-    int nt, hipStream_t s) {
-    int groups = 2048 / (M > 0 ? M : 1);  // >=2048 WGs fills 8 XCDs
-    if (groups < 1) groups = 1;
-    if (groups > 256) groups = 256;
-    const int blocks = M * groups;
-    if (seq_state) {
-        if (nt) hipLaunchKernelGGL((k3_fanout_m<1, true>), dim3(blocks), dim3(256), 0, s,
-                    buf, payload_off, mbase, mdst, 0u, seq_state, M, groups, capacity,
-                    units_per_pair, wire_len, egress);
-        else hipLaunchKernelGGL((k3_fanout_m<0, true>), dim3(blocks), dim3(256), 0, s,
-                    buf, payload_off, mbase, mdst, 0u, seq_state, M, groups, capacity,
-                    units_per_pair, wire_len, egress);
-    } else {
-        if (nt) hipLaunchKernelGGL((k3_fanout_m<1, false>), dim3(blocks), dim3(256), 0, s,
-                    buf, payload_off, mbase, mdst, seq_base, nullptr, M, groups, capacity,
-                    units_per_pair, wire_len, egress);
-        else hipLaunchKernelGGL((k3_fanout_m<0, false>), dim3(blocks), dim3(256), 0, s,
-                    buf, payload_off, mbase, mdst, seq_base, nullptr, M, groups, capacity,
-                    units_per_pair, wire_len, egress);
-    }
-}
-
 extern "C" void launch_k2b_blocks_t(
     const uint64_t* mask_t, int32_t M, int32_t W, int32_t n_users,
     int64_t ring_bytes, int32_t capacity, int32_t uniform_rec,
     uint64_t* ring_wpos, int32_t* n_pairs,
     int32_t* bcount, int32_t* pprefix, int32_t* ubase, int32_t* ufit, int64_t* udst,
     PairRec* pairs,
-    uint32_t* drops, int emit_pairs, hipStream_t s) {
+    uint32_t* drops, hipStream_t s) {
     const int NB = (M + K2B_BLK - 1) / K2B_BLK;
     const int waves = W * NB;
     const int threads = 256;
@@ -1175,10 +995,9 @@ extern "C" void launch_k2b_blocks_t(
     hipLaunchKernelGGL(k2b_p2_bases, dim3(blocks_u), dim3(threads), 0, s,
                        bcount, W, NB, n_users, ring_bytes, capacity, uniform_rec,
                        ring_wpos, n_pairs, pprefix, ubase, ufit, udst, drops);
-    if (emit_pairs)
-        hipLaunchKernelGGL(k2b_p3_emit, dim3(blocks_wb), dim3(threads), 0, s,
-                           mask_t, pprefix, ubase, ufit, udst, M, W, NB, n_users,
-                           capacity, uniform_rec, pairs);
+    hipLaunchKernelGGL(k2b_p3_emit, dim3(blocks_wb), dim3(threads), 0, s,
+                       mask_t, pprefix, ubase, ufit, udst, M, W, NB, n_users,
+                       capacity, uniform_rec, pairs);
 }
 
 void launch_k2b_fused_t(const uint64_t* mask_t, const int32_t* payload_len, int32_t M,

@@ -217,24 +217,6 @@ class GpuBrokerEngine:
         self._k2b_scratch = (NB, bufs)
         return bufs
 
-    def _m_index_scratch(self, M: int):
-        """Scratch for the message-major fan-out index: column prefixes
-        [W][M], per-message counts, mbase scan, and the 8 B/delivery dst
-        list (sized to pair_capacity)."""
-        W = (self.n_users + 63) // 64
-        cur = getattr(self, "_m_scratch", None)
-        if cur is not None and cur[0] >= M:
-            return cur[1]
-        o32 = dict(dtype=torch.int32, device=self.device)
-        bufs = (
-            torch.empty(W * M, **o32),                  # cpx
-            torch.empty(M, **o32),                      # tcount
-            torch.empty(M + 1, **o32),                  # mbase
-            torch.empty(self.pair_capacity, dtype=torch.int64, device=self.device),  # mdst
-        )
-        self._m_scratch = (M, bufs)
-        return bufs
-
     def _assign_emit(self, ops, mask_t, payload_len, rec: int, M: int) -> None:
         """Dispatch K2b: block-parallel pipeline for uniform records (fills
         the chip at any population — the one-lane-per-user fused kernel
@@ -244,7 +226,7 @@ class GpuBrokerEngine:
             ops.assign_emit_blocks_t(
                 mask_t, self.ring_wpos, self.ring_bytes, self.n_users,
                 bcount, pprefix, ubase, ufit, udst,
-                self._pairs, self._drops, self._n_pairs, rec, 1,
+                self._pairs, self._drops, self._n_pairs, rec,
             )
         else:
             ops.assign_emit_fused_t(
@@ -269,24 +251,7 @@ class GpuBrokerEngine:
         self._n_pairs.zero_()
         uniform = self.fanout_wire and uniform_wire_len is not None
         rec = ring_rec(uniform_wire_len) if uniform else 0
-        # message-major path: broadcast-only uniform records <= 4 KiB ride
-        # K3m (payload staged in registers once per wave, streamed to a
-        # recipient segment) — the flat kernel re-reads every unit per
-        # recipient, capping it ~20% below the write-only roofline
-        m_path = (uniform and not self.direct_enabled and rec <= 4096
-                  and uniform_wire_len % 16 == 0 and self.use_gpu_ops)
-        if m_path:
-            bcount, pprefix, ubase, ufit, udst = self._k2b_block_scratch(M)
-            ops.assign_emit_blocks_t(
-                mask_t, self.ring_wpos, self.ring_bytes, self.n_users,
-                bcount, pprefix, ubase, ufit, udst,
-                self._pairs, self._drops, self._n_pairs, rec, 0,  # skip P3
-            )
-            cpx, tcount, mbase, mdst = self._m_index_scratch(M)
-            ops.build_m_index(mask_t, pprefix, ufit, udst, self.n_users, rec,
-                              cpx, tcount, mbase, mdst)
-        else:
-            self._assign_emit(ops, mask_t, payload_len, rec, M)
+        self._assign_emit(ops, mask_t, payload_len, rec, M)
         if self.direct_enabled:
             # K5 lookup + K5b on-device delivery-pair emission: direct pairs
             # append to the same pair list, all consumed by the single
@@ -303,11 +268,7 @@ class GpuBrokerEngine:
         # would idle lanes on the tail pass; at >=4 KiB records a wave's 64
         # passes are already ~fully utilized and flat's per-unit index math
         # costs ~3% (measured on the 64 KiB mixed bench) — use wave there.
-        if m_path:
-            cpx, tcount, mbase, mdst = self._m_index_scratch(M)
-            ops.fanout_m(buf, payload_off, mbase, mdst, seq_base, self._seq_dev,
-                         rec // 16, uniform_wire_len, self.egress, nt, 0)
-        elif uniform and rec <= 4096:
+        if uniform and rec <= 4096:
             units = rec // 16
             ops.fanout_flat2(buf, payload_off, payload_len, self._pairs, seq_base,
                              self._n_pairs, units, self.egress, nt, 0)
@@ -329,19 +290,10 @@ class GpuBrokerEngine:
         payload_len = (offsets[1:] - offsets[:-1]).to(torch.int32).contiguous()
         self._n_pairs.zero_()
         rec = units * 16
-        wire_len = (units - 1) * 16
-        bcount, pprefix, ubase, ufit, udst = self._k2b_block_scratch(M)
-        ops.assign_emit_blocks_t(
-            mask_t, self.ring_wpos, self.ring_bytes, self.n_users,
-            bcount, pprefix, ubase, ufit, udst,
-            self._pairs, self._drops, self._n_pairs, rec, 0,
-        )
-        cpx, tcount, mbase, mdst = self._m_index_scratch(M)
-        ops.build_m_index(mask_t, pprefix, ufit, udst, self.n_users, rec,
-                          cpx, tcount, mbase, mdst)
-        ops.fanout_m(buf, payload_off, mbase, mdst, 0, self._seq_dev,
-                     units, wire_len, self.egress,
-                     1 if self.nt_fanout else 0, 1)
+        self._assign_emit(ops, mask_t, payload_len, rec, M)
+        ops.fanout_flat3(buf, payload_off, payload_len, self._pairs, self._seq_dev,
+                         self._n_pairs, units, self.egress,
+                         1 if self.nt_fanout else 0, 0)
         ops.seq_advance(self._seq_dev, M)
 
     def tick_graphed(self, buf: torch.Tensor, offsets: torch.Tensor,

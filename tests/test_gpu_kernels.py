@@ -450,7 +450,7 @@ def _run_k2b(ops, variant, mask_t, n_users, ring_bytes, cap, rec, wpos0):
     if variant == "fused":
         wlen = torch.full((M,), rec - 16, dtype=torch.int32, device="cuda")
         ops.assign_emit_fused_t(mask_t, wlen, wpos, ring_bytes, n_users,
-                                pairs, drops, n_pairs, rec, 1)
+                                pairs, drops, n_pairs, rec)
     else:
         W64 = mask_t.shape[0] * 64
         NB = (M + 31) // 32
@@ -460,7 +460,7 @@ def _run_k2b(ops, variant, mask_t, n_users, ring_bytes, cap, rec, wpos0):
             torch.empty(NB * W64, **o32), torch.empty(NB * W64, **o32),
             torch.empty(W64, **o32), torch.empty(W64, **o32),
             torch.empty(W64, dtype=torch.int64, device="cuda"),
-            pairs, drops, n_pairs, rec, 1)
+            pairs, drops, n_pairs, rec)
     torch.cuda.synchronize()
     return wpos.cpu(), pairs.cpu(), int(drops.cpu()[0]), int(n_pairs.cpu()[0])
 
