@@ -287,3 +287,25 @@ def test_cli_parser_smoke():
     except SystemExit as e:
         p_err = e.code
     assert p_err == 0
+
+
+def test_large_message_flood(tmp_path):
+    """9 MB direct + broadcast to self in a loop — the reference's
+    bad-sender load shape (bad-sender.rs:24-105) — through the memory pool
+    and framing limits."""
+
+    async def go():
+        brokers, marshal, endpoint = await start_stack(tmp_path)
+        client = make_client(endpoint, seed=9, topics=[0])
+        await client.ensure_initialized()
+        payload = bytes(9 * 1024 * 1024)
+        for i in range(3):
+            await client.send_direct_message(client.public_key, payload)
+            msg = await asyncio.wait_for(client.receive_message(), timeout=30)
+            assert isinstance(msg, m.Direct) and len(msg.message) == len(payload)
+            await client.send_broadcast_message([0], payload)
+            msg = await asyncio.wait_for(client.receive_message(), timeout=30)
+            assert isinstance(msg, m.Broadcast) and len(msg.message) == len(payload)
+        await stop_stack(brokers, marshal, client)
+
+    run(go())
