@@ -52,14 +52,27 @@ async def read_length_delimited(reader: asyncio.StreamReader, limiter: Limiter) 
 
 
 async def write_length_delimited(writer: asyncio.StreamWriter, message: Bytes) -> None:
-    data = message.data
+    await write_frames(writer, [message])
+
+
+async def write_frames(writer: asyncio.StreamWriter, messages) -> None:
+    """Write a batch of frames with ONE flush — the writer task coalesces
+    everything queued behind a connection (a burst of fan-out deliveries)
+    into a single syscall-ish drain, like the reference's buffered tokio
+    writer (protocols/mod.rs write half)."""
+    total = 0
     try:
-        writer.write(struct.pack(">I", len(data)))
-        writer.write(data)
+        bufs = []
+        for message in messages:
+            data = message.data
+            bufs.append(struct.pack(">I", len(data)))
+            bufs.append(data)
+            total += len(data)
+        writer.writelines(bufs)
         await asyncio.wait_for(writer.drain(), IO_TIMEOUT_S)
     except (asyncio.TimeoutError, ConnectionResetError, OSError) as e:
         raise ConnectionError_(f"failed to send message: {e}") from e
-    BYTES_SENT.inc(len(data))
+    BYTES_SENT.inc(total)
 
 
 class Connection:
@@ -100,16 +113,29 @@ class Connection:
             try:
                 while True:
                     item = await send_q.get()
-                    if item is None:  # soft close: flush then stop
+                    closing = False
+                    batch = []
+                    while True:  # coalesce everything already queued
+                        if item is None:  # soft close: flush then stop
+                            closing = True
+                            break
+                        batch.append(item)
+                        try:
+                            item = send_q.get_nowait()
+                        except asyncio.QueueEmpty:
+                            break
+                    if batch:
+                        try:
+                            await write_frames(writer, batch)
+                        finally:
+                            for b in batch:
+                                b.drop()
+                    if closing:
                         try:
                             await writer.drain()
                         except Exception:
                             pass
                         return
-                    try:
-                        await write_length_delimited(writer, item)
-                    finally:
-                        item.drop()
             except ConnectionError_:
                 pass
 

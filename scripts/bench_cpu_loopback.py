@@ -75,8 +75,22 @@ async def main(n_msgs: int = 2000, payload: int = 1024) -> None:
         lat.append(time.perf_counter() - s)
     dt = time.perf_counter() - t0
 
+    # burst (pipelined) phase: all messages in flight at once — exercises
+    # the writer task's frame coalescing under queue depth
+    async def blast():
+        for _ in range(n_msgs):
+            await alice.send_direct_message(bob.public_key, payload_bytes)
+
+    tb = time.perf_counter()
+    sender = asyncio.get_running_loop().create_task(blast())
+    for _ in range(n_msgs):
+        await bob.receive_message()
+    await sender
+    burst_dt = time.perf_counter() - tb
+
     print(json.dumps({
         "config": "cpu-loopback: marshal + 1 broker + 2 clients, TCP, direct echo",
+        "burst_msgs_per_sec": n_msgs / burst_dt,
         "msgs_per_sec": n_msgs / dt,
         "p50_latency_ms": statistics.median(lat) * 1000,
         "p99_latency_ms": sorted(lat)[int(len(lat) * 0.99)] * 1000,
