@@ -13,6 +13,7 @@
 #include "bls/bls.h"
 #include "wire/message.h"
 #include "state/versioned_map.h"
+#include "net/pump.h"
 
 namespace py = pybind11;
 using namespace bn254;
@@ -355,6 +356,29 @@ PYBIND11_MODULE(pushcdn_core, m) {
     m.def("wire_serialize_topics", &w_ser_topics);
     m.def("wire_serialize_sync", &w_ser_sync);
     m.def("wire_deserialize", &w_deserialize);
+    py::class_<net::Pump>(m, "Pump")
+        .def(py::init<>())
+        .def("notify_fd", &net::Pump::notify_fd)
+        .def("add", &net::Pump::add)
+        .def("send", [](net::Pump& p, int64_t id, py::bytes b) {
+            char* d; Py_ssize_t l;
+            if (PyBytes_AsStringAndSize(b.ptr(), &d, &l) != 0)
+                throw std::runtime_error("bad bytes");
+            return p.send(id, d, (size_t)l);
+        })
+        .def("send_backlog", &net::Pump::send_backlog)
+        .def("poll_dirty", &net::Pump::poll_dirty)
+        .def("recv_batch", [](net::Pump& p, int64_t id, size_t maxf) {
+            auto r = p.recv_batch(id, maxf);
+            py::list out;
+            for (auto& s : r.first) out.append(py::bytes(s));
+            return py::make_tuple(out, r.second);
+        })
+        .def("soft_close", &net::Pump::soft_close)
+        .def("hard_close", &net::Pump::hard_close)
+        .def("forget", &net::Pump::forget)
+        .def("byte_counters", &net::Pump::byte_counters)
+        .def("stop", &net::Pump::stop, py::call_guard<py::gil_scoped_release>());
     py::class_<PyVersionedMap>(m, "VersionedMap")
         .def(py::init<const std::string&>())
         .def("insert", &PyVersionedMap::insert)

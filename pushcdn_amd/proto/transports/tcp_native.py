@@ -1,0 +1,190 @@
+"""Native TCP transport: frames are read/written by the C++ epoll pump
+(csrc/net/pump.h) on its own thread; asyncio only sees WHOLE messages in
+batches through an eventfd it watches with ``loop.add_reader``.  This is
+the MI355X-native analog of the reference's tokio connection core
+(``cdn-proto/src/connection/protocols/mod.rs:139-217``): per-connection
+reader/writer actors, 4-byte-BE length framing (``:311-394``), max-size
+guard, soft-close flush.
+
+Accept/connect (the rare path) stay on asyncio's non-blocking socket
+helpers; the connected fd is then handed to the pump, which owns it from
+that point on.  Backpressure is byte-based: senders await while the
+connection's unflushed outbox exceeds the high-water mark.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import socket
+from typing import Dict, Optional
+
+from ..errors import ConnectionError_
+from ..limiter import Bytes, Limiter
+from .base import Connection, Listener, Protocol, UnfinalizedConnection
+from .tcp import parse_endpoint
+
+SEND_HWM_BYTES = 64 << 20  # per-connection unflushed-outbox high-water mark
+RECV_BATCH = 1024
+
+
+def _get_core():
+    from ...ops.build import build_core
+
+    return build_core()
+
+
+class _PumpManager:
+    """One C++ pump + one eventfd watcher per event loop."""
+
+    _by_loop: "Dict[int, _PumpManager]" = {}
+
+    def __init__(self, loop: asyncio.AbstractEventLoop) -> None:
+        self.pump = _get_core().Pump()
+        self.loop = loop
+        self.conns: Dict[int, "PumpConnection"] = {}
+        loop.add_reader(self.pump.notify_fd(), self._on_notify)
+
+    @classmethod
+    def current(cls) -> "_PumpManager":
+        loop = asyncio.get_running_loop()
+        key = id(loop)
+        mgr = cls._by_loop.get(key)
+        if mgr is None or mgr.loop.is_closed():
+            mgr = cls(loop)
+            cls._by_loop[key] = mgr
+        return mgr
+
+    def _on_notify(self) -> None:
+        for cid in self.pump.poll_dirty():
+            conn = self.conns.get(cid)
+            if conn is not None:
+                conn._pump_dirty()
+
+    def attach(self, sock: socket.socket) -> int:
+        fd = sock.detach()  # the pump owns the fd now
+        return self.pump.add(fd)
+
+
+class PumpConnection(Connection):
+    """Connection whose data path lives in the C++ pump."""
+
+    def __init__(self, mgr: _PumpManager, cid: int, limiter: Limiter) -> None:
+        self._mgr = mgr
+        self._cid = cid
+        self._limiter = limiter
+        self._closed = False
+        self._dead = False
+        self._recv_q: "asyncio.Queue[Bytes]" = asyncio.Queue()
+        self._wakeup = asyncio.Event()
+        mgr.conns[cid] = self
+
+    def _poll_inbox(self) -> bool:
+        """Drain the C++ inbox into the asyncio queue (no signaling)."""
+        frames, closed = self._mgr.pump.recv_batch(self._cid, RECV_BATCH)
+        for f in frames:
+            self._recv_q.put_nowait(Bytes(f))
+        if closed:
+            self._dead = True
+        return bool(frames) or closed
+
+    # called from the manager's eventfd callback
+    def _pump_dirty(self) -> None:
+        if self._poll_inbox():
+            self._wakeup.set()
+
+    async def send_message_raw(self, raw: Bytes) -> None:
+        try:
+            ok = self._mgr.pump.send(self._cid, raw.data)
+        finally:
+            raw.drop()
+        if not ok:
+            raise ConnectionError_("connection writer closed")
+        while self._mgr.pump.send_backlog(self._cid) > SEND_HWM_BYTES:
+            await asyncio.sleep(0.001)
+
+    async def recv_message_raw(self) -> Bytes:
+        while True:
+            if not self._recv_q.empty():
+                return self._recv_q.get_nowait()
+            if self._dead:
+                raise ConnectionError_("connection reader closed")
+            self._wakeup.clear()
+            # drain anything that raced the notify callback (e.g. frames
+            # that landed before this connection registered)
+            self._poll_inbox()
+            if self._recv_q.empty() and not self._dead:
+                await self._wakeup.wait()
+
+    async def soft_close(self) -> None:
+        if self._closed:
+            return
+        self._closed = True
+        self._mgr.pump.soft_close(self._cid)
+        self._release()
+
+    def close(self) -> None:
+        if not self._closed:
+            self._closed = True
+            self._mgr.pump.hard_close(self._cid)
+        self._release()
+
+    def _release(self) -> None:
+        self._mgr.conns.pop(self._cid, None)
+        self._mgr.pump.forget(self._cid)
+        self._dead = True
+        self._wakeup.set()
+
+
+class TcpNativeUnfinalized(UnfinalizedConnection):
+    def __init__(self, sock: socket.socket) -> None:
+        self._sock = sock
+
+    async def finalize(self, limiter: Limiter) -> Connection:
+        mgr = _PumpManager.current()
+        cid = mgr.attach(self._sock)
+        return PumpConnection(mgr, cid, limiter)
+
+
+class TcpNativeListener(Listener):
+    def __init__(self, sock: socket.socket) -> None:
+        self._sock = sock
+
+    async def accept(self) -> TcpNativeUnfinalized:
+        loop = asyncio.get_running_loop()
+        conn, _addr = await loop.sock_accept(self._sock)
+        return TcpNativeUnfinalized(conn)
+
+    async def close(self) -> None:
+        self._sock.close()
+
+    @property
+    def port(self) -> int:
+        return self._sock.getsockname()[1]
+
+
+class TcpNative(Protocol):
+    @classmethod
+    async def connect(cls, endpoint: str, use_local_authority: bool,
+                      limiter: Limiter) -> Connection:
+        host, port = parse_endpoint(endpoint)
+        loop = asyncio.get_running_loop()
+        sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+        sock.setblocking(False)
+        try:
+            await loop.sock_connect(sock, (host or "127.0.0.1", port))
+        except OSError as e:
+            sock.close()
+            raise ConnectionError_(f"failed to connect to {endpoint}: {e}") from e
+        mgr = _PumpManager.current()
+        cid = mgr.attach(sock)
+        return PumpConnection(mgr, cid, limiter)
+
+    @classmethod
+    async def bind(cls, endpoint: str, certificate=None, key=None) -> TcpNativeListener:
+        host, port = parse_endpoint(endpoint)
+        sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+        sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        sock.setblocking(False)
+        sock.bind((host or "0.0.0.0", port))
+        sock.listen(1024)
+        return TcpNativeListener(sock)

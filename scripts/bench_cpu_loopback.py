@@ -23,7 +23,11 @@ from pushcdn_amd.marshal import Marshal, MarshalConfig
 from pushcdn_amd.proto.transports.tcp import Tcp
 
 
-async def main(n_msgs: int = 2000, payload: int = 1024) -> None:
+async def main(n_msgs: int = 2000, payload: int = 1024, native: bool = False) -> None:
+    if native:
+        from pushcdn_amd.proto.transports.tcp_native import TcpNative as Proto
+    else:
+        Proto = Tcp
     db = tempfile.mktemp(suffix=".db")
     broker = Broker(BrokerConfig(
         public_bind_endpoint="127.0.0.1:0",
@@ -32,8 +36,8 @@ async def main(n_msgs: int = 2000, payload: int = 1024) -> None:
         private_advertise_endpoint="127.0.0.1:0",
         discovery_endpoint=db,
         keypair=bls.KeyPair.from_seed(1000),
-        user_protocol=Tcp,
-        broker_protocol=Tcp,
+        user_protocol=Proto,
+        broker_protocol=Proto,
     ))
     await broker.start()
     # fix up advertise endpoints with the real bound ports
@@ -48,14 +52,14 @@ async def main(n_msgs: int = 2000, payload: int = 1024) -> None:
     broker.connections.identity = broker.identity
     await broker.discovery.perform_heartbeat(0, 600)
 
-    marshal = Marshal(MarshalConfig(bind_endpoint="127.0.0.1:0", discovery_endpoint=db))
+    marshal = Marshal(MarshalConfig(bind_endpoint="127.0.0.1:0", discovery_endpoint=db, protocol=Proto))
     await marshal.start()
     marshal_ep = f"127.0.0.1:{marshal._listener.port}"
 
     alice = Client(ClientConfig(endpoint=marshal_ep, keypair=bls.KeyPair.from_seed(1),
-                                subscribed_topics=[0], protocol=Tcp))
+                                subscribed_topics=[0], protocol=Proto))
     bob = Client(ClientConfig(endpoint=marshal_ep, keypair=bls.KeyPair.from_seed(2),
-                              subscribed_topics=[0], protocol=Tcp))
+                              subscribed_topics=[0], protocol=Proto))
     await alice.ensure_initialized()
     await bob.ensure_initialized()
     await asyncio.sleep(0.2)
@@ -89,7 +93,7 @@ async def main(n_msgs: int = 2000, payload: int = 1024) -> None:
     burst_dt = time.perf_counter() - tb
 
     print(json.dumps({
-        "config": "cpu-loopback: marshal + 1 broker + 2 clients, TCP, direct echo",
+        "config": ("cpu-loopback native-pump" if native else "cpu-loopback") + ": marshal + 1 broker + 2 clients, TCP, direct echo",
         "burst_msgs_per_sec": n_msgs / burst_dt,
         "msgs_per_sec": n_msgs / dt,
         "p50_latency_ms": statistics.median(lat) * 1000,
@@ -104,4 +108,6 @@ async def main(n_msgs: int = 2000, payload: int = 1024) -> None:
 
 
 if __name__ == "__main__":
-    asyncio.run(main())
+    import sys as _sys
+    _native = "--native" in _sys.argv
+    asyncio.run(main(native=_native))
