@@ -18,6 +18,24 @@ import asyncio
 import random
 
 
+def _transport(name: str):
+    """Map a --transport flag to a Protocol class.  tcp-native is the C++
+    epoll pump (csrc/net/pump.h) — the production per-connection IO path."""
+    if name == "tcp":
+        from .proto.transports.tcp import Tcp
+
+        return Tcp
+    if name == "tcp-tls":
+        from .proto.transports.tcp_tls import TcpTls
+
+        return TcpTls
+    if name == "tcp-native":
+        from .proto.transports.tcp_native import TcpNative
+
+        return TcpNative
+    raise SystemExit(f"unknown transport {name!r}")
+
+
 def _broker_args(p: argparse.ArgumentParser) -> None:
     p.add_argument("-d", "--discovery-endpoint", default="/tmp/pushcdn-discovery.db")
     p.add_argument("--public-bind-endpoint", default="0.0.0.0:1738")
@@ -31,6 +49,9 @@ def _broker_args(p: argparse.ArgumentParser) -> None:
     p.add_argument("--global-memory-pool-size", type=int, default=1 << 30)
     p.add_argument("--data-plane", choices=["host", "gpu"], default="host")
     p.add_argument("--gpu-device", default="cuda:0")
+    p.add_argument("--user-transport", choices=["tcp", "tcp-tls", "tcp-native"],
+                   default="tcp", help="user-plane transport (tcp-native = C++ epoll pump)")
+    p.add_argument("--broker-transport", choices=["tcp", "tcp-native"], default="tcp")
 
 
 def cmd_broker(args) -> None:
@@ -50,6 +71,8 @@ def cmd_broker(args) -> None:
         ca_key_path=args.ca_key_path,
         data_plane=args.data_plane,
         gpu_device=args.gpu_device,
+        user_protocol=_transport(args.user_transport),
+        broker_protocol=_transport(args.broker_transport),
     )
     asyncio.run(Broker(cfg).run_forever())
 
@@ -93,6 +116,7 @@ def cmd_marshal(args) -> None:
         global_memory_pool_size=args.global_memory_pool_size,
         ca_cert_path=args.ca_cert_path,
         ca_key_path=args.ca_key_path,
+        protocol=_transport(getattr(args, "transport", "tcp")),
     )
     asyncio.run(Marshal(cfg).run_forever())
 
@@ -110,6 +134,7 @@ def cmd_client(args) -> None:
                 endpoint=args.marshal_endpoint,
                 keypair=bls.KeyPair.from_seed(random.randrange(2**63)),
                 subscribed_topics=[0],
+                protocol=_transport(getattr(args, "transport", "tcp")),
             )
         )
         while True:
@@ -189,6 +214,7 @@ def cmd_bad_sender(args) -> None:
                 endpoint=args.marshal_endpoint,
                 keypair=bls.KeyPair.from_seed(random.randrange(2**63)),
                 subscribed_topics=[0],
+                protocol=_transport(getattr(args, "transport", "tcp")),
             )
         )
         payload = bytes(args.message_size)
@@ -224,10 +250,12 @@ def main(argv=None) -> None:
     ms.add_argument("--ca-cert-path", default=None)
     ms.add_argument("--ca-key-path", default=None)
     ms.add_argument("--global-memory-pool-size", type=int, default=1 << 30)
+    ms.add_argument("--transport", choices=["tcp", "tcp-tls", "tcp-native"], default="tcp")
     ms.set_defaults(fn=cmd_marshal)
 
     c = sub.add_parser("client")
     c.add_argument("-m", "--marshal-endpoint", default="127.0.0.1:1737")
+    c.add_argument("--transport", choices=["tcp", "tcp-tls", "tcp-native"], default="tcp")
     c.set_defaults(fn=cmd_client)
 
     bb = sub.add_parser("bad-broker")
