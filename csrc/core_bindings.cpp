@@ -366,6 +366,11 @@ PYBIND11_MODULE(pushcdn_core, m) {
                 throw std::runtime_error("bad bytes");
             return p.send(id, d, (size_t)l);
         })
+        .def("send_ring", [](net::Pump& p, int64_t id, py::buffer ring, size_t wpos) {
+            py::buffer_info info = ring.request();
+            if (wpos > (size_t)info.size) throw std::runtime_error("wpos beyond ring");
+            return p.send_ring(id, (const uint8_t*)info.ptr, wpos);
+        })
         .def("send_backlog", &net::Pump::send_backlog)
         .def("poll_dirty", &net::Pump::poll_dirty)
         .def("recv_batch", [](net::Pump& p, int64_t id, size_t maxf) {

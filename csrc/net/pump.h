@@ -121,6 +121,36 @@ public:
         return true;
     }
 
+    // enqueue every record of a drained egress ring as one frame each —
+    // ring layout: {u32 len, u32 seq, 8B pad} + payload, 16-aligned
+    // (pushcdn_amd/broker/gpu_engine.py ring_rec / parse_ring_records).
+    // The payload IS a serialized wire message, so it goes out verbatim.
+    // Returns the number of records enqueued (0 if the conn is gone).
+    int64_t send_ring(int64_t id, const uint8_t* ring, size_t wpos) {
+        std::lock_guard<std::mutex> g(mu_);
+        auto it = conns_.find(id);
+        if (it == conns_.end() || it->second.closed || it->second.soft_closing)
+            return 0;
+        auto& c = it->second;
+        int64_t n = 0;
+        size_t pos = 0;
+        while (pos + 16 <= wpos) {
+            uint32_t len;
+            memcpy(&len, ring + pos, 4);
+            if (len > kMaxMessageSize || pos + 16 + len > wpos) break;
+            std::string framed;
+            framed.resize(4 + len);
+            uint32_t be = htonl(len);
+            memcpy(&framed[0], &be, 4);
+            memcpy(&framed[4], ring + pos + 16, len);
+            c.outbox.emplace_back(std::move(framed));
+            ++n;
+            pos += 16 + (((size_t)len + 15) & ~(size_t)15);
+        }
+        if (n) { c.want_write = true; wake(); }
+        return n;
+    }
+
     // bytes queued but not yet written (backpressure signal for Python)
     int64_t send_backlog(int64_t id) {
         std::lock_guard<std::mutex> g(mu_);

@@ -573,8 +573,18 @@ class Broker:
                 if n == 0:
                     continue
                 ring = self._engine.read_ring(slot, n)
-                for _seq, payload in parse_ring_records(ring, n):
-                    await self.try_send_to_user(pubkey, Bytes(payload))
+                handle = self.connections.users.get(pubkey)
+                sink = getattr(handle.connection, "send_ring_records", None) \
+                    if handle is not None else None
+                if sink is not None:
+                    # native pump: C++ parses + enqueues the whole ring
+                    try:
+                        sink(ring, n)
+                    except Exception:
+                        await self.remove_user(pubkey)
+                else:
+                    for _seq, payload in parse_ring_records(ring, n):
+                        await self.try_send_to_user(pubkey, Bytes(payload))
             for raw in batch:
                 raw.drop()
             await asyncio.sleep(self.config.gpu_tick_interval_s)

@@ -115,6 +115,15 @@ class PumpConnection(Connection):
             if self._recv_q.empty() and not self._dead:
                 await self._wakeup.wait()
 
+    def send_ring_records(self, ring: bytes, wpos: int) -> int:
+        """Egress fast path for the GPU broker drain: hand a drained ring
+        (16 B {len,seq} headers + wire payloads, 16-aligned records) to the
+        pump, which parses and enqueues every frame in C++ — one Python
+        call per (user, tick) instead of one per delivery."""
+        if self._closed or self._dead:
+            raise ConnectionError_("connection writer closed")
+        return self._mgr.pump.send_ring(self._cid, ring, wpos)
+
     async def soft_close(self) -> None:
         if self._closed:
             return

@@ -177,3 +177,69 @@ def test_mesh_broker_gpu_single_rank(tmp_path):
         await stop_stack([broker], marshal, alice, bob)
 
     run(go())
+
+
+def test_gpu_broker_native_tcp_full_path(tmp_path):
+    """The full production path on real hardware: native C++ TCP pump on
+    the user plane + CDNA4 kernel routing + C++ send_ring egress drain."""
+    import uuid as _uuid
+    from pushcdn_amd.broker.service import Broker, BrokerConfig
+    from pushcdn_amd.client import Client, ClientConfig
+    from pushcdn_amd.discovery import BrokerIdentifier
+    from pushcdn_amd.marshal import Marshal, MarshalConfig
+    from pushcdn_amd.proto.transports.tcp_native import TcpNative
+
+    async def go():
+        db = str(tmp_path / f"natcuda-{_uuid.uuid4().hex}.db")
+        broker = Broker(BrokerConfig(
+            public_bind_endpoint="127.0.0.1:0",
+            public_advertise_endpoint="127.0.0.1:0",
+            private_bind_endpoint="127.0.0.1:0",
+            private_advertise_endpoint="127.0.0.1:0",
+            discovery_endpoint=db,
+            keypair=bls.KeyPair.from_seed(1000),
+            user_protocol=TcpNative,
+            broker_protocol=TcpNative,
+            data_plane="gpu",
+            gpu_device="cuda:0",
+            gpu_max_users=64,
+            gpu_ring_bytes=1 << 16,
+            gpu_tick_interval_s=0.005,
+        ))
+        await broker.start()
+        pub = f"127.0.0.1:{broker._user_listener.port}"
+        priv = f"127.0.0.1:{broker._broker_listener.port}"
+        broker.config.public_advertise_endpoint = pub
+        broker.config.private_advertise_endpoint = priv
+        broker.identity = BrokerIdentifier(pub, priv)
+        broker.discovery.identity = broker.identity
+        broker.connections.identity = broker.identity
+        await broker.discovery.perform_heartbeat(0, 600)
+        marshal = Marshal(MarshalConfig(bind_endpoint="127.0.0.1:0",
+                                        discovery_endpoint=db, protocol=TcpNative))
+        await marshal.start()
+        ep = f"127.0.0.1:{marshal._listener.port}"
+
+        alice = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(41),
+                                    subscribed_topics=[7], protocol=TcpNative))
+        bob = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(42),
+                                  subscribed_topics=[7], protocol=TcpNative))
+        await alice.ensure_initialized()
+        await bob.ensure_initialized()
+        await asyncio.sleep(0.3)
+
+        for i in range(50):
+            await alice.send_broadcast_message([7], f"cuda-ring-{i}".encode())
+        for i in range(50):
+            msg = await asyncio.wait_for(bob.receive_message(), timeout=15)
+            assert msg.message == f"cuda-ring-{i}".encode()
+        await alice.send_direct_message(bob.public_key, b"cuda-direct")
+        msg = await asyncio.wait_for(bob.receive_message(), timeout=15)
+        assert msg.message == b"cuda-direct"
+
+        alice.close()
+        bob.close()
+        await marshal.close()
+        await broker.close()
+
+    run(go())

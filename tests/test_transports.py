@@ -350,3 +350,70 @@ def test_full_stack_over_native_tcp(tmp_path):
         await broker.close()
 
     run(go())
+
+
+def test_gpu_engine_drain_over_native_tcp(tmp_path):
+    """data_plane=gpu (CPU reference engine here) + native TCP user plane:
+    the egress drain rides pump.send_ring — the pump parses the drained
+    ring records and enqueues each wire frame in C++."""
+    import uuid as _uuid
+
+    from pushcdn_amd.broker.service import Broker, BrokerConfig
+    from pushcdn_amd.client import Client, ClientConfig
+    from pushcdn_amd.crypto import bls
+    from pushcdn_amd.discovery import BrokerIdentifier
+    from pushcdn_amd.marshal import Marshal, MarshalConfig
+    from pushcdn_amd.proto.transports.tcp_native import TcpNative
+
+    async def go():
+        db = str(tmp_path / f"natgpu-{_uuid.uuid4().hex}.db")
+        broker = Broker(BrokerConfig(
+            public_bind_endpoint="127.0.0.1:0",
+            public_advertise_endpoint="127.0.0.1:0",
+            private_bind_endpoint="127.0.0.1:0",
+            private_advertise_endpoint="127.0.0.1:0",
+            discovery_endpoint=db,
+            keypair=bls.KeyPair.from_seed(1000),
+            user_protocol=TcpNative,
+            broker_protocol=TcpNative,
+            data_plane="gpu",
+            gpu_device="cpu",
+            gpu_max_users=32,
+            gpu_ring_bytes=1 << 14,
+            gpu_tick_interval_s=0.01,
+        ))
+        await broker.start()
+        pub = f"127.0.0.1:{broker._user_listener.port}"
+        priv = f"127.0.0.1:{broker._broker_listener.port}"
+        broker.config.public_advertise_endpoint = pub
+        broker.config.private_advertise_endpoint = priv
+        broker.identity = BrokerIdentifier(pub, priv)
+        broker.discovery.identity = broker.identity
+        broker.connections.identity = broker.identity
+        await broker.discovery.perform_heartbeat(0, 600)
+        marshal = Marshal(MarshalConfig(bind_endpoint="127.0.0.1:0",
+                                        discovery_endpoint=db, protocol=TcpNative))
+        await marshal.start()
+        ep = f"127.0.0.1:{marshal._listener.port}"
+
+        alice = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(1),
+                                    subscribed_topics=[5], protocol=TcpNative))
+        bob = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(2),
+                                  subscribed_topics=[5], protocol=TcpNative))
+        await alice.ensure_initialized()
+        await bob.ensure_initialized()
+        await asyncio.sleep(0.2)
+
+        # burst: every delivery drains through send_ring, order preserved
+        for i in range(20):
+            await alice.send_broadcast_message([5], f"ring-{i}".encode())
+        for i in range(20):
+            msg = await asyncio.wait_for(bob.receive_message(), timeout=10)
+            assert msg.message == f"ring-{i}".encode()
+
+        alice.close()
+        bob.close()
+        await marshal.close()
+        await broker.close()
+
+    run(go())
