@@ -198,8 +198,17 @@ class MeshBroker(Broker):
                 if nbytes == 0:
                     continue
                 ring = self._engine.read_ring(slot, nbytes)
-                for _seq, payload in parse_ring_records(ring, nbytes):
-                    await self.try_send_to_user(pubkey, Bytes(payload))
+                handle = self.connections.users.get(pubkey)
+                sink = getattr(handle.connection, "send_ring_records", None) \
+                    if handle is not None else None
+                if sink is not None:
+                    try:
+                        sink(ring, nbytes)  # C++ parses + enqueues all frames
+                    except Exception:
+                        await self.remove_user(pubkey)
+                else:
+                    for _seq, payload in parse_ring_records(ring, nbytes):
+                        await self.try_send_to_user(pubkey, Bytes(payload))
             for raw in batch:
                 raw.drop()
             await asyncio.sleep(self.config.gpu_tick_interval_s)

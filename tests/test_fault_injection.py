@@ -93,3 +93,63 @@ def test_connection_churn_no_leaks(tmp_path):
         await broker.close()
 
     run(go())
+
+
+def test_connection_churn_native_transport(tmp_path):
+    """The C++ pump under connection churn: clients over tcp-native appear
+    and vanish; no fd/slot leaks, survivor stays functional."""
+    import uuid as _uuid
+
+    from pushcdn_amd.broker.service import Broker, BrokerConfig
+    from pushcdn_amd.client import Client, ClientConfig
+    from pushcdn_amd.discovery import BrokerIdentifier
+    from pushcdn_amd.marshal import Marshal, MarshalConfig
+    from pushcdn_amd.proto.transports.tcp_native import TcpNative
+
+    async def go():
+        db = str(tmp_path / f"churn-nat-{_uuid.uuid4().hex}.db")
+        broker = Broker(BrokerConfig(
+            public_bind_endpoint="127.0.0.1:0",
+            public_advertise_endpoint="127.0.0.1:0",
+            private_bind_endpoint="127.0.0.1:0",
+            private_advertise_endpoint="127.0.0.1:0",
+            discovery_endpoint=db,
+            keypair=bls.KeyPair.from_seed(1000),
+            user_protocol=TcpNative,
+            broker_protocol=TcpNative,
+        ))
+        await broker.start()
+        pub = f"127.0.0.1:{broker._user_listener.port}"
+        priv = f"127.0.0.1:{broker._broker_listener.port}"
+        broker.config.public_advertise_endpoint = pub
+        broker.config.private_advertise_endpoint = priv
+        broker.identity = BrokerIdentifier(pub, priv)
+        broker.discovery.identity = broker.identity
+        broker.connections.identity = broker.identity
+        await broker.discovery.perform_heartbeat(0, 600)
+        marshal = Marshal(MarshalConfig(bind_endpoint="127.0.0.1:0",
+                                        discovery_endpoint=db, protocol=TcpNative))
+        await marshal.start()
+        ep = f"127.0.0.1:{marshal._listener.port}"
+
+        stable = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(1),
+                                     subscribed_topics=[0], protocol=TcpNative))
+        await stable.ensure_initialized()
+        for i in range(12):
+            c = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(200 + i),
+                                    subscribed_topics=[0], protocol=TcpNative))
+            await c.ensure_initialized()
+            c.close()
+        for _ in range(50):
+            if len(broker.connections.users) == 1:
+                break
+            await asyncio.sleep(0.1)
+        assert len(broker.connections.users) == 1
+        await stable.send_direct_message(stable.public_key, b"pump-survives")
+        msg = await asyncio.wait_for(stable.receive_message(), timeout=10)
+        assert msg.message == b"pump-survives"
+        stable.close()
+        await marshal.close()
+        await broker.close()
+
+    run(go())
