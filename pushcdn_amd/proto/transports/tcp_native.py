@@ -16,7 +16,7 @@ from __future__ import annotations
 
 import asyncio
 import socket
-from typing import Dict, Optional
+from typing import Dict
 
 from ..errors import ConnectionError_
 from ..limiter import Bytes, Limiter
@@ -50,6 +50,11 @@ class _PumpManager:
         key = id(loop)
         mgr = cls._by_loop.get(key)
         if mgr is None or mgr.loop.is_closed():
+            # reap pumps whose loops are gone (tests create many loops)
+            for k, old in list(cls._by_loop.items()):
+                if old.loop.is_closed():
+                    old.pump.stop()
+                    del cls._by_loop[k]
             mgr = cls(loop)
             cls._by_loop[key] = mgr
         return mgr
