@@ -417,3 +417,92 @@ def test_gpu_engine_drain_over_native_tcp(tmp_path):
         await broker.close()
 
     run(go())
+
+
+def test_two_broker_mesh_over_native_tcp(tmp_path):
+    """Two brokers whose BROKER plane (mutual auth, CRDT syncs, forwarded
+    traffic) also rides the C++ pump; cross-broker broadcast + direct."""
+    import uuid as _uuid
+
+    from pushcdn_amd.broker.service import Broker, BrokerConfig
+    from pushcdn_amd.client import Client, ClientConfig
+    from pushcdn_amd.crypto import bls
+    from pushcdn_amd.discovery import BrokerIdentifier
+    from pushcdn_amd.marshal import Marshal, MarshalConfig
+    from pushcdn_amd.proto.transports.tcp_native import TcpNative
+
+    async def mk_broker(db, kp):
+        b = Broker(BrokerConfig(
+            public_bind_endpoint="127.0.0.1:0",
+            public_advertise_endpoint="127.0.0.1:0",
+            private_bind_endpoint="127.0.0.1:0",
+            private_advertise_endpoint="127.0.0.1:0",
+            discovery_endpoint=db,
+            keypair=kp,
+            user_protocol=TcpNative,
+            broker_protocol=TcpNative,
+            heartbeat_interval_s=0.2,
+            sync_interval_s=0.2,
+        ))
+        await b.start()
+        pub = f"127.0.0.1:{b._user_listener.port}"
+        priv = f"127.0.0.1:{b._broker_listener.port}"
+        b.config.public_advertise_endpoint = pub
+        b.config.private_advertise_endpoint = priv
+        b.identity = BrokerIdentifier(pub, priv)
+        b.discovery.identity = b.identity
+        b.connections.identity = b.identity
+        await b.discovery.perform_heartbeat(0, 600)
+        return b
+
+    async def go():
+        db = str(tmp_path / f"mesh-nat-{_uuid.uuid4().hex}.db")
+        kp = bls.KeyPair.from_seed(1000)
+        b1 = await mk_broker(db, kp)
+        b2 = await mk_broker(db, kp)
+        await asyncio.sleep(0.8)  # heartbeats dial the mesh
+        assert len(b1.connections.brokers) == 1
+        assert len(b2.connections.brokers) == 1
+
+        marshal = Marshal(MarshalConfig(bind_endpoint="127.0.0.1:0",
+                                        discovery_endpoint=db, protocol=TcpNative))
+        await marshal.start()
+        ep = f"127.0.0.1:{marshal._listener.port}"
+
+        # steer alice -> b1, bob -> b2 with artificial load reports
+        await b1.discovery.perform_heartbeat(0, 60)
+        await b2.discovery.perform_heartbeat(10, 60)
+        alice = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(11),
+                                    subscribed_topics=[3], protocol=TcpNative))
+        await alice.ensure_initialized()
+        # steering races the brokers' own 0.2 s heartbeats — retry until
+        # bob lands on b2
+        bob = None
+        for attempt in range(20):
+            await b1.discovery.perform_heartbeat(10, 60)
+            await b2.discovery.perform_heartbeat(0, 60)
+            bob = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(12),
+                                      subscribed_topics=[3], protocol=TcpNative))
+            await bob.ensure_initialized()
+            await asyncio.sleep(0.1)
+            if len(b2.connections.users) == 1:
+                break
+            bob.close()
+            await asyncio.sleep(0.2)
+        assert len(b1.connections.users) == 1 and len(b2.connections.users) == 1
+        await asyncio.sleep(0.6)  # topic/user syncs over the pump
+
+        await alice.send_broadcast_message([3], b"pump-mesh-broadcast")
+        msg = await asyncio.wait_for(bob.receive_message(), timeout=10)
+        assert msg.message == b"pump-mesh-broadcast"
+        await alice.send_direct_message(bob.public_key, b"pump-mesh-direct")
+        msg = await asyncio.wait_for(bob.receive_message(), timeout=10)
+        assert msg.message == b"pump-mesh-direct"
+
+        alice.close()
+        bob.close()
+        await marshal.close()
+        await b1.close()
+        await b2.close()
+
+    run(go())
