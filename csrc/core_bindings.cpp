@@ -369,7 +369,8 @@ PYBIND11_MODULE(pushcdn_core, m) {
         .def("send_ring", [](net::Pump& p, int64_t id, py::buffer ring, size_t wpos) {
             py::buffer_info info = ring.request();
             if (wpos > (size_t)info.size) throw std::runtime_error("wpos beyond ring");
-            return p.send_ring(id, (const uint8_t*)info.ptr, wpos);
+            auto r = p.send_ring(id, (const uint8_t*)info.ptr, wpos);
+            return py::make_tuple(r.first, r.second);
         })
         .def("send_backlog", &net::Pump::send_backlog)
         .def("poll_dirty", &net::Pump::poll_dirty)

@@ -125,14 +125,14 @@ public:
     // ring layout: {u32 len, u32 seq, 8B pad} + payload, 16-aligned
     // (pushcdn_amd/broker/gpu_engine.py ring_rec / parse_ring_records).
     // The payload IS a serialized wire message, so it goes out verbatim.
-    // Returns the number of records enqueued (0 if the conn is gone).
-    int64_t send_ring(int64_t id, const uint8_t* ring, size_t wpos) {
+    // Returns (records enqueued, payload bytes) — (0,0) if the conn is gone.
+    std::pair<int64_t, int64_t> send_ring(int64_t id, const uint8_t* ring, size_t wpos) {
         std::lock_guard<std::mutex> g(mu_);
         auto it = conns_.find(id);
         if (it == conns_.end() || it->second.closed || it->second.soft_closing)
-            return 0;
+            return {0, 0};
         auto& c = it->second;
-        int64_t n = 0;
+        int64_t n = 0, payload_bytes = 0;
         size_t pos = 0;
         while (pos + 16 <= wpos) {
             uint32_t len;
@@ -145,10 +145,11 @@ public:
             memcpy(&framed[4], ring + pos + 16, len);
             c.outbox.emplace_back(std::move(framed));
             ++n;
+            payload_bytes += (int64_t)len;
             pos += 16 + (((size_t)len + 15) & ~(size_t)15);
         }
         if (n) { c.want_write = true; wake(); }
-        return n;
+        return {n, payload_bytes};
     }
 
     // bytes queued but not yet written (backpressure signal for Python)

@@ -20,6 +20,7 @@ from typing import Dict
 
 from ..errors import ConnectionError_
 from ..limiter import Bytes, Limiter
+from ...utils.metrics import BYTES_RECV, BYTES_SENT
 from .base import Connection, Listener, Protocol, UnfinalizedConnection
 from .tcp import parse_endpoint
 
@@ -86,8 +87,12 @@ class PumpConnection(Connection):
     def _poll_inbox(self) -> bool:
         """Drain the C++ inbox into the asyncio queue (no signaling)."""
         frames, closed = self._mgr.pump.recv_batch(self._cid, RECV_BATCH)
+        total = 0
         for f in frames:
+            total += len(f)
             self._recv_q.put_nowait(Bytes(f))
+        if total:
+            BYTES_RECV.inc(total)
         if closed:
             self._dead = True
         return bool(frames) or closed
@@ -98,12 +103,14 @@ class PumpConnection(Connection):
             self._wakeup.set()
 
     async def send_message_raw(self, raw: Bytes) -> None:
+        size = len(raw.data)
         try:
             ok = self._mgr.pump.send(self._cid, raw.data)
         finally:
             raw.drop()
         if not ok:
             raise ConnectionError_("connection writer closed")
+        BYTES_SENT.inc(size)
         while self._mgr.pump.send_backlog(self._cid) > SEND_HWM_BYTES:
             await asyncio.sleep(0.001)
 
@@ -127,7 +134,10 @@ class PumpConnection(Connection):
         call per (user, tick) instead of one per delivery."""
         if self._closed or self._dead:
             raise ConnectionError_("connection writer closed")
-        return self._mgr.pump.send_ring(self._cid, ring, wpos)
+        n, payload_bytes = self._mgr.pump.send_ring(self._cid, ring, wpos)
+        if payload_bytes:
+            BYTES_SENT.inc(payload_bytes)
+        return n
 
     async def soft_close(self) -> None:
         if self._closed:
