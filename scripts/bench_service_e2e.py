@@ -85,6 +85,39 @@ async def main(args) -> None:
         await sender.send_broadcast_message([0], payload)
     await asyncio.gather(*[drain(c, 20) for c in subs])
 
+    if args.seconds:
+        # soak mode: storm rounds until the clock runs out; every message
+        # must reach every subscriber each round (loss check built in)
+        import resource
+
+        t0 = time.perf_counter()
+        rounds = 0
+        sent = 0
+        rss0 = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
+        while time.perf_counter() - t0 < args.seconds:
+            for _ in range(args.msgs):
+                await sender.send_broadcast_message([0], payload)
+            await asyncio.gather(*[drain(c, args.msgs) for c in subs])
+            sent += args.msgs
+            rounds += 1
+        dt = time.perf_counter() - t0
+        rss1 = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
+        print(json.dumps({
+            "config": ("native-tcp" if args.native else "asyncio-tcp")
+                      + ("+gpu-engine" if args.gpu else "+host-plane")
+                      + f" SOAK {args.seconds}s: {args.clients} subs",
+            "rounds": rounds,
+            "deliveries_per_sec": round(sent * args.clients / dt, 1),
+            "rss_growth_kb": rss1 - rss0,
+            "elapsed_s": round(dt, 1),
+        }))
+        sender.close()
+        for c in subs:
+            c.close()
+        await marshal.close()
+        await broker.close()
+        return
+
     t0 = time.perf_counter()
 
     async def blast():
@@ -121,5 +154,6 @@ if __name__ == "__main__":
     p.add_argument("--clients", type=int, default=20)
     p.add_argument("--msgs", type=int, default=500)
     p.add_argument("--payload", type=int, default=1024)
+    p.add_argument("--seconds", type=int, default=0, help="soak mode: run for N seconds")
     args = p.parse_args()
     asyncio.run(main(args))
