@@ -129,6 +129,28 @@ class Client:
             self._disconnect_on_error()
             raise ConnectionError_(str(e)) from e
 
+    async def receive_messages(self, max_n: int = 1024) -> "List[m.Message]":
+        """Batched receive (extension over the reference API): await the
+        first message, then drain everything already queued on the
+        connection, up to max_n.  High-rate subscribers pay one await per
+        BATCH instead of one per message."""
+        conn = await self._get_connection()
+        try:
+            out = [await conn.recv_message()]
+            q = getattr(conn, "_recv_q", None)
+            while q is not None and not q.empty() and len(out) < max_n:
+                raw = q.get_nowait()
+                try:
+                    out.append(m.deserialize(raw.data))
+                finally:
+                    raw.drop()
+            return out
+        except asyncio.CancelledError:
+            raise
+        except Exception as e:
+            self._disconnect_on_error()
+            raise ConnectionError_(str(e)) from e
+
     async def subscribe(self, topics: Sequence[int]) -> None:
         """Update the replay set first, then best-effort send
         (reference lib.rs:383-414)."""

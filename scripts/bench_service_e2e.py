@@ -77,8 +77,9 @@ async def main(args) -> None:
     payload = bytes(args.payload)
 
     async def drain(c, n):
-        for _ in range(n):
-            await c.receive_message()
+        got = 0
+        while got < n:
+            got += len(await c.receive_messages(n - got))
 
     # warmup
     for _ in range(20):

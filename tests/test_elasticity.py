@@ -147,3 +147,29 @@ def test_concurrent_sends_single_reconnect(tmp_path):
         await stop_stack([broker], marshal, client)
 
     run(go())
+
+
+def test_receive_messages_batched(tmp_path):
+    """receive_messages drains queued frames in order, one await per batch."""
+    import asyncio
+
+    from tests.test_integration import make_client, make_marshal, new_db, run, start_stack
+
+    async def go():
+        brokers, marshal, endpoint = await start_stack(tmp_path)
+        a = make_client(endpoint, seed=3, topics=[0])
+        b = make_client(endpoint, seed=4, topics=[0])
+        await a.ensure_initialized()
+        await b.ensure_initialized()
+        for i in range(30):
+            await a.send_broadcast_message([0], f"b-{i}".encode())
+        got = []
+        while len(got) < 30:
+            got.extend(msg.message for msg in await b.receive_messages())
+        assert got == [f"b-{i}".encode() for i in range(30)]
+        a.close(); b.close()
+        await marshal.close()
+        for br in brokers:
+            await br.close()
+
+    run(go())
