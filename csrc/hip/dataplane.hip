@@ -483,6 +483,7 @@ extern "C" {
 
 void launch_k4_parse(const uint8_t* buf, const int64_t* offsets, int32_t M, ParseOut out,
                      hipStream_t s) {
+    if (M <= 0) return;
     int threads = 256, blocks = (M + threads - 1) / threads;
     hipLaunchKernelGGL(k4_parse_batch, dim3(blocks), dim3(threads), 0, s, buf, offsets, M, out);
 }
@@ -491,6 +492,7 @@ void launch_k2a_topic_mask(const uint64_t* sub_bitmap, const uint8_t* buf,
                            const int64_t* topics_off, const int32_t* topics_cnt,
                            const int32_t* disc, uint64_t* mask, int32_t M, int32_t W,
                            hipStream_t s) {
+    if (M <= 0) return;
     int64_t total = (int64_t)M * W;
     int threads = 256;
     int64_t blocks = (total + threads - 1) / threads;
@@ -854,6 +856,7 @@ void launch_k2a_topic_mask_t(const uint64_t* sub_bitmap, const uint8_t* buf,
                              const int64_t* topics_off, const int32_t* topics_cnt,
                              const int32_t* disc, uint64_t* mask_t, int32_t M, int32_t W,
                              hipStream_t s) {
+    if (M <= 0) return;
     int64_t total = (int64_t)M * W;
     int threads = 256;
     int64_t blocks = (total + threads - 1) / threads;
@@ -985,6 +988,7 @@ extern "C" void launch_k2b_blocks_t(
     int32_t* bcount, int32_t* pprefix, int32_t* ubase, int32_t* ufit, int64_t* udst,
     PairRec* pairs,
     uint32_t* drops, hipStream_t s) {
+    if (M <= 0) return;
     const int NB = (M + K2B_BLK - 1) / K2B_BLK;
     const int waves = W * NB;
     const int threads = 256;
