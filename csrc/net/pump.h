@@ -28,6 +28,7 @@
 #include <mutex>
 #include <set>
 #include <string>
+#include <algorithm>
 #include <thread>
 #include <vector>
 
@@ -134,20 +135,35 @@ public:
         auto& c = it->second;
         int64_t n = 0, payload_bytes = 0;
         size_t pos = 0;
+        // collect (seq, frame): ring write order is claim order and K5b's
+        // atomic direct-delivery claims may interleave within a tick — the
+        // seq header restores per-tick arrival order (wrap-aware), same as
+        // the Python drain (gpu_engine.parse_ring_records)
+        std::vector<std::pair<uint32_t, std::string>> recs;
         while (pos + 16 <= wpos) {
-            uint32_t len;
+            uint32_t len, seq;
             memcpy(&len, ring + pos, 4);
+            memcpy(&seq, ring + pos + 4, 4);
             if (len > kMaxMessageSize || pos + 16 + len > wpos) break;
             std::string framed;
             framed.resize(4 + len);
             uint32_t be = htonl(len);
             memcpy(&framed[0], &be, 4);
             memcpy(&framed[4], ring + pos + 16, len);
-            c.outbox.emplace_back(std::move(framed));
+            recs.emplace_back(seq, std::move(framed));
             ++n;
             payload_bytes += (int64_t)len;
             pos += 16 + (((size_t)len + 15) & ~(size_t)15);
         }
+        if (recs.size() > 1) {
+            uint32_t base = recs[0].first;
+            for (auto& r : recs) if (r.first - base > 0x80000000u) base = r.first;
+            std::stable_sort(recs.begin(), recs.end(),
+                             [base](const auto& x, const auto& y) {
+                                 return (uint32_t)(x.first - base) < (uint32_t)(y.first - base);
+                             });
+        }
+        for (auto& r : recs) c.outbox.emplace_back(std::move(r.second));
         if (n) { c.want_write = true; wake(); }
         return {n, payload_bytes};
     }

@@ -379,7 +379,15 @@ class GpuBrokerEngine:
 
 
 def parse_ring_records(ring: bytes, wpos: int) -> List[Tuple[int, bytes]]:
-    """Parse delivery records out of a drained ring: [(seq, payload), ...]."""
+    """Parse delivery records out of a drained ring, ordered by sequence
+    number: [(seq, payload), ...].
+
+    Ring WRITE order is claim order: broadcasts are per-user ordered by
+    K2b, but K5b's direct-delivery claims are atomic and may interleave
+    out of order within a tick — the seq header restores the global
+    per-tick arrival order (stronger than the reference's per-connection
+    FIFO, sender.rs).  The sort is wrap-aware relative to the batch's
+    lowest seq."""
     out = []
     pos = 0
     while pos + 16 <= wpos:
@@ -388,4 +396,11 @@ def parse_ring_records(ring: bytes, wpos: int) -> List[Tuple[int, bytes]]:
         payload = ring[pos + 16 : pos + 16 + length]
         out.append((seq, payload))
         pos += ring_rec(length)
+    if len(out) > 1:
+        # circular minimum (handles u32 seq wrap mid-batch)
+        base = out[0][0]
+        for s_, _ in out:
+            if (s_ - base) & 0xFFFFFFFF > 0x80000000:
+                base = s_
+        out.sort(key=lambda r: (r[0] - base) & 0xFFFFFFFF)
     return out

@@ -243,3 +243,50 @@ def test_gpu_broker_native_tcp_full_path(tmp_path):
         await broker.close()
 
     run(go())
+
+
+def test_direct_burst_order_preserved(tmp_path):
+    """A burst of DIRECT messages from one sender in one tick must arrive
+    in send order: K5b's atomic ring claims interleave, and the drain's
+    seq sort restores arrival order (reference per-connection FIFO)."""
+    async def go():
+        db = new_db(tmp_path)
+        cfg = BrokerConfig(
+            public_bind_endpoint="dord-pub",
+            public_advertise_endpoint="dord-pub",
+            private_bind_endpoint="dord-priv",
+            private_advertise_endpoint="dord-priv",
+            discovery_endpoint=db,
+            keypair=bls.KeyPair.from_seed(1000),
+            user_protocol=Memory,
+            broker_protocol=Memory,
+            data_plane="gpu",
+            gpu_device="cuda:0",
+            gpu_max_users=16,
+            gpu_ring_bytes=1 << 18,
+            gpu_tick_interval_s=0.05,  # long tick so the burst lands in ONE batch
+        )
+        broker = Broker(cfg)
+        await broker.start()
+        await broker.discovery.perform_heartbeat(0, 60)
+        marshal, endpoint = make_marshal(db)
+        await marshal.start()
+        alice = make_client(endpoint, seed=61, topics=[])
+        bob = make_client(endpoint, seed=62, topics=[])
+        await alice.ensure_initialized()
+        await bob.ensure_initialized()
+        await asyncio.sleep(0.3)
+
+        for round_ in range(3):
+            for i in range(64):
+                await alice.send_direct_message(bob.public_key,
+                                                f"d-{round_}-{i}".encode())
+            got = [
+                (await asyncio.wait_for(bob.receive_message(), timeout=15)).message
+                for _ in range(64)
+            ]
+            assert got == [f"d-{round_}-{i}".encode() for i in range(64)], got[:8]
+
+        await stop_stack([broker], marshal, alice, bob)
+
+    run(go())
