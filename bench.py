@@ -189,8 +189,11 @@ def main() -> None:
     )
     host_batches = [builder(seed=rank * 1000 + v) for v in range(n_variants)]
     wire_len = host_batches[0][2]
-    cap = max(len(b) for b, _, _ in host_batches)
-    cap = (cap + 255) & ~255
+    # capacity EXACTLY the (uniform) batch size: the all-gathered buffer is
+    # then gap-free, so one combined offsets table covers every rank's
+    # messages and the whole step routes in ONE kernel-pipeline pass
+    cap = len(host_batches[0][0])
+    assert all(len(b) == cap for b, _, _ in host_batches)
     offsets_t = torch.tensor(host_batches[0][1], dtype=torch.int64)
     pinned = []
     for b, off, _ in host_batches:
@@ -204,6 +207,15 @@ def main() -> None:
     from pushcdn_amd.parallel.mesh import RcclMesh
 
     mesh = RcclMesh(torch.device(device), batch_capacity=cap)
+    # combined offsets across the gathered [world_size * cap] buffer:
+    # every rank's batch is identical in shape, so the combined table is
+    # just the local table tiled with a +cap stride
+    if world_size > 1:
+        combined = torch.cat(
+            [offsets_t[:-1] + r * cap for r in range(world_size)]
+            + [torch.tensor([world_size * cap], dtype=torch.int64)]
+        )
+        dev_comb_offsets = combined.to(device)
     use_graph = (not use_cpu) and args.graph and args.mode == "broadcast"
 
     # Double-buffered ingest: the NEXT step's H2D copy runs on a side stream
@@ -233,20 +245,29 @@ def main() -> None:
                     dev_bufs[nxt].copy_(pinned[(i + 1) % n_variants], non_blocking=True)
                     copy_done[nxt].record(copy_stream)
             buf = dev_bufs[cur]
-        # broker->broker mesh: all-gather this tick's batches over xGMI
-        for r, view, n_msgs, nbytes in mesh.exchange(buf, args.batch, len(host_batches[v][0])):
-            if use_graph:
-                # hipGraph-captured tick (captured per fixed buffer; the
-                # gathered views and ingest buffers are stable addresses)
-                eng.tick_graphed(view, dev_offsets, wire_len)
+        # broker->broker mesh: all-gather this tick's batches over xGMI,
+        # then route ALL ranks' messages in one pipeline pass
+        if world_size > 1:
+            gathered = mesh.exchange_flat(buf)
+            if use_cpu:
+                hb = bytes(gathered.numpy().tobytes())
+                eng.tick(gathered, dev_comb_offsets, host_batch=hb,
+                         host_offsets=[int(x) for x in combined],
+                         uniform_wire_len=wire_len)
+            elif use_graph:
+                eng.tick_graphed(gathered, dev_comb_offsets, wire_len)
             else:
-                eng.tick(
-                    view,
-                    dev_offsets,
-                    host_batch=None if not use_cpu else host_batches[v][0],
-                    host_offsets=None if not use_cpu else host_batches[v][1],
-                    uniform_wire_len=wire_len,
-                )
+                eng.tick(gathered, dev_comb_offsets, uniform_wire_len=wire_len)
+        elif use_graph:
+            eng.tick_graphed(buf, dev_offsets, wire_len)
+        else:
+            eng.tick(
+                buf,
+                dev_offsets,
+                host_batch=None if not use_cpu else host_batches[v][0],
+                host_offsets=None if not use_cpu else host_batches[v][1],
+                uniform_wire_len=wire_len,
+            )
         if not use_cpu:
             tick_done[i % 2].record(torch.cuda.current_stream())
         eng.drain_cursors()

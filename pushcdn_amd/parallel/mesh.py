@@ -44,6 +44,18 @@ class RcclMesh:
             # per-rank metadata: [n_messages, batch_bytes]
             self._meta = torch.zeros(self.world_size * 2, dtype=torch.int64, device=device)
 
+    def exchange_flat(self, batch: torch.Tensor) -> torch.Tensor:
+        """All-gather this tick's batch and return the CONTIGUOUS
+        [world_size * capacity] gathered tensor.  When every rank's batch
+        has identical uniform layout and fills `capacity` exactly, the
+        caller can route all ranks' messages in ONE kernel-pipeline pass
+        over this buffer instead of one pass per rank."""
+        if not self.enabled:
+            return batch
+        assert batch.numel() == self.capacity
+        self.dist.all_gather_into_tensor(self._gathered, batch)
+        return self._gathered
+
     def exchange(
         self, batch: torch.Tensor, n_messages: int, batch_bytes: int
     ) -> List[Tuple[int, torch.Tensor, int, int]]:
