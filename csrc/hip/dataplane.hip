@@ -10,13 +10,18 @@
 //                        rows (reference connections/mod.rs:94-124);
 //                        the _t variant writes the mask TRANSPOSED [W][M]
 //                        so K2b scans contiguous memory
-//   K2b k2b_fused_t    — the production emit: per-user ordered scan with
-//                        wave-aggregated slot claims (one atomic per wave64)
-//                        + closed-form ring math for uniform records; the
-//                        per-connection FIFO the reference gets from its
-//                        per-conn channel actors (protocols/mod.rs:139-217).
-//                        (k2b_count/k2b_emit are the straightforward
-//                        two-pass reference pair, kept for golden tests.)
+//   K2b k2b_p1_count / k2b_p2_bases / k2b_p3_emit — the production emit
+//                        for uniform records: per-(user, 32-msg block)
+//                        counts, then per-user closed-form ring math with
+//                        wave-aggregated span claims (one atomic per
+//                        wave64), then block-parallel emission — fills the
+//                        chip at any population (one-lane-per-user runs
+//                        only W waves).  k2b_fused_t is the one-kernel
+//                        variant (non-uniform records + golden tests);
+//                        k2b_count/k2b_emit the two-pass reference pair.
+//                        All preserve the per-connection FIFO the
+//                        reference gets from its per-conn channel actors
+//                        (protocols/mod.rs:139-217).
 //   K3  k3_fanout_flat_t — the production fan-out for uniform records
 //                        (unit-per-lane flat index, ~100% lane utilization,
 //                        non-temporal 16 B stores; seq from value or device

@@ -3,7 +3,7 @@
 from __future__ import annotations
 
 import time
-from typing import List, Sequence, Tuple
+from typing import Sequence, Tuple
 
 from ..crypto import bls
 from ..proto import message as m

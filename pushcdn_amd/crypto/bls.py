@@ -12,7 +12,6 @@ source-for-source with the K1 batched GPU verification kernel.
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Tuple
 
 # Namespaces (reference signature.rs:19-32)
 USER_MARSHAL_NAMESPACE = "espresso-cdn-user-marshal-auth"

@@ -7,7 +7,6 @@ from __future__ import annotations
 
 import asyncio
 import socket
-from typing import Optional
 
 from ..errors import ConnectionError_, ParseError
 from ..limiter import Limiter
