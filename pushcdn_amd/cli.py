@@ -144,7 +144,7 @@ def cmd_client(args) -> None:
             await client.send_broadcast_message([0], b"hello broadcast")
             msg = await client.receive_message()
             assert isinstance(msg, m.Broadcast) and msg.message == b"hello broadcast"
-            print("echo ok")
+            print("echo ok", flush=True)
             await asyncio.sleep(5)
 
     asyncio.run(go())
@@ -226,7 +226,7 @@ def cmd_bad_sender(args) -> None:
             await client.receive_message()
             n += 2
             if n % 10 == 0:
-                print(f"{n} messages echoed")
+                print(f"{n} messages echoed", flush=True)
 
     asyncio.run(go())
 
