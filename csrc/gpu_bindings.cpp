@@ -39,11 +39,11 @@ void launch_k1_hash_to_g1(uint8_t*, const int64_t*, int32_t, uint8_t*, hipStream
 void launch_k3_fanout_wave(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
                            const uint32_t*, const int32_t*, uint8_t*, int, int, hipStream_t);
 void launch_k3_fanout_flat2(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
-                            uint32_t, const int32_t*, int32_t, int32_t, uint8_t*, int, int,
-                            hipStream_t);
+                            uint32_t, const int32_t*, int32_t, int32_t, int32_t, uint8_t*,
+                            int, int, hipStream_t);
 void launch_k3_fanout_flat3(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
-                            const uint32_t*, const int32_t*, int32_t, int32_t, uint8_t*, int,
-                            int, hipStream_t);
+                            const uint32_t*, const int32_t*, int32_t, int32_t, int32_t,
+                            uint8_t*, int, int, hipStream_t);
 void launch_k_seq_advance(uint32_t*, int32_t, hipStream_t);
 void launch_k5b_emit_direct(const int32_t*, const int32_t*, const int64_t*, const int32_t*,
                             int32_t, int64_t, int32_t, uint64_t*, int32_t*, PairRec*,
@@ -229,27 +229,28 @@ void fanout_wave(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor pay
 void fanout_flat2(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_len,
                   torch::Tensor pairs,
                   int64_t seq_base, torch::Tensor n_pairs, int64_t units_per_pair,
-                  torch::Tensor egress, int64_t nt, int64_t grid) {
+                  torch::Tensor egress, int64_t nt, int64_t grid, int64_t uniform_len) {
     CHECK_DEV(egress); CHECK_CONTIG(egress);
     int32_t capacity = (int32_t)pairs.size(0);
     launch_k3_fanout_flat2(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
                            payload_len.data_ptr<int32_t>(), pair_ptr(pairs),
                            (uint32_t)seq_base, n_pairs.data_ptr<int32_t>(), capacity,
-                           (int32_t)units_per_pair, egress.data_ptr<uint8_t>(), (int)nt,
-                           (int)grid, cur_stream());
+                           (int32_t)units_per_pair, (int32_t)uniform_len,
+                           egress.data_ptr<uint8_t>(), (int)nt, (int)grid, cur_stream());
 }
 
 void fanout_flat3(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor payload_len,
                   torch::Tensor pairs,
                   torch::Tensor seq_state, torch::Tensor n_pairs, int64_t units_per_pair,
-                  torch::Tensor egress, int64_t nt, int64_t grid) {
+                  torch::Tensor egress, int64_t nt, int64_t grid, int64_t uniform_len) {
     CHECK_DEV(egress); CHECK_CONTIG(egress);
     int32_t capacity = (int32_t)pairs.size(0);
     launch_k3_fanout_flat3(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
                            payload_len.data_ptr<int32_t>(), pair_ptr(pairs),
                            (const uint32_t*)seq_state.data_ptr<int32_t>(),
                            n_pairs.data_ptr<int32_t>(), capacity, (int32_t)units_per_pair,
-                           egress.data_ptr<uint8_t>(), (int)nt, (int)grid, cur_stream());
+                           (int32_t)uniform_len, egress.data_ptr<uint8_t>(), (int)nt,
+                           (int)grid, cur_stream());
 }
 
 void seq_advance(torch::Tensor seq_state, int64_t m) {

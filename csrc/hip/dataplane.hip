@@ -652,6 +652,8 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
     const int32_t* __restrict__ n_pairs_ptr,
     int32_t capacity,
     int32_t units_per_pair,
+    int32_t uniform_len,                  // >0: every record is this long —
+                                          // skips the per-unit length load
     uint8_t* __restrict__ egress)
 {
     typedef unsigned int v4u __attribute__((ext_vector_type(4)));
@@ -666,16 +668,16 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
         const PairRec pr = pairs[p];   // one dword4 load
         if (pr.user < 0) continue;
         const int mi = pr.msg;
+        const int32_t len = uniform_len > 0 ? uniform_len : payload_len[mi];
         uint8_t* dst = egress + pr.dst + (size_t)unit * 16;
         if (unit == 0) {
-            uint32_t hdr[4] = {(uint32_t)payload_len[mi], seq_base + (uint32_t)mi, 0, 0};
+            uint32_t hdr[4] = {(uint32_t)len, seq_base + (uint32_t)mi, 0, 0};
             v4u h; memcpy(&h, hdr, 16);
             if (NT) __builtin_nontemporal_store(h, (v4u*)dst);
             else memcpy(dst, hdr, 16);
             continue;
         }
         const uint8_t* src = buf + payload_off[mi] + (size_t)(unit - 1) * 16;
-        const int32_t len = payload_len[mi];
         const int32_t coff = (unit - 1) * 16;
         if (coff + 16 <= len && (((uintptr_t)src) & 15) == 0) {
             v4u v = *(const v4u*)src;
@@ -702,33 +704,34 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
 extern "C" void launch_k3_fanout_flat2(
     const uint8_t* buf, const int64_t* payload_off, const int32_t* payload_len,
     const PairRec* pairs,
-    uint32_t seq_base, const int32_t* n_pairs_ptr, int32_t capacity, int32_t units_per_pair,
+    uint32_t seq_base, const int32_t* n_pairs_ptr, int32_t capacity, int32_t units_per_pair, int32_t uniform_len,
     uint8_t* egress, int nt, int grid, hipStream_t s) {
     if (grid <= 0) grid = 16384;  // swept: 16384 > 8192 > 4096 (~0.5% each)
     if (nt)
         hipLaunchKernelGGL((k3_fanout_flat_t<1, false>), dim3(grid), dim3(256), 0, s, buf,
                            payload_off, payload_len, pairs, seq_base,
-                           nullptr, n_pairs_ptr, capacity, units_per_pair, egress);
+                           nullptr, n_pairs_ptr, capacity, units_per_pair, uniform_len, egress);
     else
         hipLaunchKernelGGL((k3_fanout_flat_t<0, false>), dim3(grid), dim3(256), 0, s, buf,
                            payload_off, payload_len, pairs, seq_base,
-                           nullptr, n_pairs_ptr, capacity, units_per_pair, egress);
+                           nullptr, n_pairs_ptr, capacity, units_per_pair, uniform_len, egress);
 }
 
 extern "C" void launch_k3_fanout_flat3(
     const uint8_t* buf, const int64_t* payload_off, const int32_t* payload_len,
     const PairRec* pairs,
     const uint32_t* seq_state, const int32_t* n_pairs_ptr, int32_t capacity,
-    int32_t units_per_pair, uint8_t* egress, int nt, int grid, hipStream_t s) {
+    int32_t units_per_pair, int32_t uniform_len, uint8_t* egress, int nt, int grid,
+    hipStream_t s) {
     if (grid <= 0) grid = 16384;
     if (nt)
         hipLaunchKernelGGL((k3_fanout_flat_t<1, true>), dim3(grid), dim3(256), 0, s, buf,
                            payload_off, payload_len, pairs, 0u,
-                           seq_state, n_pairs_ptr, capacity, units_per_pair, egress);
+                           seq_state, n_pairs_ptr, capacity, units_per_pair, uniform_len, egress);
     else
         hipLaunchKernelGGL((k3_fanout_flat_t<0, true>), dim3(grid), dim3(256), 0, s, buf,
                            payload_off, payload_len, pairs, 0u,
-                           seq_state, n_pairs_ptr, capacity, units_per_pair, egress);
+                           seq_state, n_pairs_ptr, capacity, units_per_pair, uniform_len, egress);
 }
 
 extern "C" __global__ void k_seq_advance(uint32_t* seq_state, int32_t m) {

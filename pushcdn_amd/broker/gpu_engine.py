@@ -270,8 +270,11 @@ class GpuBrokerEngine:
         # costs ~3% (measured on the 64 KiB mixed bench) — use wave there.
         if uniform and rec <= 4096:
             units = rec // 16
+            # uniform records: wire length passed as a scalar — saves the
+            # per-unit payload_len load (~1 load per 16 B stored)
             ops.fanout_flat2(buf, payload_off, payload_len, self._pairs, seq_base,
-                             self._n_pairs, units, self.egress, nt, 0)
+                             self._n_pairs, units, self.egress, nt, 0,
+                             uniform_wire_len)
         else:
             seq = torch.arange(seq_base, seq_base + M, dtype=torch.int32, device=self.device)
             ops.fanout_wave(buf, payload_off, payload_len, self._pairs, seq,
@@ -293,7 +296,7 @@ class GpuBrokerEngine:
         self._assign_emit(ops, mask_t, payload_len, rec, M)
         ops.fanout_flat3(buf, payload_off, payload_len, self._pairs, self._seq_dev,
                          self._n_pairs, units, self.egress,
-                         1 if self.nt_fanout else 0, 0)
+                         1 if self.nt_fanout else 0, 0, (units - 1) * 16)
         ops.seq_advance(self._seq_dev, M)
 
     def tick_graphed(self, buf: torch.Tensor, offsets: torch.Tensor,
