@@ -652,10 +652,8 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
     const int32_t* __restrict__ n_pairs_ptr,
     int32_t capacity,
     int32_t units_per_pair,
-    int32_t uniform_len,                  // >0: every record is this long AND
-                                          // the batch is contiguous (stride =
-                                          // uniform_len) — skips the per-unit
-                                          // length and offset loads
+    int32_t uniform_len,                  // >0: every record is this long —
+                                          // skips the per-unit length load
     uint8_t* __restrict__ egress)
 {
     typedef unsigned int v4u __attribute__((ext_vector_type(4)));
@@ -679,9 +677,7 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
             else memcpy(dst, hdr, 16);
             continue;
         }
-        const int64_t src_off = uniform_len > 0 ? (int64_t)mi * uniform_len
-                                                 : payload_off[mi];
-        const uint8_t* src = buf + src_off + (size_t)(unit - 1) * 16;
+        const uint8_t* src = buf + payload_off[mi] + (size_t)(unit - 1) * 16;
         const int32_t coff = (unit - 1) * 16;
         if (coff + 16 <= len && (((uintptr_t)src) & 15) == 0) {
             v4u v = *(const v4u*)src;

@@ -192,12 +192,7 @@ class GpuBrokerEngine:
 
         uniform_wire_len: if the caller knows every message in the batch has
         this exact wire length (and fanout_wire is set), the flat-index K3
-        variant runs at ~100% lane utilization.
-        uniform_wire_len contract: when set, every message in the batch
-        is exactly that many (16-aligned) wire bytes AND the batch is
-        contiguous (offsets affine) — the flat fan-out then derives
-        source addresses and lengths from the scalar.
-        """
+        variant runs at ~100% lane utilization."""
         if self.use_gpu_ops:
             return self._tick_gpu(buf, offsets, uniform_wire_len)
         assert host_batch is not None and host_offsets is not None
