@@ -51,7 +51,8 @@ def pack_mesh_batch(msgs: List[bytes], capacity: int) -> Tuple[torch.Tensor, int
     buf[: header_words * 8] = header.view(torch.uint8)
     pos = base
     for raw in msgs:
-        buf[pos : pos + len(raw)] = torch.frombuffer(bytearray(raw), dtype=torch.uint8)
+        if raw:  # torch.frombuffer rejects empty buffers
+            buf[pos : pos + len(raw)] = torch.frombuffer(bytearray(raw), dtype=torch.uint8)
         pos += (len(raw) + 15) & ~15
     return buf, n, total
 

@@ -185,3 +185,47 @@ def test_crdt_convergence_cross_wire(ops_a, ops_b, rounds):
     a.merge(py_full(b))
     py_merge(b, a.get_full())
     assert sorted(a.items()) == sorted(b.items())
+
+
+@SET
+@given(st.lists(st.binary(min_size=0, max_size=512), max_size=32))
+def test_mesh_batch_pack_roundtrip(msgs):
+    """pack_mesh_batch -> unpack_mesh_offsets recovers every message
+    (the broker-plane wire layout for the RCCL collective)."""
+    import torch
+
+    from pushcdn_amd.broker.mesh_service import pack_mesh_batch, unpack_mesh_offsets
+
+    cap = 1 << 16
+    buf, n, used = pack_mesh_batch(msgs, cap)
+    assert n == len(msgs) and used <= cap
+    offsets = unpack_mesh_offsets(buf, n)
+    raw = bytes(buf.numpy().tobytes())
+    for i, msg in enumerate(msgs):
+        start = int(offsets[i])
+        assert raw[start : start + len(msg)] == msg
+        assert start % 16 == 0
+
+
+@SET
+@given(st.lists(st.tuples(st.integers(0, 7), st.integers(0, 7)), max_size=40))
+def test_relational_map_invariants(ops_seq):
+    """Both directions of the bidirectional multimap stay consistent under
+    arbitrary associate/dissociate sequences."""
+    from pushcdn_amd.broker.relational_map import RelationalMap
+
+    rm = RelationalMap()
+    for i, (k, v) in enumerate(ops_seq):
+        if i % 3 == 2:
+            rm.dissociate_key_from_values(f"k{k}", [v])
+        elif i % 7 == 6:
+            rm.remove_key(f"k{k}")
+        else:
+            rm.associate_key_with_values(f"k{k}", [v])
+        # invariant: by_key and by_value are exact transposes
+        fwd = {(k_, v_) for k_ in rm.get_keys() for v_ in rm.get_values_by_key(k_)}
+        rev = {(k_, v_) for v_ in rm.get_values() for k_ in rm.get_keys_by_value(v_)}
+        assert fwd == rev
+        # no empty buckets linger
+        assert all(rm.get_values_by_key(k_) for k_ in rm.get_keys())
+        assert all(rm.get_keys_by_value(v_) for v_ in rm.get_values())
