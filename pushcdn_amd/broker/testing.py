@@ -65,6 +65,8 @@ class TestDefinition:
     connected_users: List[TestUser] = field(default_factory=list)
     connected_brokers: List[TestBroker] = field(default_factory=list)
     topic_space: object = None  # TopicSpace override (default ALL_TOPICS)
+    user_message_hook: object = None
+    broker_message_hook: object = None
 
     async def into_run(self) -> TestRun:
         n = uuid.uuid4().hex[:8]
@@ -84,6 +86,8 @@ class TestDefinition:
         )
         if self.topic_space is not None:
             cfg.topic_space = self.topic_space
+        cfg.user_message_hook = self.user_message_hook
+        cfg.broker_message_hook = self.broker_message_hook
         broker = Broker(cfg)
         await broker.start()
 

@@ -197,3 +197,34 @@ def test_invalid_topic_subscribe_disconnects():
         await run_.close()
 
     asyncio.run(asyncio.wait_for(go(), timeout=30))
+
+
+def test_broker_message_hook_skip():
+    """The broker-plane hook (reference def.rs MessageHookDef on the Broker
+    ConnectionDef): SkipMessage drops an inbound peer-broker message before
+    routing; everything else flows."""
+    from pushcdn_amd.broker.service import SKIP_MESSAGE
+
+    async def go():
+        def hook(msg):
+            if isinstance(msg, m.Broadcast) and msg.message == b"censored":
+                return SKIP_MESSAGE
+            return "process"
+
+        td = TestDefinition(
+            connected_users=[TestUser(topics=[0])],
+            connected_brokers=[TestBroker(connected_users=[], topics=[0])],
+            broker_message_hook=hook,
+        )
+        tr = await td.into_run()
+
+        # peer broker pushes a broadcast: hook lets it through to the user
+        await tr.brokers[0].send_message(m.Broadcast([0], b"from-peer"))
+        await assert_received(tr.users[0], m.Broadcast([0], b"from-peer"), 1)
+
+        # censored broadcast is dropped before routing
+        await tr.brokers[0].send_message(m.Broadcast([0], b"censored"))
+        await assert_not_received(tr.users[0])
+        await tr.close()
+
+    run(go())
