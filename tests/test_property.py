@@ -229,3 +229,21 @@ def test_relational_map_invariants(ops_seq):
         # no empty buckets linger
         assert all(rm.get_values_by_key(k_) for k_ in rm.get_keys())
         assert all(rm.get_keys_by_value(v_) for v_ in rm.get_values())
+
+
+@settings(max_examples=15, deadline=None, derandomize=True)
+@given(st.text(min_size=0, max_size=32), st.binary(max_size=256),
+       st.integers(min_value=0, max_value=2**64 - 1))
+def test_bls_sign_verify_property(namespace, message, seed):
+    """sign/verify holds for arbitrary namespaces and messages; a different
+    namespace, message or key always fails (domain separation)."""
+    from pushcdn_amd.crypto import bls
+
+    kp = bls.KeyPair.from_seed(seed)
+    sig = bls.sign(kp.private_key, namespace, message)
+    assert len(sig) == 64
+    assert bls.verify(kp.public_key, namespace, message, sig)
+    assert not bls.verify(kp.public_key, namespace + "x", message, sig)
+    assert not bls.verify(kp.public_key, namespace, message + b"x", sig)
+    other = bls.KeyPair.from_seed(seed ^ 0x5A5A)
+    assert not bls.verify(other.public_key, namespace, message, sig)
