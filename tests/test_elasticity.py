@@ -173,3 +173,51 @@ def test_receive_messages_batched(tmp_path):
             await br.close()
 
     run(go())
+
+
+def test_dynamic_subscriptions_replay_after_restart(tmp_path):
+    """Subscriptions changed at runtime (subscribe/unsubscribe AFTER
+    connect) are what gets replayed on reconnect (reference lib.rs:383-414
+    updates the replay set first, then best-effort sends)."""
+
+    async def go():
+        db = new_db(tmp_path)
+        broker = make_broker(db, tag="replay")
+        await broker.start()
+        await broker.discovery.perform_heartbeat(0, 60)
+        marshal, endpoint = make_marshal(db)
+        await marshal.start()
+
+        client = make_client(endpoint, seed=6, topics=[2])
+        await client.ensure_initialized()
+        await client.subscribe([7])      # added at runtime
+        await client.unsubscribe([2])    # removed at runtime
+        await asyncio.sleep(0.2)
+
+        await broker.close()
+        broker2 = make_broker(db, tag="replay")
+        await broker2.start()
+        await broker2.discovery.perform_heartbeat(0, 60)
+
+        # after reconnect, topic 7 must deliver (replayed) ...
+        ok = False
+        for _ in range(10):
+            try:
+                await client.send_broadcast_message([7], b"on-seven")
+                msg = await asyncio.wait_for(client.receive_message(), timeout=2)
+                if isinstance(msg, m.Broadcast) and msg.message == b"on-seven":
+                    ok = True
+                    break
+            except Exception:
+                await asyncio.sleep(0.3)
+        assert ok
+        # ... and topic 2 must NOT (unsubscribed before the restart)
+        await client.send_broadcast_message([2], b"on-two")
+        try:
+            msg = await asyncio.wait_for(client.receive_message(), timeout=1)
+            raise AssertionError(f"unsubscribed topic delivered: {msg}")
+        except asyncio.TimeoutError:
+            pass
+        await stop_stack([broker2], marshal, client)
+
+    run(go())
