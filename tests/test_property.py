@@ -247,3 +247,43 @@ def test_bls_sign_verify_property(namespace, message, seed):
     assert not bls.verify(kp.public_key, namespace, message + b"x", sig)
     other = bls.KeyPair.from_seed(seed ^ 0x5A5A)
     assert not bls.verify(other.public_key, namespace, message, sig)
+
+
+@settings(max_examples=30, deadline=None, derandomize=True)
+@given(st.lists(st.integers(min_value=1, max_value=64), min_size=1, max_size=12))
+def test_limiter_pool_never_overallocates(sizes):
+    """The global byte pool's outstanding total never exceeds its budget,
+    and every release returns capacity (reference pool.rs:28-111)."""
+    import asyncio
+
+    from pushcdn_amd.proto.limiter import Limiter
+
+    async def go():
+        budget = 128
+        lim = Limiter(global_memory_pool_size=budget)
+        held = []
+        outstanding = 0
+        for n in sizes:
+            if outstanding + n > budget:
+                # must NOT be grantable right now
+                task = asyncio.ensure_future(lim.allocate_message_bytes(n))
+                await asyncio.sleep(0)
+                assert not task.done()
+                # free everything; the waiter must then proceed
+                for p in held:
+                    p.release()
+                held.clear()
+                outstanding = 0
+                held.append(await asyncio.wait_for(task, 1))
+                outstanding += n
+            else:
+                held.append(await asyncio.wait_for(lim.allocate_message_bytes(n), 1))
+                outstanding += n
+            assert outstanding <= budget
+        for p in held:
+            p.release()
+        # fully drained: a budget-size allocation succeeds immediately
+        p = await asyncio.wait_for(lim.allocate_message_bytes(budget), 1)
+        p.release()
+
+    asyncio.run(go())
