@@ -268,3 +268,31 @@ def test_native_and_python_crdt_wire_compatible():
     changed = native2.merge(pyraw)
     assert changed == [(b"carol", None, b"broker-3")]
     assert native2.get(b"carol") == b"broker-3"
+
+
+def test_embedded_permit_uniqueness_and_least_conn_determinism(tmp_path):
+    """Permits are unique across a burst of issues; least-connections
+    placement is deterministic under ties (identity order) — the marshal's
+    placement contract (reference embedded.rs:241-309, discovery trait)."""
+    import asyncio
+
+    from pushcdn_amd.discovery import BrokerIdentifier, new_discovery_client
+
+    async def go():
+        db = str(tmp_path / "uniq.db")
+        a = new_discovery_client(db, BrokerIdentifier("a-pub", "a-priv"))
+        b = new_discovery_client(db, BrokerIdentifier("b-pub", "b-priv"))
+        await a.perform_heartbeat(3, 60)
+        await b.perform_heartbeat(3, 60)
+        # tie on connections: identity order decides, consistently
+        picks = {str(await a.get_with_least_connections()) for _ in range(5)}
+        assert len(picks) == 1
+        # permit burst: all unique, all single-use
+        target = await a.get_with_least_connections()
+        permits = [await a.issue_permit(target, 30, b"user-%d" % i) for i in range(50)]
+        assert len(set(permits)) == 50
+        for i, p in enumerate(permits):
+            assert await a.validate_permit(target, p) == b"user-%d" % i
+            assert await a.validate_permit(target, p) is None  # one-time
+
+    asyncio.run(asyncio.wait_for(go(), 30))
