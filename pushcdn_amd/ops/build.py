@@ -90,11 +90,13 @@ def build_core(verbose: bool = False):
     BUILD_DIR.mkdir(exist_ok=True)
     so = BUILD_DIR / "pushcdn_core.so"
     srcs = [str(s) for s in CORE_SOURCES]
-    deps = list((CSRC / "common").glob("*.h")) + list((CSRC / "bls").glob("*.h")) + CORE_SOURCES
+    deps = (list((CSRC / "common").glob("*.h")) + list((CSRC / "bls").glob("*.h"))
+            + list((CSRC / "wire").glob("*.h")) + list((CSRC / "state").glob("*.h"))
+            + list((CSRC / "net").glob("*.h")) + CORE_SOURCES)
     newest_dep = max(p.stat().st_mtime for p in deps)
     if not so.exists() or so.stat().st_mtime < newest_dep:
         cmd = [
-            "g++", "-O2", "-std=c++17", "-shared", "-fPIC",
+            "g++", "-O3", "-std=c++17", "-shared", "-fPIC",
             f"-I{pybind11.get_include()}",
             f"-I{sysconfig.get_paths()['include']}",
             f"-I{CSRC}",
