@@ -309,3 +309,28 @@ def test_large_message_flood(tmp_path):
         await stop_stack(brokers, marshal, client)
 
     run(go())
+
+
+def test_multiple_marshals_share_discovery(tmp_path):
+    """Several marshals over one discovery namespace: permits issued by any
+    marshal validate at the broker (the reference runs marshal fleets
+    behind a load balancer; permits live in discovery, not the marshal)."""
+
+    async def go():
+        brokers, marshal1, ep1 = await start_stack(tmp_path)
+        marshal2, ep2 = make_marshal(brokers[0].config.discovery_endpoint)
+        await marshal2.start()
+
+        a = make_client(ep1, seed=31, topics=[0])
+        b = make_client(ep2, seed=32, topics=[0])  # different marshal
+        await a.ensure_initialized()
+        await b.ensure_initialized()
+        await asyncio.sleep(0.2)
+        await a.send_broadcast_message([0], b"via-marshal-1")
+        msg = await asyncio.wait_for(b.receive_message(), timeout=10)
+        assert msg.message == b"via-marshal-1"
+        a.close(); b.close()
+        await marshal2.close()
+        await stop_stack(brokers, marshal1)
+
+    run(go())
