@@ -296,3 +296,21 @@ def test_embedded_permit_uniqueness_and_least_conn_determinism(tmp_path):
             assert await a.validate_permit(target, p) is None  # one-time
 
     asyncio.run(asyncio.wait_for(go(), 30))
+
+
+def test_mnemonic_and_local_ip_helpers():
+    """Mnemonic ids are deterministic, distinct, and human-shaped
+    (reference util.rs mnemonic); local_ip substitution only rewrites the
+    placeholder (reference lib.rs:157-168)."""
+    from pushcdn_amd.broker.service import resolve_local_ip
+    from pushcdn_amd.utils.mnemonic import mnemonic
+
+    a = mnemonic(b"key-one")
+    b = mnemonic(b"key-two")
+    assert a == mnemonic(b"key-one")  # deterministic
+    assert a != b
+    assert "-" in a and a.islower()
+
+    assert resolve_local_ip("127.0.0.1:1738") == "127.0.0.1:1738"  # untouched
+    out = resolve_local_ip("local_ip:1738")
+    assert out.endswith(":1738") and "local_ip" not in out
