@@ -153,3 +153,53 @@ def test_connection_churn_native_transport(tmp_path):
         await broker.close()
 
     run(go())
+
+
+def test_injector_clis_run(tmp_path):
+    """The fault-injector binaries start against a live stack and do their
+    thing for a couple of seconds (reference bad-broker.rs / bad-connector.rs
+    / bad-sender.rs run under process-compose)."""
+    import os
+    import signal
+    import subprocess
+    import sys
+    import time
+
+    from pathlib import Path
+
+    REPO = str(Path(__file__).resolve().parent.parent)
+    db = str(tmp_path / "inj.db")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO
+    procs = []
+
+    def spawn(*args):
+        p = subprocess.Popen([sys.executable, "-m", "pushcdn_amd.cli", *args],
+                             cwd=REPO, env=env, stdout=subprocess.PIPE,
+                             stderr=subprocess.STDOUT, text=True,
+                             start_new_session=True)
+        procs.append(p)
+        return p
+
+    try:
+        spawn("marshal", "-d", db, "-b", "127.0.0.1:42737")
+        spawn("broker", "-d", db,
+              "--public-bind-endpoint", "127.0.0.1:42738",
+              "--public-advertise-endpoint", "127.0.0.1:42738",
+              "--private-bind-endpoint", "127.0.0.1:42739",
+              "--private-advertise-endpoint", "127.0.0.1:42739")
+        time.sleep(4)
+        bb = spawn("bad-broker", "-d", db)
+        bc = spawn("bad-connector", "-m", "127.0.0.1:42737")
+        time.sleep(4)
+        # injectors are alive and chaosing; stack processes alive too
+        for p in procs:
+            assert p.poll() is None, (p.args, p.stdout.read()[-500:])
+    finally:
+        for p in procs:
+            try:
+                os.killpg(p.pid, signal.SIGKILL)
+            except Exception:
+                p.kill()
+        for p in procs:
+            p.wait(timeout=10)
