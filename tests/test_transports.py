@@ -506,3 +506,41 @@ def test_two_broker_mesh_over_native_tcp(tmp_path):
         await b2.close()
 
     run(go())
+
+
+def test_tls_explicit_ca_trust(tmp_path):
+    """A client trusting an EXPLICIT CA file (not the process-local CA)
+    handshakes with a server using a leaf minted from that CA
+    (reference tls.rs:100-126 load_ca + root store wiring)."""
+    from pushcdn_amd.crypto.tls import generate_ca
+    from pushcdn_amd.proto.transports.tcp_tls import TcpTls
+
+    ca_cert, ca_key = generate_ca(str(tmp_path / "ca"))
+
+    class TlsWithCa(TcpTls):  # RunDef-style CA wiring via class attrs
+        ca_cert_path = ca_cert
+        ca_key_path = ca_key
+
+    async def go():
+        limiter = Limiter(global_memory_pool_size=1 << 20)
+        listener = await TlsWithCa.bind("127.0.0.1:0", None, None)
+        endpoint = f"127.0.0.1:{listener.port}"
+
+        async def server():
+            conn = await (await listener.accept()).finalize(limiter)
+            msg = await conn.recv_message()
+            await conn.send_message(m.Direct(b"s", msg.message[::-1]))
+            await conn.soft_close()
+
+        async def client():
+            # use_local_authority=False -> trust the EXPLICIT CA file
+            conn = await TlsWithCa.connect(endpoint, False, limiter)
+            await conn.send_message(m.Direct(b"c", b"abc"))
+            reply = await conn.recv_message()
+            assert reply.message == b"cba"
+            await conn.soft_close()
+
+        await asyncio.wait_for(asyncio.gather(server(), client()), timeout=15)
+        await listener.close()
+
+    run(go())
