@@ -126,3 +126,25 @@ def test_subscribe_all_bulk():
     wpos = eng.drain_cursors()
     delivered = sum(1 for u in range(70) if int(wpos[u]) > 0)
     assert delivered == 70
+
+
+def test_parse_ring_records_seq_sort_with_wrap():
+    """Drained ring records come back in seq order even when the u32
+    sequence counter wraps mid-batch (the ring WRITE order is claim order;
+    K5b's atomic direct claims interleave within a tick)."""
+    import struct
+
+    from pushcdn_amd.broker.gpu_engine import parse_ring_records, ring_rec
+
+    def rec(seq, payload):
+        r = struct.pack("<IIII", len(payload), seq, 0, 0) + payload
+        return r + b"\x00" * (ring_rec(len(payload)) - len(r))
+
+    ring = (rec(1, b"c") + rec(0xFFFFFFFE, b"a")
+            + rec(0, b"b2") + rec(0xFFFFFFFF, b"b1"))
+    out = parse_ring_records(ring, len(ring))
+    assert [p for _, p in out] == [b"a", b"b1", b"b2", b"c"]
+
+    # no-wrap ordinary case
+    ring2 = rec(7, b"y") + rec(5, b"x") + rec(9, b"z")
+    assert [p for _, p in parse_ring_records(ring2, len(ring2))] == [b"x", b"y", b"z"]

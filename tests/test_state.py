@@ -314,3 +314,21 @@ def test_mnemonic_and_local_ip_helpers():
     assert resolve_local_ip("127.0.0.1:1738") == "127.0.0.1:1738"  # untouched
     out = resolve_local_ip("local_ip:1738")
     assert out.endswith(":1738") and "local_ip" not in out
+
+
+def test_rundef_bundles():
+    """RunDef-style config bundles wire scheme x protocol x discovery x
+    topic space together (reference def.rs:54-168)."""
+    from pushcdn_amd.proto.rundef import production_run_def, testing_run_def
+    from pushcdn_amd.proto.topic import ALL_TOPICS, TEST_TOPIC_SPACE
+    from pushcdn_amd.proto.transports.memory import Memory
+    from pushcdn_amd.proto.transports.tcp_tls import TcpTls
+
+    prod = production_run_def("/tmp/x.db")
+    test = testing_run_def("/tmp/y.db")
+    assert prod.topic_space is ALL_TOPICS
+    assert prod.user.protocol is TcpTls
+    assert test.topic_space is TEST_TOPIC_SPACE
+    assert test.user.protocol is Memory
+    # prune semantics ride along
+    assert test.topic_space.prune([0, 1, 0]) == [0, 1]
