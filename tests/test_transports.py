@@ -544,3 +544,46 @@ def test_tls_explicit_ca_trust(tmp_path):
         await listener.close()
 
     run(go())
+
+
+def test_pump_many_connections_stress():
+    """One pump, 60 concurrent native-loopback connections, interleaved
+    bursts — every frame arrives intact, in per-connection order, no
+    cross-connection bleed."""
+    from pushcdn_amd.proto.transports.tcp_native import TcpNative
+
+    async def go():
+        limiter = Limiter(global_memory_pool_size=1 << 28)
+        listener = await TcpNative.bind("127.0.0.1:0", None, None)
+        endpoint = f"127.0.0.1:{listener.port}"
+        N, MSGS = 60, 40
+
+        async def server():
+            async def serve_one(idx):
+                conn = await (await listener.accept()).finalize(limiter)
+                ident = (await conn.recv_message()).message  # b"conn-<i>"
+                for j in range(MSGS):
+                    msg = await conn.recv_message()
+                    assert msg.message == ident + b":%d" % j
+                await conn.send_message(m.Direct(b"s", ident + b":done"))
+                await conn.soft_close()
+
+            await asyncio.gather(*[serve_one(i) for i in range(N)])
+
+        async def client(i):
+            conn = await TcpNative.connect(endpoint, True, limiter)
+            ident = b"conn-%d" % i
+            await conn.send_message(m.Direct(b"c", ident))
+            for j in range(MSGS):
+                await conn.send_message(m.Direct(b"c", ident + b":%d" % j))
+                if j % 7 == 0:
+                    await asyncio.sleep(0)  # interleave with other clients
+            reply = await conn.recv_message()
+            assert reply.message == ident + b":done"
+            await conn.soft_close()
+
+        await asyncio.wait_for(
+            asyncio.gather(server(), *[client(i) for i in range(N)]), timeout=60)
+        await listener.close()
+
+    run(go())
