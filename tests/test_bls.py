@@ -136,3 +136,26 @@ def test_fp12_fast_paths_consistent(core):
     squaring must agree with the dense/general implementations."""
     for a, b in [(3, 5), (7, 11), (123456, 654321)]:
         assert core._fp12_fastpath_ok(a, b)
+
+
+def test_gpu_batch_verifier_host_fallback():
+    """Below min_batch the micro-batching verifier answers from the host
+    path (a K1 launch costs ~100 ms; tiny batches shouldn't pay it) — and
+    bad signatures still fail."""
+    import asyncio
+
+    from pushcdn_amd.crypto import bls
+    from pushcdn_amd.crypto.gpu_verify import GpuBatchVerifier
+
+    async def go():
+        v = GpuBatchVerifier(max_wait_s=0.01, min_batch=8)
+        kp = bls.KeyPair.from_seed(5)
+        sig = bls.sign(kp.private_key, "ns", b"msg")
+        ok, bad = await asyncio.gather(
+            v.verify(kp.public_key, "ns", b"msg", sig),
+            v.verify(kp.public_key, "ns", b"other", sig),
+        )
+        assert ok is True and bad is False
+        await v.close()
+
+    asyncio.run(asyncio.wait_for(go(), 30))
