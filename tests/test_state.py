@@ -332,3 +332,14 @@ def test_rundef_bundles():
     assert test.user.protocol is Memory
     # prune semantics ride along
     assert test.topic_space.prune([0, 1, 0]) == [0, 1]
+
+
+def test_fnv1a64_known_vectors():
+    """The routing hash matches the published FNV-1a 64 vectors — it must
+    agree across Python, host C++ and the device kernels (reference
+    util.rs:19-23 uses the same function for DirectMap keys)."""
+    from pushcdn_amd.utils.keyhash import fnv1a64
+
+    assert fnv1a64(b"") == 0xCBF29CE484222325
+    assert fnv1a64(b"a") == 0xAF63DC4C8601EC8C
+    assert fnv1a64(b"foobar") == 0x85944171F73967E8
