@@ -92,6 +92,9 @@ public:
         fcntl(fd, F_SETFL, flags | O_NONBLOCK);
         int one = 1;
         setsockopt(fd, IPPROTO_TCP, 1 /*TCP_NODELAY*/, &one, sizeof(one));
+        int bufsz = 4 << 20;  // large frames: fewer syscalls per message
+        setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &bufsz, sizeof(bufsz));
+        setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &bufsz, sizeof(bufsz));
         int64_t id;
         {
             std::lock_guard<std::mutex> g(mu_);
@@ -278,9 +281,10 @@ private:
     // reads everything available; returns true if new complete frames landed
     bool do_read(Conn& c) {
         bool new_frames = false;
-        char tmp[1 << 16];
+        static thread_local std::vector<char> tmpv(1 << 20);
+        char* tmp = tmpv.data();
         while (true) {
-            ssize_t n = ::recv(c.fd, tmp, sizeof(tmp), 0);
+            ssize_t n = ::recv(c.fd, tmp, tmpv.size(), 0);
             if (n > 0) {
                 c.in_bytes += (uint64_t)n;
                 c.rbuf.insert(c.rbuf.end(), tmp, tmp + n);
@@ -297,7 +301,7 @@ private:
                     new_frames = true;
                 }
                 if (off) c.rbuf.erase(c.rbuf.begin(), c.rbuf.begin() + off);
-                if (n < (ssize_t)sizeof(tmp)) continue;  // might be more
+                if (n < (ssize_t)tmpv.size()) continue;  // might be more
             } else if (n == 0) {
                 close_locked(c);
                 return true;
