@@ -31,6 +31,8 @@ from typing import Dict, List, Optional, Tuple
 
 import torch
 
+from ..utils.keyhash import fnv1a64
+
 RING_ALIGN = 16
 
 
@@ -40,9 +42,6 @@ def ring_rec(length: int) -> int:
     ring_rec; K3 zero-pads the tail unit so each record is written with
     full vector stores."""
     return (length + 16 + (RING_ALIGN - 1)) & ~(RING_ALIGN - 1)
-
-
-from ..utils.keyhash import fnv1a64
 
 
 @dataclass
