@@ -587,3 +587,19 @@ def test_pump_many_connections_stress():
         await listener.close()
 
     run(go())
+
+
+def test_leaf_cert_san_is_espresso(tmp_path):
+    """Leaf certificates carry the fixed SAN/SNI name "espresso" — wire
+    parity with the reference's pinned name (tls.rs:63-71, tcp_tls.rs:91-95:
+    clients always connect with server_hostname=espresso)."""
+    import subprocess
+
+    from pushcdn_amd.crypto.tls import generate_ca, generate_cert_from_ca
+
+    ca_cert, ca_key = generate_ca(str(tmp_path / "ca"))
+    cert, _key = generate_cert_from_ca(ca_cert, ca_key, str(tmp_path / "leaf"))
+    text = subprocess.run(["openssl", "x509", "-in", cert, "-noout", "-text"],
+                          capture_output=True, text=True, check=True).stdout
+    assert "DNS:espresso" in text
+    assert "CN = espresso" in text or "CN=espresso" in text
