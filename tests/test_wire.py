@@ -128,3 +128,19 @@ def test_cpp_codec_rejects_garbage():
     assert core.wire_deserialize(bytes(data)) is not None
     for cut in (9, 17, len(data) - 8):
         assert core.wire_deserialize(bytes(data[:cut])) is None
+
+
+def test_error_taxonomy_hierarchy():
+    """The error taxonomy drives reconnect policy (reference error.rs:21-44):
+    connection/auth/parse errors are distinct and catchable as the shared
+    base, so callers can branch like the reference's Error enum."""
+    from pushcdn_amd.proto import errors as e
+
+    for exc in (e.ConnectionError_, e.AuthenticationError, e.ParseError,
+                e.DeserializeError, e.SerializeError, e.TopicError,
+                e.CryptoError, e.DiscoveryError):
+        assert issubclass(exc, e.CdnError)
+        try:
+            raise exc("boom")
+        except e.CdnError as caught:
+            assert "boom" in str(caught)
