@@ -86,6 +86,10 @@ class MeshBroker(Broker):
         self._mesh_executor = concurrent.futures.ThreadPoolExecutor(
             max_workers=1, thread_name_prefix="mesh-tick")
 
+    async def close(self) -> None:
+        await super().close()
+        self._mesh_executor.shutdown(wait=False, cancel_futures=True)
+
     # the framed broker mesh is replaced by the collective: no dialing, no
     # framed sync blasts; heartbeats still publish load for the marshal
     async def _heartbeat_task(self) -> None:
