@@ -603,3 +603,39 @@ def test_leaf_cert_san_is_espresso(tmp_path):
                           capture_output=True, text=True, check=True).stdout
     assert "DNS:espresso" in text
     assert "CN = espresso" in text or "CN=espresso" in text
+
+
+def test_queued_messages_drain_after_peer_close():
+    """Messages already received before the peer closed are still readable;
+    only AFTER the queue drains does recv fail (base.py recv_message_raw's
+    reader-death branch — the reference's channel close semantics)."""
+    async def go():
+        limiter = Limiter(global_memory_pool_size=1 << 20)
+        listener = await Tcp.bind("127.0.0.1:0", None, None)
+        endpoint = f"127.0.0.1:{listener.port}"
+
+        got = []
+
+        async def server():
+            conn = await (await listener.accept()).finalize(limiter)
+            await conn.send_message(m.Direct(b"s", b"one"))
+            await conn.send_message(m.Direct(b"s", b"two"))
+            await conn.soft_close()  # flush, then the peer sees EOF
+
+        async def client():
+            conn = await Tcp.connect(endpoint, True, limiter)
+            await asyncio.sleep(0.3)  # let both frames land and the reader die
+            got.append((await conn.recv_message()).message)
+            got.append((await conn.recv_message()).message)
+            try:
+                await asyncio.wait_for(conn.recv_message(), timeout=2)
+                raise AssertionError("expected ConnectionError after drain")
+            except ConnectionError_:
+                pass
+            conn.close()
+
+        await asyncio.wait_for(asyncio.gather(server(), client()), timeout=15)
+        await listener.close()
+        assert got == [b"one", b"two"]
+
+    run(go())
