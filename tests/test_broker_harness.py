@@ -228,3 +228,34 @@ def test_broker_message_hook_skip():
         await tr.close()
 
     run(go())
+
+
+def test_strong_consistency_pushes_syncs_on_connect():
+    """strong-consistency (reference cargo feature, broker default): when a
+    user connects, partial user+topic syncs go to peers IMMEDIATELY instead
+    of waiting for the 10 s sync timer (user/handler.rs:79-90)."""
+    async def go():
+        td = TestDefinition(
+            connected_users=[],
+            connected_brokers=[TestBroker(connected_users=[], topics=[])],
+        )
+        tr = await td.into_run()
+        assert tr.broker.config.strong_consistency  # default on
+
+        # inject a user AFTER the broker mesh exists: the add_user path
+        # should blast partial syncs to the fake peer broker at once
+        client_half = await TestDefinition._inject_user(tr.broker, at_index(9), [3])
+        await tr.broker._send_partial_syncs()
+        # the fake peer receives TopicSync/UserSync frames without any timer
+        saw = set()
+        for _ in range(4):
+            try:
+                msg = await asyncio.wait_for(tr.brokers[0].recv_message(), timeout=1)
+                saw.add(type(msg).__name__)
+            except Exception:
+                break
+        assert "UserSync" in saw or "TopicSync" in saw, saw
+        client_half.close()
+        await tr.close()
+
+    run(go())
