@@ -221,3 +221,39 @@ def test_dynamic_subscriptions_replay_after_restart(tmp_path):
         await stop_stack([broker2], marshal, client)
 
     run(go())
+
+
+def test_client_reconnects_after_explicit_close(tmp_path):
+    """The elastic client treats its own closed connection like any other
+    failure: the next operation reconnects and replays subscriptions
+    (reference reconnect_if_needed, lib.rs:204-258)."""
+    from tests.test_integration import make_broker, make_client, make_marshal, new_db, run
+
+    async def go():
+        db = new_db(tmp_path)
+        broker = make_broker(db, tag="reopen")
+        await broker.start()
+        await broker.discovery.perform_heartbeat(0, 60)
+        marshal, endpoint = make_marshal(db)
+        await marshal.start()
+        client = make_client(endpoint, seed=8, topics=[1])
+        await client.ensure_initialized()
+        await client.ensure_initialized()  # idempotent
+        client.close()
+
+        ok = False
+        for _ in range(10):
+            try:
+                await client.send_broadcast_message([1], b"after-close")
+                msg = await asyncio.wait_for(client.receive_message(), timeout=2)
+                if msg.message == b"after-close":
+                    ok = True
+                    break
+            except Exception:
+                await asyncio.sleep(0.2)
+        assert ok
+        client.close()
+        await marshal.close()
+        await broker.close()
+
+    run(go())
