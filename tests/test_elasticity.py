@@ -223,10 +223,11 @@ def test_dynamic_subscriptions_replay_after_restart(tmp_path):
     run(go())
 
 
-def test_client_reconnects_after_explicit_close(tmp_path):
-    """The elastic client treats its own closed connection like any other
-    failure: the next operation reconnects and replays subscriptions
-    (reference reconnect_if_needed, lib.rs:204-258)."""
+def test_client_explicit_close_is_final(tmp_path):
+    """Explicit Client.close() is FINAL: reconnect only heals FAILURES,
+    never a user-initiated shutdown — post-close operations raise instead
+    of silently reviving the session."""
+    from pushcdn_amd.proto.errors import ConnectionError_
     from tests.test_integration import make_broker, make_client, make_marshal, new_db, run
 
     async def go():
@@ -240,19 +241,12 @@ def test_client_reconnects_after_explicit_close(tmp_path):
         await client.ensure_initialized()
         await client.ensure_initialized()  # idempotent
         client.close()
-
-        ok = False
-        for _ in range(10):
-            try:
-                await client.send_broadcast_message([1], b"after-close")
-                msg = await asyncio.wait_for(client.receive_message(), timeout=2)
-                if msg.message == b"after-close":
-                    ok = True
-                    break
-            except Exception:
-                await asyncio.sleep(0.2)
-        assert ok
-        client.close()
+        assert client.is_closed
+        try:
+            await client.send_broadcast_message([1], b"after-close")
+            raise AssertionError("closed client accepted a send")
+        except ConnectionError_:
+            pass
         await marshal.close()
         await broker.close()
 
