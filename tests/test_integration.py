@@ -334,3 +334,40 @@ def test_multiple_marshals_share_discovery(tmp_path):
         await stop_stack(brokers, marshal1)
 
     run(go())
+
+
+def test_marshal_rejects_garbage_and_wrong_type(tmp_path):
+    """The marshal's one-shot handler: a non-auth first frame or undecodable
+    bytes ends the attempt cleanly (reference handlers.rs:21-37 bails), and
+    the marshal keeps serving afterwards."""
+    from pushcdn_amd.proto.limiter import Bytes, Limiter
+    from pushcdn_amd.proto.transports.memory import Memory
+
+    async def go():
+        brokers, marshal, endpoint = await start_stack(tmp_path)
+        limiter = Limiter(global_memory_pool_size=1 << 20)
+
+        # wrong first message type
+        conn = await Memory.connect(endpoint, True, limiter)
+        await conn.send_message(m.Subscribe([1]))
+        try:
+            reply = await asyncio.wait_for(conn.recv_message(), timeout=5)
+            assert getattr(reply, "permit", 0) == 0  # failure response
+        except Exception:
+            pass  # or the marshal just dropped us — also fine
+        conn.close()
+
+        # undecodable bytes
+        conn = await Memory.connect(endpoint, True, limiter)
+        await conn.send_message_raw(Bytes(b"\x00\x01\x02\x03garbage"))
+        await asyncio.sleep(0.2)
+        conn.close()
+
+        # the marshal still serves real clients
+        client = make_client(endpoint, seed=44, topics=[0])
+        await client.ensure_initialized()
+        await client.send_direct_message(client.public_key, b"still-up")
+        assert (await asyncio.wait_for(client.receive_message(), timeout=10)).message == b"still-up"
+        await stop_stack(brokers, marshal, client)
+
+    run(go())
