@@ -79,7 +79,10 @@ class Marshal:
         try:
             await asyncio.wait_for(
                 MarshalAuth.verify_user(connection, self.discovery, self._verifier), 5)
-        except asyncio.TimeoutError:
+        except (asyncio.TimeoutError, Exception):
+            # one-shot handler: any failure (slow peer, dropped socket,
+            # malformed auth) just ends this attempt; never let it become
+            # an unobserved task exception
             pass
         finally:
             await connection.soft_close()
