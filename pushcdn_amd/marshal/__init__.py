@@ -79,7 +79,7 @@ class Marshal:
         try:
             await asyncio.wait_for(
                 MarshalAuth.verify_user(connection, self.discovery, self._verifier), 5)
-        except (asyncio.TimeoutError, Exception):
+        except Exception:
             # one-shot handler: any failure (slow peer, dropped socket,
             # malformed auth) just ends this attempt; never let it become
             # an unobserved task exception
