@@ -33,6 +33,9 @@ def test_bench_world_size_2_gloo():
     assert result["n_gpus"] == 2
     assert result["value"] > 0
     assert result["config"]["subscribers_per_broker"] == 16
+    assert result["config"]["drops"] == 0
+    # whole-node deliveries: both ranks' 4-msg batches to 16 local users each
+    assert result["config"]["deliveries_per_step_node"] == 2 * 4 * 16
 
 
 def test_mesh_exchange_semantics():
