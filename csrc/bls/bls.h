@@ -98,7 +98,25 @@ BN_INLINE void verkey_serialize(const VerKey& vk, uint8_t out[128]) {
     u256_to_le(vk.y.c1.to_u256(), out + 96);
 }
 
-BN_INLINE bool verkey_deserialize(const uint8_t in[128], VerKey& vk) {
+// Prime-order subgroup membership for a non-infinity on-curve G2 point.
+// BN254 E'(Fp2) has a large cofactor, so on-curve alone is NOT enough (the
+// reference's ark deserialize_uncompressed enforces subgroup membership).
+// Fast endomorphism check: P is in the r-order subgroup iff
+//     psi(P) == [t-1]P          with t-1 = 6x^2
+// where psi is the untwist-Frobenius endomorphism (g2_frobenius).
+// Soundness: psi^2 - [t]psi + [p] = 0 holds on all of E'(Fp2); substituting
+// psi(P) = [t-1]P yields [p+1-t]P = O, and p+1-t = r exactly for BN curves
+// (G1 cofactor is 1).  ~127-bit scalar mul instead of the naive 254-bit [r]P.
+BN_BIGFUNC bool g2_in_subgroup(const Fp2& x, const Fp2& y) {
+    G2 s = G2::scalar_mul({x, y, Fp2::one()}, from_limbs(bn254c::SIX_X_SQ));
+    if (s.is_infinity()) return false;  // psi(P) of a finite point is finite
+    G2Affine psi = g2_frobenius({x, y});
+    Fp2 sx, sy;
+    s.to_affine(sx, sy);
+    return sx == psi.x && sy == psi.y;
+}
+
+BN_BIGFUNC bool verkey_deserialize(const uint8_t in[128], VerKey& vk) {
     U256 p = from_limbs(bn254c::P);
     U256 xc0 = u256_from_le(in), xc1 = u256_from_le(in + 32);
     U256 yc0 = u256_from_le(in + 64), yc1 = u256_from_le(in + 96);
@@ -107,7 +125,8 @@ BN_INLINE bool verkey_deserialize(const uint8_t in[128], VerKey& vk) {
     vk.x = {Fp::from_u256(xc0), Fp::from_u256(xc1)};
     vk.y = {Fp::from_u256(yc0), Fp::from_u256(yc1)};
     if (vk.x.is_zero() && vk.y.is_zero()) return false;  // infinity not a valid key
-    return g2_on_curve(vk.x, vk.y);
+    if (!g2_on_curve(vk.x, vk.y)) return false;
+    return g2_in_subgroup(vk.x, vk.y);
 }
 
 BN_INLINE void sig_serialize(const Fp& x, const Fp& y, uint8_t out[64]) {

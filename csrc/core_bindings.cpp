@@ -154,6 +154,14 @@ static bool subgroup_test() {
            G2::scalar_mul(g2_generator(), r).is_infinity();
 }
 
+static bool verkey_ok(const py::bytes& vk_bytes) {
+    // full deserialization validity incl. the G2 prime-order subgroup check
+    auto vkv = to_vec(vk_bytes);
+    if (vkv.size() != 128) return false;
+    bls::VerKey vk;
+    return bls::verkey_deserialize(vkv.data(), vk);
+}
+
 static bool hard_exp_chain_matches_generic(uint64_t a, uint64_t b) {
     // Miller-loop output for random-ish points, then compare the two hard-part
     // implementations after the shared easy part.
@@ -342,6 +350,7 @@ PYBIND11_MODULE(pushcdn_core, m) {
     m.def("_g1_mul", &g1_mul_test);
     m.def("_pairing_bilinear", &pairing_bilinearity_test);
     m.def("_subgroup_ok", &subgroup_test);
+    m.def("_verkey_ok", &verkey_ok, "verkey bytes pass deserialize incl. subgroup check");
     m.def("_sha256", &sha256_test);
     m.def("_hash_to_g1", &hash_to_g1_test);
     m.def("_hard_exp_chain_ok", &hard_exp_chain_matches_generic);

@@ -21,7 +21,7 @@ struct ParseOut {
 struct alignas(16) PairRec { int32_t user; int32_t msg; int64_t dst; };
 
 extern "C" {
-void launch_k4_parse(const uint8_t*, const int64_t*, int32_t, ParseOut, hipStream_t);
+void launch_k4_parse(const uint8_t*, const int64_t*, int32_t, uint64_t, ParseOut, hipStream_t);
 void launch_k2a_topic_mask(const uint64_t*, const uint8_t*, const int64_t*, const int32_t*,
                            const int32_t*, uint64_t*, int32_t, int32_t, hipStream_t);
 void launch_k2b_count(const uint64_t*, int32_t, int32_t, int32_t, int32_t*, hipStream_t);
@@ -71,7 +71,8 @@ static inline hipStream_t cur_stream() {
     return at::hip::getCurrentHIPStream().stream();
 }
 
-std::vector<torch::Tensor> parse_batch(torch::Tensor buf, torch::Tensor offsets) {
+std::vector<torch::Tensor> parse_batch(torch::Tensor buf, torch::Tensor offsets,
+                                       uint64_t hash_seed) {
     CHECK_DEV(buf); CHECK_CONTIG(buf); CHECK_DEV(offsets); CHECK_CONTIG(offsets);
     TORCH_CHECK(buf.dtype() == torch::kUInt8 && offsets.dtype() == torch::kInt64);
     int32_t M = (int32_t)offsets.size(0) - 1;
@@ -91,8 +92,8 @@ std::vector<torch::Tensor> parse_batch(torch::Tensor buf, torch::Tensor offsets)
                      topics_cnt.data_ptr<int32_t>(),
                      (uint64_t*)recip_hash.data_ptr<int64_t>(),
                      (uint64_t*)timestamp.data_ptr<int64_t>()};
-        launch_k4_parse(buf.data_ptr<uint8_t>(), offsets.data_ptr<int64_t>(), M, out,
-                        cur_stream());
+        launch_k4_parse(buf.data_ptr<uint8_t>(), offsets.data_ptr<int64_t>(), M, hash_seed,
+                        out, cur_stream());
     }
     return {disc, payload_off, payload_len, topics_off, topics_cnt, recip_hash, timestamp};
 }
@@ -331,7 +332,8 @@ void emit_direct(torch::Tensor disc, torch::Tensor owner, torch::Tensor payload_
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-    m.def("parse_batch", &parse_batch, "K4: on-device capnp parse of a message batch");
+    m.def("parse_batch", &parse_batch, "K4: on-device capnp parse of a message batch",
+          py::arg("buf"), py::arg("offsets"), py::arg("hash_seed") = 0);
     m.def("topic_mask", &topic_mask, "K2a: per-message subscriber mask");
     m.def("assign_emit", &assign_emit, "K2b: per-user FIFO ring assignment + pair list");
     m.def("fanout", &fanout, "K3: N-way payload fan-out into egress rings");

@@ -82,10 +82,14 @@ __device__ inline void store_pair(PairRec* p, int32_t u, int32_t m, int64_t d) {
 
 // ---------------------------------------------------------------------------
 // 64-bit FNV-1a — the routing hash for user public keys. Host mirror in
-// csrc/common/hash.h / pushcdn_amd/utils/keyhash.py must match bit-for-bit.
+// pushcdn_amd/utils/keyhash.py must match bit-for-bit.
+// `seed` XORs into the offset basis: brokers derive it from the cluster
+// private key, so an attacker cannot grind out hash collisions offline
+// (FNV alone is byte-invertible). seed=0 == classic FNV-1a.
 // ---------------------------------------------------------------------------
-__host__ __device__ inline uint64_t fnv1a64(const uint8_t* data, uint32_t len) {
-    uint64_t h = 0xcbf29ce484222325ull;
+__host__ __device__ inline uint64_t fnv1a64(const uint8_t* data, uint32_t len,
+                                            uint64_t seed = 0) {
+    uint64_t h = 0xcbf29ce484222325ull ^ seed;
     for (uint32_t i = 0; i < len; ++i) {
         h ^= (uint64_t)data[i];
         h *= 0x100000001b3ull;
@@ -162,7 +166,7 @@ __device__ inline bool read_byte_list(const uint8_t* seg, int64_t nwords, int64_
 extern "C" __global__ void k4_parse_batch(
     const uint8_t* __restrict__ buf,
     const int64_t* __restrict__ offsets,  // [M+1] byte offsets into buf
-    int32_t M, ParseOut out)
+    int32_t M, uint64_t hash_seed, ParseOut out)
 {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= M) return;
@@ -235,7 +239,7 @@ extern "C" __global__ void k4_parse_batch(
         if (ipw < 2) return;
         int64_t ro; int32_t rl;
         if (!read_byte_list(seg, nwords, it + idw, &ro, &rl)) return;
-        o_rhash = fnv1a64(seg + ro, rl);
+        o_rhash = fnv1a64(seg + ro, rl, hash_seed);
         int64_t mo; int32_t ml;
         if (!read_byte_list(seg, nwords, it + idw + 1, &mo, &ml)) return;
         o_poff = ml ? seg_base + mo : 0; o_plen = ml;
@@ -486,11 +490,12 @@ extern "C" __global__ void k2c_apply_subs(
 // ---------------------------------------------------------------------------
 extern "C" {
 
-void launch_k4_parse(const uint8_t* buf, const int64_t* offsets, int32_t M, ParseOut out,
-                     hipStream_t s) {
+void launch_k4_parse(const uint8_t* buf, const int64_t* offsets, int32_t M, uint64_t hash_seed,
+                     ParseOut out, hipStream_t s) {
     if (M <= 0) return;
     int threads = 256, blocks = (M + threads - 1) / threads;
-    hipLaunchKernelGGL(k4_parse_batch, dim3(blocks), dim3(threads), 0, s, buf, offsets, M, out);
+    hipLaunchKernelGGL(k4_parse_batch, dim3(blocks), dim3(threads), 0, s, buf, offsets, M,
+                       hash_seed, out);
 }
 
 void launch_k2a_topic_mask(const uint64_t* sub_bitmap, const uint8_t* buf,

@@ -90,7 +90,8 @@ class BrokerAuth:
         ):
             await _fail(connection, "failed to verify signature")
             return False
-        if int(time.time()) - msg.timestamp > TIMESTAMP_WINDOW_S:
+        # abs(): future timestamps rejected too (reference u64 wrap semantics)
+        if abs(int(time.time()) - msg.timestamp) > TIMESTAMP_WINDOW_S:
             await _fail(connection, "timestamp is too old")
             return False
         await connection.send_message(

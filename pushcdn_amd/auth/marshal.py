@@ -52,7 +52,10 @@ class MarshalAuth:
         if not sig_ok:
             await _fail(connection, "failed to verify signature")
             return None
-        if int(time.time()) - msg.timestamp > TIMESTAMP_WINDOW_S:
+        # abs(): also reject FUTURE timestamps — the reference's unsigned u64
+        # subtraction wraps for ts > now and rejects them; a signed check
+        # alone would let pre-signed future timestamps stay replayable.
+        if abs(int(time.time()) - msg.timestamp) > TIMESTAMP_WINDOW_S:
             await _fail(connection, "timestamp is too old")
             return None
 

@@ -65,6 +65,7 @@ class GpuBrokerEngine:
         pair_capacity: int = 4 << 20,
         nt_fanout: bool = True,
         direct_enabled: bool = True,
+        hash_seed: int = 0,
     ) -> None:
         # pair_capacity bounds the preallocated delivery-pair buffers (the
         # sync-free pipeline drops + counts beyond it); nt_fanout uses
@@ -73,6 +74,11 @@ class GpuBrokerEngine:
         self.pair_capacity = pair_capacity
         self.nt_fanout = nt_fanout
         self.direct_enabled = direct_enabled
+        # hash_seed keys the routing hash (keyhash.derive_routing_seed from
+        # the cluster private key in live services): without it FNV-1a
+        # collisions are offline-grindable and a crafted pubkey could siphon
+        # a victim's Direct messages (K5 matches on the 64-bit hash only).
+        self.hash_seed = hash_seed & ((1 << 64) - 1)
         # fanout_wire=True: fan out the WHOLE serialized wire message (the
         # reference's raw-bytes-forwarded-verbatim invariant);
         # False: fan out only the payload field (kernel golden tests).
@@ -101,6 +107,7 @@ class GpuBrokerEngine:
         self.direct_keys = torch.zeros(direct_table_size, dtype=torch.int64, device=dev)
         self.direct_vals = torch.zeros(direct_table_size, dtype=torch.int32, device=dev)
         self._direct_entries: Dict[int, int] = {}
+        self._direct_pubkeys: Dict[int, bytes] = {}  # hash -> full key (collision refusal)
         self.seq = 0
         self.total = TickStats()
         if self.use_gpu_ops:
@@ -137,17 +144,40 @@ class GpuBrokerEngine:
         for t in topics:
             self.sub_bitmap[t & 0xFF] = full.to(self.device)
 
+    def _direct_hash(self, pubkey: bytes) -> int:
+        """Routing hash of a pubkey, with collision refusal: if a DIFFERENT
+        pubkey already registered the same 64-bit hash, registering this one
+        would silently misroute one user's Direct traffic to the other —
+        refuse instead (accidental probability 2^-64/pair; adversarial
+        construction requires the secret seed)."""
+        h = fnv1a64(pubkey, self.hash_seed)
+        prev = self._direct_pubkeys.get(h)
+        if prev is not None and prev != pubkey:
+            raise ValueError("direct routing-hash collision; refusing registration")
+        return h
+
     def register_direct(self, pubkey: bytes, owner: int) -> None:
         """owner >= 0: local user index. owner < 0: -(broker_rank+2)."""
-        h = fnv1a64(pubkey)
+        h = self._direct_hash(pubkey)
+        self._direct_pubkeys[h] = pubkey
         self._direct_entries[h] = owner
         self._rebuild_direct_table()
 
     def register_direct_bulk(self, entries) -> None:
         """Register many (pubkey, owner) pairs with one table rebuild."""
         for pubkey, owner in entries:
-            self._direct_entries[fnv1a64(pubkey)] = owner
+            h = self._direct_hash(pubkey)
+            self._direct_pubkeys[h] = pubkey
+            self._direct_entries[h] = owner
         self._rebuild_direct_table()
+
+    def unregister_direct(self, pubkey: bytes) -> None:
+        """Remove a departed user's entry so a Direct to their key is dropped
+        instead of delivered to whoever reuses the slot."""
+        h = fnv1a64(pubkey, self.hash_seed)
+        if self._direct_entries.pop(h, None) is not None:
+            self._direct_pubkeys.pop(h, None)
+            self._rebuild_direct_table()
 
     def subscribe_modulo(self, n_topics: int) -> None:
         """Bulk: user u subscribes to topic (u % n_topics) — the mixed-bench
@@ -238,7 +268,7 @@ class GpuBrokerEngine:
         ops = self._ops
         M = offsets.shape[0] - 1
         disc, payload_off, payload_len, topics_off, topics_cnt, recip_hash, _ts = ops.parse_batch(
-            buf, offsets
+            buf, offsets, self.hash_seed
         )
         mask_t = ops.topic_mask_t(self.sub_bitmap, buf, topics_off, topics_cnt, disc)
         if self.fanout_wire:
@@ -286,7 +316,9 @@ class GpuBrokerEngine:
         fan-out -> device seq bump.  All tensors fixed-address."""
         ops = self._ops
         M = offsets.shape[0] - 1
-        disc, _po, _pl, topics_off, topics_cnt, _rh, _ts = ops.parse_batch(buf, offsets)
+        disc, _po, _pl, topics_off, topics_cnt, _rh, _ts = ops.parse_batch(
+            buf, offsets, self.hash_seed
+        )
         mask_t = ops.topic_mask_t(self.sub_bitmap, buf, topics_off, topics_cnt, disc)
         payload_off = offsets[:-1].contiguous()
         payload_len = (offsets[1:] - offsets[:-1]).to(torch.int32).contiguous()
@@ -331,7 +363,7 @@ class GpuBrokerEngine:
     def _tick_cpu(self, batch: bytes, offsets: List[int]) -> TickStats:
         from ..ops import reference as ref
 
-        pr = ref.parse_batch(batch, offsets)
+        pr = ref.parse_batch(batch, offsets, self.hash_seed)
         M = len(offsets) - 1
         mask = ref.topic_mask(self.sub_bitmap, batch, pr.topics_off, pr.topics_cnt, pr.disc)
         if self.fanout_wire:

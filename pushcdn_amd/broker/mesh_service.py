@@ -133,7 +133,9 @@ class MeshBroker(Broker):
                     for t in raw[r["topics_off"] : r["topics_off"] + r["topics_cnt"]]:
                         batch_topics |= 1 << t
                 elif r["disc"] == 3:
-                    direct_bits |= 1 << (fnv1a64(r["recipient"]) & 63)
+                    # same keyed hash as the engine's K5 table (seed is
+                    # cluster-shared, so digests agree across brokers)
+                    direct_bits |= 1 << (fnv1a64(r["recipient"], self._engine.hash_seed) & 63)
             exchanged = self.mesh.exchange_interest(
                 send_buf, n_local, used_bytes, batch_topics, interests,
                 direct_bits=direct_bits, owned_bits=owned_bits,
@@ -194,7 +196,7 @@ class MeshBroker(Broker):
                 # 64b digest of the direct users owned (connected) here —
                 # the mesh-plane analog of the reference's DirectMap
                 for pubkey in self._gpu_user_by_slot.values():
-                    owned_bits |= 1 << (fnv1a64(pubkey) & 63)
+                    owned_bits |= 1 << (fnv1a64(pubkey, self._engine.hash_seed) & 63)
             wpos = await asyncio.get_running_loop().run_in_executor(
                 self._mesh_executor, self._blocking_mesh_tick,
                 msgs, dev_buf, interests, owned_bits)

@@ -129,11 +129,16 @@ class Broker:
         if config.data_plane == "gpu":
             from .gpu_engine import GpuBrokerEngine
 
+            from ..utils.keyhash import derive_routing_seed
+
             self._engine = GpuBrokerEngine(
                 device=config.gpu_device,
                 n_users=config.gpu_max_users,
                 ring_bytes=config.gpu_ring_bytes,
                 fanout_wire=True,  # forward raw wire bytes verbatim
+                # keyed routing hash: cluster-wide secret derived from the
+                # shared broker private key (see keyhash.derive_routing_seed)
+                hash_seed=derive_routing_seed(self.keypair.private_key),
             )
             self._gpu_queue = asyncio.Queue()
             self._free_gpu_slots = list(range(config.gpu_max_users - 1, -1, -1))
@@ -277,9 +282,18 @@ class Broker:
             if self._engine is not None and old.gpu_index is not None:
                 self._release_gpu_slot(old.gpu_index)
         if self._engine is not None:
-            handle.gpu_index = self._claim_gpu_slot(pubkey)
-            self._engine.subscribe(handle.gpu_index, topics)
-            self._engine.register_direct(pubkey, handle.gpu_index)
+            # A capacity or routing-hash-collision refusal must clean up the
+            # just-registered user — round 1 raised here AFTER add_user,
+            # leaving a registered user with no receive loop and no
+            # eviction (reference eviction semantics connections/mod.rs:278-304).
+            try:
+                handle.gpu_index = self._claim_gpu_slot(pubkey)
+                self._engine.subscribe(handle.gpu_index, topics)
+                self._engine.register_direct(pubkey, handle.gpu_index)
+            except (RuntimeError, ValueError) as exc:
+                log.warning("refusing user %s: %s", ident(pubkey), exc)
+                await self.remove_user(pubkey)
+                return
         handle.task = asyncio.get_running_loop().create_task(
             self._user_receive_loop(pubkey, handle)
         )
@@ -545,7 +559,11 @@ class Broker:
         return slot
 
     def _release_gpu_slot(self, slot: int) -> None:
-        self._gpu_user_by_slot.pop(slot, None)
+        pubkey = self._gpu_user_by_slot.pop(slot, None)
+        if pubkey is not None:
+            # drop the direct-table entry: a Direct to the departed key must
+            # be dropped, not delivered to the slot's next occupant
+            self._engine.unregister_direct(pubkey)
         # clear all subscriptions for the slot
         self._engine.unsubscribe(slot, list(range(256)))
         self._free_gpu_slots.append(slot)

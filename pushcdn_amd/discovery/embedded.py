@@ -12,7 +12,7 @@ reference's integration tests fake a cluster (tests/src/tests/mod.rs:62-143).
 from __future__ import annotations
 
 import asyncio
-import random
+import secrets
 import sqlite3
 import time
 from typing import List, Optional, Set
@@ -99,8 +99,11 @@ class EmbeddedDiscovery(DiscoveryClient):
         async with self._lock:
             self._prune()
             for _ in range(16):
-                # permit > 1: 0 = failed, 1 = success-flag (reference message.rs:338-345)
-                permit = random.randrange(2, 2**63)
+                # permit > 1: 0 = failed, 1 = success-flag (reference message.rs:338-345).
+                # CSPRNG (secrets): permits are a credential; Mersenne Twister
+                # output observable via a client's own permits must not let it
+                # predict others' (reference StdRng::from_entropy()).
+                permit = secrets.randbelow(2**63 - 2) + 2
                 try:
                     self._conn.execute(
                         "INSERT INTO permits(permit, broker, user_pubkey, expiry) "
@@ -119,20 +122,34 @@ class EmbeddedDiscovery(DiscoveryClient):
         """One-time validation (GETDEL): returns the user pubkey or None."""
         async with self._lock:
             self._prune()
-            row = self._conn.execute(
-                "SELECT user_pubkey, broker FROM permits WHERE permit = ?", (permit,)
-            ).fetchone()
-            if row is None:
-                self._conn.commit()
-                return None
-            self._conn.execute("DELETE FROM permits WHERE permit = ?", (permit,))
+            # Permits are broker-bound unless global_permits is on (reference
+            # redis.rs:219-265).  The broker binding is part of the SELECT and
+            # DELETE predicates so that a permit presented to the WRONG broker
+            # is rejected without being consumed — it stays redeemable at the
+            # broker it was issued for, matching the redis backend's
+            # broker-scoped key.
+            if self.global_permits:
+                row = self._conn.execute(
+                    "SELECT user_pubkey FROM permits WHERE permit = ?", (permit,)
+                ).fetchone()
+                if row is None:
+                    self._conn.commit()
+                    return None
+                self._conn.execute("DELETE FROM permits WHERE permit = ?", (permit,))
+            else:
+                row = self._conn.execute(
+                    "SELECT user_pubkey FROM permits WHERE permit = ? AND broker = ?",
+                    (permit, str(broker)),
+                ).fetchone()
+                if row is None:
+                    self._conn.commit()
+                    return None
+                self._conn.execute(
+                    "DELETE FROM permits WHERE permit = ? AND broker = ?",
+                    (permit, str(broker)),
+                )
             self._conn.commit()
-        pubkey, issued_broker = row
-        # permits are broker-bound unless global_permits is on
-        # (reference redis.rs:219-265)
-        if not self.global_permits and issued_broker != str(broker):
-            return None
-        return bytes(pubkey)
+        return bytes(row[0])
 
     async def set_whitelist(self, users: List[bytes]) -> None:
         async with self._lock:

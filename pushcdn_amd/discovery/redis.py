@@ -12,7 +12,7 @@ set (redis.rs:271-326; empty set = allow all).
 from __future__ import annotations
 
 import asyncio
-import random
+import secrets
 from typing import List, Optional, Set
 from urllib.parse import urlparse
 
@@ -122,7 +122,10 @@ class RedisDiscovery(DiscoveryClient):
     async def issue_permit(
         self, broker: BrokerIdentifier, expiry_s: float, user_pubkey: bytes
     ) -> int:
-        permit = random.randrange(2, 2**63)
+        # CSPRNG: permits are the sole broker-side credential (the reference
+        # uses StdRng::from_entropy(), redis.rs:207-214) — a predictable RNG
+        # would let one client forecast other users' permits.
+        permit = secrets.randbelow(2**63 - 2) + 2
         scope = "any" if self.global_permits else str(broker)
         await self._r.cmd("SET", f"permit:{scope}:{permit}", user_pubkey,
                           "EX", int(max(1, expiry_s)))

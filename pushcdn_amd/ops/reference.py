@@ -36,7 +36,7 @@ def _i64(x: int) -> int:
     return x - (1 << 64) if x >= (1 << 63) else x
 
 
-def parse_batch(buf: bytes, offsets: List[int]) -> ParseResult:
+def parse_batch(buf: bytes, offsets: List[int], hash_seed: int = 0) -> ParseResult:
     """Mirror of k4_parse_batch. `buf` is concatenated serialized Messages."""
     from ..proto import message as msglib
 
@@ -63,7 +63,7 @@ def parse_batch(buf: bytes, offsets: List[int]) -> ParseResult:
         topics_cnt[i] = r["topics_cnt"]
         timestamp[i] = _i64(r["timestamp"])
         if r["disc"] == 3:
-            recip_hash[i] = _i64(fnv1a64(r["recipient"]))
+            recip_hash[i] = _i64(fnv1a64(r["recipient"], hash_seed))
     return ParseResult(disc, payload_off, payload_len, topics_off, topics_cnt, recip_hash, timestamp)
 
 
