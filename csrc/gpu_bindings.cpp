@@ -36,6 +36,9 @@ void launch_k2c_apply_subs(uint64_t*, const uint8_t*, const int64_t*, const int3
 void launch_k1_bls_verify(const uint8_t*, const uint8_t*, uint8_t*, const int64_t*, int32_t,
                           int32_t*, hipStream_t);
 void launch_k1_hash_to_g1(uint8_t*, const int64_t*, int32_t, uint8_t*, hipStream_t);
+void launch_k1_precompute_g2_lines(uint8_t*, int32_t*, hipStream_t);
+void launch_k1_bls_verify2(const uint8_t*, const uint8_t*, uint8_t*, const int64_t*,
+                           const uint8_t*, int32_t, int32_t*, hipStream_t);
 void launch_k3_fanout_wave(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
                            const uint32_t*, const int32_t*, uint8_t*, int, int, hipStream_t);
 void launch_k3_fanout_flat2(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
@@ -203,6 +206,36 @@ torch::Tensor bls_verify_batch(torch::Tensor vks, torch::Tensor sigs, torch::Ten
     return ok;
 }
 
+torch::Tensor precompute_g2_lines(torch::Tensor device_probe) {
+    // fixed-g2 Miller-loop line coefficients (128 records x 192 B); computed
+    // once per process by a 1-thread kernel, consumed by bls_verify_batch2
+    auto out = torch::zeros({128 * 192},
+                            torch::TensorOptions().dtype(torch::kUInt8)
+                                .device(device_probe.device()));
+    auto n = torch::zeros({1}, torch::TensorOptions().dtype(torch::kInt32)
+                                   .device(device_probe.device()));
+    launch_k1_precompute_g2_lines(out.data_ptr<uint8_t>(), n.data_ptr<int32_t>(),
+                                  cur_stream());
+    return out;
+}
+
+torch::Tensor bls_verify_batch2(torch::Tensor vks, torch::Tensor sigs, torch::Tensor msgs,
+                                torch::Tensor moff, torch::Tensor g2_lines) {
+    CHECK_DEV(vks); CHECK_DEV(sigs); CHECK_DEV(msgs); CHECK_DEV(moff); CHECK_DEV(g2_lines);
+    CHECK_CONTIG(vks); CHECK_CONTIG(sigs); CHECK_CONTIG(msgs); CHECK_CONTIG(moff);
+    CHECK_CONTIG(g2_lines);
+    int32_t N = (int32_t)moff.size(0) - 1;
+    TORCH_CHECK(vks.numel() == (int64_t)N * 128 && sigs.numel() == (int64_t)N * 64);
+    auto ok = torch::zeros({N}, torch::TensorOptions().dtype(torch::kInt32).device(vks.device()));
+    if (N > 0) {
+        launch_k1_bls_verify2(vks.data_ptr<uint8_t>(), sigs.data_ptr<uint8_t>(),
+                              msgs.data_ptr<uint8_t>(), moff.data_ptr<int64_t>(),
+                              g2_lines.data_ptr<uint8_t>(), N, ok.data_ptr<int32_t>(),
+                              cur_stream());
+    }
+    return ok;
+}
+
 torch::Tensor hash_to_g1_batch(torch::Tensor msgs, torch::Tensor moff) {
     CHECK_DEV(msgs); CHECK_DEV(moff);
     int32_t N = (int32_t)moff.size(0) - 1;
@@ -340,6 +373,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("direct_lookup", &direct_lookup, "K5: batched direct-route hash probe");
     m.def("apply_subs", &apply_subs, "K2c: apply subscribe/unsubscribe batch to bitmap");
     m.def("bls_verify_batch", &bls_verify_batch, "K1: batched BLS-over-BN254 verification");
+    m.def("precompute_g2_lines", &precompute_g2_lines,
+          "fixed-g2 Miller line coefficients for K1 v2 (once per process)");
+    m.def("bls_verify_batch2", &bls_verify_batch2,
+          "K1 v2: 2-lane Fp2-decomposed batched BLS verification");
     m.def("hash_to_g1_batch", &hash_to_g1_batch, "K1 helper: batched hash-to-G1");
     m.def("fanout_wave", &fanout_wave, "K3v2: wave-per-pair fan-out (nt flag, device count)");
     m.def("fanout_flat2", &fanout_flat2, "K3v4: flat fan-out, seq from base, capacity clamp");

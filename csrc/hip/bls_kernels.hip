@@ -18,6 +18,7 @@
 #include <stdint.h>
 
 #include "../bls/bls.h"
+#include "bn254_pair2.h"
 
 using namespace bn254;
 
@@ -45,6 +46,92 @@ k1_bls_verify(
     uint8_t* scratch = msgs + moff[i];
     uint32_t msg_len = (uint32_t)(moff[i + 1] - moff[i] - 1);  // spare byte excluded
     ok[i] = bls::verify_core(vk, scratch, msg_len, sx, sy) ? 1 : 0;
+}
+
+// ---------------------------------------------------------------------------
+// K1 v2: 2-lane Fp2-decomposed verification (see bn254_pair2.h header
+// comment) — lane pair (2i, 2i+1) verifies signature i; the fixed-g2 side
+// of the pairing product reads precomputed Miller-loop line coefficients.
+// ---------------------------------------------------------------------------
+
+// Runs the single-lane Miller-loop R-evolution ONCE for Q = g2 generator and
+// serializes every line's raw Montgomery limbs in NAF-walk order (the exact
+// order miller_loop2_lines replays). One thread; launched once per process.
+extern "C" __global__ void k1_precompute_g2_lines(uint8_t* __restrict__ out,
+                                                  int32_t* __restrict__ n_out) {
+    if (blockIdx.x != 0 || threadIdx.x != 0) return;
+    Fp two_inv = Fp::from_u64(2).inv();
+    G2 gen = g2_generator();
+    G2Affine q{gen.X, gen.Y};
+    G2Affine negq{q.x, Fp2::neg(q.y)};
+    G2Proj r{q.x, q.y, Fp2::one()};
+    int cur = 0;
+    auto emit = [&](const LineCoeffs& l) {
+        u64* rec = (u64*)(out + (size_t)cur * bn254p2::LINE_REC_BYTES);
+        for (int i = 0; i < 4; ++i) rec[i] = l.c0.c0.n.v[i];
+        for (int i = 0; i < 4; ++i) rec[4 + i] = l.c0.c1.n.v[i];
+        for (int i = 0; i < 4; ++i) rec[8 + i] = l.c1.c0.n.v[i];
+        for (int i = 0; i < 4; ++i) rec[12 + i] = l.c1.c1.n.v[i];
+        for (int i = 0; i < 4; ++i) rec[16 + i] = l.c2.c0.n.v[i];
+        for (int i = 0; i < 4; ++i) rec[20 + i] = l.c2.c1.n.v[i];
+        ++cur;
+    };
+    BN_NOUNROLL for (int i = bn254c::ATE_NAF_LEN - 2; i >= 0; --i) {
+        emit(doubling_step(r, two_inv));
+        int8_t d = bn254c::ATE_NAF[i];
+        if (d == 1) emit(addition_step(r, q));
+        else if (d == -1) emit(addition_step(r, negq));
+    }
+    G2Affine q1 = g2_frobenius(q);
+    G2Affine q2 = g2_frobenius(q1);
+    q2.y = Fp2::neg(q2.y);
+    emit(addition_step(r, q1));
+    emit(addition_step(r, q2));
+    *n_out = cur;
+}
+
+extern "C" __global__ void __launch_bounds__(64, 1)
+k1_bls_verify2(
+    const uint8_t* __restrict__ vks,      // [N][128]
+    const uint8_t* __restrict__ sigs,     // [N][64]
+    uint8_t* __restrict__ msgs,           // namespaced messages + 1 spare byte each
+    const int64_t* __restrict__ moff,     // [N+1]
+    const uint8_t* __restrict__ g2_lines, // precomputed fixed-g2 line coeffs
+    int32_t N,
+    int32_t* __restrict__ ok)
+{
+    int lane = blockIdx.x * blockDim.x + threadIdx.x;
+    int v = lane >> 1;  // verification index: one PAIR of lanes per signature
+    if (v >= N) return;
+    bn254p2::PL L = bn254p2::PL::self();
+
+    // verkey: this lane loads its Fp2 components; checks combined pairwise
+    Fp vkx, vky;
+    if (!bn254p2::verkey_load2(L, vks + (size_t)v * 128, vkx, vky)) {
+        if (!L.hi) ok[v] = 0;
+        return;
+    }
+    // signature (G1): plain Fp, duplicated across the pair (lockstep-free)
+    Fp sx, sy;
+    if (!bls::sig_deserialize(sigs + (size_t)v * 64, sx, sy)) {
+        if (!L.hi) ok[v] = 0;
+        return;
+    }
+    // hash-to-G1: identical data on both lanes of the pair
+    uint8_t* scratch = msgs + moff[v];
+    uint32_t msg_len = (uint32_t)(moff[v + 1] - moff[v] - 1);
+    Fp hx, hy;
+    if (!bls::hash_to_g1_with_scratch(scratch, msg_len, hx, hy)) {
+        if (!L.hi) ok[v] = 0;
+        return;
+    }
+    // e(H, pk) * e(-sig, g2) == 1, one shared final exponentiation;
+    // the fixed-g2 loop replays precomputed lines (no G2 arithmetic)
+    bn254p2::F12 ml1 = bn254p2::miller_loop2(L, hx, hy, {vkx, vky});
+    bn254p2::F12 ml2 = bn254p2::miller_loop2_lines(L, sx, Fp::neg(sy), g2_lines);
+    bn254p2::F12 f = L.f12mul(ml1, ml2);
+    bool good = L.f12is_one(bn254p2::final_exponentiation2(L, f));
+    if (!L.hi) ok[v] = good ? 1 : 0;
 }
 
 // Device self-test: hash_to_g1 + sign-shaped scalar mul, for golden tests.
@@ -80,6 +167,20 @@ void launch_k1_hash_to_g1(uint8_t* msgs, const int64_t* moff, int32_t N, uint8_t
     int threads = 64;
     int blocks = (N + threads - 1) / threads;
     hipLaunchKernelGGL(k1_hash_to_g1, dim3(blocks), dim3(threads), 0, s, msgs, moff, N, out);
+}
+
+void launch_k1_precompute_g2_lines(uint8_t* out, int32_t* n_out, hipStream_t s) {
+    hipLaunchKernelGGL(k1_precompute_g2_lines, dim3(1), dim3(1), 0, s, out, n_out);
+}
+
+void launch_k1_bls_verify2(const uint8_t* vks, const uint8_t* sigs, uint8_t* msgs,
+                           const int64_t* moff, const uint8_t* g2_lines, int32_t N,
+                           int32_t* ok, hipStream_t s) {
+    int threads = 64;                       // 32 verifications per wave (2 lanes each)
+    int lanes = 2 * N;
+    int blocks = (lanes + threads - 1) / threads;
+    hipLaunchKernelGGL(k1_bls_verify2, dim3(blocks), dim3(threads), 0, s, vks, sigs, msgs,
+                       moff, g2_lines, N, ok);
 }
 
 }  // extern "C"

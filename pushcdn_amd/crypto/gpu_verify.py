@@ -91,6 +91,16 @@ class GpuBatchVerifier:
                 if not fut.done():
                     fut.set_result(bool(ok))
 
+    def _g2_lines(self):
+        """Precomputed fixed-g2 Miller-loop lines for the v2 kernel (lazy,
+        once per verifier)."""
+        lines = getattr(self, "_g2_lines_t", None)
+        if lines is None:
+            probe = self._torch.zeros(1, dtype=self._torch.uint8, device=self.device)
+            lines = self._ops.precompute_g2_lines(probe)
+            self._g2_lines_t = lines
+        return lines
+
     def _verify_batch_gpu(self, batch) -> List[bool]:
         torch = self._torch
         vks = bytearray()
@@ -107,10 +117,13 @@ class GpuBatchVerifier:
             msgs += ns.encode() + msg + b"\x00"  # spare counter byte
             offsets.append(len(msgs))
         dev = self.device
-        ok = self._ops.bls_verify_batch(
+        # v2: 2-lane Fp2-decomposed kernel (bn254_pair2.h) — ~2 lanes per
+        # verification, half the per-lane live state, fixed-g2 lines
+        ok = self._ops.bls_verify_batch2(
             torch.frombuffer(vks, dtype=torch.uint8).to(dev),
             torch.frombuffer(sigs, dtype=torch.uint8).to(dev),
             torch.frombuffer(msgs, dtype=torch.uint8).to(dev),
             torch.tensor(offsets, dtype=torch.int64, device=dev),
+            self._g2_lines(),
         )
         return [bool(x) for x in ok.cpu().tolist()]

@@ -244,6 +244,57 @@ def test_k1_bls_verify_batch_matches_host(ops):
     assert ok.cpu().tolist() == want
 
 
+def test_k1v2_bls_verify_matches_host(ops):
+    """K1 v2 (2-lane Fp2 decomposition, bn254_pair2.h) golden vs host BLS:
+    valid sigs, corrupted sigs, wrong namespace, malformed verkeys, and an
+    on-curve-but-out-of-subgroup verkey must all match the host verdicts."""
+    from pushcdn_amd.crypto import bls
+    from pushcdn_amd.ops.build import build_core
+    from tests.test_round2_fixes import _find_cofactor_point
+
+    core = build_core()
+    ns = bls.USER_MARSHAL_NAMESPACE
+    N = 65  # odd count exercises the tail pair
+    vks, sigs, msgs, offsets, want = [], [], bytearray(), [0], []
+    cx, cy = _find_cofactor_point(7)
+    cof_vk = b"".join(c.to_bytes(32, "little") for c in (cx[0], cx[1], cy[0], cy[1]))
+    for i in range(N):
+        kp = bls.KeyPair.from_seed(1000 + i)
+        msg = f"ts2-{i}".encode()
+        use_ns = ns if i % 7 else "wrong-ns"
+        sig = bls.sign(kp.private_key, use_ns, msg)
+        vk = kp.public_key
+        if i % 3 == 0:
+            sig = bytearray(sig)
+            sig[2] ^= 0x20
+            sig = bytes(sig)
+        if i % 11 == 5:
+            vk = cof_vk  # on-curve, outside the r-order subgroup
+        if i % 13 == 6:
+            vk = b"\xff" * 128  # coordinates >= p: malformed
+        vks.append(vk)
+        sigs.append(sig)
+        msgs += _namespaced(ns, msg)
+        offsets.append(len(msgs))
+        want.append(1 if core.verify(vk, ns, msg, sig) else 0)
+    assert 1 in want and 0 in want
+
+    vks_t = torch.frombuffer(bytearray(b"".join(vks)), dtype=torch.uint8).to("cuda")
+    sigs_t = torch.frombuffer(bytearray(b"".join(sigs)), dtype=torch.uint8).to("cuda")
+    msgs_t = torch.frombuffer(bytearray(msgs), dtype=torch.uint8).to("cuda")
+    moff_t = torch.tensor(offsets, dtype=torch.int64, device="cuda")
+    probe = torch.zeros(1, dtype=torch.uint8, device="cuda")
+    lines = ops.precompute_g2_lines(probe)
+    ok = ops.bls_verify_batch2(vks_t, sigs_t, msgs_t, moff_t, lines)
+    torch.cuda.synchronize()
+    assert ok.cpu().tolist() == want
+
+    # v1 and v2 agree on the same batch
+    ok1 = ops.bls_verify_batch(vks_t, sigs_t, msgs_t, moff_t)
+    torch.cuda.synchronize()
+    assert ok1.cpu().tolist() == want
+
+
 def test_k1_hash_to_g1_matches_host(ops):
     from pushcdn_amd.crypto import bls
     from pushcdn_amd.ops.build import build_core
