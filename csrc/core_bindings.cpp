@@ -381,6 +381,18 @@ PYBIND11_MODULE(pushcdn_core, m) {
             auto r = p.send_ring(id, (const uint8_t*)info.ptr, wpos);
             return py::make_tuple(r.first, r.second);
         })
+        .def("send_rings_batch",
+             [](net::Pump& p, py::buffer base, const std::vector<int64_t>& ids,
+                const std::vector<int64_t>& starts, const std::vector<int64_t>& ends) {
+                 py::buffer_info info = base.request();
+                 if (ids.size() != starts.size() || ids.size() != ends.size())
+                     throw std::runtime_error("send_rings_batch: length mismatch");
+                 for (size_t j = 0; j < ids.size(); ++j)
+                     if (starts[j] < 0 || ends[j] < starts[j] || ends[j] > info.size)
+                         throw std::runtime_error("send_rings_batch: range beyond buffer");
+                 return p.send_rings_batch((const uint8_t*)info.ptr, ids, starts, ends);
+             },
+             "batched tick drain: one call for all users' compacted rings")
         .def("send_backlog", &net::Pump::send_backlog)
         .def("poll_dirty", &net::Pump::poll_dirty)
         .def("recv_batch", [](net::Pump& p, int64_t id, size_t maxf) {

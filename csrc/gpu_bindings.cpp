@@ -37,6 +37,8 @@ void launch_k1_bls_verify(const uint8_t*, const uint8_t*, uint8_t*, const int64_
                           int32_t*, hipStream_t);
 void launch_k1_hash_to_g1(uint8_t*, const int64_t*, int32_t, uint8_t*, hipStream_t);
 void launch_k1_precompute_g2_lines(uint8_t*, int32_t*, hipStream_t);
+void launch_k7_compact_rings(const uint8_t*, int64_t, const int64_t*, const int64_t*,
+                             uint8_t*, int32_t, int32_t, hipStream_t);
 void launch_k1_bls_verify2(const uint8_t*, const uint8_t*, uint8_t*, const int64_t*,
                            const uint8_t*, int32_t, int32_t*, hipStream_t);
 void launch_k3_fanout_wave(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
@@ -206,6 +208,18 @@ torch::Tensor bls_verify_batch(torch::Tensor vks, torch::Tensor sigs, torch::Ten
     return ok;
 }
 
+void compact_rings(torch::Tensor egress, int64_t ring_bytes, torch::Tensor wpos,
+                   torch::Tensor dst_off, torch::Tensor staging, int64_t max_chunks) {
+    CHECK_DEV(egress); CHECK_DEV(wpos); CHECK_DEV(dst_off); CHECK_DEV(staging);
+    CHECK_CONTIG(egress); CHECK_CONTIG(wpos); CHECK_CONTIG(dst_off); CHECK_CONTIG(staging);
+    TORCH_CHECK(wpos.dtype() == torch::kInt64 && dst_off.dtype() == torch::kInt64);
+    int32_t n = (int32_t)wpos.size(0);
+    launch_k7_compact_rings(egress.data_ptr<uint8_t>(), ring_bytes,
+                            wpos.data_ptr<int64_t>(), dst_off.data_ptr<int64_t>(),
+                            staging.data_ptr<uint8_t>(), n, (int32_t)max_chunks,
+                            cur_stream());
+}
+
 torch::Tensor precompute_g2_lines(torch::Tensor device_probe) {
     // fixed-g2 Miller-loop line coefficients (128 records x 192 B); computed
     // once per process by a 1-thread kernel, consumed by bls_verify_batch2
@@ -373,6 +387,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("direct_lookup", &direct_lookup, "K5: batched direct-route hash probe");
     m.def("apply_subs", &apply_subs, "K2c: apply subscribe/unsubscribe batch to bitmap");
     m.def("bls_verify_batch", &bls_verify_batch, "K1: batched BLS-over-BN254 verification");
+    m.def("compact_rings", &compact_rings,
+          "K7: gather used egress-ring prefixes into one staging buffer");
     m.def("precompute_g2_lines", &precompute_g2_lines,
           "fixed-g2 Miller line coefficients for K1 v2 (once per process)");
     m.def("bls_verify_batch2", &bls_verify_batch2,

@@ -66,12 +66,23 @@ struct PL {
 
     __device__ static PL self() { return PL{(threadIdx.x & 1) != 0}; }
 
+    // exchange a 32-bit value with the partner lane: DPP quad_perm(1,0,3,2)
+    // is a single full-rate VALU op vs ds_bpermute (LDS pipeline) for
+    // __shfl_xor — the exchange sits on f2mul's critical path.
+    __device__ static uint32_t xchg32(uint32_t v) {
+        return (uint32_t)__builtin_amdgcn_mov_dpp((int)v, 0xB1 /*quad_perm 1,0,3,2*/,
+                                                  0xF, 0xF, true);
+    }
+
     // exchange an Fp with the partner lane
     __device__ static Fp xchg(const Fp& a) {
         Fp r;
 #pragma unroll
-        for (int i = 0; i < 4; ++i)
-            r.n.v[i] = (u64)__shfl_xor((unsigned long long)a.n.v[i], 1, 64);
+        for (int i = 0; i < 4; ++i) {
+            uint32_t lo = xchg32((uint32_t)a.n.v[i]);
+            uint32_t hi32 = xchg32((uint32_t)(a.n.v[i] >> 32));
+            r.n.v[i] = ((u64)hi32 << 32) | lo;
+        }
         return r;
     }
 
@@ -81,12 +92,12 @@ struct PL {
 
     __device__ bool f2is_zero(const Fp& a) const {
         unsigned z = a.is_zero() ? 1u : 0u;
-        return (z & (unsigned)__shfl_xor(z, 1, 64)) != 0;
+        return (z & xchg32(z)) != 0;
     }
 
     __device__ bool f2eq(const Fp& a, const Fp& b) const {
         unsigned e = (a == b) ? 1u : 0u;
-        return (e & (unsigned)__shfl_xor(e, 1, 64)) != 0;
+        return (e & xchg32(e)) != 0;
     }
 
     // Karatsuba across the pair: lane0 computes v0=a0b0, lane1 v1=a1b1,
@@ -202,7 +213,7 @@ struct PL {
                       a.c1.c0.is_zero() & a.c1.c1.is_zero() & a.c1.c2.is_zero())
                          ? 1u
                          : 0u;
-        return (e & (unsigned)__shfl_xor(e, 1, 64)) != 0;
+        return (e & xchg32(e)) != 0;
     }
 
     __device__ F12 f12mul(const F12& a, const F12& b) const {
@@ -546,7 +557,7 @@ __device__ inline bool verkey_load2(const PL& L, const uint8_t* in, Fp& x, Fp& y
     U256 xv = bls::u256_from_le(in + (L.hi ? 32 : 0));
     U256 yv = bls::u256_from_le(in + 64 + (L.hi ? 32 : 0));
     unsigned ok = (!u256_gte(xv, p) && !u256_gte(yv, p)) ? 1u : 0u;
-    if ((ok & (unsigned)__shfl_xor(ok, 1, 64)) == 0) return false;
+    if ((ok & PL::xchg32(ok)) == 0) return false;
     x = Fp::from_u256(xv);
     y = Fp::from_u256(yv);
     if (L.f2is_zero(x) && L.f2is_zero(y)) return false;  // infinity

@@ -1082,3 +1082,45 @@ extern "C" void launch_k5b_emit_direct(
                        payload_off, payload_len, M, ring_bytes, capacity, ring_wpos, n_pairs,
                        pairs, drops);
 }
+
+// ---------------------------------------------------------------------------
+// K7: egress-ring compaction. Gathers every ring's used prefix into one
+// contiguous staging buffer so the per-tick drain is ONE D2H copy + ONE
+// C++ pump call instead of one of each per user (the socket-path
+// bottleneck: reference Arc-clone forwarding has no per-recipient copies
+// either, sender.rs:16-33).  wpos/dst_off are device tensors; records are
+// 16-aligned so 16 B vector copies cover exactly the used bytes.
+//   grid.x = ring (user), grid.y = 64 KiB chunk of that ring
+// ---------------------------------------------------------------------------
+#define K7_CHUNK (64 * 1024)
+
+extern "C" __global__ void k7_compact_rings(
+    const uint8_t* __restrict__ egress,
+    int64_t ring_bytes,
+    const int64_t* __restrict__ wpos,     // [N] used bytes per ring
+    const int64_t* __restrict__ dst_off,  // [N] exclusive scan of wpos
+    uint8_t* __restrict__ staging)
+{
+    int u = blockIdx.x;
+    int64_t used = wpos[u];
+    int64_t chunk = (int64_t)blockIdx.y * K7_CHUNK;
+    if (chunk >= used) return;
+    int64_t end = used < chunk + K7_CHUNK ? used : chunk + K7_CHUNK;
+    const uint8_t* src = egress + (int64_t)u * ring_bytes;
+    uint8_t* dst = staging + dst_off[u];
+    for (int64_t pos = chunk + (int64_t)threadIdx.x * 16; pos < end;
+         pos += (int64_t)blockDim.x * 16) {
+        cdn_v4u v;
+        memcpy(&v, src + pos, 16);
+        *(cdn_v4u*)(dst + pos) = v;
+    }
+}
+
+extern "C" void launch_k7_compact_rings(const uint8_t* egress, int64_t ring_bytes,
+                                        const int64_t* wpos, const int64_t* dst_off,
+                                        uint8_t* staging, int32_t n_rings,
+                                        int32_t max_chunks, hipStream_t s) {
+    if (n_rings <= 0 || max_chunks <= 0) return;
+    hipLaunchKernelGGL(k7_compact_rings, dim3(n_rings, max_chunks), dim3(256), 0, s,
+                       egress, ring_bytes, wpos, dst_off, staging);
+}

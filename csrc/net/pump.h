@@ -135,40 +135,40 @@ public:
         auto it = conns_.find(id);
         if (it == conns_.end() || it->second.closed || it->second.soft_closing)
             return {0, 0};
-        auto& c = it->second;
-        int64_t n = 0, payload_bytes = 0;
-        size_t pos = 0;
-        // collect (seq, frame): ring write order is claim order and K5b's
-        // atomic direct-delivery claims may interleave within a tick — the
-        // seq header restores per-tick arrival order (wrap-aware), same as
-        // the Python drain (gpu_engine.parse_ring_records)
-        std::vector<std::pair<uint32_t, std::string>> recs;
-        while (pos + 16 <= wpos) {
-            uint32_t len, seq;
-            memcpy(&len, ring + pos, 4);
-            memcpy(&seq, ring + pos + 4, 4);
-            if (len > kMaxMessageSize || pos + 16 + len > wpos) break;
-            std::string framed;
-            framed.resize(4 + len);
-            uint32_t be = htonl(len);
-            memcpy(&framed[0], &be, 4);
-            memcpy(&framed[4], ring + pos + 16, len);
-            recs.emplace_back(seq, std::move(framed));
-            ++n;
-            payload_bytes += (int64_t)len;
-            pos += 16 + (((size_t)len + 15) & ~(size_t)15);
+        auto r = enqueue_ring_locked(it->second, ring, wpos);
+        if (r.first) wake();
+        return r;
+    }
+
+    // Batched egress drain: ONE call + ONE lock + ONE wake for a whole
+    // tick's worth of users.  `base` is the compacted staging buffer (K7
+    // gather of every used ring prefix, D2H'd in one copy); user j's
+    // records occupy [starts[j], ends[j]).  Per-user record counts are
+    // returned; -1 marks a connection that is gone (caller evicts), so the
+    // eviction-on-error contract (reference user/sender.rs:16-33) survives
+    // the batching.
+    std::vector<int64_t> send_rings_batch(const uint8_t* base,
+                                          const std::vector<int64_t>& ids,
+                                          const std::vector<int64_t>& starts,
+                                          const std::vector<int64_t>& ends) {
+        std::vector<int64_t> counts(ids.size(), 0);
+        bool any = false;
+        {
+            std::lock_guard<std::mutex> g(mu_);
+            for (size_t j = 0; j < ids.size(); ++j) {
+                auto it = conns_.find(ids[j]);
+                if (it == conns_.end() || it->second.closed || it->second.soft_closing) {
+                    counts[j] = -1;
+                    continue;
+                }
+                auto r = enqueue_ring_locked(it->second, base + starts[j],
+                                             (size_t)(ends[j] - starts[j]));
+                counts[j] = r.first;
+                any |= r.first > 0;
+            }
         }
-        if (recs.size() > 1) {
-            uint32_t base = recs[0].first;
-            for (auto& r : recs) if (r.first - base > 0x80000000u) base = r.first;
-            std::stable_sort(recs.begin(), recs.end(),
-                             [base](const auto& x, const auto& y) {
-                                 return (uint32_t)(x.first - base) < (uint32_t)(y.first - base);
-                             });
-        }
-        for (auto& r : recs) c.outbox.emplace_back(std::move(r.second));
-        if (n) { c.want_write = true; wake(); }
-        return {n, payload_bytes};
+        if (any) wake();
+        return counts;
     }
 
     // bytes queued but not yet written (backpressure signal for Python)
@@ -259,6 +259,51 @@ private:
         uint64_t one = 1;
         ssize_t r = write(evfd_, &one, 8);
         (void)r;
+    }
+
+    // parse one drained ring, restore per-tick arrival order, and COALESCE
+    // all frames into a single outbox entry (one big send() instead of one
+    // syscall per delivery).  Ring write order is claim order and K5b's
+    // atomic direct-delivery claims may interleave within a tick — the seq
+    // header restores arrival order (wrap-aware), same as the Python drain
+    // (gpu_engine.parse_ring_records).
+    std::pair<int64_t, int64_t> enqueue_ring_locked(Conn& c, const uint8_t* ring,
+                                                    size_t wpos) {
+        int64_t n = 0, payload_bytes = 0;
+        size_t pos = 0;
+        std::vector<std::pair<uint32_t, std::pair<size_t, uint32_t>>> recs;  // seq -> (off, len)
+        while (pos + 16 <= wpos) {
+            uint32_t len, seq;
+            memcpy(&len, ring + pos, 4);
+            memcpy(&seq, ring + pos + 4, 4);
+            if (len > kMaxMessageSize || pos + 16 + len > wpos) break;
+            recs.push_back({seq, {pos + 16, len}});
+            ++n;
+            payload_bytes += (int64_t)len;
+            pos += 16 + (((size_t)len + 15) & ~(size_t)15);
+        }
+        if (recs.size() > 1) {
+            uint32_t base = recs[0].first;
+            for (auto& r : recs) if (r.first - base > 0x80000000u) base = r.first;
+            std::stable_sort(recs.begin(), recs.end(),
+                             [base](const auto& x, const auto& y) {
+                                 return (uint32_t)(x.first - base) < (uint32_t)(y.first - base);
+                             });
+        }
+        if (n) {
+            std::string coalesced;
+            coalesced.resize((size_t)payload_bytes + 4 * (size_t)n);
+            size_t w = 0;
+            for (auto& r : recs) {
+                uint32_t be = htonl(r.second.second);
+                memcpy(&coalesced[w], &be, 4);
+                memcpy(&coalesced[w + 4], ring + r.second.first, r.second.second);
+                w += 4 + r.second.second;
+            }
+            c.outbox.emplace_back(std::move(coalesced));
+            c.want_write = true;
+        }
+        return {n, payload_bytes};
     }
 
     void close_locked(Conn& c) {

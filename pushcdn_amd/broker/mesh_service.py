@@ -155,14 +155,12 @@ class MeshBroker(Broker):
                     host_batch=host_bytes,
                     host_offsets=[int(x) for x in offsets],
                 )
-        return self._engine.drain_cursors()
+        return self._engine.drain_compact()
 
     async def _gpu_tick_task(self) -> None:
         """Fixed-cadence mesh tick: pack queued local messages (possibly
         zero), exchange with every peer, route every rank's batch locally,
         drain egress back to user connections."""
-        from .gpu_engine import parse_ring_records
-
         self.mesh = RcclMesh(self._engine.device, self.batch_capacity)
         dev_buf = (
             torch.zeros(self.batch_capacity, dtype=torch.uint8, device=self._engine.device)
@@ -197,25 +195,10 @@ class MeshBroker(Broker):
                 # the mesh-plane analog of the reference's DirectMap
                 for pubkey in self._gpu_user_by_slot.values():
                     owned_bits |= 1 << (fnv1a64(pubkey, self._engine.hash_seed) & 63)
-            wpos = await asyncio.get_running_loop().run_in_executor(
+            wpos, offsets, staging = await asyncio.get_running_loop().run_in_executor(
                 self._mesh_executor, self._blocking_mesh_tick,
                 msgs, dev_buf, interests, owned_bits)
-            for slot, pubkey in list(self._gpu_user_by_slot.items()):
-                nbytes = int(wpos[slot])
-                if nbytes == 0:
-                    continue
-                ring = self._engine.read_ring(slot, nbytes)
-                handle = self.connections.users.get(pubkey)
-                sink = getattr(handle.connection, "send_ring_records", None) \
-                    if handle is not None else None
-                if sink is not None:
-                    try:
-                        sink(ring, nbytes)  # C++ parses + enqueues all frames
-                    except Exception:
-                        await self.remove_user(pubkey)
-                else:
-                    for _seq, payload in parse_ring_records(ring, nbytes):
-                        await self.try_send_to_user(pubkey, Bytes(payload))
+            await self._dispatch_egress(wpos, offsets, staging)
             for raw in batch:
                 raw.drop()
             await asyncio.sleep(self.config.gpu_tick_interval_s)

@@ -568,11 +568,69 @@ class Broker:
         self._engine.unsubscribe(slot, list(range(256)))
         self._free_gpu_slots.append(slot)
 
+    async def _drain_egress(self) -> None:
+        """One tick's egress: K7-compact every used ring into one staging
+        buffer (one D2H), then hand the WHOLE tick to the C++ pump in one
+        send_rings_batch call.  Per-user fallback (memory/asyncio transports
+        or a foreign pump) parses records in Python as before.  Eviction on
+        delivery failure is preserved (reference user/sender.rs:16-33)."""
+        wpos, offsets, staging = self._engine.drain_compact()
+        await self._dispatch_egress(wpos, offsets, staging)
+
+    async def _dispatch_egress(self, wpos, offsets, staging) -> None:
+        from .gpu_engine import parse_ring_records
+
+        batch_entries = []  # (pubkey, cid, start, end)
+        fallback = []       # (slot, pubkey, nbytes)
+        pump = None
+        for slot, pubkey in list(self._gpu_user_by_slot.items()):
+            n = int(wpos[slot])
+            if n == 0:
+                continue
+            handle = self.connections.users.get(pubkey)
+            ph = getattr(handle.connection, "pump_handle", None) \
+                if handle is not None else None
+            if ph is not None:
+                try:
+                    p, cid = ph()
+                except Exception:
+                    await self.remove_user(pubkey)
+                    continue
+                if pump is None:
+                    pump = p
+                if p is pump:
+                    batch_entries.append(
+                        (pubkey, cid, int(offsets[slot]), int(offsets[slot + 1])))
+                    continue
+            fallback.append((slot, pubkey, n))
+        if batch_entries:
+            counts = pump.send_rings_batch(
+                staging.numpy(),
+                [e[1] for e in batch_entries],
+                [e[2] for e in batch_entries],
+                [e[3] for e in batch_entries],
+            )
+            for (pubkey, _cid, _s, _e), cnt in zip(batch_entries, counts):
+                if cnt < 0:
+                    await self.remove_user(pubkey)
+        for slot, pubkey, n in fallback:
+            ring = bytes(staging[int(offsets[slot]):int(offsets[slot + 1])].numpy()
+                         .tobytes())
+            handle = self.connections.users.get(pubkey)
+            sink = getattr(handle.connection, "send_ring_records", None) \
+                if handle is not None else None
+            if sink is not None:
+                try:
+                    sink(ring, n)
+                except Exception:
+                    await self.remove_user(pubkey)
+            else:
+                for _seq, payload in parse_ring_records(ring, n):
+                    await self.try_send_to_user(pubkey, Bytes(payload))
+
     async def _gpu_tick_task(self) -> None:
         """Batch queued user messages through the kernel pipeline each tick,
         then drain egress rings back to the user connections."""
-        from .gpu_engine import parse_ring_records
-
         while True:
             item = await self._gpu_queue.get()
             batch: List[Bytes] = [item[0]]
@@ -585,24 +643,7 @@ class Broker:
                 offsets.append(len(buf))
             dbuf, doff = self._engine.ingest(bytes(buf), offsets)
             self._engine.tick(dbuf, doff, host_batch=bytes(buf), host_offsets=offsets)
-            wpos = self._engine.drain_cursors()
-            for slot, pubkey in list(self._gpu_user_by_slot.items()):
-                n = int(wpos[slot])
-                if n == 0:
-                    continue
-                ring = self._engine.read_ring(slot, n)
-                handle = self.connections.users.get(pubkey)
-                sink = getattr(handle.connection, "send_ring_records", None) \
-                    if handle is not None else None
-                if sink is not None:
-                    # native pump: C++ parses + enqueues the whole ring
-                    try:
-                        sink(ring, n)
-                    except Exception:
-                        await self.remove_user(pubkey)
-                else:
-                    for _seq, payload in parse_ring_records(ring, n):
-                        await self.try_send_to_user(pubkey, Bytes(payload))
+            await self._drain_egress()
             for raw in batch:
                 raw.drop()
             await asyncio.sleep(self.config.gpu_tick_interval_s)

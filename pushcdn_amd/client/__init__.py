@@ -151,6 +151,29 @@ class Client:
             self._disconnect_on_error()
             raise ConnectionError_(str(e)) from e
 
+    async def receive_raw_batch(self, max_n: int = 4096) -> "List[bytes]":
+        """Batched receive of RAW wire frames (no deserialization): the
+        high-rate benchmark/relay path — callers that only count, forward,
+        or peek fixed offsets skip the per-message parse entirely.  The
+        reference forwards raw bytes verbatim the same way
+        (user/handler.rs:109 keeps the raw alongside the parsed form)."""
+        conn = await self._get_connection()
+        try:
+            raw = await conn.recv_message_raw()
+            out = [raw.data]
+            raw.drop()
+            q = getattr(conn, "_recv_q", None)
+            while q is not None and not q.empty() and len(out) < max_n:
+                r = q.get_nowait()
+                out.append(r.data)
+                r.drop()
+            return out
+        except asyncio.CancelledError:
+            raise
+        except Exception as e:
+            self._disconnect_on_error()
+            raise ConnectionError_(str(e)) from e
+
     async def subscribe(self, topics: Sequence[int]) -> None:
         """Update the replay set first, then best-effort send
         (reference lib.rs:383-414)."""

@@ -127,6 +127,14 @@ class PumpConnection(Connection):
             if self._recv_q.empty() and not self._dead:
                 await self._wakeup.wait()
 
+    def pump_handle(self):
+        """(pump, conn id) for the batched tick drain (Pump.send_rings_batch):
+        all connections of one event loop share one pump, so a whole tick's
+        egress goes out in ONE C++ call."""
+        if self._closed or self._dead:
+            raise ConnectionError_("connection writer closed")
+        return self._mgr.pump, self._cid
+
     def send_ring_records(self, ring: bytes, wpos: int) -> int:
         """Egress fast path for the GPU broker drain: hand a drained ring
         (16 B {len,seq} headers + wire payloads, 16-aligned records) to the
