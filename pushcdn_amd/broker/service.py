@@ -301,8 +301,78 @@ class Broker:
             # immediate partial syncs on connect (user/handler.rs:79-90)
             await self._send_partial_syncs()
 
+    async def _user_receive_loop_fast(self, pubkey: bytes, handle: UserHandle) -> None:
+        """Batched ingest fast path for the GPU data plane on the native
+        pump: drain every frame already queued on the connection in one
+        pass and route by a STRUCTURAL parse (proto.message.parse_offsets —
+        the host mirror of K4) instead of a full deserialize per message.
+        Semantics match the slow loop: Broadcast with no valid topic or any
+        malformed/unexpected frame disconnects (reference
+        user/handler.rs:95-163); Subscribe/Unsubscribe are handled inline.
+        Used only when no user message hook is installed — hooks see
+        deserialized Messages, so they take the general loop."""
+        from ..proto.message import parse_offsets
+
+        connection = handle.connection
+        q = connection._recv_q
+        valid_topic = [False] * 256
+        for t in getattr(self.config.topic_space, "valid", range(256)):
+            valid_topic[t & 0xFF] = True
+        try:
+            while True:
+                raw = await connection.recv_message_raw()
+                frames = [raw]
+                while not q.empty() and len(frames) < 2048:
+                    frames.append(q.get_nowait())
+                for raw in frames:
+                    try:
+                        r = parse_offsets(raw.data)
+                    except Exception:
+                        raise ConnectionError_("malformed frame")
+                    disc = r["disc"]
+                    if disc == 4:  # Broadcast
+                        toff, tcnt = r["topics_off"], r["topics_cnt"]
+                        topics = raw.data[toff:toff + tcnt]
+                        # reference prune semantics: error (disconnect) if no
+                        # valid topic remains, including the empty list
+                        if not any(valid_topic[t] for t in topics):
+                            raise ConnectionError_("no valid topics")
+                        await self._gpu_queue.put((raw, None))
+                    elif disc == 3:  # Direct
+                        await self._gpu_queue.put((raw, r["recipient"]))
+                    elif disc in (5, 6):  # Subscribe / Unsubscribe
+                        toff, tcnt = r["topics_off"], r["topics_cnt"]
+                        try:
+                            topics = self.config.topic_space.prune(
+                                list(raw.data[toff:toff + tcnt]))
+                        except TopicError:
+                            raise ConnectionError_("no valid topics")
+                        if disc == 5:
+                            self.connections.subscribe_user(pubkey, topics)
+                            if handle.gpu_index is not None:
+                                self._engine.subscribe(handle.gpu_index, topics)
+                        else:
+                            self.connections.unsubscribe_user(pubkey, topics)
+                            if handle.gpu_index is not None:
+                                self._engine.unsubscribe(handle.gpu_index, topics)
+                        raw.drop()
+                    else:
+                        raise ConnectionError_("unexpected message type")
+        except (ConnectionError_, asyncio.CancelledError):
+            pass
+        finally:
+            if self.connections.users.get(pubkey) is handle:
+                await self.remove_user(pubkey)
+            else:
+                connection.close()
+
     async def _user_receive_loop(self, pubkey: bytes, handle: UserHandle) -> None:
         """The per-user hot loop (reference user/handler.rs:95-163)."""
+        if (self._engine is not None
+                and self.config.user_message_hook is None
+                and hasattr(handle.connection, "_recv_q")):
+            await self._user_receive_loop_fast(pubkey, handle)
+            return
         connection = handle.connection
         try:
             while True:
