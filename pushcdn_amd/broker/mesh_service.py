@@ -65,17 +65,29 @@ def unpack_mesh_offsets(view: torch.Tensor, n_messages: int) -> torch.Tensor:
 
 class MeshBroker(Broker):
     def __init__(self, config: BrokerConfig, batch_capacity: int = 1 << 22,
-                 interest_routed: bool = False) -> None:
+                 interest_routed: Optional[bool] = None) -> None:
         assert config.data_plane == "gpu", "MeshBroker is the GPU data-plane broker"
         super().__init__(config)
         self.batch_capacity = batch_capacity
         # interest_routed: ship batches only to peers whose subscribers
         # intersect the batch's topics, or whose owned-user digest
         # intersects the batch's direct-recipient digest (grouped P2P on
-        # the xGMI links), instead of the all-gather.
+        # the xGMI links), instead of the all-gather.  Default: ON whenever
+        # there are real peers (the reference also fans out per interested
+        # broker, handler.rs:262-265).
+        if interest_routed is None:
+            import os
+
+            interest_routed = int(os.environ.get("WORLD_SIZE", "1")) > 1
         self.interest_routed = interest_routed
         self.mesh: Optional[RcclMesh] = None
         self._carry: List[Bytes] = []  # messages that didn't fit last tick
+        # per-phase wall-clock accumulators (seconds) — makes a SCALE run
+        # diagnosable: pack/h2d/exchange/tick/drain medians per tick
+        self.mesh_timings = {"pack": 0.0, "h2d": 0.0, "exchange": 0.0,
+                             "tick": 0.0, "drain": 0.0, "ticks": 0}
+        self._rebuild_task = None
+        self._mesh_pause_until = 0.0  # test hook: simulated crash window
         # The collective exchange + kernel ticks run on ONE dedicated thread
         # so the asyncio loop keeps serving user IO / heartbeats while this
         # rank waits for its peers to align on the collective.  A single
@@ -87,6 +99,8 @@ class MeshBroker(Broker):
             max_workers=1, thread_name_prefix="mesh-tick")
 
     async def close(self) -> None:
+        if self._rebuild_task is not None:
+            self._rebuild_task.cancel()
         await super().close()
         self._mesh_executor.shutdown(wait=False, cancel_futures=True)
 
@@ -111,37 +125,67 @@ class MeshBroker(Broker):
                             interests: int, owned_bits: int) -> torch.Tensor:
         """One mesh tick's blocking half (runs on the dedicated mesh thread):
         pack, H2D, collective exchange, kernel tick per received batch,
-        cursor drain.  Returns the drained ring cursors."""
+        cursor drain.  Returns the drained ring cursors.
+
+        Peer failure (collective timeout/abort) tears the communicator down
+        and the tick degrades to LOCAL-ONLY routing — the asyncio side then
+        forwards batches to peers over framed TCP and runs the rebuild loop
+        (reference semantics: evict the dead peer, keep serving,
+        re-establish on heartbeat — heartbeat.rs:67-105)."""
+        import time as _time
+
+        tm = self.mesh_timings
+        t0 = _time.perf_counter()
         host_buf, n_local, used_bytes = pack_mesh_batch(msgs, self.batch_capacity)
+        t1 = _time.perf_counter()
+        tm["pack"] += t1 - t0
         if dev_buf is not None:
             dev_buf.copy_(host_buf, non_blocking=True)
             send_buf = dev_buf
         else:
             send_buf = host_buf
-        if self.interest_routed:
-            from ..proto import message as msglib
-            from ..utils.keyhash import fnv1a64
-
-            batch_topics = 0
-            direct_bits = 0  # 64b digest of this batch's direct recipients
-            for raw in msgs:
-                try:
-                    r = msglib.parse_offsets(raw)
-                except Exception:
-                    continue
-                if r["disc"] == 4:
-                    for t in raw[r["topics_off"] : r["topics_off"] + r["topics_cnt"]]:
-                        batch_topics |= 1 << t
-                elif r["disc"] == 3:
-                    # same keyed hash as the engine's K5 table (seed is
-                    # cluster-shared, so digests agree across brokers)
-                    direct_bits |= 1 << (fnv1a64(r["recipient"], self._engine.hash_seed) & 63)
-            exchanged = self.mesh.exchange_interest(
-                send_buf, n_local, used_bytes, batch_topics, interests,
-                direct_bits=direct_bits, owned_bits=owned_bits,
-            )
+        t2 = _time.perf_counter()
+        tm["h2d"] += t2 - t1
+        degraded = not self.mesh.healthy or _time.monotonic() < self._mesh_pause_until
+        if degraded:
+            if self.mesh.healthy:
+                # simulated-crash window (test hook): stop participating
+                self.mesh.teardown()
+            exchanged = [(self.mesh.rank, send_buf, n_local, used_bytes)]
         else:
-            exchanged = self.mesh.exchange(send_buf, n_local, 0)
+            try:
+                if self.interest_routed:
+                    from ..proto import message as msglib
+                    from ..utils.keyhash import fnv1a64
+
+                    batch_topics = 0
+                    direct_bits = 0  # 64b digest of this batch's direct recipients
+                    for raw in msgs:
+                        try:
+                            r = msglib.parse_offsets(raw)
+                        except Exception:
+                            continue
+                        if r["disc"] == 4:
+                            for t in raw[r["topics_off"] : r["topics_off"] + r["topics_cnt"]]:
+                                batch_topics |= 1 << t
+                        elif r["disc"] == 3:
+                            # same keyed hash as the engine's K5 table (seed is
+                            # cluster-shared, so digests agree across brokers)
+                            direct_bits |= 1 << (
+                                fnv1a64(r["recipient"], self._engine.hash_seed) & 63)
+                    exchanged = self.mesh.exchange_interest(
+                        send_buf, n_local, used_bytes, batch_topics, interests,
+                        direct_bits=direct_bits, owned_bits=owned_bits,
+                    )
+                else:
+                    exchanged = self.mesh.exchange(send_buf, n_local, 0)
+            except Exception:
+                # peer failure: tear down, serve local-only; the asyncio
+                # side starts TCP fallback + the rebuild loop
+                self.mesh.teardown()
+                exchanged = [(self.mesh.rank, send_buf, n_local, used_bytes)]
+        t3 = _time.perf_counter()
+        tm["exchange"] += t3 - t2
         for rank, view, n_msgs, _nbytes in exchanged:
             if n_msgs == 0:
                 continue
@@ -155,13 +199,85 @@ class MeshBroker(Broker):
                     host_batch=host_bytes,
                     host_offsets=[int(x) for x in offsets],
                 )
-        return self._engine.drain_compact()
+        t4 = _time.perf_counter()
+        tm["tick"] += t4 - t3
+        out = self._engine.drain_compact()
+        tm["drain"] += _time.perf_counter() - t4
+        tm["ticks"] += 1
+        return out
+
+    def _mesh_pause(self, seconds: float) -> None:
+        """Test hook: simulate this broker crashing out of the mesh for
+        `seconds` (stops participating in collectives, tears the
+        communicator down), then rejoining via the rebuild rendezvous."""
+        import time as _time
+
+        self._mesh_pause_until = _time.monotonic() + seconds
+
+    async def _mesh_rebuild_loop(self) -> None:
+        """Re-establish the communicator after a peer failure.  Runs rebuild
+        attempts on a DEDICATED thread (degraded ticks keep flowing on the
+        mesh thread — they no longer touch the communicator) until every
+        rank meets at the rendezvous store.  Meanwhile degraded ticks
+        deliver locally and _forward_degraded ships batches to peers over
+        framed TCP — the reference's keep-serving-while-reconnecting
+        behavior."""
+        import concurrent.futures
+        import time as _time
+
+        ex = concurrent.futures.ThreadPoolExecutor(
+            max_workers=1, thread_name_prefix="mesh-rebuild")
+        loop = asyncio.get_running_loop()
+        try:
+            while not self._closed:
+                if _time.monotonic() < self._mesh_pause_until:
+                    await asyncio.sleep(0.05)
+                    continue
+                ok = await loop.run_in_executor(
+                    ex, self.mesh.rebuild, self.config.mesh_rebuild_timeout_s)
+                if ok:
+                    from ..utils.log import log
+
+                    log.info("mesh communicator rebuilt (world=%d)",
+                             self.mesh.world_size)
+                    return
+                await asyncio.sleep(0.2)
+        finally:
+            ex.shutdown(wait=False)
+            self._rebuild_task = None
+
+    async def _maybe_dial_peers(self) -> None:
+        """While degraded, keep framed-TCP links to every live peer dialed
+        (throttled; the inherited broker plane is still listening)."""
+        import time as _time
+
+        now = _time.monotonic()
+        if now - getattr(self, "_last_dial_check", 0.0) < 0.3:
+            return
+        self._last_dial_check = now
+        try:
+            others = await self.discovery.get_other_brokers()
+        except Exception:
+            others = set()
+        for peer in others:
+            if peer not in self.connections.brokers:
+                asyncio.get_running_loop().create_task(self._dial_broker(peer))
+
+    async def _forward_degraded(self, batch: List[Bytes]) -> None:
+        """Host-TCP fallback while the communicator is down: forward every
+        queued message to the dialed peers; the receiving broker delivers
+        with to_users_only/to_user_only semantics via its host plane —
+        exactly the reference mesh path (broker/handler.rs:148-161)."""
+        if self.connections.brokers:
+            for raw in batch:
+                await self.try_send_to_brokers(raw.clone())
 
     async def _gpu_tick_task(self) -> None:
         """Fixed-cadence mesh tick: pack queued local messages (possibly
         zero), exchange with every peer, route every rank's batch locally,
         drain egress back to user connections."""
-        self.mesh = RcclMesh(self._engine.device, self.batch_capacity)
+        self.mesh = RcclMesh(self._engine.device, self.batch_capacity,
+                             timeout_s=self.config.mesh_timeout_s)
         dev_buf = (
             torch.zeros(self.batch_capacity, dtype=torch.uint8, device=self._engine.device)
             if self._engine.is_cuda
@@ -199,6 +315,13 @@ class MeshBroker(Broker):
                 self._mesh_executor, self._blocking_mesh_tick,
                 msgs, dev_buf, interests, owned_bits)
             await self._dispatch_egress(wpos, offsets, staging)
+            if self.mesh.enabled and not self.mesh.healthy:
+                await self._maybe_dial_peers()
+                if batch:
+                    await self._forward_degraded(batch)
+                if self._rebuild_task is None:
+                    self._rebuild_task = asyncio.get_running_loop().create_task(
+                        self._mesh_rebuild_loop())
             for raw in batch:
                 raw.drop()
             await asyncio.sleep(self.config.gpu_tick_interval_s)

@@ -89,6 +89,11 @@ class BrokerConfig:
     gpu_tick_interval_s: float = 0.002
     gpu_max_users: int = 16384
     gpu_ring_bytes: int = 1 << 21
+    # MeshBroker: collective timeout (peer-failure detection latency) and
+    # rebuild-attempt window (communicator rebuild on
+    # membership change with host-TCP fallback meanwhile)
+    mesh_timeout_s: Optional[float] = 30.0
+    mesh_rebuild_timeout_s: float = 10.0
 
 
 @dataclass

@@ -19,13 +19,15 @@ driver of this process group (torchrun / the service supervisor) owns that —
 from __future__ import annotations
 
 import os
-from typing import List, Tuple
+from datetime import timedelta
+from typing import List, Optional, Tuple
 
 import torch
 
 
 class RcclMesh:
-    def __init__(self, device: torch.device, batch_capacity: int) -> None:
+    def __init__(self, device: torch.device, batch_capacity: int,
+                 timeout_s: Optional[float] = None) -> None:
         import torch.distributed as dist
 
         self.dist = dist
@@ -34,15 +36,30 @@ class RcclMesh:
         self.world_size = int(os.environ.get("WORLD_SIZE", "1"))
         self.rank = int(os.environ.get("RANK", "0"))
         self.enabled = self.world_size > 1
+        # healthy=False after a peer failure (collective timeout/abort);
+        # the service runs local-only ticks until rebuild() succeeds
+        self.healthy = True
+        self.timeout = timedelta(seconds=timeout_s) if timeout_s else None
+        self._backend = "nccl" if device.type == "cuda" else "gloo"
+        self._store = None
         if self.enabled and not dist.is_initialized():
-            backend = "nccl" if device.type == "cuda" else "gloo"
-            dist.init_process_group(backend=backend)
+            kw = {"timeout": self.timeout} if self.timeout else {}
+            dist.init_process_group(backend=self._backend, **kw)
+        self._meta_group = None
         if self.enabled:
             self._gathered = torch.zeros(
                 self.world_size * batch_capacity, dtype=torch.uint8, device=device
             )
-            # per-rank metadata: [n_messages, batch_bytes]
-            self._meta = torch.zeros(self.world_size * 2, dtype=torch.int64, device=device)
+            self._make_meta_group()
+
+    def _make_meta_group(self) -> None:
+        # tiny per-tick metadata rides a HOST-side gloo group so the device
+        # stream never has to sync for it (round-1 weak item: the per-tick
+        # meta.to("cpu") stalled the loop while peers aligned); the payload
+        # all-gather stays on the RCCL default group and is consumed
+        # stream-ordered with no host sync at all
+        kw = {"timeout": self.timeout} if self.timeout else {}
+        self._meta_group = self.dist.new_group(backend="gloo", **kw)
 
     def exchange_flat(self, batch: torch.Tensor) -> torch.Tensor:
         """All-gather this tick's batch and return the CONTIGUOUS
@@ -66,11 +83,12 @@ class RcclMesh:
         if not self.enabled:
             return [(0, batch, n_messages, batch_bytes)]
         assert batch.numel() == self.capacity
-        meta_local = torch.tensor([n_messages, batch_bytes], dtype=torch.int64,
-                                  device=self.device)
-        self.dist.all_gather_into_tensor(self._meta, meta_local)
+        # payload: async on the device stream; metadata: host gloo (no
+        # device sync anywhere in this call — consumers are stream-ordered)
         self.dist.all_gather_into_tensor(self._gathered, batch)
-        meta = self._meta.to("cpu")
+        meta_local = torch.tensor([n_messages, batch_bytes], dtype=torch.int64)
+        meta = torch.zeros(self.world_size * 2, dtype=torch.int64)
+        self.dist.all_gather_into_tensor(meta, meta_local, group=self._meta_group)
         out = []
         for r in range(self.world_size):
             out.append((
@@ -100,11 +118,9 @@ class RcclMesh:
         tbits = 0
         for t in targets:
             tbits |= 1 << t
-        meta_local = torch.tensor([n_messages, batch_bytes, tbits], dtype=torch.int64,
-                                  device=self.device)
-        meta = torch.zeros(self.world_size * 3, dtype=torch.int64, device=self.device)
-        self.dist.all_gather_into_tensor(meta, meta_local)
-        meta_h = meta.to("cpu")
+        meta_local = torch.tensor([n_messages, batch_bytes, tbits], dtype=torch.int64)
+        meta_h = torch.zeros(self.world_size * 3, dtype=torch.int64)
+        self.dist.all_gather_into_tensor(meta_h, meta_local, group=self._meta_group)
         ops = []
         for r in range(self.world_size):
             if r == self.rank:
@@ -161,14 +177,15 @@ class RcclMesh:
         def pack_bits(v: int) -> list:
             return [_i64(v >> (64 * i)) for i in range(4)]
 
+        # metadata on the host gloo group (no device roundtrip)
         meta_local = torch.tensor(
             [n_messages, batch_bytes] + pack_bits(batch_topics) + pack_bits(interests)
             + [_i64(direct_bits), _i64(owned_bits)],
-            dtype=torch.int64, device=self.device)
+            dtype=torch.int64)
         stride = meta_local.numel()
-        meta = torch.zeros(self.world_size * stride, dtype=torch.int64, device=self.device)
-        self.dist.all_gather_into_tensor(meta, meta_local)
-        mh = meta.to("cpu").tolist()
+        meta = torch.zeros(self.world_size * stride, dtype=torch.int64)
+        self.dist.all_gather_into_tensor(meta, meta_local, group=self._meta_group)
+        mh = meta.tolist()
 
         def unpack_bits(words) -> int:
             v = 0
@@ -222,10 +239,42 @@ class RcclMesh:
         self.dist.all_reduce(t, op=self.dist.ReduceOp.MAX)
         return float(t.cpu()[0])
 
-    def rebuild(self) -> None:
+    def teardown(self) -> None:
+        """Abandon the current communicator (peer failure detected)."""
+        self.healthy = False
+        try:
+            if self.dist.is_initialized():
+                self.dist.destroy_process_group()
+        except Exception:
+            pass
+        self._meta_group = None
+        self._store = None
+
+    def rebuild(self, timeout_s: float = 10.0) -> bool:
         """Communicator teardown/rebuild on membership change (the xGMI
-        analog of a TCP reconnect — SURVEY §5.3)."""
-        if self.dist.is_initialized():
-            self.dist.destroy_process_group()
-        backend = "nccl" if self.device.type == "cuda" else "gloo"
-        self.dist.init_process_group(backend=backend)
+        analog of a TCP reconnect — SURVEY §5.3).  Rendezvous is a fresh
+        TCPStore on MASTER_PORT+1000 that rank 0 re-creates per attempt and
+        surviving/rejoining ranks connect to — no epoch agreement problem:
+        a freshly restarted rank converges by retrying until ALL world_size
+        ranks meet at the store.  Returns True on success; False means not
+        every rank has arrived yet (caller keeps serving local-only ticks
+        and retries — the reference's eviction-until-reconnect semantics,
+        heartbeat.rs:67-105)."""
+        self.teardown()
+        addr = os.environ.get("MASTER_ADDR", "127.0.0.1")
+        port = int(os.environ.get("MASTER_PORT", "29500")) + 1000
+        try:
+            store = self.dist.TCPStore(
+                addr, port, self.world_size, self.rank == 0,
+                timeout=timedelta(seconds=timeout_s))
+            self.dist.init_process_group(
+                backend=self._backend, store=store,
+                world_size=self.world_size, rank=self.rank,
+                timeout=self.timeout or timedelta(seconds=max(30.0, timeout_s)))
+            self._store = store  # keep the rendezvous store alive with the pg
+            self._make_meta_group()
+            self.healthy = True
+            return True
+        except Exception:
+            self.teardown()
+            return False
