@@ -342,9 +342,9 @@ class Broker:
                         # valid topic remains, including the empty list
                         if not any(valid_topic[t] for t in topics):
                             raise ConnectionError_("no valid topics")
-                        await self._gpu_queue.put((raw, None))
+                        await self._gpu_queue.put((raw, ("b", bytes(topics))))
                     elif disc == 3:  # Direct
-                        await self._gpu_queue.put((raw, r["recipient"]))
+                        await self._gpu_queue.put((raw, ("d", r["recipient"])))
                     elif disc in (5, 6):  # Subscribe / Unsubscribe
                         toff, tcnt = r["topics_off"], r["topics_cnt"]
                         try:
@@ -400,12 +400,12 @@ class Broker:
                     except TopicError:
                         break
                     if self._engine is not None:
-                        await self._gpu_queue.put((raw, None))
+                        await self._gpu_queue.put((raw, ("b", bytes(topics))))
                     else:
                         await self.handle_broadcast_message(topics, raw, to_users_only=False)
                 elif isinstance(msg, m.Direct):
                     if self._engine is not None:
-                        await self._gpu_queue.put((raw, msg.recipient))
+                        await self._gpu_queue.put((raw, ("d", msg.recipient)))
                     else:
                         await self.handle_direct_message(msg.recipient, raw, to_user_only=False)
                 elif isinstance(msg, m.Subscribe):
@@ -545,10 +545,20 @@ class Broker:
                     except TopicError:
                         raw.drop()
                         continue
-                    # single-hop mesh: deliver only to local users
-                    await self.handle_broadcast_message(topics, raw, to_users_only=True)
+                    # single-hop mesh: deliver only to local users.  On the
+                    # GPU plane that means routing through the ENGINE with
+                    # no re-forwarding (fwd=None) — the kernels only ever
+                    # write local rings, which IS to_users_only.
+                    if self._engine is not None:
+                        await self._gpu_queue.put((raw, None))
+                    else:
+                        await self.handle_broadcast_message(topics, raw, to_users_only=True)
                 elif isinstance(msg, m.Direct):
-                    await self.handle_direct_message(msg.recipient, raw, to_user_only=True)
+                    if self._engine is not None:
+                        # K5 drops non-local recipients = to_user_only
+                        await self._gpu_queue.put((raw, None))
+                    else:
+                        await self.handle_direct_message(msg.recipient, raw, to_user_only=True)
                 elif isinstance(msg, m.UserSync):
                     to_kick = self.connections.apply_user_sync(msg.data)
                     for pubkey in to_kick:
@@ -703,22 +713,44 @@ class Broker:
                 for _seq, payload in parse_ring_records(ring, n):
                     await self.try_send_to_user(pubkey, Bytes(payload))
 
+    async def _forward_to_mesh(self, raw: Bytes, fwd) -> None:
+        """Broker-plane forwarding for a LOCAL-origin message on the GPU
+        data plane — the framed-TCP side of the unified broker plane
+        (reference broker/handler.rs:197-272): Broadcasts go to every peer
+        with subscribers on the message's topics, Directs to the DirectMap
+        owner; remote-origin messages (fwd=None) are never re-forwarded
+        (single-hop mesh)."""
+        kind, data = fwd
+        if kind == "b":
+            _users, brokers = self.connections.get_interested_by_topic(
+                list(data), to_users_only=False)
+            for broker in brokers:
+                await self.try_send_to_broker(broker, raw.clone())
+        else:
+            owner = self.connections.get_broker_identifier_of_user(data)
+            if owner is not None and owner != self.identity:
+                await self.try_send_to_broker(owner, raw.clone())
+
     async def _gpu_tick_task(self) -> None:
         """Batch queued user messages through the kernel pipeline each tick,
         then drain egress rings back to the user connections."""
         while True:
             item = await self._gpu_queue.get()
-            batch: List[Bytes] = [item[0]]
+            batch = [item]
             while not self._gpu_queue.empty() and len(batch) < 4096:
-                batch.append(self._gpu_queue.get_nowait()[0])
+                batch.append(self._gpu_queue.get_nowait())
+            if self.connections.brokers:
+                for raw, fwd in batch:
+                    if fwd is not None:
+                        await self._forward_to_mesh(raw, fwd)
             buf = bytearray()
             offsets = [0]
-            for raw in batch:
+            for raw, _fwd in batch:
                 buf += raw.data
                 offsets.append(len(buf))
             dbuf, doff = self._engine.ingest(bytes(buf), offsets)
             self._engine.tick(dbuf, doff, host_batch=bytes(buf), host_offsets=offsets)
             await self._drain_egress()
-            for raw in batch:
+            for raw, _fwd in batch:
                 raw.drop()
             await asyncio.sleep(self.config.gpu_tick_interval_s)
