@@ -207,10 +207,25 @@ class GpuBrokerEngine:
 
     # ---------------------------- the hot tick ----------------------------
 
-    def ingest(self, batch: bytes, offsets: List[int]) -> Tuple[torch.Tensor, torch.Tensor]:
-        """H2D-copy one batch of serialized messages. Returns device (buf, offsets)."""
+    def ingest(self, batch: bytes, offsets: List[int],
+               staging: Optional[torch.Tensor] = None
+               ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """H2D-copy one batch of serialized messages. Returns device (buf, offsets).
+
+        `staging`: an HBM-pool slice (hbm_pool.PoolBytes.tensor) to land the
+        batch in instead of a fresh allocation — the bounded/backpressured
+        ingest path (reference limiter semantics).  Safe to reuse after the
+        caller drops the allocation because all consumers are stream-ordered
+        behind this copy on the engine's stream."""
         host = torch.frombuffer(bytearray(batch), dtype=torch.uint8)
         off = torch.tensor(offsets, dtype=torch.int64)
+        if staging is not None:
+            assert staging.numel() >= host.numel()
+            dst = staging[:host.numel()]
+            dst.copy_(host, non_blocking=self.is_cuda)
+            if self.is_cuda:
+                return dst, off.to(self.device, non_blocking=True)
+            return dst, off
         if self.is_cuda:
             return host.to(self.device, non_blocking=True), off.to(self.device, non_blocking=True)
         return host, off

@@ -94,6 +94,9 @@ class BrokerConfig:
     # membership change with host-TCP fallback meanwhile)
     mesh_timeout_s: Optional[float] = 30.0
     mesh_rebuild_timeout_s: float = 10.0
+    # HBM message-pool budget for ingest staging (reference
+    # --global-memory-pool-size default 1 GiB, broker.rs:71-73); 0 = off
+    gpu_pool_bytes: int = 1 << 30
 
 
 @dataclass
@@ -148,6 +151,20 @@ class Broker:
             self._gpu_queue = asyncio.Queue()
             self._free_gpu_slots = list(range(config.gpu_max_users - 1, -1, -1))
             self._gpu_user_by_slot: Dict[int, bytes] = {}
+            if config.gpu_pool_bytes:
+                from .hbm_pool import HbmMessagePool
+
+                # the CPU reference engine (tests) caps the arena so suites
+                # that stand up many brokers stay light; on cuda the full
+                # budget is carved out of the 288 GB HBM up front
+                pool_bytes = config.gpu_pool_bytes
+                if not self._engine.is_cuda:
+                    pool_bytes = min(pool_bytes, 32 << 20)
+                self._hbm_pool = HbmMessagePool(pool_bytes, device=config.gpu_device)
+            else:
+                self._hbm_pool = None
+        else:
+            self._hbm_pool = None
 
     # ------------------------------ lifecycle ------------------------------
 
@@ -748,9 +765,16 @@ class Broker:
             for raw, _fwd in batch:
                 buf += raw.data
                 offsets.append(len(buf))
-            dbuf, doff = self._engine.ingest(bytes(buf), offsets)
+            # ingest staging from the bounded HBM pool: exhaustion WAITS
+            # here (backpressure up through the tick queue to the sockets),
+            # matching the reference limiter (protocols/mod.rs:328)
+            pb = await self._hbm_pool.alloc(len(buf)) if self._hbm_pool else None
+            dbuf, doff = self._engine.ingest(
+                bytes(buf), offsets, staging=pb.tensor if pb else None)
             self._engine.tick(dbuf, doff, host_batch=bytes(buf), host_offsets=offsets)
             await self._drain_egress()
+            if pb is not None:
+                pb.drop()
             for raw, _fwd in batch:
                 raw.drop()
             await asyncio.sleep(self.config.gpu_tick_interval_s)

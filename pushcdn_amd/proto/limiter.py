@@ -6,8 +6,9 @@ optional bounded per-connection message channels.  Receivers block until the
 global pool has room, which backpressures the socket (``protocols/mod.rs:328``).
 
 On the GPU data plane the same semantics are provided by the HBM message pool
-(``pushcdn_amd.broker.gpu_engine.MessagePool``): bounded total bytes,
-refcounted release, allocation backpressure.
+(``pushcdn_amd.broker.hbm_pool.HbmMessagePool``): ingest staging comes out of
+one pre-sized HBM arena with bounded total bytes, refcounted release, and
+allocation backpressure.
 """
 
 from __future__ import annotations
