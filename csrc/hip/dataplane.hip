@@ -1058,19 +1058,29 @@ extern "C" __global__ void k5b_emit_direct(
     if (u < 0) return;
     int32_t len = payload_len[i];
     uint64_t rec = ring_rec(len);
-    uint64_t old = atomicAdd((unsigned long long*)&ring_wpos[u], (unsigned long long)rec);
     int slot = atomicAdd(n_pairs, 1);
-    if (old + rec > (uint64_t)ring_bytes || slot >= capacity) {
-        // roll the cursor back (transient overshoot may make a concurrent
-        // claim drop spuriously — benign; never corrupts a neighbor ring)
-        if (old + rec > (uint64_t)ring_bytes)
-            atomicAdd((unsigned long long*)&ring_wpos[u],
-                      (unsigned long long)(-(long long)rec));
-        if (slot < capacity) store_pair(pairs + slot, -1, i, 0);
+    if (slot >= capacity) {
         atomicAdd(drops, 1u);
         return;
     }
-    store_pair(pairs + slot, u, i, (int64_t)u * ring_bytes + (int64_t)old);
+    // CAS claim: the round-1 add-then-rollback scheme could, with two
+    // concurrent overshooters and an interleaved accept, leave an accepted
+    // interval ABOVE a rolled-back one and let later claims overlap it.
+    // CAS never inflates the cursor, so accepted intervals are exact and
+    // a full ring drops cleanly (counted) with no spurious rollback race.
+    unsigned long long cur = atomicAdd((unsigned long long*)&ring_wpos[u], 0ull);
+    while (true) {
+        if (cur + rec > (uint64_t)ring_bytes) {
+            store_pair(pairs + slot, -1, i, 0);  // K3 skips user<0 pairs
+            atomicAdd(drops, 1u);
+            return;
+        }
+        unsigned long long prev = atomicCAS((unsigned long long*)&ring_wpos[u], cur,
+                                            cur + (unsigned long long)rec);
+        if (prev == cur) break;
+        cur = prev;
+    }
+    store_pair(pairs + slot, u, i, (int64_t)u * ring_bytes + (int64_t)cur);
 }
 
 extern "C" void launch_k5b_emit_direct(

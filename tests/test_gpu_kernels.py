@@ -564,3 +564,43 @@ def test_k2b_blocks_matches_fused(ops):
     assert int(wf.sum()) == int(wb.sum()) == cap * rec  # every slot emitted
     for pr in (prf, prb):
         assert bool((pr[:cap, 0] >= 0).all())         # all in-capacity slots real
+
+
+def test_k5b_direct_contention_ring_exhaustion(ops):
+    """K5b claim correctness under heavy same-user contention with a ring
+    too small for the burst: delivered + dropped == total, the delivered
+    prefix parses cleanly (no overlapped/garbled records — the round-1
+    add-then-rollback scheme could overlap an accepted claim), and every
+    delivered payload is one of the sent messages."""
+    from pushcdn_amd.broker.gpu_engine import GpuBrokerEngine, parse_ring_records, ring_rec
+    from pushcdn_amd.proto import message as m
+
+    N = 2048
+    raw = m.serialize(m.Direct(recipient=b"X" * 64, message=b"c" * 100))
+    wire = (len(raw) + 15) & ~15
+    rec = ring_rec(wire)
+    fit = 300  # ring holds only 300 of the 2048 concurrent claims
+    ring_bytes = ((fit * rec + 15) & ~15)
+    eng = GpuBrokerEngine(device="cuda:0", n_users=2, ring_bytes=ring_bytes,
+                          fanout_wire=True, direct_enabled=True)
+    eng.register_direct(b"X" * 64, 0)
+    buf = bytearray()
+    offsets = [0]
+    for _ in range(N):
+        buf += raw + b"\x00" * (wire - len(raw))
+        offsets.append(len(buf))
+    dbuf, doff = eng.ingest(bytes(buf), offsets)
+    eng.tick(dbuf, doff, uniform_wire_len=wire)
+    torch.cuda.synchronize()
+    drops = int(eng._drops[0])
+    wpos = eng.drain_cursors()
+    delivered = int(wpos[0]) // rec
+    assert int(wpos[1]) == 0  # neighbor ring untouched
+    assert int(wpos[0]) % rec == 0
+    assert delivered + drops == N, (delivered, drops)
+    assert delivered == ring_bytes // rec  # filled exactly to capacity
+    recs = parse_ring_records(eng.read_ring(0, int(wpos[0])), int(wpos[0]))
+    assert len(recs) == delivered
+    want = raw + b"\x00" * (wire - len(raw))
+    for _seq, payload in recs:
+        assert payload == want
