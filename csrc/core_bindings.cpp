@@ -393,6 +393,25 @@ PYBIND11_MODULE(pushcdn_core, m) {
                  return p.send_rings_batch((const uint8_t*)info.ptr, ids, starts, ends);
              },
              "batched tick drain: one call for all users' compacted rings")
+        .def("set_ingest", &net::Pump::set_ingest)
+        .def("recv_ingest", [](net::Pump& p, int64_t id) {
+            auto b = p.recv_ingest(id);
+            py::list offs, disc, toff, tcnt, roff, rlen;
+            for (auto o : b.offs) offs.append(o);
+            for (auto& m : b.meta) {
+                disc.append(m.disc);
+                toff.append(m.topics_off);
+                tcnt.append(m.topics_cnt);
+                roff.append(m.recip_off);
+                rlen.append(m.recip_len);
+            }
+            return py::make_tuple(py::bytes(b.blob), offs, disc, toff, tcnt, roff,
+                                  rlen, b.closed);
+        })
+        .def("send_raw", [](net::Pump& p, int64_t id, py::buffer data) {
+            py::buffer_info info = data.request();
+            return p.send_raw(id, (const char*)info.ptr, (size_t)info.size);
+        })
         .def("send_backlog", &net::Pump::send_backlog)
         .def("poll_dirty", &net::Pump::poll_dirty)
         .def("recv_batch", [](net::Pump& p, int64_t id, size_t maxf) {

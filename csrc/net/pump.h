@@ -32,16 +32,98 @@
 #include <thread>
 #include <vector>
 
+#include "../wire/message.h"
+
 namespace net {
 
 // wire framing limit — mirrors proto/transports/base.py MAX_MESSAGE_SIZE
 constexpr uint64_t kMaxMessageSize = (0xFFFFFFFFull) / 8;
+
+// Structural classification of one wire frame for the C++ ingest path —
+// the data-plane fields the broker routes on, WITHOUT copying the payload
+// (offset/length views into the frame; host mirror of the K4 kernel's
+// output shape). disc = -1 marks a malformed frame (caller disconnects,
+// reference user/handler.rs:109 semantics).
+struct FrameMeta {
+    int32_t disc = -1;
+    uint32_t topics_off = 0, topics_cnt = 0;   // Broadcast/Sub/Unsub, frame-relative
+    uint32_t recip_off = 0, recip_len = 0;     // Direct, frame-relative
+};
+
+inline FrameMeta classify_frame(const uint8_t* buf, size_t len) {
+    FrameMeta m;
+    if (len < 16) return m;
+    uint32_t seg_m1, nw;
+    memcpy(&seg_m1, buf, 4);
+    memcpy(&nw, buf + 4, 4);
+    if (seg_m1 != 0 || 8 + (uint64_t)nw * 8 > len) return m;
+    wire::Reader r{buf + 8, nw};
+    size_t mt;
+    uint16_t mdw, mpw;
+    if (!r.struct_ptr(0, &mt, &mdw, &mpw) || mdw < 1 || mpw < 1) return m;
+    uint64_t w0;
+    r.u64(mt, &w0);
+    uint16_t disc = (uint16_t)w0;
+    size_t up = mt + mdw;
+    // offset-returning byte-list decode (no copy)
+    auto list_view = [&](size_t pw, uint32_t* off, uint32_t* n) -> bool {
+        uint64_t v;
+        if (!r.u64(pw, &v)) return false;
+        if (v == 0) { *off = 0; *n = 0; return true; }
+        if ((v & 3) != 1) return false;
+        int64_t b = (v >> 2) & 0x3fffffff;
+        if (b & 0x20000000) b -= 0x40000000;
+        if (((v >> 32) & 7) != 2) return false;
+        uint64_t count = (v >> 35) & 0x1fffffff;
+        int64_t t = (int64_t)pw + 1 + b;
+        if (t < 0 || (uint64_t)t * 8 + count > (uint64_t)nw * 8) return false;
+        *off = (uint32_t)(8 + t * 8);
+        *n = (uint32_t)count;
+        return true;
+    };
+    switch (disc) {
+    case wire::DIRECT: {
+        size_t it; uint16_t idw, ipw;
+        if (!r.struct_ptr(up, &it, &idw, &ipw) || ipw < 2) return m;
+        if (!list_view(it + idw, &m.recip_off, &m.recip_len)) return m;
+        break;
+    }
+    case wire::BROADCAST: {
+        size_t it; uint16_t idw, ipw;
+        if (!r.struct_ptr(up, &it, &idw, &ipw) || ipw < 2) return m;
+        if (!list_view(it + idw, &m.topics_off, &m.topics_cnt)) return m;
+        break;
+    }
+    case wire::SUBSCRIBE:
+    case wire::UNSUBSCRIBE:
+        if (!list_view(up, &m.topics_off, &m.topics_cnt)) return m;
+        break;
+    case wire::AUTHENTICATE_WITH_KEY:
+    case wire::AUTHENTICATE_WITH_PERMIT:
+    case wire::AUTHENTICATE_RESPONSE:
+    case wire::USER_SYNC:
+    case wire::TOPIC_SYNC:
+        break;
+    default:
+        return m;
+    }
+    m.disc = disc;
+    return m;
+}
 
 struct Conn {
     int fd = -1;
     // inbound: frame assembly
     std::vector<uint8_t> rbuf;           // partial wire bytes
     std::deque<std::string> inbox;       // complete frames (payload only)
+    // C++ ingest mode (GPU broker data plane): complete frames accumulate
+    // in ONE contiguous buffer with offsets + routing metadata; Python
+    // pulls a whole tick's worth in one call (recv_ingest)
+    bool ingest = false;
+    bool paused = false;                 // EPOLLIN parked: ibuf over budget
+    std::string ibuf;
+    std::vector<int64_t> ioffs;          // frame end offsets into ibuf
+    std::vector<FrameMeta> imeta;
     // outbound
     std::deque<std::string> outbox;      // framed bytes (header+payload)
     size_t out_off = 0;                  // offset into outbox.front()
@@ -51,6 +133,11 @@ struct Conn {
     bool forget_pending = false;         // erase once flushed + closed
     uint64_t in_bytes = 0, out_bytes = 0;
 };
+
+// ingest-mode read backpressure threshold: above this, stop reading the
+// socket until Python drains (TCP window then backpressures the sender —
+// the reference's limiter-blocks-the-reader behavior, protocols/mod.rs:328)
+constexpr size_t kIngestPauseBytes = 32u << 20;
 
 class Pump {
 public:
@@ -191,6 +278,68 @@ public:
         return out;
     }
 
+    // enable C++ ingest mode for a connection (GPU broker user plane):
+    // frames accumulate contiguously with routing metadata; Python pulls a
+    // tick's worth at a time with recv_ingest
+    void set_ingest(int64_t id) {
+        std::lock_guard<std::mutex> g(mu_);
+        auto it = conns_.find(id);
+        if (it == conns_.end()) return;
+        auto& c = it->second;
+        c.ingest = true;
+        // frames that raced in through the normal inbox re-route
+        while (!c.inbox.empty()) {
+            std::string& f = c.inbox.front();
+            c.imeta.push_back(classify_frame((const uint8_t*)f.data(), f.size()));
+            c.ibuf.append(f);
+            c.ioffs.push_back((int64_t)c.ibuf.size());
+            c.inbox.pop_front();
+        }
+    }
+
+    struct IngestBatch {
+        std::string blob;                // concatenated frame bytes
+        std::vector<int64_t> offs;       // frame END offsets
+        std::vector<FrameMeta> meta;
+        bool closed = false;
+    };
+
+    // pull everything the pump has classified for this connection (ONE
+    // Python call per tick per connection instead of one per message)
+    IngestBatch recv_ingest(int64_t id) {
+        IngestBatch out;
+        std::lock_guard<std::mutex> g(mu_);
+        auto it = conns_.find(id);
+        if (it == conns_.end()) {
+            out.closed = true;
+            return out;
+        }
+        auto& c = it->second;
+        out.blob.swap(c.ibuf);
+        out.offs.swap(c.ioffs);
+        out.meta.swap(c.imeta);
+        out.closed = c.closed;
+        if (c.paused && c.fd >= 0) {
+            c.paused = false;
+            update_interest(id, c);
+            wake();
+        }
+        return out;
+    }
+
+    // enqueue PRE-FRAMED bytes verbatim (a batch of [4B len][frame] records
+    // built by the caller) — the sender-side batch analog of send()
+    bool send_raw(int64_t id, const char* data, size_t len) {
+        std::lock_guard<std::mutex> g(mu_);
+        auto it = conns_.find(id);
+        if (it == conns_.end() || it->second.closed || it->second.soft_closing)
+            return false;
+        it->second.outbox.emplace_back(data, len);
+        it->second.want_write = true;
+        wake();
+        return true;
+    }
+
     // drain up to max_frames complete inbound frames; empty vector + closed
     // flag tells Python the peer is gone
     std::pair<std::vector<std::string>, bool> recv_batch(int64_t id, size_t max_frames) {
@@ -318,7 +467,7 @@ private:
     void update_interest(int64_t id, Conn& c) {
         if (c.fd < 0) return;
         struct epoll_event ev {};
-        ev.events = EPOLLIN | (c.want_write ? EPOLLOUT : 0);
+        ev.events = (c.paused ? 0 : EPOLLIN) | (c.want_write ? EPOLLOUT : 0);
         ev.data.u64 = (uint64_t)id;
         epoll_ctl(epfd_, EPOLL_CTL_MOD, c.fd, &ev);
     }
@@ -341,11 +490,22 @@ private:
                     uint64_t len = ntohl(be);
                     if (len > kMaxMessageSize) { close_locked(c); return true; }
                     if (c.rbuf.size() - off - 4 < len) break;
-                    c.inbox.emplace_back((const char*)c.rbuf.data() + off + 4, len);
+                    if (c.ingest) {
+                        const uint8_t* f = c.rbuf.data() + off + 4;
+                        c.imeta.push_back(classify_frame(f, len));
+                        c.ibuf.append((const char*)f, len);
+                        c.ioffs.push_back((int64_t)c.ibuf.size());
+                    } else {
+                        c.inbox.emplace_back((const char*)c.rbuf.data() + off + 4, len);
+                    }
                     off += 4 + len;
                     new_frames = true;
                 }
                 if (off) c.rbuf.erase(c.rbuf.begin(), c.rbuf.begin() + off);
+                if (c.ingest && c.ibuf.size() > kIngestPauseBytes) {
+                    c.paused = true;  // re-armed by recv_ingest's drain
+                    break;
+                }
                 if (n < (ssize_t)tmpv.size()) continue;  // might be more
             } else if (n == 0) {
                 close_locked(c);
@@ -409,6 +569,7 @@ private:
                 if (evs[i].events & (EPOLLIN | EPOLLHUP | EPOLLERR)) {
                     bool was_closed = c.closed;
                     bool frames = do_read(c);
+                    if (c.fd >= 0 && c.paused) update_interest(id, c);  // park EPOLLIN
                     if (frames || (c.closed && !was_closed)) {
                         notify = true;
                         dirty_.insert(id);

@@ -64,6 +64,10 @@ def unpack_mesh_offsets(view: torch.Tensor, n_messages: int) -> torch.Tensor:
 
 
 class MeshBroker(Broker):
+    # the collective pack path consumes per-message Bytes; the C++ blob
+    # ingest is the single-broker socket fast path
+    BLOB_INGEST = False
+
     def __init__(self, config: BrokerConfig, batch_capacity: int = 1 << 22,
                  interest_routed: Optional[bool] = None) -> None:
         assert config.data_plane == "gpu", "MeshBroker is the GPU data-plane broker"

@@ -323,6 +323,67 @@ class Broker:
             # immediate partial syncs on connect (user/handler.rs:79-90)
             await self._send_partial_syncs()
 
+    # MeshBroker overrides to False: its collective pack path consumes
+    # per-message Bytes, not ingest blobs
+    BLOB_INGEST = True
+
+    async def _user_receive_loop_ingest(self, pubkey: bytes, handle: UserHandle) -> None:
+        """C++-ingest receive loop: the pump accumulates classified frames
+        in one contiguous buffer; Python pulls a TICK's worth per call and
+        enqueues the whole blob — no per-message interpreter work at all on
+        the Broadcast/Direct hot path.  Semantics match the reference loop
+        (user/handler.rs:95-163): malformed/unexpected frames and
+        no-valid-topic broadcasts disconnect; Subscribe/Unsubscribe apply
+        inline."""
+        connection = handle.connection
+        connection.enable_ingest()
+        valid_topic = [False] * 256
+        for t in getattr(self.config.topic_space, "valid", range(256)):
+            valid_topic[t & 0xFF] = True
+        try:
+            while True:
+                blob, ends, discs, toffs, tcnts, roffs, rlens = \
+                    await connection.recv_ingest_batch()
+                fwds = []
+                start = 0
+                for i, d in enumerate(discs):
+                    end = ends[i]
+                    if d == 4:  # Broadcast
+                        ab = start + toffs[i]
+                        if not any(valid_topic[b] for b in blob[ab:ab + tcnts[i]]):
+                            raise ConnectionError_("no valid topics")
+                        fwds.append(("b", ab, ab + tcnts[i]))
+                    elif d == 3:  # Direct
+                        ab = start + roffs[i]
+                        fwds.append(("d", ab, ab + rlens[i]))
+                    elif d in (5, 6):  # Subscribe / Unsubscribe (rare, inline)
+                        ab = start + toffs[i]
+                        try:
+                            topics = self.config.topic_space.prune(
+                                list(blob[ab:ab + tcnts[i]]))
+                        except TopicError:
+                            raise ConnectionError_("no valid topics")
+                        if d == 5:
+                            self.connections.subscribe_user(pubkey, topics)
+                            if handle.gpu_index is not None:
+                                self._engine.subscribe(handle.gpu_index, topics)
+                        else:
+                            self.connections.unsubscribe_user(pubkey, topics)
+                            if handle.gpu_index is not None:
+                                self._engine.unsubscribe(handle.gpu_index, topics)
+                        fwds.append(None)
+                    else:
+                        raise ConnectionError_("malformed or unexpected frame")
+                    start = end
+                await self._gpu_queue.put(("blob", blob, ends, fwds))
+        except (ConnectionError_, asyncio.CancelledError):
+            pass
+        finally:
+            if self.connections.users.get(pubkey) is handle:
+                await self.remove_user(pubkey)
+            else:
+                connection.close()
+
     async def _user_receive_loop_fast(self, pubkey: bytes, handle: UserHandle) -> None:
         """Batched ingest fast path for the GPU data plane on the native
         pump: drain every frame already queued on the connection in one
@@ -390,6 +451,12 @@ class Broker:
 
     async def _user_receive_loop(self, pubkey: bytes, handle: UserHandle) -> None:
         """The per-user hot loop (reference user/handler.rs:95-163)."""
+        if (self._engine is not None
+                and self.config.user_message_hook is None
+                and type(self).BLOB_INGEST
+                and hasattr(handle.connection, "enable_ingest")):
+            await self._user_receive_loop_ingest(pubkey, handle)
+            return
         if (self._engine is not None
                 and self.config.user_message_hook is None
                 and hasattr(handle.connection, "_recv_q")):
@@ -750,21 +817,51 @@ class Broker:
 
     async def _gpu_tick_task(self) -> None:
         """Batch queued user messages through the kernel pipeline each tick,
-        then drain egress rings back to the user connections."""
+        then drain egress rings back to the user connections.  Queue items
+        are either per-message (Bytes, fwd) pairs (asyncio transports,
+        broker-plane inbound) or whole ingest blobs
+        ("blob", bytes, end_offsets, fwds) from the C++ pump path."""
         while True:
             item = await self._gpu_queue.get()
             batch = [item]
             while not self._gpu_queue.empty() and len(batch) < 4096:
                 batch.append(self._gpu_queue.get_nowait())
-            if self.connections.brokers:
-                for raw, fwd in batch:
-                    if fwd is not None:
-                        await self._forward_to_mesh(raw, fwd)
             buf = bytearray()
             offsets = [0]
-            for raw, _fwd in batch:
-                buf += raw.data
-                offsets.append(len(buf))
+            have_peers = bool(self.connections.brokers)
+            legacy = []  # Bytes to drop after the tick
+            for item in batch:
+                if item[0] == "blob":
+                    _tag, blob, ends, fwds = item
+                    if have_peers:
+                        fs = 0
+                        for i, fwd in enumerate(fwds):
+                            fe = ends[i]
+                            if fwd is not None:
+                                kind, s, e = fwd
+                                if kind == "b":
+                                    _u, brokers = self.connections.get_interested_by_topic(
+                                        list(blob[s:e]), to_users_only=False)
+                                    for b in brokers:
+                                        await self.try_send_to_broker(
+                                            b, Bytes(blob[fs:fe]))
+                                else:
+                                    owner = self.connections.get_broker_identifier_of_user(
+                                        blob[s:e])
+                                    if owner is not None and owner != self.identity:
+                                        await self.try_send_to_broker(
+                                            owner, Bytes(blob[fs:fe]))
+                            fs = fe
+                    base = len(buf)
+                    buf += blob
+                    offsets.extend(base + e for e in ends)
+                else:
+                    raw, fwd = item
+                    legacy.append(raw)
+                    if have_peers and fwd is not None:
+                        await self._forward_to_mesh(raw, fwd)
+                    buf += raw.data
+                    offsets.append(len(buf))
             # ingest staging from the bounded HBM pool: exhaustion WAITS
             # here (backpressure up through the tick queue to the sockets),
             # matching the reference limiter (protocols/mod.rs:328)
@@ -775,6 +872,6 @@ class Broker:
             await self._drain_egress()
             if pb is not None:
                 pb.drop()
-            for raw, _fwd in batch:
+            for raw in legacy:
                 raw.drop()
             await asyncio.sleep(self.config.gpu_tick_interval_s)

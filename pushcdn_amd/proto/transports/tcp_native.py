@@ -80,6 +80,7 @@ class PumpConnection(Connection):
         self._limiter = limiter
         self._closed = False
         self._dead = False
+        self._ingest = False
         self._recv_q: "asyncio.Queue[Bytes]" = asyncio.Queue()
         self._wakeup = asyncio.Event()
         mgr.conns[cid] = self
@@ -99,8 +100,43 @@ class PumpConnection(Connection):
 
     # called from the manager's eventfd callback
     def _pump_dirty(self) -> None:
+        if self._ingest:
+            self._wakeup.set()
+            return
         if self._poll_inbox():
             self._wakeup.set()
+
+    def enable_ingest(self) -> None:
+        """Switch this connection to the C++ ingest path: the pump
+        accumulates classified frames in one contiguous buffer and Python
+        pulls a whole tick's worth per call (recv_ingest_batch) — the
+        per-message interpreter round-trip disappears from the broker's
+        user plane."""
+        self._ingest = True
+        self._mgr.pump.set_ingest(self._cid)
+
+    async def recv_ingest_batch(self):
+        """(blob, end_offsets, discs, topics_off, topics_cnt, recip_off,
+        recip_len) — awaits until at least one frame arrived; raises when
+        the peer is gone and everything is drained."""
+        while True:
+            blob, offs, disc, toff, tcnt, roff, rlen, closed = \
+                self._mgr.pump.recv_ingest(self._cid)
+            if blob:
+                return blob, offs, disc, toff, tcnt, roff, rlen
+            if closed or self._dead or self._closed:
+                self._dead = True
+                raise ConnectionError_("connection reader closed")
+            self._wakeup.clear()
+            # re-check: a frame may have raced the notify
+            blob, offs, disc, toff, tcnt, roff, rlen, closed = \
+                self._mgr.pump.recv_ingest(self._cid)
+            if blob:
+                return blob, offs, disc, toff, tcnt, roff, rlen
+            if closed or self._dead or self._closed:
+                self._dead = True
+                raise ConnectionError_("connection reader closed")
+            await self._wakeup.wait()
 
     async def send_message_raw(self, raw: Bytes) -> None:
         size = len(raw.data)
