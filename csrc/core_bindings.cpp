@@ -378,7 +378,11 @@ PYBIND11_MODULE(pushcdn_core, m) {
         .def("send_ring", [](net::Pump& p, int64_t id, py::buffer ring, size_t wpos) {
             py::buffer_info info = ring.request();
             if (wpos > (size_t)info.size) throw std::runtime_error("wpos beyond ring");
-            auto r = p.send_ring(id, (const uint8_t*)info.ptr, wpos);
+            std::pair<int64_t, int64_t> r;
+            {
+                py::gil_scoped_release nogil;
+                r = p.send_ring(id, (const uint8_t*)info.ptr, wpos);
+            }
             return py::make_tuple(r.first, r.second);
         })
         .def("send_rings_batch",
@@ -390,6 +394,9 @@ PYBIND11_MODULE(pushcdn_core, m) {
                  for (size_t j = 0; j < ids.size(); ++j)
                      if (starts[j] < 0 || ends[j] < starts[j] || ends[j] > info.size)
                          throw std::runtime_error("send_rings_batch: range beyond buffer");
+                 // the coalescing memcpy can be hundreds of MB per tick —
+                 // never hold the GIL for it (it stalled the event loop)
+                 py::gil_scoped_release nogil;
                  return p.send_rings_batch((const uint8_t*)info.ptr, ids, starts, ends);
              },
              "batched tick drain: one call for all users' compacted rings")
@@ -410,6 +417,7 @@ PYBIND11_MODULE(pushcdn_core, m) {
         })
         .def("send_raw", [](net::Pump& p, int64_t id, py::buffer data) {
             py::buffer_info info = data.request();
+            py::gil_scoped_release nogil;
             return p.send_raw(id, (const char*)info.ptr, (size_t)info.size);
         })
         .def("send_backlog", &net::Pump::send_backlog)

@@ -218,40 +218,57 @@ public:
     // The payload IS a serialized wire message, so it goes out verbatim.
     // Returns (records enqueued, payload bytes) — (0,0) if the conn is gone.
     std::pair<int64_t, int64_t> send_ring(int64_t id, const uint8_t* ring, size_t wpos) {
+        std::string buf;
+        auto r = build_ring_frames(ring, wpos, &buf);
         std::lock_guard<std::mutex> g(mu_);
         auto it = conns_.find(id);
         if (it == conns_.end() || it->second.closed || it->second.soft_closing)
             return {0, 0};
-        auto r = enqueue_ring_locked(it->second, ring, wpos);
-        if (r.first) wake();
+        if (r.first) {
+            it->second.outbox.emplace_back(std::move(buf));
+            it->second.want_write = true;
+            wake();
+        }
         return r;
     }
 
-    // Batched egress drain: ONE call + ONE lock + ONE wake for a whole
-    // tick's worth of users.  `base` is the compacted staging buffer (K7
-    // gather of every used ring prefix, D2H'd in one copy); user j's
-    // records occupy [starts[j], ends[j]).  Per-user record counts are
-    // returned; -1 marks a connection that is gone (caller evicts), so the
-    // eviction-on-error contract (reference user/sender.rs:16-33) survives
-    // the batching.
+    // Batched egress drain: ONE call for a whole tick's worth of users.
+    // `base` is the compacted staging buffer (K7 gather of every used ring
+    // prefix, D2H'd in one copy); user j's records occupy
+    // [starts[j], ends[j]).  The coalesced per-user frame buffers are
+    // built OUTSIDE the pump mutex (and the caller releases the GIL), so
+    // neither the epoll thread nor the event loop stalls behind the
+    // memcpy; the lock is held only to push finished buffers.  Per-user
+    // record counts are returned; -1 marks a connection that is gone
+    // (caller evicts — reference user/sender.rs:16-33).
     std::vector<int64_t> send_rings_batch(const uint8_t* base,
                                           const std::vector<int64_t>& ids,
                                           const std::vector<int64_t>& starts,
                                           const std::vector<int64_t>& ends) {
-        std::vector<int64_t> counts(ids.size(), 0);
+        size_t n_users = ids.size();
+        std::vector<int64_t> counts(n_users, 0);
+        std::vector<std::string> bufs(n_users);
+        std::vector<int64_t> payload(n_users, 0);
+        for (size_t j = 0; j < n_users; ++j) {
+            auto r = build_ring_frames(base + starts[j],
+                                       (size_t)(ends[j] - starts[j]), &bufs[j]);
+            counts[j] = r.first;
+            payload[j] = r.second;
+        }
         bool any = false;
         {
             std::lock_guard<std::mutex> g(mu_);
-            for (size_t j = 0; j < ids.size(); ++j) {
+            for (size_t j = 0; j < n_users; ++j) {
                 auto it = conns_.find(ids[j]);
                 if (it == conns_.end() || it->second.closed || it->second.soft_closing) {
                     counts[j] = -1;
                     continue;
                 }
-                auto r = enqueue_ring_locked(it->second, base + starts[j],
-                                             (size_t)(ends[j] - starts[j]));
-                counts[j] = r.first;
-                any |= r.first > 0;
+                if (counts[j] > 0) {
+                    it->second.outbox.emplace_back(std::move(bufs[j]));
+                    it->second.want_write = true;
+                    any = true;
+                }
             }
         }
         if (any) wake();
@@ -411,27 +428,32 @@ private:
     }
 
     // parse one drained ring, restore per-tick arrival order, and COALESCE
-    // all frames into a single outbox entry (one big send() instead of one
+    // all frames into a single buffer (one big send() instead of one
     // syscall per delivery).  Ring write order is claim order and K5b's
     // atomic direct-delivery claims may interleave within a tick — the seq
     // header restores arrival order (wrap-aware), same as the Python drain
-    // (gpu_engine.parse_ring_records).
-    std::pair<int64_t, int64_t> enqueue_ring_locked(Conn& c, const uint8_t* ring,
-                                                    size_t wpos) {
+    // (gpu_engine.parse_ring_records).  Broadcast-only rings are already
+    // in order, so the common case skips the sort entirely.  No lock
+    // needed: reads caller memory, writes caller-owned `out`.
+    static std::pair<int64_t, int64_t> build_ring_frames(const uint8_t* ring, size_t wpos,
+                                                         std::string* out) {
         int64_t n = 0, payload_bytes = 0;
         size_t pos = 0;
         std::vector<std::pair<uint32_t, std::pair<size_t, uint32_t>>> recs;  // seq -> (off, len)
+        bool ordered = true;
         while (pos + 16 <= wpos) {
             uint32_t len, seq;
             memcpy(&len, ring + pos, 4);
             memcpy(&seq, ring + pos + 4, 4);
             if (len > kMaxMessageSize || pos + 16 + len > wpos) break;
+            if (!recs.empty() && (uint32_t)(seq - recs.back().first) > 0x80000000u)
+                ordered = false;
             recs.push_back({seq, {pos + 16, len}});
             ++n;
             payload_bytes += (int64_t)len;
             pos += 16 + (((size_t)len + 15) & ~(size_t)15);
         }
-        if (recs.size() > 1) {
+        if (!ordered && recs.size() > 1) {
             uint32_t base = recs[0].first;
             for (auto& r : recs) if (r.first - base > 0x80000000u) base = r.first;
             std::stable_sort(recs.begin(), recs.end(),
@@ -440,17 +462,14 @@ private:
                              });
         }
         if (n) {
-            std::string coalesced;
-            coalesced.resize((size_t)payload_bytes + 4 * (size_t)n);
+            out->resize((size_t)payload_bytes + 4 * (size_t)n);
             size_t w = 0;
             for (auto& r : recs) {
                 uint32_t be = htonl(r.second.second);
-                memcpy(&coalesced[w], &be, 4);
-                memcpy(&coalesced[w + 4], ring + r.second.first, r.second.second);
+                memcpy(&(*out)[w], &be, 4);
+                memcpy(&(*out)[w + 4], ring + r.second.first, r.second.second);
                 w += 4 + r.second.second;
             }
-            c.outbox.emplace_back(std::move(coalesced));
-            c.want_write = true;
         }
         return {n, payload_bytes};
     }

@@ -773,7 +773,11 @@ class Broker:
                     continue
             fallback.append((slot, pubkey, n))
         if batch_entries:
-            counts = pump.send_rings_batch(
+            # off-loop: the coalescing memcpy releases the GIL in C++, so
+            # running it on an executor thread keeps the event loop (and
+            # with it the ingest notify path) fully responsive
+            counts = await asyncio.get_running_loop().run_in_executor(
+                None, pump.send_rings_batch,
                 staging.numpy(),
                 [e[1] for e in batch_entries],
                 [e[2] for e in batch_entries],
@@ -821,8 +825,13 @@ class Broker:
         are either per-message (Bytes, fwd) pairs (asyncio transports,
         broker-plane inbound) or whole ingest blobs
         ("blob", bytes, end_offsets, fwds) from the C++ pump path."""
+        import time as _time
+
+        ts = self.tick_stats = {"assemble": 0.0, "ingest": 0.0, "tick": 0.0,
+                                "drain": 0.0, "sleep": 0.0, "msgs": 0, "ticks": 0}
         while True:
             item = await self._gpu_queue.get()
+            t0 = _time.perf_counter()
             batch = [item]
             while not self._gpu_queue.empty() and len(batch) < 4096:
                 batch.append(self._gpu_queue.get_nowait())
@@ -862,16 +871,28 @@ class Broker:
                         await self._forward_to_mesh(raw, fwd)
                     buf += raw.data
                     offsets.append(len(buf))
+            t1 = _time.perf_counter()
             # ingest staging from the bounded HBM pool: exhaustion WAITS
             # here (backpressure up through the tick queue to the sockets),
             # matching the reference limiter (protocols/mod.rs:328)
             pb = await self._hbm_pool.alloc(len(buf)) if self._hbm_pool else None
+            host = bytes(buf)
             dbuf, doff = self._engine.ingest(
-                bytes(buf), offsets, staging=pb.tensor if pb else None)
-            self._engine.tick(dbuf, doff, host_batch=bytes(buf), host_offsets=offsets)
+                host, offsets, staging=pb.tensor if pb else None)
+            t2 = _time.perf_counter()
+            self._engine.tick(dbuf, doff, host_batch=host, host_offsets=offsets)
+            t3 = _time.perf_counter()
             await self._drain_egress()
+            t4 = _time.perf_counter()
             if pb is not None:
                 pb.drop()
             for raw in legacy:
                 raw.drop()
+            ts["assemble"] += t1 - t0
+            ts["ingest"] += t2 - t1
+            ts["tick"] += t3 - t2
+            ts["drain"] += t4 - t3
+            ts["msgs"] += len(offsets) - 1
+            ts["ticks"] += 1
             await asyncio.sleep(self.config.gpu_tick_interval_s)
+            ts["sleep"] += _time.perf_counter() - t4

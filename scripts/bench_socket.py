@@ -236,6 +236,12 @@ async def run_coordinator(args) -> None:
         "lat_samples": len(lats),
         "seconds": args.seconds,
     }
+    ts = getattr(broker, "tick_stats", None)
+    if ts and ts["ticks"]:
+        result["broker_tick_stats"] = {
+            k: (round(v / ts["ticks"] * 1e3, 3) if isinstance(v, float) else v)
+            for k, v in ts.items()
+        }
     print(json.dumps(result), flush=True)
     out = Path("gpurun_out")
     out.mkdir(exist_ok=True)
