@@ -85,8 +85,9 @@ async def run_subscriber(args) -> None:
     for c in clients:
         c.close()
     lats.sort()
-    step = max(1, len(lats) // 2000)
-    print(json.dumps({"count": count, "lats": lats[::step]}), flush=True)
+    step = max(1, len(lats) // 1000)
+    sampled = [round(x, 6) for x in lats[::step]][:1200]
+    print(json.dumps({"count": count, "lats": sampled}), flush=True)
 
 
 async def run_sender(args) -> None:
@@ -180,7 +181,8 @@ async def run_coordinator(args) -> None:
         p = await asyncio.create_subprocess_exec(
             sys.executable, __file__, "--role", "sub", "--endpoint", ep,
             "--clients", str(n), "--seed", str(seed),
-            stdin=asyncio.subprocess.PIPE, stdout=asyncio.subprocess.PIPE)
+            stdin=asyncio.subprocess.PIPE, stdout=asyncio.subprocess.PIPE,
+            limit=32 << 20)
         sub_procs.append(p)
         seed += n
 
