@@ -452,18 +452,34 @@ class GpuBrokerEngine:
         if total == 0:
             self.ring_wpos.zero_()
             return wpos, offsets, torch.empty(0, dtype=torch.uint8)
+        # DOUBLE-buffered host staging: the caller may still be writing the
+        # previous drain's frames to sockets (pipelined drains) while this
+        # tick compacts into the other buffer; callers must not reuse a
+        # buffer until its drain completed (see service._drain_egress)
+        idx = self.next_staging_index()
+        self._staging_flip = idx ^ 1
         if self._staging_dev is None or self._staging_dev.numel() < total:
             cap = max(total, 1 << 22)
             self._staging_dev = torch.empty(cap, dtype=torch.uint8, device=self.device)
-            self._staging_host = torch.empty(cap, dtype=torch.uint8,
-                                             pin_memory=self.is_cuda)
+            self._staging_host = [
+                torch.empty(cap, dtype=torch.uint8, pin_memory=self.is_cuda)
+                for _ in range(2)
+            ]
         dst_off = offsets[: self.n_users].to(self.device, non_blocking=True)
         max_chunks = (int(wpos.max()) + (64 << 10) - 1) // (64 << 10)
         self._ops.compact_rings(self.egress, self.ring_bytes, wpos_dev, dst_off,
                                 self._staging_dev, max_chunks)
         self.ring_wpos.zero_()  # stream-ordered after K7's reads
-        self._staging_host[:total].copy_(self._staging_dev[:total])  # sync D2H
-        return wpos, offsets, self._staging_host[:total]
+        host = self._staging_host[idx]
+        if host.numel() < total:  # staging grew since this buffer was made
+            host = self._staging_host[idx] = torch.empty(
+                self._staging_dev.numel(), dtype=torch.uint8, pin_memory=self.is_cuda)
+        host[:total].copy_(self._staging_dev[:total])  # sync D2H
+        return wpos, offsets, host[:total]
+
+    def next_staging_index(self) -> int:
+        """Which host staging buffer the NEXT drain_compact will fill."""
+        return getattr(self, "_staging_flip", 0)
 
     def read_ring(self, user_idx: int, nbytes: Optional[int] = None) -> bytes:
         n = self.ring_bytes if nbytes is None else nbytes

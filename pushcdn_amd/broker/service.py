@@ -134,6 +134,8 @@ class Broker:
         self._engine = None
         self._gpu_queue: Optional[asyncio.Queue] = None
         self._free_gpu_slots: List[int] = []
+        self._drain_by_buffer: Dict[int, asyncio.Task] = {}
+        self._last_drain: Optional[asyncio.Task] = None
         if config.data_plane == "gpu":
             from .gpu_engine import GpuBrokerEngine
 
@@ -742,9 +744,30 @@ class Broker:
         buffer (one D2H), then hand the WHOLE tick to the C++ pump in one
         send_rings_batch call.  Per-user fallback (memory/asyncio transports
         or a foreign pump) parses records in Python as before.  Eviction on
-        delivery failure is preserved (reference user/sender.rs:16-33)."""
+        delivery failure is preserved (reference user/sender.rs:16-33).
+
+        Drains are PIPELINED: the socket dispatch of tick N runs as a
+        background task (chained after N-1 so per-user frame order holds)
+        while tick N+1 ingests and routes; host staging is double-buffered
+        and a buffer is only reused once its drain completed."""
+        idx = self._engine.next_staging_index()
+        busy = self._drain_by_buffer.get(idx)
+        if busy is not None:
+            await asyncio.shield(busy)  # buffer still being written out
         wpos, offsets, staging = self._engine.drain_compact()
-        await self._dispatch_egress(wpos, offsets, staging)
+        prev = self._last_drain
+
+        async def _dispatch_chained():
+            if prev is not None:
+                try:
+                    await asyncio.shield(prev)
+                except Exception:
+                    pass
+            await self._dispatch_egress(wpos, offsets, staging)
+
+        task = asyncio.get_running_loop().create_task(_dispatch_chained())
+        self._drain_by_buffer[idx] = task
+        self._last_drain = task
 
     async def _dispatch_egress(self, wpos, offsets, staging) -> None:
         from .gpu_engine import parse_ring_records
@@ -773,19 +796,27 @@ class Broker:
                     continue
             fallback.append((slot, pubkey, n))
         if batch_entries:
-            # off-loop: the coalescing memcpy releases the GIL in C++, so
-            # running it on an executor thread keeps the event loop (and
-            # with it the ingest notify path) fully responsive
-            counts = await asyncio.get_running_loop().run_in_executor(
-                None, pump.send_rings_batch,
-                staging.numpy(),
-                [e[1] for e in batch_entries],
-                [e[2] for e in batch_entries],
-                [e[3] for e in batch_entries],
-            )
-            for (pubkey, _cid, _s, _e), cnt in zip(batch_entries, counts):
-                if cnt < 0:
-                    await self.remove_user(pubkey)
+            # off-loop AND parallel: the coalescing memcpy releases the GIL
+            # in C++ (frame building is lock-free), so sharding the users
+            # across executor threads scales the drain with cores while the
+            # event loop stays responsive
+            loop = asyncio.get_running_loop()
+            view = staging.numpy()
+            nshards = min(4, len(batch_entries))
+            per = (len(batch_entries) + nshards - 1) // nshards
+            chunks = [batch_entries[i:i + per]
+                      for i in range(0, len(batch_entries), per)]
+            results = await asyncio.gather(*(
+                loop.run_in_executor(
+                    None, pump.send_rings_batch, view,
+                    [e[1] for e in chunk],
+                    [e[2] for e in chunk],
+                    [e[3] for e in chunk],
+                ) for chunk in chunks))
+            for chunk, counts in zip(chunks, results):
+                for (pubkey, _cid, _s, _e), cnt in zip(chunk, counts):
+                    if cnt < 0:
+                        await self.remove_user(pubkey)
         for slot, pubkey, n in fallback:
             ring = bytes(staging[int(offsets[slot]):int(offsets[slot + 1])].numpy()
                          .tobytes())
