@@ -772,9 +772,8 @@ class Broker:
     async def _dispatch_egress(self, wpos, offsets, staging) -> None:
         from .gpu_engine import parse_ring_records
 
-        batch_entries = []  # (pubkey, cid, start, end)
+        by_pump = {}        # pump -> [(pubkey, cid, start, end)]
         fallback = []       # (slot, pubkey, nbytes)
-        pump = None
         for slot, pubkey in list(self._gpu_user_by_slot.items()):
             n = int(wpos[slot])
             if n == 0:
@@ -788,32 +787,32 @@ class Broker:
                 except Exception:
                     await self.remove_user(pubkey)
                     continue
-                if pump is None:
-                    pump = p
-                if p is pump:
-                    batch_entries.append(
-                        (pubkey, cid, int(offsets[slot]), int(offsets[slot + 1])))
-                    continue
+                by_pump.setdefault(p, []).append(
+                    (pubkey, cid, int(offsets[slot]), int(offsets[slot + 1])))
+                continue
             fallback.append((slot, pubkey, n))
-        if batch_entries:
+        if by_pump:
             # off-loop AND parallel: the coalescing memcpy releases the GIL
             # in C++ (frame building is lock-free), so sharding the users
-            # across executor threads scales the drain with cores while the
-            # event loop stays responsive
+            # across executor threads — and across pump shards
+            # (PUSHCDN_PUMP_SHARDS epoll threads) — scales the drain with
+            # cores while the event loop stays responsive
             loop = asyncio.get_running_loop()
             view = staging.numpy()
-            nshards = min(4, len(batch_entries))
-            per = (len(batch_entries) + nshards - 1) // nshards
-            chunks = [batch_entries[i:i + per]
-                      for i in range(0, len(batch_entries), per)]
-            results = await asyncio.gather(*(
-                loop.run_in_executor(
-                    None, pump.send_rings_batch, view,
-                    [e[1] for e in chunk],
-                    [e[2] for e in chunk],
-                    [e[3] for e in chunk],
-                ) for chunk in chunks))
-            for chunk, counts in zip(chunks, results):
+            jobs = []   # (chunk, future)
+            for pump, entries in by_pump.items():
+                nshards = min(4, len(entries))
+                per = (len(entries) + nshards - 1) // nshards
+                for i in range(0, len(entries), per):
+                    chunk = entries[i:i + per]
+                    jobs.append((chunk, loop.run_in_executor(
+                        None, pump.send_rings_batch, view,
+                        [e[1] for e in chunk],
+                        [e[2] for e in chunk],
+                        [e[3] for e in chunk],
+                    )))
+            results = await asyncio.gather(*(f for _c, f in jobs))
+            for (chunk, _f), counts in zip(jobs, results):
                 for (pubkey, _cid, _s, _e), cnt in zip(chunk, counts):
                     if cnt < 0:
                         await self.remove_user(pubkey)

@@ -35,15 +35,32 @@ def _get_core():
 
 
 class _PumpManager:
-    """One C++ pump + one eventfd watcher per event loop."""
+    """C++ pump shard(s) + one eventfd watcher each, per event loop.
+
+    PUSHCDN_PUMP_SHARDS (default 1) sets how many epoll threads serve this
+    process's connections, round-robin assigned: a broker pushing multiple
+    GB/s of egress spreads the send() work across shards; client processes
+    keep the single-thread default."""
 
     _by_loop: "Dict[int, _PumpManager]" = {}
 
     def __init__(self, loop: asyncio.AbstractEventLoop) -> None:
-        self.pump = _get_core().Pump()
+        import os
+
+        n = max(1, int(os.environ.get("PUSHCDN_PUMP_SHARDS", "1")))
+        core = _get_core()
+        self.pumps = [core.Pump() for _ in range(n)]
         self.loop = loop
-        self.conns: Dict[int, "PumpConnection"] = {}
-        loop.add_reader(self.pump.notify_fd(), self._on_notify)
+        # conns keyed per shard: connection ids are per-pump
+        self.conns: "list[Dict[int, PumpConnection]]" = [{} for _ in range(n)]
+        self._next_shard = 0
+        for i, p in enumerate(self.pumps):
+            loop.add_reader(p.notify_fd(), self._on_notify, i)
+
+    # single-shard compatibility accessor
+    @property
+    def pump(self):
+        return self.pumps[0]
 
     @classmethod
     def current(cls) -> "_PumpManager":
@@ -54,28 +71,35 @@ class _PumpManager:
             # reap pumps whose loops are gone (tests create many loops)
             for k, old in list(cls._by_loop.items()):
                 if old.loop.is_closed():
-                    old.pump.stop()
+                    for p in old.pumps:
+                        p.stop()
                     del cls._by_loop[k]
             mgr = cls(loop)
             cls._by_loop[key] = mgr
         return mgr
 
-    def _on_notify(self) -> None:
-        for cid in self.pump.poll_dirty():
-            conn = self.conns.get(cid)
+    def _on_notify(self, shard: int) -> None:
+        for cid in self.pumps[shard].poll_dirty():
+            conn = self.conns[shard].get(cid)
             if conn is not None:
                 conn._pump_dirty()
 
-    def attach(self, sock: socket.socket) -> int:
-        fd = sock.detach()  # the pump owns the fd now
-        return self.pump.add(fd)
+    def attach(self, sock: socket.socket):
+        """-> (shard index, conn id); the pump owns the fd now."""
+        fd = sock.detach()
+        shard = self._next_shard
+        self._next_shard = (shard + 1) % len(self.pumps)
+        return shard, self.pumps[shard].add(fd)
 
 
 class PumpConnection(Connection):
     """Connection whose data path lives in the C++ pump."""
 
-    def __init__(self, mgr: _PumpManager, cid: int, limiter: Limiter) -> None:
+    def __init__(self, mgr: _PumpManager, shard: int, cid: int,
+                 limiter: Limiter) -> None:
         self._mgr = mgr
+        self._pump = mgr.pumps[shard]
+        self._shard = shard
         self._cid = cid
         self._limiter = limiter
         self._closed = False
@@ -83,11 +107,11 @@ class PumpConnection(Connection):
         self._ingest = False
         self._recv_q: "asyncio.Queue[Bytes]" = asyncio.Queue()
         self._wakeup = asyncio.Event()
-        mgr.conns[cid] = self
+        mgr.conns[shard][cid] = self
 
     def _poll_inbox(self) -> bool:
         """Drain the C++ inbox into the asyncio queue (no signaling)."""
-        frames, closed = self._mgr.pump.recv_batch(self._cid, RECV_BATCH)
+        frames, closed = self._pump.recv_batch(self._cid, RECV_BATCH)
         total = 0
         for f in frames:
             total += len(f)
@@ -113,7 +137,7 @@ class PumpConnection(Connection):
         per-message interpreter round-trip disappears from the broker's
         user plane."""
         self._ingest = True
-        self._mgr.pump.set_ingest(self._cid)
+        self._pump.set_ingest(self._cid)
 
     async def recv_ingest_batch(self):
         """(blob, end_offsets, discs, topics_off, topics_cnt, recip_off,
@@ -121,7 +145,7 @@ class PumpConnection(Connection):
         the peer is gone and everything is drained."""
         while True:
             blob, offs, disc, toff, tcnt, roff, rlen, closed = \
-                self._mgr.pump.recv_ingest(self._cid)
+                self._pump.recv_ingest(self._cid)
             if blob:
                 return blob, offs, disc, toff, tcnt, roff, rlen
             if closed or self._dead or self._closed:
@@ -130,7 +154,7 @@ class PumpConnection(Connection):
             self._wakeup.clear()
             # re-check: a frame may have raced the notify
             blob, offs, disc, toff, tcnt, roff, rlen, closed = \
-                self._mgr.pump.recv_ingest(self._cid)
+                self._pump.recv_ingest(self._cid)
             if blob:
                 return blob, offs, disc, toff, tcnt, roff, rlen
             if closed or self._dead or self._closed:
@@ -141,13 +165,13 @@ class PumpConnection(Connection):
     async def send_message_raw(self, raw: Bytes) -> None:
         size = len(raw.data)
         try:
-            ok = self._mgr.pump.send(self._cid, raw.data)
+            ok = self._pump.send(self._cid, raw.data)
         finally:
             raw.drop()
         if not ok:
             raise ConnectionError_("connection writer closed")
         BYTES_SENT.inc(size)
-        while self._mgr.pump.send_backlog(self._cid) > SEND_HWM_BYTES:
+        while self._pump.send_backlog(self._cid) > SEND_HWM_BYTES:
             await asyncio.sleep(0.001)
 
     async def recv_message_raw(self) -> Bytes:
@@ -169,7 +193,7 @@ class PumpConnection(Connection):
         egress goes out in ONE C++ call."""
         if self._closed or self._dead:
             raise ConnectionError_("connection writer closed")
-        return self._mgr.pump, self._cid
+        return self._pump, self._cid
 
     def send_ring_records(self, ring: bytes, wpos: int) -> int:
         """Egress fast path for the GPU broker drain: hand a drained ring
@@ -178,7 +202,7 @@ class PumpConnection(Connection):
         call per (user, tick) instead of one per delivery."""
         if self._closed or self._dead:
             raise ConnectionError_("connection writer closed")
-        n, payload_bytes = self._mgr.pump.send_ring(self._cid, ring, wpos)
+        n, payload_bytes = self._pump.send_ring(self._cid, ring, wpos)
         if payload_bytes:
             BYTES_SENT.inc(payload_bytes)
         return n
@@ -187,18 +211,18 @@ class PumpConnection(Connection):
         if self._closed:
             return
         self._closed = True
-        self._mgr.pump.soft_close(self._cid)
+        self._pump.soft_close(self._cid)
         self._release()
 
     def close(self) -> None:
         if not self._closed:
             self._closed = True
-            self._mgr.pump.hard_close(self._cid)
+            self._pump.hard_close(self._cid)
         self._release()
 
     def _release(self) -> None:
-        self._mgr.conns.pop(self._cid, None)
-        self._mgr.pump.forget(self._cid)
+        self._mgr.conns[self._shard].pop(self._cid, None)
+        self._pump.forget(self._cid)
         self._dead = True
         self._wakeup.set()
 
@@ -209,8 +233,8 @@ class TcpNativeUnfinalized(UnfinalizedConnection):
 
     async def finalize(self, limiter: Limiter) -> Connection:
         mgr = _PumpManager.current()
-        cid = mgr.attach(self._sock)
-        return PumpConnection(mgr, cid, limiter)
+        shard, cid = mgr.attach(self._sock)
+        return PumpConnection(mgr, shard, cid, limiter)
 
 
 class TcpNativeListener(Listener):
@@ -244,8 +268,8 @@ class TcpNative(Protocol):
             sock.close()
             raise ConnectionError_(f"failed to connect to {endpoint}: {e}") from e
         mgr = _PumpManager.current()
-        cid = mgr.attach(sock)
-        return PumpConnection(mgr, cid, limiter)
+        shard, cid = mgr.attach(sock)
+        return PumpConnection(mgr, shard, cid, limiter)
 
     @classmethod
     async def bind(cls, endpoint: str, certificate=None, key=None) -> TcpNativeListener:

@@ -130,6 +130,9 @@ async def run_sender(args) -> None:
 
 
 async def run_coordinator(args) -> None:
+    import os
+
+    os.environ.setdefault("PUSHCDN_PUMP_SHARDS", str(args.pump_shards))
     import torch
 
     from pushcdn_amd.broker.service import Broker, BrokerConfig
@@ -271,6 +274,7 @@ def main():
     p.add_argument("--t0", type=float, default=0)
     p.add_argument("--t1", type=float, default=0)
     p.add_argument("--tag", default="")
+    p.add_argument("--pump-shards", type=int, default=4)
     args = p.parse_args()
     if args.role == "sub":
         asyncio.run(run_subscriber(args))
