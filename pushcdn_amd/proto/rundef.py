@@ -48,6 +48,21 @@ def production_run_def(discovery_endpoint: str) -> RunDef:
     )
 
 
+def quic_run_def(discovery_endpoint: str) -> RunDef:
+    """BLS + TCP (brokers) + QUIC-profile (users) + Redis/KeyDB — the
+    reference's QUIC-capable generics (def.rs:101-136 are generic over
+    Protocol; quic.rs is a first-class user transport).  See
+    transports/quic.py for the QUIC-profile scope note."""
+    from .transports.quic import Quic
+
+    return RunDef(
+        broker=ConnectionDef(protocol=Tcp),
+        user=ConnectionDef(protocol=Quic),
+        discovery_endpoint=discovery_endpoint,
+        topic_space=ALL_TOPICS,
+    )
+
+
 def testing_run_def(discovery_endpoint: str) -> RunDef:
     """Memory transports + embedded SQLite (reference TestingRunDef,
     def.rs:140-159)."""

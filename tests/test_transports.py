@@ -639,3 +639,158 @@ def test_queued_messages_drain_after_peer_close():
         assert got == [b"one", b"two"]
 
     run(go())
+
+
+def test_quic_conformance():
+    """The QUIC-profile transport (reliable TLS stream over UDP) passes the
+    same conformance contract (reference quic.rs:279-298)."""
+    from pushcdn_amd.proto.transports.quic import Quic
+
+    run(_conformance(Quic, "127.0.0.1:0"))
+
+
+def test_quic_rejects_untrusted_ca(tmp_path):
+    """A client with a DIFFERENT trust root must refuse the server's cert —
+    same trust model as TcpTls (reference tls.rs:100-126)."""
+    from pushcdn_amd.crypto import tls as tlslib
+    from pushcdn_amd.proto.transports.quic import Quic
+
+    async def go():
+        limiter = Limiter(global_memory_pool_size=1 << 20)
+        listener = await Quic.bind("127.0.0.1:0", None, None)
+        ep = f"127.0.0.1:{listener.port}"
+
+        async def server():
+            try:
+                unf = await asyncio.wait_for(listener.accept(), 8)
+                await unf.finalize(limiter)
+            except Exception:
+                pass
+
+        st = asyncio.get_running_loop().create_task(server())
+        other_ca_cert, other_ca_key = tlslib.generate_ca(str(tmp_path / "otherca"))
+
+        class UntrustingClient(Quic):
+            ca_cert_path = other_ca_cert
+            ca_key_path = other_ca_key
+
+        from pushcdn_amd.proto.errors import ConnectionError_
+        try:
+            await UntrustingClient.connect(ep, False, limiter)
+            raise AssertionError("handshake against the wrong CA succeeded")
+        except ConnectionError_:
+            pass
+        st.cancel()
+        await listener.close()
+
+    run(go())
+
+
+def test_quic_packet_loss_recovery():
+    """Retransmission: drop 20% of datagrams in each direction and the
+    stream still delivers everything in order."""
+    import random
+
+    from pushcdn_amd.proto.transports import quic as quicmod
+
+    async def go():
+        rng = random.Random(42)
+        orig = quicmod._QuicEndpoint.send_pkt
+
+        def lossy(self, addr, ptype, cid, payload):
+            # never drop the handshake/teardown control packets, only data
+            if ptype in (quicmod.PKT_STREAM, quicmod.PKT_ACK) and rng.random() < 0.2:
+                return
+            orig(self, addr, ptype, cid, payload)
+
+        quicmod._QuicEndpoint.send_pkt = lossy
+        try:
+            limiter = Limiter(global_memory_pool_size=1 << 22)
+            listener = await quicmod.Quic.bind("127.0.0.1:0", None, None)
+            ep = f"127.0.0.1:{listener.port}"
+
+            async def server():
+                unf = await listener.accept()
+                conn = await unf.finalize(limiter)
+                for i in range(20):
+                    msg = await conn.recv_message()
+                    assert msg.message == bytes([i]) * 3000, f"msg {i}"
+                await conn.send_message(m.Direct(b"s", b"all-received"))
+                await conn.soft_close()
+
+            st = asyncio.get_running_loop().create_task(server())
+            conn = await quicmod.Quic.connect(ep, True, limiter)
+            for i in range(20):
+                await conn.send_message(m.Direct(b"c", bytes([i]) * 3000))
+            reply = await asyncio.wait_for(conn.recv_message(), 30)
+            assert reply.message == b"all-received"
+            await conn.soft_close()
+            await st
+            await listener.close()
+        finally:
+            quicmod._QuicEndpoint.send_pkt = orig
+
+    run(go())
+
+
+def test_quic_full_service_stack(tmp_path):
+    """marshal + broker + two clients entirely over the QUIC transport:
+    auth/permits, subscribe, broadcast and direct round-trips."""
+    import uuid as _uuid
+
+    from pushcdn_amd.broker.service import Broker, BrokerConfig
+    from pushcdn_amd.client import Client, ClientConfig
+    from pushcdn_amd.crypto import bls
+    from pushcdn_amd.discovery import BrokerIdentifier
+    from pushcdn_amd.marshal import Marshal, MarshalConfig
+    from pushcdn_amd.proto.transports.quic import Quic
+
+    async def go():
+        db = str(tmp_path / f"quic-{_uuid.uuid4().hex}.db")
+        broker = Broker(BrokerConfig(
+            public_bind_endpoint="127.0.0.1:0",
+            public_advertise_endpoint="127.0.0.1:0",
+            private_bind_endpoint="127.0.0.1:0",
+            private_advertise_endpoint="127.0.0.1:0",
+            discovery_endpoint=db,
+            keypair=bls.KeyPair.from_seed(1000),
+            user_protocol=Quic,
+            broker_protocol=Quic,
+        ))
+        await broker.start()
+        pub = f"127.0.0.1:{broker._user_listener.port}"
+        priv = f"127.0.0.1:{broker._broker_listener.port}"
+        broker.config.public_advertise_endpoint = pub
+        broker.config.private_advertise_endpoint = priv
+        broker.identity = BrokerIdentifier(pub, priv)
+        broker.discovery.identity = broker.identity
+        broker.connections.identity = broker.identity
+        await broker.discovery.perform_heartbeat(0, 600)
+        marshal = Marshal(MarshalConfig(bind_endpoint="127.0.0.1:0",
+                                        discovery_endpoint=db, protocol=Quic))
+        await marshal.start()
+        ep = f"127.0.0.1:{marshal._listener.port}"
+
+        alice = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(91),
+                                    subscribed_topics=[2], protocol=Quic))
+        bob = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(92),
+                                  subscribed_topics=[2], protocol=Quic))
+        await alice.ensure_initialized()
+        await bob.ensure_initialized()
+        await asyncio.sleep(0.2)
+
+        await alice.send_broadcast_message([2], b"quic-broadcast")
+        msg = await asyncio.wait_for(bob.receive_message(), timeout=10)
+        assert msg.message == b"quic-broadcast"
+        await bob.send_direct_message(alice.public_key, b"quic-direct")
+        got = await asyncio.wait_for(alice.receive_message(), timeout=10)
+        while got.message == b"quic-broadcast":  # alice's own echo first
+            got = await asyncio.wait_for(alice.receive_message(), timeout=10)
+        assert got.message == b"quic-direct"
+
+        alice.close()
+        bob.close()
+        await marshal.close()
+        await broker.close()
+
+    run(go())
