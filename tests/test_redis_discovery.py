@@ -6,6 +6,9 @@ heartbeats, least-connections placement counting outstanding permits,
 one-time permits, whitelist (empty set = allow all)."""
 
 import asyncio
+import shutil
+
+import pytest
 import time
 
 
@@ -214,3 +217,78 @@ def test_whitelist_semantics():
         await srv.close()
 
     run(go())
+
+
+@pytest.mark.skipif(
+    not (shutil.which("redis-server") or shutil.which("keydb-server")),
+    reason="no redis/keydb server binary in this image (opt-in lane; the "
+           "deploy compose stands up eqalpha/keydb)")
+def test_redis_discovery_against_real_server(tmp_path):
+    """Drives RedisDiscovery against a REAL redis/keydb server (opt-in):
+    heartbeat TTL expiry, least-connections scan, broker-scoped GETDEL
+    permits, and whitelist semantics (reference redis.rs:81-326)."""
+    import socket as _socket
+    import subprocess
+    import time as _time
+
+    server_bin = shutil.which("keydb-server") or shutil.which("redis-server")
+    s = _socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    proc = subprocess.Popen(
+        [server_bin, "--port", str(port), "--save", "", "--appendonly", "no",
+         "--dir", str(tmp_path)],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    try:
+        deadline = _time.time() + 10
+        while _time.time() < deadline:
+            try:
+                probe = _socket.create_connection(("127.0.0.1", port), timeout=0.5)
+                probe.close()
+                break
+            except OSError:
+                _time.sleep(0.1)
+
+        async def go():
+            from pushcdn_amd.discovery import BrokerIdentifier
+            from pushcdn_amd.discovery.redis import RedisDiscovery
+
+            a = BrokerIdentifier("a-pub", "a-priv")
+            b = BrokerIdentifier("b-pub", "b-priv")
+            da = RedisDiscovery(f"redis://127.0.0.1:{port}", a)
+            db = RedisDiscovery(f"redis://127.0.0.1:{port}", b)
+
+            # heartbeat + least-connections
+            await da.perform_heartbeat(5, 60)
+            await db.perform_heartbeat(1, 60)
+            assert await da.get_with_least_connections() == b
+            assert await da.get_other_brokers() == {b}
+
+            # permits: broker-scoped, one-shot GETDEL
+            permit = await da.issue_permit(b, 30, b"user-pk")
+            assert permit > 1
+            assert await db.validate_permit(a, permit) is None  # wrong broker
+            assert await db.validate_permit(b, permit) == b"user-pk"
+            assert await db.validate_permit(b, permit) is None  # consumed
+
+            # permit load influences placement (conns + outstanding permits)
+            for _ in range(10):
+                await da.issue_permit(b, 30, b"u")
+            assert await da.get_with_least_connections() == a
+
+            # whitelist: empty = allow all; set restricts
+            assert await da.check_whitelist(b"anyone")
+            await da.set_whitelist([b"alice"])
+            assert await da.check_whitelist(b"alice")
+            assert not await da.check_whitelist(b"mallory")
+
+            # heartbeat TTL expiry prunes dead brokers
+            await da.perform_heartbeat(0, 1)
+            await asyncio.sleep(1.5)
+            assert a not in await db.get_other_brokers()
+
+        asyncio.run(asyncio.wait_for(go(), 30))
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)
