@@ -41,6 +41,9 @@ void launch_k7_compact_rings(const uint8_t*, int64_t, const int64_t*, const int6
                              uint8_t*, int32_t, int32_t, hipStream_t);
 void launch_k1_bls_verify2(const uint8_t*, const uint8_t*, uint8_t*, const int64_t*,
                            const uint8_t*, int32_t, int32_t*, hipStream_t);
+void launch_k1_bls_verify_wave(const uint8_t*, const uint8_t*, uint8_t*, const int64_t*,
+                               const uint8_t*, const uint64_t*, int32_t, int32_t*,
+                               hipStream_t);
 void launch_k3_fanout_wave(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
                            const uint32_t*, const int32_t*, uint8_t*, int, int, hipStream_t);
 void launch_k3_fanout_flat2(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
@@ -250,6 +253,27 @@ torch::Tensor bls_verify_batch2(torch::Tensor vks, torch::Tensor sigs, torch::Te
     return ok;
 }
 
+torch::Tensor bls_verify_batch_wave(torch::Tensor vks, torch::Tensor sigs,
+                                    torch::Tensor msgs, torch::Tensor moff,
+                                    torch::Tensor g2_lines, torch::Tensor rand_r) {
+    CHECK_DEV(vks); CHECK_DEV(sigs); CHECK_DEV(msgs); CHECK_DEV(moff); CHECK_DEV(g2_lines);
+    CHECK_DEV(rand_r);
+    CHECK_CONTIG(vks); CHECK_CONTIG(sigs); CHECK_CONTIG(msgs); CHECK_CONTIG(moff);
+    CHECK_CONTIG(g2_lines); CHECK_CONTIG(rand_r);
+    int32_t N = (int32_t)moff.size(0) - 1;
+    TORCH_CHECK(vks.numel() == (int64_t)N * 128 && sigs.numel() == (int64_t)N * 64);
+    TORCH_CHECK(rand_r.numel() >= N && rand_r.dtype() == torch::kInt64);
+    auto ok = torch::zeros({N}, torch::TensorOptions().dtype(torch::kInt32).device(vks.device()));
+    if (N > 0) {
+        launch_k1_bls_verify_wave(vks.data_ptr<uint8_t>(), sigs.data_ptr<uint8_t>(),
+                                  msgs.data_ptr<uint8_t>(), moff.data_ptr<int64_t>(),
+                                  g2_lines.data_ptr<uint8_t>(),
+                                  (const uint64_t*)rand_r.data_ptr<int64_t>(), N,
+                                  ok.data_ptr<int32_t>(), cur_stream());
+    }
+    return ok;
+}
+
 torch::Tensor hash_to_g1_batch(torch::Tensor msgs, torch::Tensor moff) {
     CHECK_DEV(msgs); CHECK_DEV(moff);
     int32_t N = (int32_t)moff.size(0) - 1;
@@ -393,6 +417,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fixed-g2 Miller line coefficients for K1 v2 (once per process)");
     m.def("bls_verify_batch2", &bls_verify_batch2,
           "K1 v2: 2-lane Fp2-decomposed batched BLS verification");
+    m.def("bls_verify_batch_wave", &bls_verify_batch_wave,
+          "K1 v3: wave-batched product verification (shared final exp, "
+          "exact per-item fallback)");
     m.def("hash_to_g1_batch", &hash_to_g1_batch, "K1 helper: batched hash-to-G1");
     m.def("fanout_wave", &fanout_wave, "K3v2: wave-per-pair fan-out (nt flag, device count)");
     m.def("fanout_flat2", &fanout_flat2, "K3v4: flat fan-out, seq from base, capacity clamp");

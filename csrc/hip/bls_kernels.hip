@@ -134,6 +134,77 @@ k1_bls_verify2(
     if (!L.hi) ok[v] = good ? 1 : 0;
 }
 
+// ---------------------------------------------------------------------------
+// K1 v3: wave-batched product verification.  Standard small-exponent batch
+// verification: with per-item secret random coefficients r_i in [1, 2^64),
+//     prod_i [ e([r_i]H_i, pk_i) * e(-[r_i]sig_i, g2) ] == 1
+// holds iff every item verifies, except with probability ~2^-64 (an
+// adversary cannot craft cancelling cofactors without predicting r_i).
+// Each WAVE (32 lane-pairs) multiplies its items' Miller products via a
+// shuffle butterfly and runs ONE shared final exponentiation; a wave that
+// fails the batched check falls back to exact per-item final
+// exponentiations, so per-item verdicts are always exact.  Marshal-side
+// auth storms are overwhelmingly valid, so the common path amortizes the
+// final exponentiation 32x.  (Reference semantics stay per-item:
+// marshal.rs:66-72 accepts/rejects each connection individually.)
+// ---------------------------------------------------------------------------
+extern "C" __global__ void __launch_bounds__(64, 1)
+k1_bls_verify_wave(
+    const uint8_t* __restrict__ vks,      // [N][128]
+    const uint8_t* __restrict__ sigs,     // [N][64]
+    uint8_t* __restrict__ msgs,
+    const int64_t* __restrict__ moff,     // [N+1]
+    const uint8_t* __restrict__ g2_lines,
+    const uint64_t* __restrict__ rand_r,  // [N] secret batch coefficients
+    int32_t N,
+    int32_t* __restrict__ ok)
+{
+    int lane = blockIdx.x * blockDim.x + threadIdx.x;
+    int v = lane >> 1;
+    bn254p2::PL L = bn254p2::PL::self();
+    bool active = v < N;
+    bool item_ok = false;
+    Fp vkx, vky, hx, hy, sx, sy;
+    if (active) {
+        item_ok = bn254p2::verkey_load2(L, vks + (size_t)v * 128, vkx, vky);
+        if (item_ok) item_ok = bls::sig_deserialize(sigs + (size_t)v * 64, sx, sy);
+        if (item_ok) {
+            uint8_t* scratch = msgs + moff[v];
+            uint32_t msg_len = (uint32_t)(moff[v + 1] - moff[v] - 1);
+            item_ok = bls::hash_to_g1_with_scratch(scratch, msg_len, hx, hy);
+        }
+    }
+    bn254p2::F12 fi = L.f12one();
+    if (item_ok) {
+        // scale BOTH G1 inputs by the item's secret coefficient
+        uint64_t r = rand_r[v] | 1;  // never zero
+        U256 k{{r, 0, 0, 0}};
+        G1 Hj = G1::scalar_mul({hx, hy, Fp::one()}, k);
+        G1 Sj = G1::scalar_mul({sx, Fp::neg(sy), Fp::one()}, k);
+        Fp shx, shy, ssx, ssy;
+        Hj.to_affine(shx, shy);
+        Sj.to_affine(ssx, ssy);
+        bn254p2::F12 ml1 = bn254p2::miller_loop2(L, shx, shy, {vkx, vky});
+        bn254p2::F12 ml2 = bn254p2::miller_loop2_lines(L, ssx, ssy, g2_lines);
+        fi = L.f12mul(ml1, ml2);
+    }
+    // butterfly product across the wave's 32 pairs (all lanes converged;
+    // masks >= 2 keep the c0/c1 pair parity aligned)
+    bn254p2::F12 f = fi;
+    for (int mask = 2; mask <= 32; mask <<= 1)
+        f = L.f12mul(f, bn254p2::shfl_f12(f, mask));
+    bool wave_ok = L.f12is_one(bn254p2::final_exponentiation2(L, f));
+    int verdict;
+    if (wave_ok) {
+        verdict = item_ok ? 1 : 0;
+    } else {
+        // rare path: exact per-item check (invalid signature in the wave)
+        verdict = (item_ok &&
+                   L.f12is_one(bn254p2::final_exponentiation2(L, fi))) ? 1 : 0;
+    }
+    if (active && !L.hi) ok[v] = verdict;
+}
+
 // Device self-test: hash_to_g1 + sign-shaped scalar mul, for golden tests.
 extern "C" __global__ void __launch_bounds__(64)
 k1_hash_to_g1(
@@ -181,6 +252,17 @@ void launch_k1_bls_verify2(const uint8_t* vks, const uint8_t* sigs, uint8_t* msg
     int blocks = (lanes + threads - 1) / threads;
     hipLaunchKernelGGL(k1_bls_verify2, dim3(blocks), dim3(threads), 0, s, vks, sigs, msgs,
                        moff, g2_lines, N, ok);
+}
+
+void launch_k1_bls_verify_wave(const uint8_t* vks, const uint8_t* sigs, uint8_t* msgs,
+                               const int64_t* moff, const uint8_t* g2_lines,
+                               const uint64_t* rand_r, int32_t N, int32_t* ok,
+                               hipStream_t s) {
+    int threads = 64;
+    int lanes = 2 * N;
+    int blocks = (lanes + threads - 1) / threads;
+    hipLaunchKernelGGL(k1_bls_verify_wave, dim3(blocks), dim3(threads), 0, s, vks, sigs,
+                       msgs, moff, g2_lines, rand_r, N, ok);
 }
 
 }  // extern "C"

@@ -569,4 +569,29 @@ __device__ inline bool verkey_load2(const PL& L, const uint8_t* in, Fp& x, Fp& y
     return g2_in_subgroup2(L, x, y);
 }
 
+// ---------------------------------------------------------------------------
+// wave-level helpers for batched product verification
+// ---------------------------------------------------------------------------
+
+// exchange an Fp with the lane at (lane ^ mask) — used by the wave butterfly
+// product (mask >= 2 keeps pair parity aligned, so c0/c1 lanes line up)
+__device__ inline Fp xchg_mask(const Fp& a, int mask) {
+    Fp r;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+        r.n.v[i] = (u64)__shfl_xor((unsigned long long)a.n.v[i], mask, 64);
+    return r;
+}
+
+__device__ inline F12 shfl_f12(const F12& a, int mask) {
+    F12 r;
+    r.c0.c0 = xchg_mask(a.c0.c0, mask);
+    r.c0.c1 = xchg_mask(a.c0.c1, mask);
+    r.c0.c2 = xchg_mask(a.c0.c2, mask);
+    r.c1.c0 = xchg_mask(a.c1.c0, mask);
+    r.c1.c1 = xchg_mask(a.c1.c1, mask);
+    r.c1.c2 = xchg_mask(a.c1.c2, mask);
+    return r;
+}
+
 }  // namespace bn254p2

@@ -604,3 +604,55 @@ def test_k5b_direct_contention_ring_exhaustion(ops):
     want = raw + b"\x00" * (wire - len(raw))
     for _seq, payload in recs:
         assert payload == want
+
+
+def test_k1v3_wave_batch_matches_host(ops):
+    """K1 v3 (wave-batched product verification) must give EXACT per-item
+    verdicts — the shared-final-exp fast path for all-valid waves and the
+    per-item fallback for waves containing invalid items — across valid,
+    corrupted, wrong-namespace, malformed and out-of-subgroup inputs,
+    at a non-multiple-of-32 batch size."""
+    import secrets as _secrets
+
+    from pushcdn_amd.crypto import bls
+    from pushcdn_amd.ops.build import build_core
+    from tests.test_round2_fixes import _find_cofactor_point
+
+    core = build_core()
+    ns = bls.USER_MARSHAL_NAMESPACE
+    N = 77  # 2 full waves + a 13-item tail wave
+    vks, sigs, msgs, offsets, want = [], [], bytearray(), [0], []
+    cx, cy = _find_cofactor_point(9)
+    cof_vk = b"".join(c.to_bytes(32, "little") for c in (cx[0], cx[1], cy[0], cy[1]))
+    for i in range(N):
+        kp = bls.KeyPair.from_seed(2000 + i)
+        msg = f"w-{i}".encode()
+        sig = bls.sign(kp.private_key, ns, msg)
+        vk = kp.public_key
+        if 32 <= i < 64:
+            # second wave carries failures -> exercises the fallback path
+            if i % 3 == 0:
+                sig = bytes([sig[0] ^ 4]) + sig[1:]
+            if i == 40:
+                vk = cof_vk
+            if i == 45:
+                vk = b"\xff" * 128
+        vks.append(vk)
+        sigs.append(sig)
+        msgs += _namespaced(ns, msg)
+        offsets.append(len(msgs))
+        want.append(1 if core.verify(vk, ns, msg, sig) else 0)
+    assert all(want[:32]) and 0 in want[32:64] and all(want[64:])
+
+    vks_t = torch.frombuffer(bytearray(b"".join(vks)), dtype=torch.uint8).to("cuda")
+    sigs_t = torch.frombuffer(bytearray(b"".join(sigs)), dtype=torch.uint8).to("cuda")
+    msgs_t = torch.frombuffer(bytearray(msgs), dtype=torch.uint8).to("cuda")
+    moff_t = torch.tensor(offsets, dtype=torch.int64, device="cuda")
+    probe = torch.zeros(1, dtype=torch.uint8, device="cuda")
+    lines = ops.precompute_g2_lines(probe)
+    for trial in range(3):  # fresh coefficients each time
+        rand_r = torch.frombuffer(bytearray(_secrets.token_bytes(8 * N)),
+                                  dtype=torch.int64).to("cuda")
+        ok = ops.bls_verify_batch_wave(vks_t, sigs_t, msgs_t, moff_t, lines, rand_r)
+        torch.cuda.synchronize()
+        assert ok.cpu().tolist() == want, f"trial {trial}"
