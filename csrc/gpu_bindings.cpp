@@ -44,6 +44,9 @@ void launch_k1_bls_verify2(const uint8_t*, const uint8_t*, uint8_t*, const int64
 void launch_k1_bls_verify_wave(const uint8_t*, const uint8_t*, uint8_t*, const int64_t*,
                                const uint8_t*, const uint64_t*, int32_t, int32_t*,
                                hipStream_t);
+void launch_k1_dbg_wave(const uint8_t*, const uint8_t*, uint8_t*, const int64_t*,
+                        const uint8_t*, const uint64_t*, int32_t, int32_t, int32_t*,
+                        hipStream_t);
 void launch_k3_fanout_wave(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
                            const uint32_t*, const int32_t*, uint8_t*, int, int, hipStream_t);
 void launch_k3_fanout_flat2(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
@@ -417,6 +420,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fixed-g2 Miller line coefficients for K1 v2 (once per process)");
     m.def("bls_verify_batch2", &bls_verify_batch2,
           "K1 v2: 2-lane Fp2-decomposed batched BLS verification");
+    m.def("_k1_dbg_wave",
+          [](torch::Tensor vks, torch::Tensor sigs, torch::Tensor msgs, torch::Tensor moff,
+             torch::Tensor g2_lines, torch::Tensor rand_r, int64_t mode) {
+              int32_t N = (int32_t)moff.size(0) - 1;
+              auto ok = torch::zeros({N}, torch::TensorOptions().dtype(torch::kInt32)
+                                              .device(vks.device()));
+              launch_k1_dbg_wave(vks.data_ptr<uint8_t>(), sigs.data_ptr<uint8_t>(),
+                                 msgs.data_ptr<uint8_t>(), moff.data_ptr<int64_t>(),
+                                 g2_lines.data_ptr<uint8_t>(),
+                                 (const uint64_t*)rand_r.data_ptr<int64_t>(), N,
+                                 (int32_t)mode, ok.data_ptr<int32_t>(), cur_stream());
+              return ok;
+          });
     m.def("bls_verify_batch_wave", &bls_verify_batch_wave,
           "K1 v3: wave-batched product verification (shared final exp, "
           "exact per-item fallback)");

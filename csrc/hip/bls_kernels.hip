@@ -178,12 +178,9 @@ k1_bls_verify_wave(
     if (item_ok) {
         // scale BOTH G1 inputs by the item's secret coefficient
         uint64_t r = rand_r[v] | 1;  // never zero
-        U256 k{{r, 0, 0, 0}};
-        G1 Hj = G1::scalar_mul({hx, hy, Fp::one()}, k);
-        G1 Sj = G1::scalar_mul({sx, Fp::neg(sy), Fp::one()}, k);
         Fp shx, shy, ssx, ssy;
-        Hj.to_affine(shx, shy);
-        Sj.to_affine(ssx, ssy);
+        bn254p2::g1_smul_affine(hx, hy, r, shx, shy);
+        bn254p2::g1_smul_affine(sx, Fp::neg(sy), r, ssx, ssy);
         bn254p2::F12 ml1 = bn254p2::miller_loop2(L, shx, shy, {vkx, vky});
         bn254p2::F12 ml2 = bn254p2::miller_loop2_lines(L, ssx, ssy, g2_lines);
         fi = L.f12mul(ml1, ml2);
@@ -203,6 +200,90 @@ k1_bls_verify_wave(
                    L.f12is_one(bn254p2::final_exponentiation2(L, fi))) ? 1 : 0;
     }
     if (active && !L.hi) ok[v] = verdict;
+}
+
+// debug bisect of the v3 pipeline (mode gates how far each lane goes):
+// 1 = parse+hash, 2 = +G1 scalar muls/to_affine, 3 = +millers,
+// 4 = +per-item FE (no butterfly), 5 = full v3
+extern "C" __global__ void __launch_bounds__(64, 1)
+k1_dbg_wave(const uint8_t* vks, const uint8_t* sigs, uint8_t* msgs,
+            const int64_t* moff, const uint8_t* g2_lines, const uint64_t* rand_r,
+            int32_t N, int32_t mode, int32_t* ok)
+{
+    int lane = blockIdx.x * blockDim.x + threadIdx.x;
+    int v = lane >> 1;
+    bn254p2::PL L = bn254p2::PL::self();
+    bool active = v < N;
+    bool item_ok = false;
+    Fp vkx, vky, hx, hy, sx, sy;
+    if (active) {
+        item_ok = bn254p2::verkey_load2(L, vks + (size_t)v * 128, vkx, vky);
+        if (item_ok) item_ok = bls::sig_deserialize(sigs + (size_t)v * 64, sx, sy);
+        if (item_ok) {
+            uint8_t* scratch = msgs + moff[v];
+            uint32_t msg_len = (uint32_t)(moff[v + 1] - moff[v] - 1);
+            item_ok = bls::hash_to_g1_with_scratch(scratch, msg_len, hx, hy);
+        }
+    }
+    if (mode <= 1) { if (active && !L.hi) ok[v] = item_ok ? 1 : 0; return; }
+    if (mode == 20 || mode == 21) {  // scalar_mul only, no to_affine
+        Fp outb = Fp::zero();
+        if (item_ok) {
+            uint64_t r = (mode == 20) ? 3ull : (rand_r[v] | 1);
+            Fp ax, ay;
+            bn254p2::g1_smul_affine(hx, hy, r, ax, ay);
+            outb = ax;
+        }
+        if (active && !L.hi) ok[v] = (int32_t)(outb.n.v[0] & 1);
+        return;
+    }
+    if (mode == 22) {  // to_affine only (inv of a known-good Z=1 point? use hx)
+        Fp outb = Fp::zero();
+        if (item_ok) {
+            G1 p{hx, hy, Fp::from_u64(2)};
+            Fp ax, ay;
+            p.to_affine(ax, ay);
+            outb = ax;
+        }
+        if (active && !L.hi) ok[v] = (int32_t)(outb.n.v[0] & 1);
+        return;
+    }
+    bn254p2::F12 fi = L.f12one();
+    Fp shx = hx, shy = hy, ssx = sx, ssy = Fp::neg(sy);
+    if (item_ok) {
+        uint64_t r = rand_r[v] | 1;
+        bn254p2::g1_smul_affine(hx, hy, r, shx, shy);
+        bn254p2::g1_smul_affine(sx, Fp::neg(sy), r, ssx, ssy);
+    }
+    if (mode <= 2) { if (active && !L.hi) ok[v] = (int32_t)(shx.n.v[0] & 1); return; }
+    if (item_ok) {
+        bn254p2::F12 ml1 = bn254p2::miller_loop2(L, shx, shy, {vkx, vky});
+        bn254p2::F12 ml2 = bn254p2::miller_loop2_lines(L, ssx, ssy, g2_lines);
+        fi = L.f12mul(ml1, ml2);
+    }
+    if (mode <= 3) { if (active && !L.hi) ok[v] = (int32_t)(fi.c0.c0.n.v[0] & 1); return; }
+    if (mode <= 4) {
+        bool one = L.f12is_one(bn254p2::final_exponentiation2(L, fi));
+        if (active && !L.hi) ok[v] = (item_ok && one) ? 1 : 0;
+        return;
+    }
+    bn254p2::F12 f = fi;
+    for (int mask = 2; mask <= 32; mask <<= 1)
+        f = L.f12mul(f, bn254p2::shfl_f12(f, mask));
+    bool wave_ok = L.f12is_one(bn254p2::final_exponentiation2(L, f));
+    int verdict;
+    if (wave_ok) verdict = item_ok ? 1 : 0;
+    else verdict = (item_ok && L.f12is_one(bn254p2::final_exponentiation2(L, fi))) ? 1 : 0;
+    if (active && !L.hi) ok[v] = verdict;
+}
+
+extern "C" void launch_k1_dbg_wave(const uint8_t* vks, const uint8_t* sigs, uint8_t* msgs,
+                                   const int64_t* moff, const uint8_t* g2_lines,
+                                   const uint64_t* rand_r, int32_t N, int32_t mode,
+                                   int32_t* ok, hipStream_t s) {
+    int blocks = (2 * N + 63) / 64;
+    hipLaunchKernelGGL(k1_dbg_wave, dim3(blocks), dim3(64), 0, s, vks, sigs, msgs, moff,
+                       g2_lines, rand_r, N, mode, ok);
 }
 
 // Device self-test: hash_to_g1 + sign-shaped scalar mul, for golden tests.

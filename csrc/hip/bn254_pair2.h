@@ -570,6 +570,64 @@ __device__ inline bool verkey_load2(const PL& L, const uint8_t* in, Fp& x, Fp& y
 }
 
 // ---------------------------------------------------------------------------
+// G1 scalar multiplication by a SMALL positive scalar (the batch
+// coefficients): left-to-right jacobian double-and-add with NO infinity
+// branches — impossible here because 1 <= r < 2^64 << group order and the
+// base has prime order, so no intermediate is ever the identity and the
+// mixed add never sees equal points after the first doubling.  (The
+// generic Point<Fp>::scalar_mul template hangs on gfx950 in this kernel —
+// compiler pathology around its data-dependent early returns under
+// divergence; this straight-line form also drops ~2x the work.)
+// ---------------------------------------------------------------------------
+
+__device__ inline void g1_dbl_j(Fp& X, Fp& Y, Fp& Z) {
+    // dbl-2009-l
+    Fp A = Fp::sqr(X);
+    Fp B = Fp::sqr(Y);
+    Fp C = Fp::sqr(B);
+    Fp t = Fp::sqr(Fp::add(X, B));
+    Fp D = Fp::dbl(Fp::sub(Fp::sub(t, A), C));
+    Fp E = Fp::add(Fp::dbl(A), A);
+    Fp F2 = Fp::sqr(E);
+    Fp X3 = Fp::sub(F2, Fp::dbl(D));
+    Fp eight_c = Fp::dbl(Fp::dbl(Fp::dbl(C)));
+    Fp Y3 = Fp::sub(Fp::mul(E, Fp::sub(D, X3)), eight_c);
+    Fp Z3 = Fp::dbl(Fp::mul(Y, Z));
+    X = X3; Y = Y3; Z = Z3;
+}
+
+__device__ inline void g1_addmixed_j(Fp& X, Fp& Y, Fp& Z, const Fp& px, const Fp& py) {
+    // madd-2007-bl (Z2 = 1)
+    Fp Z1Z1 = Fp::sqr(Z);
+    Fp U2 = Fp::mul(px, Z1Z1);
+    Fp S2 = Fp::mul(Fp::mul(py, Z), Z1Z1);
+    Fp H = Fp::sub(U2, X);
+    Fp HH = Fp::sqr(H);
+    Fp I = Fp::dbl(Fp::dbl(HH));
+    Fp J = Fp::mul(H, I);
+    Fp rr = Fp::dbl(Fp::sub(S2, Y));
+    Fp V = Fp::mul(X, I);
+    Fp X3 = Fp::sub(Fp::sub(Fp::sqr(rr), J), Fp::dbl(V));
+    Fp Y3 = Fp::sub(Fp::mul(rr, Fp::sub(V, X3)), Fp::dbl(Fp::mul(Y, J)));
+    Fp Z3 = Fp::sub(Fp::sub(Fp::sqr(Fp::add(Z, H)), Z1Z1), HH);
+    X = X3; Y = Y3; Z = Z3;
+}
+
+__device__ __attribute__((noinline)) void g1_smul_affine(const Fp& px, const Fp& py,
+                                                         uint64_t r, Fp& ox, Fp& oy) {
+    int top = 63 - __builtin_clzll(r | 1);
+    Fp X = px, Y = py, Z = Fp::one();
+    BN_NOUNROLL for (int b = top - 1; b >= 0; --b) {
+        g1_dbl_j(X, Y, Z);
+        if ((r >> b) & 1) g1_addmixed_j(X, Y, Z, px, py);
+    }
+    Fp zi = Z.inv();
+    Fp zi2 = Fp::sqr(zi);
+    ox = Fp::mul(X, zi2);
+    oy = Fp::mul(Y, Fp::mul(zi2, zi));
+}
+
+// ---------------------------------------------------------------------------
 // wave-level helpers for batched product verification
 // ---------------------------------------------------------------------------
 
