@@ -271,6 +271,7 @@ k1_dbg_wave(const uint8_t* vks, const uint8_t* sigs, uint8_t* msgs,
     for (int mask = 2; mask <= 32; mask <<= 1)
         f = L.f12mul(f, bn254p2::shfl_f12(f, mask));
     bool wave_ok = L.f12is_one(bn254p2::final_exponentiation2(L, f));
+    if (mode == 6) { if (active && !L.hi) ok[v] = wave_ok ? 1 : 0; return; }
     int verdict;
     if (wave_ok) verdict = item_ok ? 1 : 0;
     else verdict = (item_ok && L.f12is_one(bn254p2::final_exponentiation2(L, fi))) ? 1 : 0;

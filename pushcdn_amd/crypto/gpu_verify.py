@@ -117,20 +117,17 @@ class GpuBatchVerifier:
             msgs += ns.encode() + msg + b"\x00"  # spare counter byte
             offsets.append(len(msgs))
         dev = self.device
-        # v3: wave-batched product verification (2-lane Fp2 pairing from
-        # bn254_pair2.h + per-wave shared final exponentiation behind
-        # SECRET random coefficients; exact per-item fallback in-kernel)
-        import secrets as _secrets
-
-        rand_r = torch.frombuffer(
-            bytearray(_secrets.token_bytes(8 * len(batch))), dtype=torch.int64
-        ).to(dev)
-        ok = self._ops.bls_verify_batch_wave(
+        # v2: 2-lane Fp2-decomposed kernel (bn254_pair2.h).  The v3
+        # wave-batched product check (bls_verify_batch_wave) was measured
+        # NOT to win here: its shared final exponentiation runs in lockstep
+        # on lanes that in v2 were already doing their own FE in parallel,
+        # so there is no wall-clock amortization at per-pair granularity —
+        # see profiles/k1_ab_r02.txt and scripts/k1_bench.py.
+        ok = self._ops.bls_verify_batch2(
             torch.frombuffer(vks, dtype=torch.uint8).to(dev),
             torch.frombuffer(sigs, dtype=torch.uint8).to(dev),
             torch.frombuffer(msgs, dtype=torch.uint8).to(dev),
             torch.tensor(offsets, dtype=torch.int64, device=dev),
             self._g2_lines(),
-            rand_r,
         )
         return [bool(x) for x in ok.cpu().tolist()]
