@@ -63,12 +63,25 @@ async def write_frames(writer: asyncio.StreamWriter, messages) -> None:
     total = 0
     try:
         bufs = []
+        pending = 0
         for message in messages:
             data = message.data
             bufs.append(struct.pack(">I", len(data)))
             bufs.append(data)
             total += len(data)
-        writer.writelines(bufs)
+            pending += len(data)
+            # the reference's 5 s timeout is PER MESSAGE (mod.rs:368) —
+            # applying it to a whole coalesced burst would kill connections
+            # that are making healthy progress through a big backlog, so
+            # drain (with the per-message bound) whenever the accumulated
+            # chunk gets large
+            if pending >= (4 << 20):
+                writer.writelines(bufs)
+                await asyncio.wait_for(writer.drain(), IO_TIMEOUT_S)
+                bufs = []
+                pending = 0
+        if bufs:
+            writer.writelines(bufs)
         await asyncio.wait_for(writer.drain(), IO_TIMEOUT_S)
     except (asyncio.TimeoutError, ConnectionResetError, OSError) as e:
         raise ConnectionError_(f"failed to send message: {e}") from e

@@ -37,7 +37,13 @@ from ..limiter import Limiter
 from .base import Connection, Listener, Protocol, UnfinalizedConnection
 from .tcp import parse_endpoint
 
-MTU = 1200                    # stream bytes per datagram
+MTU = 32768                   # stream bytes per datagram (UDP max 65507;
+                              # per-datagram Python work is the throughput
+                              # bound, so bigger beats 1500-wire-MTU purism
+                              # for this loopback/intra-DC profile)
+CWND = 1 << 20                # in-flight cap: bounds kernel-buffer loss
+ACK_EVERY = 16                # coalesce in-order acks (gaps ack immediately)
+UDP_BUF = 8 << 20             # SO_RCVBUF/SO_SNDBUF on the datagram socket
 RETX_S = 0.2                  # retransmission timer
 HANDSHAKE_TIMEOUT_S = 5.0
 LINGER_S = 5.0                # soft-close flush bound (reference 5 s timeouts)
@@ -53,6 +59,18 @@ PKT_FIN = 4        # [cid 8][u64 final offset]
 PKT_CLOSE = 5      # [cid 8]
 
 
+def _bump_udp_buffers(transport) -> None:
+    import socket as _socket
+
+    sock = transport.get_extra_info("socket")
+    if sock is not None:
+        for opt in (_socket.SO_RCVBUF, _socket.SO_SNDBUF):
+            try:
+                sock.setsockopt(_socket.SOL_SOCKET, opt, UDP_BUF)
+            except OSError:
+                pass
+
+
 class _QuicConn:
     """Reliability state for one connection's single bidi stream."""
 
@@ -66,6 +84,7 @@ class _QuicConn:
         self.tx_base = 0
         self.tx_next = 0
         self.unacked = bytearray()
+        self.tx_trim = 0  # acked prefix not yet physically removed
         # rx
         self.rx_off = 0
         self.rx_fin: Optional[int] = None
@@ -81,6 +100,9 @@ class _QuicConn:
 
     # ------------------------------ tx ------------------------------
 
+    def _tx_len(self) -> int:
+        return len(self.unacked) - self.tx_trim
+
     def stream_write(self, data: bytes) -> None:
         if self.closed or self.closing:
             return
@@ -89,49 +111,71 @@ class _QuicConn:
         self._watermarks()
 
     def _pump_tx(self) -> None:
-        end = self.tx_base + len(self.unacked)
+        # window-limited: never more than CWND bytes in flight — unpaced
+        # blasts overflow the peer's UDP buffer and collapse into
+        # retransmission storms
+        end = min(self.tx_base + self._tx_len(), self.tx_base + CWND)
         while self.tx_next < end:
             off = self.tx_next
-            chunk = bytes(self.unacked[off - self.tx_base:off - self.tx_base + MTU])
+            p = off - self.tx_base + self.tx_trim
+            chunk = bytes(self.unacked[p:p + MTU])
             self.ep.send_pkt(self.addr, PKT_STREAM, self.cid,
                              struct.pack("<Q", off) + chunk)
             self.tx_next = off + len(chunk)
 
     def on_ack(self, cum: int) -> None:
         if cum > self.tx_base:
-            del self.unacked[:cum - self.tx_base]
+            # deferred trim: del-from-front per ack is quadratic on large
+            # buffers (a 10 MiB message generates ~8,000 acks)
+            self.tx_trim += cum - self.tx_base
             self.tx_base = cum
+            if self.tx_trim > (1 << 20):
+                del self.unacked[:self.tx_trim]
+                self.tx_trim = 0
             if self.tx_next < cum:
                 self.tx_next = cum
             self._last_progress = self.loop.time()
+            self._dup_acks = 0
+            self._pump_tx()  # window opened
             self._watermarks()
-        if self.closing and not self.unacked:
+        elif cum == self.tx_base and self.tx_next > self.tx_base:
+            # duplicate ack: receiver is missing the segment at tx_base —
+            # fast retransmit after 3 instead of waiting the 200 ms timer
+            self._dup_acks = getattr(self, "_dup_acks", 0) + 1
+            if self._dup_acks >= 3:
+                self._dup_acks = 0
+                self.tx_next = self.tx_base
+                self._pump_tx()
+        if self.closing and self._tx_len() == 0:
             self._finish_close()
 
     def _watermarks(self) -> None:
+        # standard asyncio lower-transport flow control: the transport
+        # (a _FlowControlMixin) pauses/resumes its protocol based on
+        # get_write_buffer_size() — under TLS that propagates to the app's
+        # drain() through sslproto's _protocol_paused delegation
         t = self.transport
         if t is None:
             return
-        if len(self.unacked) > TX_HIGH and not t._paused_writing:
-            t._paused_writing = True
-            t._protocol.pause_writing()
-        elif len(self.unacked) < TX_LOW and t._paused_writing:
-            t._paused_writing = False
-            t._protocol.resume_writing()
+        t._maybe_pause_protocol()
+        t._maybe_resume_protocol()
 
     def _on_timer(self) -> None:
         if self.closed:
             return
-        if self.unacked:
-            # no ack progress: go-back-N from tx_base
+        if self._tx_len() and self.loop.time() - self._last_progress >= RETX_S:
+            # a full RTO with NO ack progress: go-back-N from tx_base.
+            # (Rewinding while acks are flowing would resend the whole
+            # window every tick and collapse throughput.)
             self.tx_next = self.tx_base
             self._pump_tx()
             if self.closing and self.loop.time() - self._last_progress > LINGER_S:
                 self._finish_close()   # peer gone; stop lingering
         elif self.closing:
             self._finish_close()
-        if self.rx_fin is not None or self.closing:
-            # re-ack so a retransmitting peer converges
+        # flush any coalesced ack + re-ack so a retransmitting peer converges
+        if getattr(self, "_ack_pending", 0) or self.rx_fin is not None or self.closing:
+            self._ack_pending = 0
             self.ep.send_pkt(self.addr, PKT_ACK, self.cid,
                              struct.pack("<Q", self.rx_off))
         if not self.closed:
@@ -153,7 +197,12 @@ class _QuicConn:
                 nxt = self.reorder.pop(self.rx_off)
                 self.reorder_bytes -= len(nxt)
                 self._deliver(nxt)
-        self.ep.send_pkt(self.addr, PKT_ACK, self.cid, struct.pack("<Q", self.rx_off))
+        # ack coalescing: gaps (duplicate acks drive fast retransmit) and
+        # fin-adjacent packets ack immediately; in-order flow acks 1-in-N
+        self._ack_pending = getattr(self, "_ack_pending", 0) + 1
+        if off > self.rx_off or self.rx_fin is not None or self._ack_pending >= ACK_EVERY:
+            self._ack_pending = 0
+            self.ep.send_pkt(self.addr, PKT_ACK, self.cid, struct.pack("<Q", self.rx_off))
         self._check_fin()
 
     def _deliver(self, data: bytes) -> None:
@@ -161,8 +210,8 @@ class _QuicConn:
         t = self.transport
         if t is None:
             self.pre_buf += data  # before finalize wires the transport
-        elif t._paused_reading:
-            t._rx_pending += data
+        elif t._paused and not t._started:
+            t._rx_pending += data  # pre-TLS window only
         else:
             t._protocol.data_received(data)
 
@@ -192,13 +241,13 @@ class _QuicConn:
             return
         self.closing = True
         self._last_progress = self.loop.time()
-        if not self.unacked:
+        if self._tx_len() == 0:
             self._finish_close()
 
     def _finish_close(self) -> None:
         if self.closed:
             return
-        fin = struct.pack("<Q", self.tx_base + len(self.unacked))
+        fin = struct.pack("<Q", self.tx_base + self._tx_len())
         for _ in range(3):
             self.ep.send_pkt(self.addr, PKT_FIN, self.cid, fin)
         self._teardown()
@@ -224,28 +273,46 @@ class _QuicConn:
                 pass
 
 
-class _QuicStreamTransport(asyncio.Transport):
+from asyncio import transports as _transports
+
+
+class _QuicStreamTransport(_transports._FlowControlMixin, asyncio.Transport):
     """asyncio Transport facade over a _QuicConn's stream — the layer
-    loop.start_tls wraps with SSLProtocol (real TLS 1.3 on the stream)."""
+    loop.start_tls wraps with SSLProtocol (real TLS 1.3 on the stream).
+
+    Subclasses _FlowControlMixin so the standard protocol write-pause
+    machinery (and 3.10 sslproto's pokes at `_paused`, `_protocol_paused`
+    and `get_write_buffer_size`) all behave like a real socket transport."""
 
     # loop.start_tls gates on this marker: it means the transport honors
     # pause_reading/resume_reading/set_protocol during the protocol swap
     _start_tls_compatible = True
 
     def __init__(self, conn: _QuicConn, protocol: asyncio.BaseProtocol) -> None:
-        super().__init__()
+        super().__init__(extra=None, loop=asyncio.get_running_loop())
         self._conn = conn
         self._protocol = protocol
-        self._paused_writing = False
         # born PAUSED: bytes that raced the accept/finalize split (e.g. the
         # peer's TLS ClientHello) must reach the SSLProtocol that
         # loop.start_tls installs, never the plain StreamReaderProtocol —
-        # start_tls resumes reading after it swaps the protocol
-        self._paused_reading = True
+        # start_tls resumes reading after it swaps the protocol.  After
+        # that FIRST resume the transport always delivers immediately:
+        # honoring reader pauses by buffering here deadlocks (the reader's
+        # resume only fires on new feed_data wakeups), and asyncio readers
+        # tolerate over-limit buffers — the same unbounded-buffer posture
+        # 3.10's sslproto itself has on the write side.
+        self._paused = True
+        self._started = False
         self._rx_pending = bytearray(conn.pre_buf)
         conn.pre_buf.clear()
         self._lost = False
         conn.transport = self
+
+    def get_write_buffer_size(self) -> int:
+        return self._conn._tx_len()
+
+    def set_write_buffer_limits(self, high=None, low=None):
+        self._set_write_buffer_limits(high=high, low=low)
 
     def get_extra_info(self, name, default=None):
         if name == "peername":
@@ -273,10 +340,11 @@ class _QuicStreamTransport(asyncio.Transport):
         return False
 
     def pause_reading(self) -> None:
-        self._paused_reading = True
+        self._paused = True
 
     def resume_reading(self) -> None:
-        self._paused_reading = False
+        self._paused = False
+        self._started = True
         if self._rx_pending:
             data = bytes(self._rx_pending)
             self._rx_pending.clear()
@@ -411,10 +479,11 @@ class Quic(Protocol):
         loop = asyncio.get_running_loop()
         ep = _QuicEndpoint(server=False)
         try:
-            await loop.create_datagram_endpoint(
+            transport, _ = await loop.create_datagram_endpoint(
                 lambda: ep, remote_addr=(host or "127.0.0.1", port))
         except OSError as e:
             raise ConnectionError_(f"failed to connect to {endpoint}: {e}") from e
+        _bump_udp_buffers(transport)
         cid = secrets.token_bytes(8)
         conn = _QuicConn(ep, cid, None)  # connected socket: sendto(None)
         ep.conns[cid] = conn
@@ -438,6 +507,7 @@ class Quic(Protocol):
         host, port = parse_endpoint(endpoint)
         loop = asyncio.get_running_loop()
         ep = _QuicEndpoint(server=True)
-        await loop.create_datagram_endpoint(
+        transport, _ = await loop.create_datagram_endpoint(
             lambda: ep, local_addr=(host or "0.0.0.0", port))
+        _bump_udp_buffers(transport)
         return QuicListener(ep)

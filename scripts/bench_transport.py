@@ -50,8 +50,12 @@ async def run_one(proto, payload_bytes: int, n_msgs: int) -> float:
     return n_msgs * payload_bytes / dt
 
 
-async def main(native: bool) -> None:
-    if native:
+async def main(native: bool, proto_name: str = "") -> None:
+    if proto_name == "quic":
+        from pushcdn_amd.proto.transports.quic import Quic as proto
+    elif proto_name == "tls":
+        from pushcdn_amd.proto.transports.tcp_tls import TcpTls as proto
+    elif native or proto_name == "native":
         from pushcdn_amd.proto.transports.tcp_native import TcpNative as proto
     else:
         from pushcdn_amd.proto.transports.tcp import Tcp as proto
@@ -61,9 +65,9 @@ async def main(native: bool) -> None:
         key = f"{size}B" if size < 1024 else (
             f"{size >> 10}KiB" if size < (1 << 20) else f"{size >> 20}MiB")
         out[key] = round(bps / 1e9, 3)
+    name = proto_name or ("tcp-native" if native else "tcp-asyncio")
     print(json.dumps({
-        "config": ("tcp-native" if native else "tcp-asyncio")
-                  + " raw transfer, loopback (GB/s by message size)",
+        "config": name + " raw transfer, loopback (GB/s by message size)",
         "gbps": out,
     }))
 
@@ -71,4 +75,6 @@ async def main(native: bool) -> None:
 if __name__ == "__main__":
     p = argparse.ArgumentParser()
     p.add_argument("--native", action="store_true")
-    asyncio.run(main(p.parse_args().native))
+    p.add_argument("--proto", default="", choices=["", "tcp", "native", "tls", "quic"])
+    a = p.parse_args()
+    asyncio.run(main(a.native, a.proto))
