@@ -64,9 +64,9 @@ def unpack_mesh_offsets(view: torch.Tensor, n_messages: int) -> torch.Tensor:
 
 
 class MeshBroker(Broker):
-    # the collective pack path consumes per-message Bytes; the C++ blob
-    # ingest is the single-broker socket fast path
-    BLOB_INGEST = False
+    # C++ blob ingest is supported: the tick normalizes blob items into
+    # per-message views for the collective pack
+    BLOB_INGEST = True
 
     def __init__(self, config: BrokerConfig, batch_capacity: int = 1 << 22,
                  interest_routed: Optional[bool] = None) -> None:
@@ -125,8 +125,9 @@ class MeshBroker(Broker):
     async def _send_partial_syncs(self) -> None:
         pass
 
-    def _blocking_mesh_tick(self, msgs: List[bytes], dev_buf: Optional[torch.Tensor],
-                            interests: int, owned_bits: int) -> torch.Tensor:
+    def _blocking_mesh_tick(self, msgs, dev_buf: Optional[torch.Tensor],
+                            interests: int, owned_bits: int,
+                            batch_topics: int, direct_bits: int) -> torch.Tensor:
         """One mesh tick's blocking half (runs on the dedicated mesh thread):
         pack, H2D, collective exchange, kernel tick per received batch,
         cursor drain.  Returns the drained ring cursors.
@@ -159,24 +160,6 @@ class MeshBroker(Broker):
         else:
             try:
                 if self.interest_routed:
-                    from ..proto import message as msglib
-                    from ..utils.keyhash import fnv1a64
-
-                    batch_topics = 0
-                    direct_bits = 0  # 64b digest of this batch's direct recipients
-                    for raw in msgs:
-                        try:
-                            r = msglib.parse_offsets(raw)
-                        except Exception:
-                            continue
-                        if r["disc"] == 4:
-                            for t in raw[r["topics_off"] : r["topics_off"] + r["topics_cnt"]]:
-                                batch_topics |= 1 << t
-                        elif r["disc"] == 3:
-                            # same keyed hash as the engine's K5 table (seed is
-                            # cluster-shared, so digests agree across brokers)
-                            direct_bits |= 1 << (
-                                fnv1a64(r["recipient"], self._engine.hash_seed) & 63)
                     exchanged = self.mesh.exchange_interest(
                         send_buf, n_local, used_bytes, batch_topics, interests,
                         direct_bits=direct_bits, owned_bits=owned_bits,
@@ -267,14 +250,14 @@ class MeshBroker(Broker):
             if peer not in self.connections.brokers:
                 asyncio.get_running_loop().create_task(self._dial_broker(peer))
 
-    async def _forward_degraded(self, batch: List[Bytes]) -> None:
+    async def _forward_degraded(self, entries) -> None:
         """Host-TCP fallback while the communicator is down: forward every
         queued message to the dialed peers; the receiving broker delivers
         with to_users_only/to_user_only semantics via its host plane —
         exactly the reference mesh path (broker/handler.rs:148-161)."""
         if self.connections.brokers:
-            for raw in batch:
-                await self.try_send_to_brokers(raw.clone())
+            for data, _owner, _fwd in entries:
+                await self.try_send_to_brokers(Bytes(bytes(data)))
 
     async def _gpu_tick_task(self) -> None:
         """Fixed-cadence mesh tick: pack queued local messages (possibly
@@ -287,45 +270,100 @@ class MeshBroker(Broker):
             if self._engine.is_cuda
             else None
         )
+        from ..proto import message as msglib
+        from ..utils.keyhash import fnv1a64
+
         while True:
             # collect up to a capacity-bounded batch; an oversize tick must
-            # NEVER raise here — that would stall every peer's collective
-            batch: List[Bytes] = self._carry
+            # NEVER raise here — that would stall every peer's collective.
+            # Queue items are either (Bytes, fwd) pairs or C++ ingest blobs
+            # ("blob", bytes, end_offsets, fwds); both normalize to
+            # (data_view, owner_or_None, fwd) entries.
+            entries = self._carry
             self._carry = []
-            used = sum((len(r.data) + 15) & ~15 for r in batch)
+            used = sum((len(e[0]) + 15) & ~15 for e in entries)
             budget = self.batch_capacity - 16 * 4096  # header headroom
-            while not self._gpu_queue.empty() and len(batch) < 4096:
-                raw = self._gpu_queue.get_nowait()[0]
-                padded = (len(raw.data) + 15) & ~15
-                if used + padded > budget:
-                    self._carry.append(raw)  # next tick
-                    break
-                batch.append(raw)
-                used += padded
-            msgs = [raw.data for raw in batch]
+            overflow = False
+            while not overflow and not self._gpu_queue.empty() and len(entries) < 4096:
+                item = self._gpu_queue.get_nowait()
+                if item[0] == "blob":
+                    _tag, blob, ends, fwds = item
+                    fs = 0
+                    for i, fe in enumerate(ends):
+                        view = memoryview(blob)[fs:fe]
+                        fs = fe
+                        padded = (len(view) + 15) & ~15
+                        if used + padded > budget:
+                            self._carry.append((view, None, fwds[i]))
+                            overflow = True
+                            continue  # keep normalizing the rest into carry
+                        if overflow:
+                            self._carry.append((view, None, fwds[i]))
+                        else:
+                            entries.append((view, None, fwds[i]))
+                            used += padded
+                else:
+                    raw, fwd = item
+                    padded = (len(raw.data) + 15) & ~15
+                    if used + padded > budget:
+                        self._carry.append((raw.data, raw, fwd))
+                        overflow = True
+                    else:
+                        entries.append((raw.data, raw, fwd))
+                        used += padded
+            msgs = [e[0] for e in entries]
             # digest inputs come from loop-owned state (connections maps),
             # so compute them HERE; the exchange + ticks then run off-loop
-            interests = owned_bits = 0
+            interests = owned_bits = batch_topics = direct_bits = 0
             if self.interest_routed:
-                from ..utils.keyhash import fnv1a64
-
                 for t in self.connections.user_topics.get_values():
                     interests |= 1 << (t & 0xFF)
                 # 64b digest of the direct users owned (connected) here —
                 # the mesh-plane analog of the reference's DirectMap
                 for pubkey in self._gpu_user_by_slot.values():
                     owned_bits |= 1 << (fnv1a64(pubkey, self._engine.hash_seed) & 63)
+                # this batch's topic bitmap + direct-recipient digest come
+                # from the ingest classification (frame-relative ranges);
+                # items without fwd info (degraded-TCP arrivals) parse here
+                for data, _owner, fwd in entries:
+                    if fwd is None:
+                        try:
+                            r = msglib.parse_offsets(bytes(data))
+                        except Exception:
+                            continue
+                        if r["disc"] == 4:
+                            for t in data[r["topics_off"]:r["topics_off"] + r["topics_cnt"]]:
+                                batch_topics |= 1 << t
+                        elif r["disc"] == 3:
+                            direct_bits |= 1 << (
+                                fnv1a64(r["recipient"], self._engine.hash_seed) & 63)
+                    else:
+                        # ("b"/"d", value_bytes) from the asyncio receive
+                        # loop, or ("b"/"d", start, end) frame-relative
+                        # ranges from the C++ ingest classification
+                        if len(fwd) == 2:
+                            kind, val = fwd
+                        else:
+                            kind, fo, fe = fwd
+                            val = bytes(data[fo:fe])
+                        if kind == "b":
+                            for t in val:
+                                batch_topics |= 1 << t
+                        else:
+                            direct_bits |= 1 << (
+                                fnv1a64(val, self._engine.hash_seed) & 63)
             wpos, offsets, staging = await asyncio.get_running_loop().run_in_executor(
                 self._mesh_executor, self._blocking_mesh_tick,
-                msgs, dev_buf, interests, owned_bits)
+                msgs, dev_buf, interests, owned_bits, batch_topics, direct_bits)
             await self._dispatch_egress(wpos, offsets, staging)
             if self.mesh.enabled and not self.mesh.healthy:
                 await self._maybe_dial_peers()
-                if batch:
-                    await self._forward_degraded(batch)
+                if entries:
+                    await self._forward_degraded(entries)
                 if self._rebuild_task is None:
                     self._rebuild_task = asyncio.get_running_loop().create_task(
                         self._mesh_rebuild_loop())
-            for raw in batch:
-                raw.drop()
+            for _data, owner, _fwd in entries:
+                if owner is not None:
+                    owner.drop()
             await asyncio.sleep(self.config.gpu_tick_interval_s)

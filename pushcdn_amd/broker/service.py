@@ -354,10 +354,11 @@ class Broker:
                         ab = start + toffs[i]
                         if not any(valid_topic[b] for b in blob[ab:ab + tcnts[i]]):
                             raise ConnectionError_("no valid topics")
-                        fwds.append(("b", ab, ab + tcnts[i]))
+                        # FRAME-RELATIVE field ranges: consumers (framed
+                        # forwarding, mesh digests) slice per-message views
+                        fwds.append(("b", toffs[i], toffs[i] + tcnts[i]))
                     elif d == 3:  # Direct
-                        ab = start + roffs[i]
-                        fwds.append(("d", ab, ab + rlens[i]))
+                        fwds.append(("d", roffs[i], roffs[i] + rlens[i]))
                     elif d in (5, 6):  # Subscribe / Unsubscribe (rare, inline)
                         ab = start + toffs[i]
                         try:
@@ -877,16 +878,16 @@ class Broker:
                         for i, fwd in enumerate(fwds):
                             fe = ends[i]
                             if fwd is not None:
-                                kind, s, e = fwd
+                                kind, s, e = fwd  # frame-relative
                                 if kind == "b":
                                     _u, brokers = self.connections.get_interested_by_topic(
-                                        list(blob[s:e]), to_users_only=False)
+                                        list(blob[fs + s:fs + e]), to_users_only=False)
                                     for b in brokers:
                                         await self.try_send_to_broker(
                                             b, Bytes(blob[fs:fe]))
                                 else:
                                     owner = self.connections.get_broker_identifier_of_user(
-                                        blob[s:e])
+                                        blob[fs + s:fs + e])
                                     if owner is not None and owner != self.identity:
                                         await self.try_send_to_broker(
                                             owner, Bytes(blob[fs:fe]))

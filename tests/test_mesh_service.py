@@ -121,3 +121,76 @@ def test_mesh_broker_interest_routed_gloo(tmp_path):
     assert out.returncode == 0, (out.stdout[-2000:], out.stderr[-3000:])
     assert "rank 0 mesh-service OK" in out.stdout
     assert "rank 1 mesh-service OK" in out.stdout
+
+
+def test_mesh_broker_blob_ingest_single_rank(tmp_path):
+    """MeshBroker consumes C++ ingest blobs: native-TCP users on a
+    single-rank mesh (the collective degenerates to self-exchange), with
+    interest digests computed from the pump's frame classification."""
+    import asyncio
+    import uuid
+
+    from pushcdn_amd.broker.mesh_service import MeshBroker
+    from pushcdn_amd.broker.service import BrokerConfig
+    from pushcdn_amd.client import Client, ClientConfig
+    from pushcdn_amd.crypto import bls
+    from pushcdn_amd.discovery import BrokerIdentifier
+    from pushcdn_amd.marshal import Marshal, MarshalConfig
+    from pushcdn_amd.proto import message as m
+    from pushcdn_amd.proto.transports.tcp_native import TcpNative
+
+    async def go():
+        os.environ.pop("WORLD_SIZE", None)
+        db = str(tmp_path / f"meshblob-{uuid.uuid4().hex}.db")
+        broker = MeshBroker(BrokerConfig(
+            public_bind_endpoint="127.0.0.1:0",
+            public_advertise_endpoint="127.0.0.1:0",
+            private_bind_endpoint="127.0.0.1:0",
+            private_advertise_endpoint="127.0.0.1:0",
+            discovery_endpoint=db,
+            keypair=bls.KeyPair.from_seed(1000),
+            user_protocol=TcpNative,
+            broker_protocol=TcpNative,
+            data_plane="gpu",
+            gpu_device="cpu",
+            gpu_max_users=16,
+            gpu_ring_bytes=1 << 14,
+            gpu_tick_interval_s=0.01,
+        ), batch_capacity=1 << 16, interest_routed=True)
+        await broker.start()
+        pub = f"127.0.0.1:{broker._user_listener.port}"
+        priv = f"127.0.0.1:{broker._broker_listener.port}"
+        broker.config.public_advertise_endpoint = pub
+        broker.config.private_advertise_endpoint = priv
+        broker.identity = BrokerIdentifier(pub, priv)
+        broker.discovery.identity = broker.identity
+        broker.connections.identity = broker.identity
+        await broker.discovery.perform_heartbeat(0, 600)
+        marshal = Marshal(MarshalConfig(bind_endpoint="127.0.0.1:0",
+                                        discovery_endpoint=db, protocol=TcpNative))
+        await marshal.start()
+        ep = f"127.0.0.1:{marshal._listener.port}"
+
+        a = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(71),
+                                subscribed_topics=[6], protocol=TcpNative))
+        b = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(72),
+                                subscribed_topics=[6], protocol=TcpNative))
+        await a.ensure_initialized()
+        await b.ensure_initialized()
+        await asyncio.sleep(0.3)
+
+        for i in range(30):
+            await a.send_broadcast_message([6], f"mb-{i}".encode())
+        got = [(await asyncio.wait_for(b.receive_message(), timeout=15)).message
+               for _ in range(30)]
+        assert got == [f"mb-{i}".encode() for i in range(30)]
+        await a.send_direct_message(b.public_key, b"mb-direct")
+        msg = await asyncio.wait_for(b.receive_message(), timeout=15)
+        assert msg.message == b"mb-direct"
+
+        a.close()
+        b.close()
+        await marshal.close()
+        await broker.close()
+
+    asyncio.run(asyncio.wait_for(go(), timeout=90))
