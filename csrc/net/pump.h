@@ -248,6 +248,7 @@ public:
     // memcpy; the lock is held only to push finished buffers.  Per-user
     // record counts are returned; -1 marks a connection that is gone
     // (caller evicts — reference user/sender.rs:16-33).
+    // returns per-user [count, payload_bytes] interleaved; count -1 = gone
     std::vector<int64_t> send_rings_batch(const uint8_t* base,
                                           const std::vector<int64_t>& ids,
                                           const std::vector<int64_t>& starts,
@@ -279,7 +280,12 @@ public:
             }
         }
         if (any) wake();
-        return counts;
+        std::vector<int64_t> out(2 * n_users);
+        for (size_t j = 0; j < n_users; ++j) {
+            out[2 * j] = counts[j];
+            out[2 * j + 1] = counts[j] < 0 ? 0 : payload[j];
+        }
+        return out;
     }
 
     // bytes queued but not yet written (backpressure signal for Python)

@@ -337,6 +337,8 @@ class Broker:
         (user/handler.rs:95-163): malformed/unexpected frames and
         no-valid-topic broadcasts disconnect; Subscribe/Unsubscribe apply
         inline."""
+        from ..utils.metrics import BYTES_RECV
+
         connection = handle.connection
         connection.enable_ingest()
         valid_topic = [False] * 256
@@ -346,6 +348,7 @@ class Broker:
             while True:
                 blob, ends, discs, toffs, tcnts, roffs, rlens = \
                     await connection.recv_ingest_batch()
+                BYTES_RECV.inc(len(blob))
                 fwds = []
                 start = 0
                 for i, d in enumerate(discs):
@@ -813,10 +816,17 @@ class Broker:
                         [e[3] for e in chunk],
                     )))
             results = await asyncio.gather(*(f for _c, f in jobs))
-            for (chunk, _f), counts in zip(jobs, results):
-                for (pubkey, _cid, _s, _e), cnt in zip(chunk, counts):
+            from ..utils.metrics import BYTES_SENT
+
+            total_payload = 0
+            for (chunk, _f), flat in zip(jobs, results):
+                for i, (pubkey, _cid, _s, _e) in enumerate(chunk):
+                    cnt = flat[2 * i]
+                    total_payload += flat[2 * i + 1]
                     if cnt < 0:
                         await self.remove_user(pubkey)
+            if total_payload:
+                BYTES_SENT.inc(total_payload)
         for slot, pubkey, n in fallback:
             ring = bytes(staging[int(offsets[slot]):int(offsets[slot + 1])].numpy()
                          .tobytes())
