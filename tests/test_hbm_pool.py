@@ -98,3 +98,53 @@ def test_engine_pooled_ingest_routes():
     for u in (0, 1):
         recs = parse_ring_records(eng.read_ring(u), int(wpos[u]))
         assert len(recs) == 1 and recs[0][1] == raw
+
+
+def test_pool_model_property():
+    """Hypothesis: random alloc/clone/drop sequences against a simple model
+    — used_bytes always equals the ring-semantics expectation, allocations
+    never overlap live ones, and the pool always returns to empty."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=60, deadline=None)
+    @given(st.lists(st.tuples(st.sampled_from(["alloc", "drop", "clone_drop"]),
+                              st.integers(1, 300)), min_size=1, max_size=120))
+    def check(ops):
+        pool = HbmMessagePool(1024, device="cpu")
+        live = []  # (PoolBytes, logical span) in FIFO order
+        model_used = 0
+        for op, arg in ops:
+            if op == "alloc":
+                b = pool.try_alloc(arg)
+                if b is None:
+                    # refusal must be justified: accounting for a possible
+                    # wrap gap, the request genuinely doesn't fit
+                    span = arg + ((1024 - pool._head) if pool._head + arg > 1024 else 0)
+                    assert pool.used_bytes + span > 1024, "spurious exhaustion"
+                    continue
+                # no overlap with any live allocation
+                for other, _ in live:
+                    if other.length > 0:
+                        assert (b.offset + b.length <= other.offset
+                                or other.offset + other.length <= b.offset), \
+                            "overlapping live allocations"
+                live.append((b, b.span))
+                model_used += b.span
+            elif op == "drop" and live:
+                idx = arg % len(live)
+                b, _span = live[idx]
+                if b.length > 0:
+                    b.drop()
+            elif op == "clone_drop" and live:
+                idx = arg % len(live)
+                b, _span = live[idx]
+                if b.length > 0:
+                    c = b.clone()
+                    c.drop()  # refcount returns to 1; still live
+                    assert b.length > 0
+        for b, _ in live:
+            if b.length > 0:
+                b.drop()
+        assert pool.used_bytes == 0, "pool must drain to empty"
+
+    check()

@@ -794,3 +794,65 @@ def test_quic_full_service_stack(tmp_path):
         await broker.close()
 
     run(go())
+
+
+def test_quic_packet_chaos_drop_dup_reorder():
+    """Reliability under combined datagram loss, duplication, AND
+    reordering: random per-packet chaos on both directions; the stream
+    must still deliver every message, in order, byte-exact."""
+    import random
+
+    from pushcdn_amd.proto.transports import quic as quicmod
+
+    async def go():
+        rng = random.Random(1234)
+        orig = quicmod._QuicEndpoint.send_pkt
+        delayed = []
+
+        def chaotic(self, addr, ptype, cid, payload):
+            if ptype in (quicmod.PKT_STREAM, quicmod.PKT_ACK):
+                r = rng.random()
+                if r < 0.10:
+                    return  # drop
+                if r < 0.18:
+                    orig(self, addr, ptype, cid, payload)  # duplicate
+                if r < 0.30:
+                    # delay/reorder: hold the packet, release a previous one
+                    delayed.append((self, addr, ptype, cid, payload))
+                    if len(delayed) > 4:
+                        args = delayed.pop(rng.randrange(len(delayed)))
+                        orig(*args)
+                    return
+            orig(self, addr, ptype, cid, payload)
+
+        quicmod._QuicEndpoint.send_pkt = chaotic
+        try:
+            limiter = Limiter(global_memory_pool_size=1 << 22)
+            listener = await quicmod.Quic.bind("127.0.0.1:0", None, None)
+            ep = f"127.0.0.1:{listener.port}"
+            rng2 = random.Random(99)
+            payloads = [bytes(rng2.randrange(256) for _ in range(rng2.randrange(1, 5000)))
+                        for _ in range(25)]
+
+            async def server():
+                unf = await listener.accept()
+                conn = await unf.finalize(limiter)
+                for i, want in enumerate(payloads):
+                    msg = await conn.recv_message()
+                    assert msg.message == want, f"msg {i} corrupted"
+                await conn.send_message(m.Direct(b"s", b"chaos-ok"))
+                await conn.soft_close()
+
+            st = asyncio.get_running_loop().create_task(server())
+            conn = await quicmod.Quic.connect(ep, True, limiter)
+            for p in payloads:
+                await conn.send_message(m.Direct(b"c", p))
+            reply = await asyncio.wait_for(conn.recv_message(), 60)
+            assert reply.message == b"chaos-ok"
+            await conn.soft_close()
+            await st
+            await listener.close()
+        finally:
+            quicmod._QuicEndpoint.send_pkt = orig
+
+    run(go())
