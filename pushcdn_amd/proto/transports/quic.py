@@ -271,6 +271,10 @@ class _QuicConn:
                 t._protocol.connection_lost(None)
             except Exception:
                 pass
+        # a client endpoint serves exactly one connection: close its UDP
+        # socket with it (long-lived reconnecting clients must not leak fds)
+        if not self.ep.server and not self.ep.conns:
+            self.ep.close()
 
 
 from asyncio import transports as _transports
