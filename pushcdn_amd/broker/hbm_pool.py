@@ -80,6 +80,7 @@ class HbmMessagePool:
         self._used = 0          # live bytes (incl. wrap gaps)
         self._live: "deque[PoolBytes]" = deque()  # FIFO of outstanding allocs
         self._cond = asyncio.Condition()
+        self._loop = None  # owning loop, learned on first loop-side release
 
     @property
     def used_bytes(self) -> int:
@@ -111,6 +112,7 @@ class HbmMessagePool:
     async def alloc(self, n: int) -> PoolBytes:
         """Blocking allocation: waits for releases when the pool is full —
         the reference's semaphore acquire (``pool.rs:60-68``)."""
+        self._loop = asyncio.get_running_loop()
         async with self._cond:
             while True:
                 b = self.try_alloc(n)
@@ -126,11 +128,17 @@ class HbmMessagePool:
             self._used -= head.span
             freed = True
         if freed:
-            # wake blocked allocators (non-async context safe: schedule)
+            # wake blocked allocators; safe from any thread (e.g. a drop on
+            # the mesh executor thread) via the owning loop
             try:
                 loop = asyncio.get_running_loop()
             except RuntimeError:
+                loop = self._loop
+                if loop is not None and not loop.is_closed():
+                    loop.call_soon_threadsafe(
+                        lambda: loop.create_task(self._notify()))
                 return
+            self._loop = loop
             loop.create_task(self._notify())
 
     async def _notify(self) -> None:
