@@ -173,6 +173,13 @@ class _QuicConn:
                 self._finish_close()   # peer gone; stop lingering
         elif self.closing:
             self._finish_close()
+        # purge reorder entries the cumulative stream has passed (chunk
+        # boundaries are retransmit-stable so this shouldn't trigger, but a
+        # stale entry must never pin the reorder budget)
+        if self.reorder:
+            stale = [k for k in self.reorder if k <= self.rx_off]
+            for k in stale:
+                self.reorder_bytes -= len(self.reorder.pop(k))
         # flush any coalesced ack + re-ack so a retransmitting peer converges
         if getattr(self, "_ack_pending", 0) or self.rx_fin is not None or self.closing:
             self._ack_pending = 0
