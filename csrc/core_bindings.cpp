@@ -422,6 +422,11 @@ PYBIND11_MODULE(pushcdn_core, m) {
         })
         .def("send_backlog", &net::Pump::send_backlog)
         .def("poll_dirty", &net::Pump::poll_dirty)
+        .def("recv_drain", [](net::Pump& p, int64_t id) {
+            auto r = p.recv_drain(id);
+            return py::make_tuple(std::get<0>(r), std::get<1>(r),
+                                  py::bytes(std::get<2>(r)), std::get<3>(r));
+        })
         .def("recv_batch", [](net::Pump& p, int64_t id, size_t maxf) {
             auto r = p.recv_batch(id, maxf);
             py::list out;

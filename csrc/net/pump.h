@@ -30,6 +30,7 @@
 #include <string>
 #include <algorithm>
 #include <thread>
+#include <tuple>
 #include <vector>
 
 #include "../wire/message.h"
@@ -368,6 +369,25 @@ public:
         it->second.want_write = true;
         wake();
         return true;
+    }
+
+    // drain EVERYTHING, returning only (count, bytes, last frame, closed) —
+    // the counting-subscriber path for benchmarks/relays where per-frame
+    // Python objects are pure overhead
+    std::tuple<int64_t, int64_t, std::string, bool> recv_drain(int64_t id) {
+        std::lock_guard<std::mutex> g(mu_);
+        auto it = conns_.find(id);
+        if (it == conns_.end()) return {0, 0, std::string(), true};
+        auto& c = it->second;
+        int64_t n = 0, bytes = 0;
+        std::string last;
+        while (!c.inbox.empty()) {
+            last = std::move(c.inbox.front());
+            c.inbox.pop_front();
+            ++n;
+            bytes += (int64_t)last.size();
+        }
+        return {n, bytes, std::move(last), c.closed};
     }
 
     // drain up to max_frames complete inbound frames; empty vector + closed

@@ -105,6 +105,7 @@ class PumpConnection(Connection):
         self._closed = False
         self._dead = False
         self._ingest = False
+        self._count_mode = False
         self._recv_q: "asyncio.Queue[Bytes]" = asyncio.Queue()
         self._wakeup = asyncio.Event()
         mgr.conns[shard][cid] = self
@@ -124,11 +125,35 @@ class PumpConnection(Connection):
 
     # called from the manager's eventfd callback
     def _pump_dirty(self) -> None:
-        if self._ingest:
+        if self._ingest or self._count_mode:
             self._wakeup.set()
             return
         if self._poll_inbox():
             self._wakeup.set()
+
+    def enable_count_mode(self) -> None:
+        """Counting-subscriber path (benchmarks/relays): frames stay in
+        C++; recv_drain returns (count, bytes, last_frame) per call."""
+        self._count_mode = True
+
+    async def recv_drain(self):
+        """(count, bytes, last_frame) — awaits until >=1 frame arrived;
+        raises when the peer is gone and drained."""
+        while True:
+            n, nbytes, last, closed = self._pump.recv_drain(self._cid)
+            if n:
+                return n, nbytes, last
+            if closed or self._dead or self._closed:
+                self._dead = True
+                raise ConnectionError_("connection reader closed")
+            self._wakeup.clear()
+            n, nbytes, last, closed = self._pump.recv_drain(self._cid)
+            if n:
+                return n, nbytes, last
+            if closed or self._dead or self._closed:
+                self._dead = True
+                raise ConnectionError_("connection reader closed")
+            await self._wakeup.wait()
 
     def enable_ingest(self) -> None:
         """Switch this connection to the C++ ingest path: the pump

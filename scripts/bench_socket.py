@@ -44,6 +44,8 @@ def _client(endpoint, seed, topics):
 
 
 async def run_subscriber(args) -> None:
+    import os
+
     from pushcdn_amd.proto import message as m
 
     clients = [_client(args.endpoint, args.seed + i, [TOPIC])
@@ -58,12 +60,21 @@ async def run_subscriber(args) -> None:
     count = 0
     lats = []
 
+    count_mode = bool(int(os.environ.get("SOCKBENCH_COUNT_MODE", "0")))
+
     async def drain(c):
         nonlocal count
         payload_off = None
+        conn = await c._get_connection()
+        if count_mode and hasattr(conn, "enable_count_mode"):
+            conn.enable_count_mode()
         while time.time() < t1 + 0.5:
             try:
-                frames = await asyncio.wait_for(c.receive_raw_batch(), timeout=0.5)
+                if count_mode:
+                    n, _nb, f = await asyncio.wait_for(conn.recv_drain(), timeout=0.5)
+                else:
+                    frames = await asyncio.wait_for(c.receive_raw_batch(), timeout=0.5)
+                    n, f = len(frames), frames[0]
             except asyncio.TimeoutError:
                 continue
             except Exception:
@@ -71,8 +82,7 @@ async def run_subscriber(args) -> None:
             now = time.time()
             if now < t0 or now >= t1:
                 continue
-            count += len(frames)
-            f = frames[0]
+            count += n
             if payload_off is None:
                 try:
                     payload_off = m.parse_offsets(f)["payload_off"]
@@ -133,6 +143,8 @@ async def run_coordinator(args) -> None:
     import os
 
     os.environ.setdefault("PUSHCDN_PUMP_SHARDS", str(args.pump_shards))
+    if args.count_mode:
+        os.environ["SOCKBENCH_COUNT_MODE"] = "1"
     import torch
 
     from pushcdn_amd.broker.service import Broker, BrokerConfig
@@ -275,6 +287,7 @@ def main():
     p.add_argument("--t1", type=float, default=0)
     p.add_argument("--tag", default="")
     p.add_argument("--pump-shards", type=int, default=4)
+    p.add_argument("--count-mode", action="store_true")
     args = p.parse_args()
     if args.role == "sub":
         asyncio.run(run_subscriber(args))
