@@ -874,8 +874,15 @@ class Broker:
             item = await self._gpu_queue.get()
             t0 = _time.perf_counter()
             batch = [item]
-            while not self._gpu_queue.empty() and len(batch) < 4096:
-                batch.append(self._gpu_queue.get_nowait())
+            # bound the tick by ITEMS and BYTES: blob items can each carry
+            # thousands of messages, and an unbounded catch-up tick would
+            # balloon the ingest staging + drain
+            tick_bytes = len(item[1]) if item[0] == "blob" else len(item[0].data)
+            while (not self._gpu_queue.empty() and len(batch) < 4096
+                   and tick_bytes < (64 << 20)):
+                nxt = self._gpu_queue.get_nowait()
+                tick_bytes += len(nxt[1]) if nxt[0] == "blob" else len(nxt[0].data)
+                batch.append(nxt)
             buf = bytearray()
             offsets = [0]
             have_peers = bool(self.connections.brokers)
