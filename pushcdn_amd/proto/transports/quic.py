@@ -29,7 +29,7 @@ import asyncio
 import secrets
 import ssl
 import struct
-from typing import Dict, Optional, Tuple
+from typing import Dict, Optional
 
 from ...crypto import tls as tlslib
 from ..errors import ConnectionError_
