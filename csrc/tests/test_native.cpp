@@ -152,6 +152,97 @@ static int test_pump_frames_and_close() {
     return 0;
 }
 
+
+static int test_pump_ingest_and_batch_drain() {
+    // round-2 paths under ASan/TSan: frame classification + contiguous
+    // ingest accumulation, recv_drain counting, send_rings_batch with the
+    // lock-free frame build, and send_raw pre-framed bursts
+    net::Pump pump;
+    int fds[2];
+    CHECK(socketpair(AF_UNIX, SOCK_STREAM, 0, fds) == 0);
+    int64_t a = pump.add(fds[0]);
+    int64_t b = pump.add(fds[1]);
+    pump.set_ingest(b);
+
+    // a Broadcast wire message, framed; classify_frame must see disc=4
+    uint8_t topics[2] = {3, 9};
+    auto bc = wire::serialize_broadcast(topics, 2, (const uint8_t*)"payload!", 8);
+    std::string burst;
+    for (int i = 0; i < 64; ++i) {
+        uint32_t be = htonl((uint32_t)bc.size());
+        burst.append((const char*)&be, 4);
+        burst.append((const char*)bc.data(), bc.size());
+    }
+    CHECK(pump.send_raw(a, burst.data(), burst.size()));
+    net::Pump::IngestBatch ib;
+    for (int spins = 0; spins < 2000 && ib.offs.size() < 64; ++spins) {
+        auto got = pump.recv_ingest(b);
+        if (!got.blob.empty()) {
+            ib.blob += got.blob;
+            for (auto o : got.offs) ib.offs.push_back((int64_t)ib.blob.size() -
+                                                      (int64_t)got.blob.size() + o);
+            for (auto& m : got.meta) ib.meta.push_back(m);
+        }
+        usleep(1000);
+    }
+    CHECK(ib.offs.size() == 64);
+    CHECK(ib.blob.size() == 64 * bc.size());
+    for (auto& m : ib.meta) {
+        CHECK(m.disc == wire::BROADCAST);
+        CHECK(m.topics_cnt == 2);
+        CHECK(ib.blob[m.topics_off] == 3 && ib.blob[m.topics_off + 1] == 9);
+    }
+
+    // batched egress drain: two fake rings (16 B headers + padded payloads)
+    // through send_rings_batch; receiver counts them via recv_drain
+    auto make_rec = [](uint32_t seq, const std::string& p) {
+        std::string r(16, '\0');
+        uint32_t len = (uint32_t)p.size();
+        memcpy(&r[0], &len, 4);
+        memcpy(&r[4], &seq, 4);
+        r += p;
+        r.resize(16 + ((p.size() + 15) & ~(size_t)15), '\0');
+        return r;
+    };
+    std::string ring1 = make_rec(0, "alpha") + make_rec(1, "beta");
+    std::string ring2 = make_rec(7, "gamma-longer-payload");
+    std::string base = ring1 + ring2;
+    int fds2[2];
+    CHECK(socketpair(AF_UNIX, SOCK_STREAM, 0, fds2) == 0);
+    int64_t c = pump.add(fds2[0]);
+    int64_t d = pump.add(fds2[1]);
+    auto counts = pump.send_rings_batch(
+        (const uint8_t*)base.data(), {c, c},
+        {0, (int64_t)ring1.size()},
+        {(int64_t)ring1.size(), (int64_t)base.size()});
+    CHECK(counts.size() == 4);  // [count, payload] x 2
+    CHECK(counts[0] == 2 && counts[1] == (int64_t)(5 + 4));
+    CHECK(counts[2] == 1 && counts[3] == 20);
+    int64_t n_total = 0;
+    std::string last;
+    for (int spins = 0; spins < 2000 && n_total < 3; ++spins) {
+        auto r = pump.recv_drain(d);
+        n_total += std::get<0>(r);
+        if (std::get<0>(r)) last = std::get<2>(r);
+        usleep(1000);
+    }
+    CHECK(n_total == 3);
+    CHECK(last == "gamma-longer-payload");
+    pump.hard_close(c);
+    pump.forget(c);
+    pump.forget(d);
+
+    // gone connection reports -1 in the batch result
+    pump.hard_close(b);
+    auto counts2 = pump.send_rings_batch((const uint8_t*)base.data(), {b},
+                                         {0}, {(int64_t)ring1.size()});
+    CHECK(counts2[0] == -1);
+    pump.forget(a);
+    pump.forget(b);
+    pump.stop();
+    return 0;
+}
+
 static int test_crdt_delta_fuzz() {
     // corrupted sync payloads under ASan: bit flips, truncations and raw
     // garbage must be rejected without any out-of-bounds access
@@ -189,6 +280,7 @@ int main() {
     if (test_field_arithmetic()) return 1;
     if (test_bls_end_to_end()) return 1;
     if (test_pump_frames_and_close()) return 1;
+    if (test_pump_ingest_and_batch_drain()) return 1;
     if (test_crdt_delta_fuzz()) return 1;
     printf("native tests OK (%d checks)\n", checks);
     return 0;
