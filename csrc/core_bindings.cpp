@@ -14,6 +14,7 @@
 #include "wire/message.h"
 #include "state/versioned_map.h"
 #include "net/pump.h"
+#include "net/udp_stream.h"
 
 namespace py = pybind11;
 using namespace bn254;
@@ -438,6 +439,38 @@ PYBIND11_MODULE(pushcdn_core, m) {
         .def("forget", &net::Pump::forget)
         .def("byte_counters", &net::Pump::byte_counters)
         .def("stop", &net::Pump::stop, py::call_guard<py::gil_scoped_release>());
+    py::class_<net::UdpPump>(m, "UdpPump")
+        .def(py::init<>())
+        .def("notify_fd", &net::UdpPump::notify_fd)
+        .def("bind", &net::UdpPump::bind)
+        .def("connect", &net::UdpPump::connect)
+        .def("port", &net::UdpPump::port)
+        .def("accept_poll", &net::UdpPump::accept_poll)
+        .def("client_status", &net::UdpPump::client_status)
+        .def("stream_write", [](net::UdpPump& p, uint64_t cid, py::buffer data) {
+            py::buffer_info info = data.request();
+            py::gil_scoped_release nogil;
+            return p.stream_write(cid, (const char*)info.ptr, (size_t)info.size);
+        })
+        .def("recv_stream", [](net::UdpPump& p, uint64_t cid) {
+            std::tuple<std::string, bool, bool> r;
+            {
+                py::gil_scoped_release nogil;
+                r = p.recv_stream(cid);
+            }
+            return py::make_tuple(py::bytes(std::get<0>(r)), std::get<1>(r),
+                                  std::get<2>(r));
+        })
+        .def("tx_backlog", &net::UdpPump::tx_backlog)
+        .def("graceful_close", &net::UdpPump::graceful_close)
+        .def("abort_conn", &net::UdpPump::abort_conn)
+        .def("forget", &net::UdpPump::forget)
+        .def("n_conns", &net::UdpPump::n_conns)
+        .def("poll_events", &net::UdpPump::poll_events,
+             py::call_guard<py::gil_scoped_release>())
+        .def("debug_set_loss", &net::UdpPump::debug_set_loss)
+        .def("debug_stats", &net::UdpPump::debug_stats)
+        .def("stop", &net::UdpPump::stop, py::call_guard<py::gil_scoped_release>());
     py::class_<PyVersionedMap>(m, "VersionedMap")
         .def(py::init<const std::string&>())
         .def("insert", &PyVersionedMap::insert)

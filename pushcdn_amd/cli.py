@@ -33,6 +33,14 @@ def _transport(name: str):
         from .proto.transports.tcp_native import TcpNative
 
         return TcpNative
+    if name == "quic":
+        from .proto.transports.quic import Quic
+
+        return Quic
+    if name == "quic-native":
+        from .proto.transports.quic import QuicNative
+
+        return QuicNative
     raise SystemExit(f"unknown transport {name!r}")
 
 
@@ -49,8 +57,9 @@ def _broker_args(p: argparse.ArgumentParser) -> None:
     p.add_argument("--global-memory-pool-size", type=int, default=1 << 30)
     p.add_argument("--data-plane", choices=["host", "gpu"], default="host")
     p.add_argument("--gpu-device", default="cuda:0")
-    p.add_argument("--user-transport", choices=["tcp", "tcp-tls", "tcp-native"],
-                   default="tcp", help="user-plane transport (tcp-native = C++ epoll pump)")
+    p.add_argument("--user-transport", choices=["tcp", "tcp-tls", "tcp-native", "quic", "quic-native"],
+                   default="tcp", help="user-plane transport (tcp-native = C++ epoll "
+                                       "pump; quic[-native] = QUIC-profile over UDP)")
     p.add_argument("--broker-transport", choices=["tcp", "tcp-native"], default="tcp")
 
 
@@ -250,12 +259,12 @@ def main(argv=None) -> None:
     ms.add_argument("--ca-cert-path", default=None)
     ms.add_argument("--ca-key-path", default=None)
     ms.add_argument("--global-memory-pool-size", type=int, default=1 << 30)
-    ms.add_argument("--transport", choices=["tcp", "tcp-tls", "tcp-native"], default="tcp")
+    ms.add_argument("--transport", choices=["tcp", "tcp-tls", "tcp-native", "quic", "quic-native"], default="tcp")
     ms.set_defaults(fn=cmd_marshal)
 
     c = sub.add_parser("client")
     c.add_argument("-m", "--marshal-endpoint", default="127.0.0.1:1737")
-    c.add_argument("--transport", choices=["tcp", "tcp-tls", "tcp-native"], default="tcp")
+    c.add_argument("--transport", choices=["tcp", "tcp-tls", "tcp-native", "quic", "quic-native"], default="tcp")
     c.set_defaults(fn=cmd_client)
 
     bb = sub.add_parser("bad-broker")

@@ -140,12 +140,18 @@ class _QuicConn:
             self._watermarks()
         elif cum == self.tx_base and self.tx_next > self.tx_base:
             # duplicate ack: receiver is missing the segment at tx_base —
-            # fast retransmit after 3 instead of waiting the 200 ms timer
+            # fast retransmit after 3 instead of waiting the 200 ms timer.
+            # Resend ONLY the head segment: rewinding the whole window per
+            # dup-ack trio re-blasts CWND bytes and storms under loss (the
+            # full rewind belongs to the RTO path).
             self._dup_acks = getattr(self, "_dup_acks", 0) + 1
             if self._dup_acks >= 3:
                 self._dup_acks = 0
-                self.tx_next = self.tx_base
-                self._pump_tx()
+                p = self.tx_trim
+                chunk = bytes(self.unacked[p:p + MTU])
+                if chunk:
+                    self.ep.send_pkt(self.addr, PKT_STREAM, self.cid,
+                                     struct.pack("<Q", self.tx_base) + chunk)
         if self.closing and self._tx_len() == 0:
             self._finish_close()
 
@@ -437,6 +443,11 @@ async def _wire_tls(conn: _QuicConn, ctx: ssl.SSLContext, *, server_side: bool,
     reader = asyncio.StreamReader(limit=1 << 20)
     protocol = asyncio.StreamReaderProtocol(reader)
     plain = _QuicStreamTransport(conn, protocol)
+    # asyncio's default 64 KiB high watermark makes the TLS writer
+    # stop-and-go: it pauses whenever >64 KiB is unacked and resumes only
+    # after a notify round-trip, capping throughput at ~high/wakeup-latency.
+    # Size it to the reliability window instead.
+    plain.set_write_buffer_limits(high=TX_HIGH, low=TX_LOW)
     protocol.connection_made(plain)
     try:
         tls_transport = await asyncio.wait_for(
@@ -522,3 +533,248 @@ class Quic(Protocol):
             lambda: ep, local_addr=(host or "0.0.0.0", port))
         _bump_udp_buffers(transport)
         return QuicListener(ep)
+
+
+# --------------------------------------------------------------------------
+# Native-datapath variant: the SAME QUIC profile with the reliability layer
+# (datagram parsing, ACK/retransmission, reordering, windowing) run by the
+# C++ UdpPump epoll thread (csrc/net/udp_stream.h) instead of per-datagram
+# Python.  Packet-for-packet interoperable with the pure-Python endpoint
+# above (same wire format + policy constants); TLS 1.3 still runs in
+# Python over the reliable stream via the same _QuicStreamTransport +
+# loop.start_tls path, so trust model and framing are identical.  Python
+# is woken through an eventfd and exchanges whole stream CHUNKS with the
+# pump, so per-datagram work never enters the interpreter.
+# --------------------------------------------------------------------------
+
+import os as _os
+
+
+def _core():
+    from ...ops.build import build_core
+
+    return build_core()
+
+
+class _NativeConn:
+    """Python-side face of one UdpPump connection — the same surface
+    _QuicStreamTransport drives on a pure-Python _QuicConn."""
+
+    def __init__(self, ep: "_NativeEndpoint", cid: int, addr) -> None:
+        self.ep = ep
+        self.cid = cid
+        self.addr = addr
+        self.pre_buf = bytearray()
+        self.transport: Optional[_QuicStreamTransport] = None
+        self.closed = False
+        self.closing = False
+        self.established = asyncio.Event()
+        self._eof_seen = False
+
+    # ---- surface used by _QuicStreamTransport ----
+    def _tx_len(self) -> int:
+        return 0 if self.closed else self.ep.pump.tx_backlog(self.cid)
+
+    def stream_write(self, data: bytes) -> None:
+        if self.closed or self.closing:
+            return
+        self.ep.pump.stream_write(self.cid, bytes(data))
+        self._watermarks()
+
+    def graceful_close(self) -> None:
+        if self.closed or self.closing:
+            return
+        self.closing = True
+        # the pump flushes unacked bytes (lingering + retransmitting), FINs,
+        # then flags closed; service() turns that into connection_lost
+        self.ep.pump.graceful_close(self.cid)
+
+    def abort(self) -> None:
+        if not self.closed:
+            self.ep.pump.abort_conn(self.cid)
+        self._teardown()
+
+    # ---- pump-event servicing (called on eventfd wakeups) ----
+    def service(self, flags: int = 0xFF) -> None:
+        # flags: 1 = rx bytes ready, 2 = ack progress, 4 = state change
+        if self.closed:
+            return
+        if flags & 6 and not self.ep.server and not self.established.is_set():
+            if self.ep.pump.client_status(self.cid) == 1:
+                self.established.set()
+        eof = closed = False
+        if flags & 5:
+            data, eof, closed = self.ep.pump.recv_stream(self.cid)
+            if data:
+                self._deliver(data)
+        if flags & 2:
+            self._watermarks()
+        if eof and not self._eof_seen:
+            self._eof_seen = True
+            t = self.transport
+            if t is not None:
+                try:
+                    t._protocol.eof_received()
+                except Exception:
+                    pass
+            # one loop tick of grace so the SSL layer can flush its
+            # close_notify through us before connection_lost clears it
+            self.ep.loop.call_soon(self._teardown)
+        elif closed:
+            self._teardown()
+
+    def _deliver(self, data: bytes) -> None:
+        t = self.transport
+        if t is None:
+            self.pre_buf += data  # before finalize wires the transport
+        elif t._paused and not t._started:
+            t._rx_pending += data  # pre-TLS window only
+        else:
+            t._protocol.data_received(data)
+
+    def _watermarks(self) -> None:
+        t = self.transport
+        if t is None:
+            return
+        t._maybe_pause_protocol()
+        t._maybe_resume_protocol()
+
+    def _teardown(self) -> None:
+        if self.closed:
+            return
+        self.closed = True
+        self.ep.conns.pop(self.cid, None)
+        self.ep.pump.forget(self.cid)
+        t = self.transport
+        if t is not None and not t._lost:
+            t._lost = True
+            try:
+                t._protocol.connection_lost(None)
+            except Exception:
+                pass
+        # a client endpoint serves exactly one connection: stop its pump
+        # (and epoll thread) with it
+        if not self.ep.server and not self.ep.conns:
+            self.ep.close()
+
+
+class _NativeEndpoint:
+    """One UdpPump + the asyncio reader that dispatches its wakeups."""
+
+    def __init__(self, server: bool) -> None:
+        self.pump = _core().UdpPump()
+        self.server = server
+        self.conns: Dict[int, _NativeConn] = {}
+        self.accept_q: "asyncio.Queue" = asyncio.Queue()
+        self.loop = asyncio.get_running_loop()
+        self._fd = self.pump.notify_fd()
+        self.loop.add_reader(self._fd, self._on_notify)
+        self._closed = False
+
+    def _on_notify(self) -> None:
+        try:
+            _os.read(self._fd, 8)
+        except (BlockingIOError, OSError):
+            pass
+        if self.server:
+            for cid, bootstrap in self.pump.accept_poll():
+                conn = _NativeConn(self, cid, None)
+                self.conns[cid] = conn
+                self.accept_q.put_nowait(
+                    QuicNativeUnfinalized(self, conn, bootstrap))
+        # service ONLY the connections with events (the pump tags each with
+        # why: rx bytes / ack progress / state change), not a flat O(conns)
+        # sweep of pybind crossings per wakeup
+        for cid, flags in self.pump.poll_events():
+            conn = self.conns.get(cid)
+            if conn is not None:
+                conn.service(flags)
+
+    def close(self) -> None:
+        if self._closed:
+            return
+        self._closed = True
+        try:
+            self.loop.remove_reader(self._fd)
+        except Exception:
+            pass
+        for conn in list(self.conns.values()):
+            conn._teardown()
+        self.pump.stop()
+
+
+class QuicNativeUnfinalized(UnfinalizedConnection):
+    def __init__(self, ep: _NativeEndpoint, conn: _NativeConn,
+                 bootstrap: int) -> None:
+        self._ep = ep
+        self._conn = conn
+        self.bootstrap = bootstrap
+
+    async def finalize(self, limiter: Limiter) -> Connection:
+        ctx = tlslib.server_context(QuicNative.ca_cert_path,
+                                    QuicNative.ca_key_path)
+        return await _wire_tls(self._conn, ctx, server_side=True,
+                               server_hostname=None, limiter=limiter)
+
+
+class QuicNativeListener(Listener):
+    def __init__(self, ep: _NativeEndpoint) -> None:
+        self._ep = ep
+
+    async def accept(self) -> QuicNativeUnfinalized:
+        return await self._ep.accept_q.get()
+
+    async def close(self) -> None:
+        self._ep.close()
+
+    @property
+    def port(self) -> int:
+        return self._ep.pump.port()
+
+
+def _resolve(host: Optional[str]) -> str:
+    import socket as _socket
+
+    if not host:
+        return "127.0.0.1"
+    try:
+        return _socket.gethostbyname(host)
+    except OSError:
+        return host
+
+
+class QuicNative(Protocol):
+    """QUIC profile with the C++ reliability datapath (see module note)."""
+
+    ca_cert_path: Optional[str] = None
+    ca_key_path: Optional[str] = None
+
+    @classmethod
+    async def connect(cls, endpoint: str, use_local_authority: bool,
+                      limiter: Limiter) -> Connection:
+        host, port = parse_endpoint(endpoint)
+        ep = _NativeEndpoint(server=False)
+        cid = int.from_bytes(secrets.token_bytes(8), "little")
+        conn = _NativeConn(ep, cid, (host or "127.0.0.1", port))
+        ep.conns[cid] = conn
+        if not ep.pump.connect(_resolve(host), port, cid, 0):
+            ep.close()
+            raise ConnectionError_(f"failed to connect to {endpoint}")
+        try:
+            await asyncio.wait_for(conn.established.wait(), HANDSHAKE_TIMEOUT_S)
+        except asyncio.TimeoutError:
+            ep.close()
+            raise ConnectionError_(f"QUIC handshake timeout to {endpoint}")
+        ctx = tlslib.client_context(use_local_authority, cls.ca_cert_path)
+        return await _wire_tls(conn, ctx, server_side=False,
+                               server_hostname=tlslib.CERT_NAME, limiter=limiter)
+
+    @classmethod
+    async def bind(cls, endpoint: str, certificate=None, key=None) -> QuicNativeListener:
+        host, port = parse_endpoint(endpoint)
+        ep = _NativeEndpoint(server=True)
+        bound = ep.pump.bind("" if not host else _resolve(host), port)
+        if bound < 0:
+            ep.close()
+            raise ConnectionError_(f"failed to bind {endpoint}")
+        return QuicNativeListener(ep)

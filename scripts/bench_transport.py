@@ -53,6 +53,8 @@ async def run_one(proto, payload_bytes: int, n_msgs: int) -> float:
 async def main(native: bool, proto_name: str = "") -> None:
     if proto_name == "quic":
         from pushcdn_amd.proto.transports.quic import Quic as proto
+    elif proto_name == "quic-native":
+        from pushcdn_amd.proto.transports.quic import QuicNative as proto
     elif proto_name == "tls":
         from pushcdn_amd.proto.transports.tcp_tls import TcpTls as proto
     elif native or proto_name == "native":
@@ -75,6 +77,7 @@ async def main(native: bool, proto_name: str = "") -> None:
 if __name__ == "__main__":
     p = argparse.ArgumentParser()
     p.add_argument("--native", action="store_true")
-    p.add_argument("--proto", default="", choices=["", "tcp", "native", "tls", "quic"])
+    p.add_argument("--proto", default="",
+                   choices=["", "tcp", "native", "tls", "quic", "quic-native"])
     a = p.parse_args()
     asyncio.run(main(a.native, a.proto))
