@@ -990,3 +990,68 @@ def test_quic_native_large_transfer_integrity():
         await listener.close()
 
     run(go())
+
+
+def test_quic_native_full_service_stack(tmp_path):
+    """marshal + broker + two clients over the NATIVE QUIC datapath
+    (C++ UdpPump reliability; same auth/permit/subscribe/route flows as
+    test_quic_full_service_stack)."""
+    import uuid as _uuid
+
+    from pushcdn_amd.broker.service import Broker, BrokerConfig
+    from pushcdn_amd.client import Client, ClientConfig
+    from pushcdn_amd.crypto import bls
+    from pushcdn_amd.discovery import BrokerIdentifier
+    from pushcdn_amd.marshal import Marshal, MarshalConfig
+    from pushcdn_amd.proto.transports.quic import QuicNative
+
+    async def go():
+        db = str(tmp_path / f"quicn-{_uuid.uuid4().hex}.db")
+        broker = Broker(BrokerConfig(
+            public_bind_endpoint="127.0.0.1:0",
+            public_advertise_endpoint="127.0.0.1:0",
+            private_bind_endpoint="127.0.0.1:0",
+            private_advertise_endpoint="127.0.0.1:0",
+            discovery_endpoint=db,
+            keypair=bls.KeyPair.from_seed(1001),
+            user_protocol=QuicNative,
+            broker_protocol=QuicNative,
+        ))
+        await broker.start()
+        pub = f"127.0.0.1:{broker._user_listener.port}"
+        priv = f"127.0.0.1:{broker._broker_listener.port}"
+        broker.config.public_advertise_endpoint = pub
+        broker.config.private_advertise_endpoint = priv
+        broker.identity = BrokerIdentifier(pub, priv)
+        broker.discovery.identity = broker.identity
+        broker.connections.identity = broker.identity
+        await broker.discovery.perform_heartbeat(0, 600)
+        marshal = Marshal(MarshalConfig(bind_endpoint="127.0.0.1:0",
+                                        discovery_endpoint=db,
+                                        protocol=QuicNative))
+        await marshal.start()
+        ep = f"127.0.0.1:{marshal._listener.port}"
+
+        alice = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(93),
+                                    subscribed_topics=[3], protocol=QuicNative))
+        bob = Client(ClientConfig(endpoint=ep, keypair=bls.KeyPair.from_seed(94),
+                                  subscribed_topics=[3], protocol=QuicNative))
+        await alice.ensure_initialized()
+        await bob.ensure_initialized()
+        await asyncio.sleep(0.2)
+
+        await alice.send_broadcast_message([3], b"quicn-broadcast")
+        msg = await asyncio.wait_for(bob.receive_message(), timeout=10)
+        assert msg.message == b"quicn-broadcast"
+        await bob.send_direct_message(alice.public_key, b"quicn-direct")
+        got = await asyncio.wait_for(alice.receive_message(), timeout=10)
+        while got.message == b"quicn-broadcast":
+            got = await asyncio.wait_for(alice.receive_message(), timeout=10)
+        assert got.message == b"quicn-direct"
+
+        alice.close()
+        bob.close()
+        await marshal.close()
+        await broker.close()
+
+    run(go())
