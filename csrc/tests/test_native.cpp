@@ -342,6 +342,47 @@ static int test_udp_stream_reliability() {
     return 0;
 }
 
+
+// the single-pass serializers must emit BYTE-IDENTICAL wire to the Builder
+// path for every padding case, and the view parser must agree with the
+// copying parser
+static int test_wire_single_pass_equivalence() {
+    std::mt19937 rng(77);
+    for (int trial = 0; trial < 300; ++trial) {
+        size_t rlen = rng() % 40;
+        size_t mlen = (trial < 50) ? rng() % 9 : rng() % 5000;
+        std::vector<uint8_t> rcpt(rlen), msg(mlen);
+        for (auto& b : rcpt) b = (uint8_t)rng();
+        for (auto& b : msg) b = (uint8_t)rng();
+        for (uint16_t disc : {wire::DIRECT, wire::BROADCAST}) {
+            auto old_bytes = disc == wire::DIRECT
+                ? wire::serialize_direct(rcpt.data(), rlen, msg.data(), mlen)
+                : wire::serialize_broadcast(rcpt.data(), rlen, msg.data(), mlen);
+            std::vector<uint8_t> neu(wire::payload_msg_wire_bytes(rlen, mlen), 0xAB);
+            wire::serialize_payload_msg_into(neu.data(), disc, rcpt.data(), rlen,
+                                             msg.data(), mlen);
+            CHECK(old_bytes == neu);
+            wire::Parsed p1;
+            wire::ParsedView p2;
+            CHECK(wire::deserialize(neu.data(), neu.size(), &p1));
+            CHECK(wire::deserialize_views(neu.data(), neu.size(), &p2));
+            CHECK(p1.disc == p2.disc);
+            CHECK(p1.payload.size() == p2.payload_len);
+            CHECK(p1.payload == std::vector<uint8_t>(p2.payload,
+                                                     p2.payload + p2.payload_len));
+            if (disc == wire::DIRECT) {
+                CHECK(p1.recipient ==
+                      std::vector<uint8_t>(p2.recipient, p2.recipient + p2.recipient_len));
+            } else {
+                CHECK(p1.topics ==
+                      std::vector<uint8_t>(p2.topics, p2.topics + p2.topics_len));
+            }
+        }
+    }
+    ++checks;
+    return 0;
+}
+
 int main() {
     if (test_wire_roundtrip()) return 1;
     if (test_wire_fuzz_no_crash()) return 1;
@@ -351,6 +392,7 @@ int main() {
     if (test_pump_ingest_and_batch_drain()) return 1;
     if (test_crdt_delta_fuzz()) return 1;
     if (test_udp_stream_reliability()) return 1;
+    if (test_wire_single_pass_equivalence()) return 1;
     printf("native tests OK (%d checks)\n", checks);
     return 0;
 }
