@@ -303,7 +303,7 @@ public:
                 (uint64_t)c.rx_ready.size(), (uint64_t)c.reorder.size()};
     }
 
-    // test hook: drop 1-in-`kill_every` outgoing STREAM/ACK datagrams
+    // test hook: drop `permille`/1000 of outgoing STREAM/ACK datagrams
     // (deterministic LCG — exercises retransmission without a lossy proxy)
     void debug_set_loss(uint32_t permille) { loss_permille_ = permille; }
 
@@ -351,9 +351,10 @@ private:
     }
 
     bool lossy_drop() {
-        if (loss_permille_ == 0) return false;
+        uint32_t lp = loss_permille_.load(std::memory_order_relaxed);
+        if (lp == 0) return false;
         lcg_ = lcg_ * 6364136223846793005ull + 1442695040888963407ull;
-        return (uint32_t)(lcg_ >> 33) % 1000 < loss_permille_;
+        return (uint32_t)(lcg_ >> 33) % 1000 < lp;
     }
 
     // mu_ held.  Datagram = [ptype][cid 8][payload].
@@ -750,7 +751,7 @@ private:
     std::map<uint64_t, UConn> conns_;
     std::map<uint64_t, uint32_t> dirty_;   // cid -> event flags
     std::vector<std::pair<uint64_t, int>> accepted_;
-    uint32_t loss_permille_ = 0;
+    std::atomic<uint32_t> loss_permille_{0};  // set from Python, read on pump thread
     uint64_t lcg_ = 0x9e3779b97f4a7c15ull;
 };
 
