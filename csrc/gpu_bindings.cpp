@@ -48,7 +48,8 @@ void launch_k1_dbg_wave(const uint8_t*, const uint8_t*, uint8_t*, const int64_t*
                         const uint8_t*, const uint64_t*, int32_t, int32_t, int32_t*,
                         hipStream_t);
 void launch_k3_fanout_wave(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
-                           const uint32_t*, const int32_t*, uint8_t*, int, int, hipStream_t);
+                           const uint32_t*, const int32_t*, int32_t, uint8_t*, int, int,
+                           hipStream_t);
 void launch_k3_fanout_flat2(const uint8_t*, const int64_t*, const int32_t*, const PairRec*,
                             uint32_t, const int32_t*, int32_t, int32_t, int32_t, uint8_t*,
                             int, int, hipStream_t);
@@ -297,7 +298,8 @@ void fanout_wave(torch::Tensor buf, torch::Tensor payload_off, torch::Tensor pay
     launch_k3_fanout_wave(buf.data_ptr<uint8_t>(), payload_off.data_ptr<int64_t>(),
                           payload_len.data_ptr<int32_t>(), pair_ptr(pairs),
                           (const uint32_t*)msg_seq.data_ptr<int32_t>(),
-                          n_pairs.data_ptr<int32_t>(), egress.data_ptr<uint8_t>(), (int)nt,
+                          n_pairs.data_ptr<int32_t>(), (int32_t)pairs.size(0),
+                          egress.data_ptr<uint8_t>(), (int)nt,
                           (int)grid, cur_stream());
 }
 

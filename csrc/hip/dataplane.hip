@@ -570,9 +570,15 @@ __global__ void __launch_bounds__(256) k3_fanout_wave_t(
     const PairRec* __restrict__ pairs,
     const uint32_t* __restrict__ msg_seq,
     const int32_t* __restrict__ n_pairs_ptr,
+    int32_t capacity,
     uint8_t* __restrict__ egress)
 {
-    const int n_pairs = *n_pairs_ptr;
+    // clamp: the claim counter keeps counting past the pair buffer when a
+    // tick oversubscribes pair_capacity (claims beyond it are recorded as
+    // drops, never stored) — reading pairs[] to the raw counter walked off
+    // the buffer and faulted (batch 1024 x 12.5k-subscriber mixed config)
+    int n_pairs = *n_pairs_ptr;
+    if (n_pairs > capacity) n_pairs = capacity;
     const int lane = threadIdx.x & 63;
     const int wave_in_wg = threadIdx.x >> 6;
     const int waves_total = gridDim.x * 4;
@@ -621,15 +627,16 @@ extern "C" {
 void launch_k3_fanout_wave(const uint8_t* buf, const int64_t* payload_off,
                            const int32_t* payload_len, const PairRec* pairs,
                            const uint32_t* msg_seq, const int32_t* n_pairs_ptr,
+                           int32_t capacity,
                            uint8_t* egress, int nt, int grid, hipStream_t s) {
     if (grid <= 0) grid = 4096;  // 4096 WGs x 4 waves = 16384 concurrent pairs
     if (nt)
         hipLaunchKernelGGL((k3_fanout_wave_t<1>), dim3(grid), dim3(256), 0, s, buf, payload_off,
-                           payload_len, pairs, msg_seq, n_pairs_ptr,
+                           payload_len, pairs, msg_seq, n_pairs_ptr, capacity,
                            egress);
     else
         hipLaunchKernelGGL((k3_fanout_wave_t<0>), dim3(grid), dim3(256), 0, s, buf, payload_off,
-                           payload_len, pairs, msg_seq, n_pairs_ptr,
+                           payload_len, pairs, msg_seq, n_pairs_ptr, capacity,
                            egress);
 }
 

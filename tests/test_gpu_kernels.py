@@ -656,3 +656,35 @@ def test_k1v3_wave_batch_matches_host(ops):
         ok = ops.bls_verify_batch_wave(vks_t, sigs_t, msgs_t, moff_t, lines, rand_r)
         torch.cuda.synchronize()
         assert ok.cpu().tolist() == want, f"trial {trial}"
+
+
+def test_pair_capacity_overflow_wave_path_is_safe():
+    """Regression: a tick that claims MORE delivery pairs than
+    pair_capacity must complete with the excess counted as drops — the
+    wave fan-out used to walk pairs[] to the raw claim counter and fault
+    (found at batch 1024 x 12.5k-subscriber 64 KiB mixed on MI355X)."""
+    from pushcdn_amd.broker.gpu_engine import GpuBrokerEngine, parse_ring_records
+
+    n_users, cap = 128, 64
+    payload = bytes(range(256)) * 32  # 8 KiB -> wave fan-out (rec > 4096)
+    eng = GpuBrokerEngine(device="cuda:0", n_users=n_users, ring_bytes=1 << 14,
+                          pair_capacity=cap, fanout_wire=True)
+    eng.subscribe_all([5])
+    raw = m.serialize(m.Broadcast([5], payload))
+    wire_len = (len(raw) + 15) & ~15
+    buf = raw + b"\x00" * (wire_len - len(raw))
+    dbuf, doff = eng.ingest(buf, [0, len(buf)])
+    eng.tick(dbuf, doff, uniform_wire_len=wire_len)
+    torch.cuda.synchronize()
+    wpos = eng.drain_cursors()
+    drops = int(eng._drops.cpu()[0])
+    delivered = sum(
+        len(parse_ring_records(eng.read_ring(u), int(wpos[u]))) for u in range(n_users)
+    )
+    # every subscriber claim beyond pair_capacity is a counted drop, and
+    # every record that WAS stored is intact
+    assert delivered <= cap
+    assert delivered + drops >= n_users
+    for u in range(n_users):
+        for _, rec in parse_ring_records(eng.read_ring(u), int(wpos[u])):
+            assert rec == buf
