@@ -383,6 +383,45 @@ static int test_wire_single_pass_equivalence() {
     return 0;
 }
 
+
+// garbage datagrams at a live server endpoint: truncated headers, huge
+// bogus offsets, unknown ptypes, STREAM floods for unknown cids — the
+// parser must neither crash nor accept a connection
+static int test_udp_pump_datagram_fuzz() {
+    net::UdpPump srv;
+    int port = srv.bind("127.0.0.1", 0);
+    CHECK(port > 0);
+    int fd = ::socket(AF_INET, SOCK_DGRAM, 0);
+    CHECK(fd >= 0);
+    sockaddr_in a{};
+    a.sin_family = AF_INET;
+    a.sin_port = htons((uint16_t)port);
+    a.sin_addr.s_addr = inet_addr("127.0.0.1");
+    std::mt19937 rng(99);
+    std::vector<uint8_t> pkt;
+    for (int i = 0; i < 3000; ++i) {
+        size_t len = rng() % 128;
+        if (i % 7 == 0) len = rng() % 2000;
+        pkt.resize(len);
+        for (auto& b : pkt) b = (uint8_t)rng();
+        if (i % 3 == 0 && len >= 9) pkt[0] = (uint8_t)(rng() % 8);  // plausible ptype
+        (void)::sendto(fd, pkt.data(), pkt.size(), 0, (sockaddr*)&a, sizeof(a));
+    }
+    usleep(50 * 1000);
+    // INIT-shaped garbage DOES accept (that is the wire contract); pure
+    // garbage must not have wedged the endpoint: a real handshake works
+    net::UdpPump cli;
+    CHECK(cli.connect("127.0.0.1", port, 0xF00Dull, 1));
+    for (int spins = 0; spins < 4000 && cli.client_status(0xF00Dull) != 1; ++spins)
+        usleep(1000);
+    CHECK(cli.client_status(0xF00Dull) == 1);
+    ++checks;
+    ::close(fd);
+    cli.stop();
+    srv.stop();
+    return 0;
+}
+
 int main() {
     if (test_wire_roundtrip()) return 1;
     if (test_wire_fuzz_no_crash()) return 1;
@@ -393,6 +432,7 @@ int main() {
     if (test_crdt_delta_fuzz()) return 1;
     if (test_udp_stream_reliability()) return 1;
     if (test_wire_single_pass_equivalence()) return 1;
+    if (test_udp_pump_datagram_fuzz()) return 1;
     printf("native tests OK (%d checks)\n", checks);
     return 0;
 }
