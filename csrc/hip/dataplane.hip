@@ -664,6 +664,10 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
     const int32_t* __restrict__ n_pairs_ptr,
     int32_t capacity,
     int32_t units_per_pair,
+    uint32_t div_magic,                   // Granlund-Montgomery multiplier for
+    int32_t div_shift,                    // f/units_per_pair (exact for f<2^31,
+                                          // d<=2^21; a per-unit 64-bit idiv was
+                                          // ~25 VALU ops on the hottest loop)
     int32_t uniform_len,                  // >0: every record is this long —
                                           // skips the per-unit length load
     uint8_t* __restrict__ egress)
@@ -675,8 +679,9 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
     const int64_t n_units = (int64_t)np * units_per_pair;
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t f = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; f < n_units; f += stride) {
-        const int p = (int)(f / units_per_pair);
-        const int unit = (int)(f - (int64_t)p * units_per_pair);
+        const uint32_t f32 = (uint32_t)f;  // n_units < 2^31 (capacity*units)
+        const int p = (int)(((uint64_t)f32 * div_magic) >> div_shift);
+        const int unit = (int)(f32 - (uint32_t)p * (uint32_t)units_per_pair);
         const PairRec pr = pairs[p];   // one dword4 load
         if (pr.user < 0) continue;
         const int mi = pr.msg;
@@ -713,20 +718,34 @@ __global__ void __launch_bounds__(256) k3_fanout_flat_t(
     }
 }
 
+// exact u32 division-by-constant: p = (f * m) >> sh for all f < 2^31,
+// 1 <= d <= 2^21 (Granlund-Montgomery with one headroom bit: l = ceil_log2 d,
+// m = ceil(2^(31+l)/d) < 2^32, sh = 31+l; m*d - 2^(31+l) < d <= 2^l)
+static inline void u32_div_magic(int32_t d, uint32_t* m, int32_t* sh) {
+    int l = 0;
+    while ((1u << l) < (uint32_t)d) ++l;  // ceil_log2(d), d >= 1
+    *m = (uint32_t)(((1ull << (31 + l)) + (uint64_t)d - 1) / (uint64_t)d);
+    *sh = 31 + l;
+}
+
 extern "C" void launch_k3_fanout_flat2(
     const uint8_t* buf, const int64_t* payload_off, const int32_t* payload_len,
     const PairRec* pairs,
     uint32_t seq_base, const int32_t* n_pairs_ptr, int32_t capacity, int32_t units_per_pair, int32_t uniform_len,
     uint8_t* egress, int nt, int grid, hipStream_t s) {
     if (grid <= 0) grid = 16384;  // swept: 16384 > 8192 > 4096 (~0.5% each)
+    uint32_t dm; int32_t dsh;
+    u32_div_magic(units_per_pair, &dm, &dsh);
     if (nt)
         hipLaunchKernelGGL((k3_fanout_flat_t<1, false>), dim3(grid), dim3(256), 0, s, buf,
                            payload_off, payload_len, pairs, seq_base,
-                           nullptr, n_pairs_ptr, capacity, units_per_pair, uniform_len, egress);
+                           nullptr, n_pairs_ptr, capacity, units_per_pair, dm, dsh,
+                           uniform_len, egress);
     else
         hipLaunchKernelGGL((k3_fanout_flat_t<0, false>), dim3(grid), dim3(256), 0, s, buf,
                            payload_off, payload_len, pairs, seq_base,
-                           nullptr, n_pairs_ptr, capacity, units_per_pair, uniform_len, egress);
+                           nullptr, n_pairs_ptr, capacity, units_per_pair, dm, dsh,
+                           uniform_len, egress);
 }
 
 extern "C" void launch_k3_fanout_flat3(
@@ -736,14 +755,18 @@ extern "C" void launch_k3_fanout_flat3(
     int32_t units_per_pair, int32_t uniform_len, uint8_t* egress, int nt, int grid,
     hipStream_t s) {
     if (grid <= 0) grid = 16384;
+    uint32_t dm; int32_t dsh;
+    u32_div_magic(units_per_pair, &dm, &dsh);
     if (nt)
         hipLaunchKernelGGL((k3_fanout_flat_t<1, true>), dim3(grid), dim3(256), 0, s, buf,
                            payload_off, payload_len, pairs, 0u,
-                           seq_state, n_pairs_ptr, capacity, units_per_pair, uniform_len, egress);
+                           seq_state, n_pairs_ptr, capacity, units_per_pair, dm, dsh,
+                           uniform_len, egress);
     else
         hipLaunchKernelGGL((k3_fanout_flat_t<0, true>), dim3(grid), dim3(256), 0, s, buf,
                            payload_off, payload_len, pairs, 0u,
-                           seq_state, n_pairs_ptr, capacity, units_per_pair, uniform_len, egress);
+                           seq_state, n_pairs_ptr, capacity, units_per_pair, dm, dsh,
+                           uniform_len, egress);
 }
 
 extern "C" __global__ void k_seq_advance(uint32_t* seq_state, int32_t m) {
