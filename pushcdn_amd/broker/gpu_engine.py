@@ -26,6 +26,8 @@ the engine's semantics are testable without a GPU.
 
 from __future__ import annotations
 
+import os
+
 from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
@@ -72,6 +74,10 @@ class GpuBrokerEngine:
         # non-temporal egress stores; direct_enabled can be turned off for
         # broadcast-only workloads to skip the per-tick K5 host check.
         self.pair_capacity = pair_capacity
+        # flat-vs-wave fan-out cutover (bytes/record): flat wins on small
+        # records (wave idles tail lanes), wave won by ~3% at >=4 KiB
+        # before the magic-division address math — env knob for A/B
+        self._flat_max_rec = int(os.environ.get("PUSHCDN_FLAT_MAX_REC", "4096"))
         self.nt_fanout = nt_fanout
         self.direct_enabled = direct_enabled
         # hash_seed keys the routing hash (keyhash.derive_routing_seed from
@@ -314,7 +320,7 @@ class GpuBrokerEngine:
         # would idle lanes on the tail pass; at >=4 KiB records a wave's 64
         # passes are already ~fully utilized and flat's per-unit index math
         # costs ~3% (measured on the 64 KiB mixed bench) — use wave there.
-        if uniform and rec <= 4096:
+        if uniform and rec <= self._flat_max_rec:
             units = rec // 16
             # uniform records: wire length passed as a scalar — saves the
             # per-unit payload_len load (~1 load per 16 B stored)
