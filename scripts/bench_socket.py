@@ -102,7 +102,6 @@ async def run_subscriber(args) -> None:
 
 async def run_sender(args) -> None:
     from pushcdn_amd.proto import message as m
-    from pushcdn_amd.proto.limiter import Bytes
 
     c = _client(args.endpoint, args.seed, [])
     await c.ensure_initialized()
