@@ -34,13 +34,38 @@ sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 TOPIC = 0
 
 
-def _client(endpoint, seed, topics):
+def _seed_shared_ca() -> None:
+    # TLS-bearing transports (quic profiles): every process must trust ONE
+    # CA; the coordinator exports its local CA and workers adopt it
+    import os
+
+    cert, key = os.environ.get("SOCKBENCH_CA_CERT"), os.environ.get("SOCKBENCH_CA_KEY")
+    if cert and key:
+        from pushcdn_amd.crypto import tls as tlslib
+
+        tlslib._LOCAL_CA = (cert, key)
+
+
+def _proto(name: str):
+    if name == "quic":
+        from pushcdn_amd.proto.transports.quic import Quic
+        return Quic
+    if name == "quic-native":
+        from pushcdn_amd.proto.transports.quic import QuicNative
+        return QuicNative
+    from pushcdn_amd.proto.transports.tcp_native import TcpNative
+    return TcpNative
+
+
+def _client(endpoint, seed, topics, transport="tcp-native"):
     from pushcdn_amd.client import Client, ClientConfig
     from pushcdn_amd.crypto import bls
-    from pushcdn_amd.proto.transports.tcp_native import TcpNative
+
+    _seed_shared_ca()
 
     return Client(ClientConfig(endpoint=endpoint, keypair=bls.KeyPair.from_seed(seed),
-                               subscribed_topics=list(topics), protocol=TcpNative))
+                               subscribed_topics=list(topics),
+                               protocol=_proto(transport)))
 
 
 async def run_subscriber(args) -> None:
@@ -48,7 +73,7 @@ async def run_subscriber(args) -> None:
 
     from pushcdn_amd.proto import message as m
 
-    clients = [_client(args.endpoint, args.seed + i, [TOPIC])
+    clients = [_client(args.endpoint, args.seed + i, [TOPIC], args.transport)
                for i in range(args.clients)]
     for c in clients:
         await c.ensure_initialized()
@@ -103,7 +128,7 @@ async def run_subscriber(args) -> None:
 async def run_sender(args) -> None:
     from pushcdn_amd.proto import message as m
 
-    c = _client(args.endpoint, args.seed, [])
+    c = _client(args.endpoint, args.seed, [], args.transport)
     await c.ensure_initialized()
     # pre-serialize one FRAMED Broadcast; each burst patches timestamps into
     # a repeated template blob and hands the whole burst to the C++ pump in
@@ -114,7 +139,9 @@ async def run_sender(args) -> None:
     payload_off = m.parse_offsets(wire)["payload_off"]
     framed = struct.pack(">I", len(wire)) + wire
     conn = await c._get_connection()
-    pump, cid = conn.pump_handle()
+    pump = cid = None
+    if hasattr(conn, "pump_handle"):   # tcp-native fast path (send_raw bursts)
+        pump, cid = conn.pump_handle()
     now = time.time()
     if args.t0 > now:
         await asyncio.sleep(args.t0 - now)
@@ -126,11 +153,21 @@ async def run_sender(args) -> None:
         deadline = time.time() + burst / args.rate
         for j in range(burst):
             struct.pack_into("<d", blob, j * stride + 4 + payload_off, time.time())
-        if not pump.send_raw(cid, bytes(blob)):
-            break
-        sent += burst
-        while pump.send_backlog(cid) > (64 << 20):
-            await asyncio.sleep(0.001)
+        if pump is not None:
+            if not pump.send_raw(cid, bytes(blob)):
+                break
+            sent += burst
+            while pump.send_backlog(cid) > (64 << 20):
+                await asyncio.sleep(0.001)
+        else:
+            # transport-generic path (QUIC profiles): client API sends
+            try:
+                for j in range(burst):
+                    pl = struct.pack("<d", time.time()) + b"\x00" * max(0, args.payload - 8)
+                    await c.send_broadcast_message([TOPIC], pl)
+            except Exception:
+                break
+            sent += burst
         dt = deadline - time.time()
         if dt > 0:
             await asyncio.sleep(dt)
@@ -152,6 +189,7 @@ async def run_coordinator(args) -> None:
     from pushcdn_amd.marshal import Marshal, MarshalConfig
     from pushcdn_amd.proto.transports.tcp_native import TcpNative
 
+    user_proto = _proto(args.transport)
     device = args.device or ("cuda:0" if torch.cuda.is_available() else "cpu")
     db = tempfile.mktemp(suffix=".db")
     broker = Broker(BrokerConfig(
@@ -161,7 +199,7 @@ async def run_coordinator(args) -> None:
         private_advertise_endpoint="127.0.0.1:0",
         discovery_endpoint=db,
         keypair=bls.KeyPair.from_seed(1000),
-        user_protocol=TcpNative,
+        user_protocol=user_proto,
         broker_protocol=TcpNative,
         data_plane="gpu",
         gpu_device=device,
@@ -179,8 +217,16 @@ async def run_coordinator(args) -> None:
     broker.connections.identity = broker.identity
     await broker.discovery.perform_heartbeat(0, 600)
     marshal = Marshal(MarshalConfig(bind_endpoint="127.0.0.1:0",
-                                    discovery_endpoint=db, protocol=TcpNative))
+                                    discovery_endpoint=db, protocol=user_proto))
     await marshal.start()
+    if args.transport != "tcp-native":
+        import os as _os
+
+        from pushcdn_amd.crypto import tls as tlslib
+
+        _ca_cert, _ca_key = tlslib.local_ca()
+        _os.environ["SOCKBENCH_CA_CERT"] = _ca_cert
+        _os.environ["SOCKBENCH_CA_KEY"] = _ca_key
     ep = f"127.0.0.1:{marshal._listener.port}"
 
     # subscriber workers
@@ -195,6 +241,7 @@ async def run_coordinator(args) -> None:
         p = await asyncio.create_subprocess_exec(
             sys.executable, __file__, "--role", "sub", "--endpoint", ep,
             "--clients", str(n), "--seed", str(seed),
+            "--transport", args.transport,
             stdin=asyncio.subprocess.PIPE, stdout=asyncio.subprocess.PIPE,
             limit=32 << 20)
         sub_procs.append(p)
@@ -217,6 +264,7 @@ async def run_coordinator(args) -> None:
             sys.executable, __file__, "--role", "send", "--endpoint", ep,
             "--seed", str(5000 + s), "--rate", str(args.rate / args.senders),
             "--payload", str(args.payload), "--t0", str(t0), "--t1", str(t1),
+            "--transport", args.transport,
             stdout=asyncio.subprocess.PIPE)
         send_procs.append(p)
 
@@ -287,6 +335,9 @@ def main():
     p.add_argument("--tag", default="")
     p.add_argument("--pump-shards", type=int, default=4)
     p.add_argument("--count-mode", action="store_true")
+    p.add_argument("--transport", default="tcp-native",
+                   choices=["tcp-native", "quic", "quic-native"],
+                   help="user-plane transport (broker mesh stays tcp-native)")
     args = p.parse_args()
     if args.role == "sub":
         asyncio.run(run_subscriber(args))
