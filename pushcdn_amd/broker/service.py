@@ -757,7 +757,12 @@ class Broker:
         idx = self._engine.next_staging_index()
         busy = self._drain_by_buffer.get(idx)
         if busy is not None:
-            await asyncio.shield(busy)  # buffer still being written out
+            try:
+                await asyncio.shield(busy)  # buffer still being written out
+            except Exception:
+                # a failed dispatch (e.g. eviction errors) must not kill the
+                # tick loop; the buffer is no longer being written either way
+                pass
         wpos, offsets, staging = self._engine.drain_compact()
         prev = self._last_drain
 
