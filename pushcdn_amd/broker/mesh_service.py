@@ -127,10 +127,10 @@ class MeshBroker(Broker):
 
     def _blocking_mesh_tick(self, msgs, dev_buf: Optional[torch.Tensor],
                             interests: int, owned_bits: int,
-                            batch_topics: int, direct_bits: int) -> torch.Tensor:
+                            batch_topics: int, direct_bits: int):
         """One mesh tick's blocking half (runs on the dedicated mesh thread):
         pack, H2D, collective exchange, kernel tick per received batch,
-        cursor drain.  Returns the drained ring cursors.
+        compacted drain.  Returns drain_compact's (wpos, offsets, staging).
 
         Peer failure (collective timeout/abort) tears the communicator down
         and the tick degrades to LOCAL-ONLY routing — the asyncio side then
