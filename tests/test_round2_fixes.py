@@ -476,10 +476,10 @@ def test_socket_bench_quic_native_cpu_smoke():
     root = Path(__file__).resolve().parent.parent
     out = subprocess.run(
         [sys.executable, "scripts/bench_socket.py", "--subs", "2",
-         "--sub-procs", "1", "--senders", "1", "--rate", "60",
-         "--seconds", "2", "--transport", "quic-native", "--device", "cpu",
+         "--sub-procs", "1", "--senders", "1", "--rate", "100",
+         "--seconds", "6", "--transport", "quic-native", "--device", "cpu",
          "--tag", "pysmoke"],
-        capture_output=True, text=True, timeout=120, cwd=root)
+        capture_output=True, text=True, timeout=180, cwd=root)
     lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
     assert lines, out.stdout[-2000:] + out.stderr[-2000:]
     d = json.loads(lines[-1])
