@@ -65,9 +65,16 @@ class Client:
         broker_conn = await self.protocol.connect(
             broker_endpoint, self.config.use_local_authority, self.limiter
         )
-        async with self._topics_lock:
-            topics = sorted(self._topics)
-        await UserAuth.authenticate_with_broker(broker_conn, permit, topics)
+        try:
+            async with self._topics_lock:
+                topics = sorted(self._topics)
+            await UserAuth.authenticate_with_broker(broker_conn, permit, topics)
+        except BaseException:
+            # auth failure / attempt-timeout cancellation after the broker
+            # socket opened: a reconnecting client must not leak one
+            # connection (and, for QUIC, one endpoint thread) per retry
+            broker_conn.close()
+            raise
         return broker_conn
 
     async def _get_connection(self) -> Connection:
