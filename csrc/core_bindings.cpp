@@ -309,7 +309,11 @@ static py::object w_deserialize(const py::bytes& raw) {
     d["timestamp"] = p.timestamp;
     d["public_key"] = py::bytes((const char*)p.public_key, p.public_key_len);
     d["signature"] = py::bytes((const char*)p.signature, p.signature_len);
-    d["context"] = p.context;
+    // context as BYTES: invalid UTF-8 from a malicious peer must surface
+    // as DeserializeError in Python (py::str conversion would raise
+    // UnicodeDecodeError out of this call), matching the pure-Python
+    // parser's read_text
+    d["context"] = py::bytes(p.context);
     d["recipient"] = py::bytes((const char*)p.recipient, p.recipient_len);
     d["topics"] = py::bytes((const char*)p.topics, p.topics_len);
     d["payload"] = py::bytes((const char*)p.payload, p.payload_len);

@@ -224,7 +224,11 @@ def deserialize(data: bytes) -> Message:
     if disc == 1:
         return AuthenticateWithPermit(d["timestamp"])
     if disc == 2:
-        return AuthenticateResponse(d["timestamp"], d["context"])
+        try:
+            context = d["context"].decode("utf-8")
+        except UnicodeDecodeError as e:
+            raise DeserializeError(f"invalid utf-8 in Text: {e}") from e
+        return AuthenticateResponse(d["timestamp"], context)
     if disc == 3:
         return Direct(d["recipient"], d["payload"])
     if disc == 4:

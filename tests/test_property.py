@@ -26,7 +26,13 @@ from pushcdn_amd.broker import versioned_map as vm
 
 pytestmark = pytest.mark.skipif(not HAVE_HYP, reason="hypothesis not installed")
 
-SET = settings(max_examples=80, deadline=None, derandomize=True)
+# HYP_EXAMPLES overrides the per-test example count (CI default 80;
+# deep local runs: HYP_EXAMPLES=2000 with HYP_RANDOM=1 for fresh seeds)
+import os as _os
+
+SET = settings(max_examples=int(_os.environ.get("HYP_EXAMPLES", "80")),
+               deadline=None,
+               derandomize=not bool(_os.environ.get("HYP_RANDOM")))
 
 u64 = st.integers(min_value=0, max_value=2**64 - 1)
 payload = st.binary(max_size=4096)
