@@ -422,6 +422,108 @@ static int test_udp_pump_datagram_fuzz() {
     return 0;
 }
 
+
+// adversarial lanes: the pump's structural classifier, the ring-record
+// parser, and the two wire parsers must be memory-safe on garbage AND the
+// copying/view parsers must AGREE on accept/reject + field bytes
+static int test_adversarial_parsers() {
+    std::mt19937 rng(4242);
+
+    // 1) classify_frame on random/mutated buffers
+    {
+        auto valid = wire::serialize_broadcast((const uint8_t*)"\x01\x02", 2,
+                                               (const uint8_t*)"payload", 7);
+        for (int trial = 0; trial < 4000; ++trial) {
+            std::vector<uint8_t> b;
+            if (trial % 2 == 0) {
+                b.assign(valid.begin(), valid.end());
+                for (int f = 0; f < 5; ++f)
+                    if (!b.empty()) b[rng() % b.size()] ^= 1u << (rng() % 8);
+                b.resize(rng() % (b.size() + 1));
+            } else {
+                b.resize(rng() % 96);
+                for (auto& x : b) x = (uint8_t)rng();
+            }
+            net::FrameMeta m = net::classify_frame(b.data(), b.size());
+            if (m.disc >= 0) {  // any views handed out must stay in-bounds
+                CHECK((size_t)m.topics_off + m.topics_cnt <= b.size());
+                CHECK((size_t)m.recip_off + m.recip_len <= b.size());
+            }
+        }
+        ++checks;
+    }
+
+    // 2) build_ring_frames on mutated ring bytes (via a pump batch drain
+    //    against a closed-conn id: builds frames, then drops them)
+    //    — exercised directly through the static helper's caller
+    //    send_rings_batch with id -1 (gone): counts become -1, but the
+    //    frame build runs first on the adversarial bytes.
+    {
+        net::Pump pump;
+        for (int trial = 0; trial < 600; ++trial) {
+            std::string ring;
+            int recs = rng() % 4;
+            for (int r = 0; r < recs; ++r) {
+                uint32_t len = rng() % 64;
+                uint32_t seq = rng();
+                std::string payload(len, (char)(rng() & 0xFF));
+                char hdr[16] = {0};
+                memcpy(hdr, &len, 4);
+                memcpy(hdr + 4, &seq, 4);
+                ring += std::string(hdr, 16) + payload;
+                ring += std::string((16 - (payload.size() % 16)) % 16, '\0');
+            }
+            // mutate: flip bytes incl. the length fields
+            std::vector<char> buf(ring.begin(), ring.end());
+            for (int f = 0; f < 6 && !buf.empty(); ++f)
+                buf[rng() % buf.size()] ^= 1 << (rng() % 8);
+            size_t wpos = buf.empty() ? 0 : rng() % (buf.size() + 1);
+            auto counts = pump.send_rings_batch(
+                (const uint8_t*)buf.data(), {99999},
+                {0}, {(int64_t)wpos});
+            CHECK(counts.size() == 2 && counts[0] == -1);  // conn unknown
+        }
+        ++checks;
+    }
+
+    // 3) deserialize vs deserialize_views agreement on mutated wire
+    {
+        auto base = wire::serialize_direct((const uint8_t*)"user-1", 6,
+                                           (const uint8_t*)"msg-payload", 11);
+        for (int trial = 0; trial < 6000; ++trial) {
+            std::vector<uint8_t> b(base.begin(), base.end());
+            int kind = rng() % 3;
+            if (kind == 0) {
+                for (int f = 0; f < 4; ++f)
+                    b[rng() % b.size()] ^= 1u << (rng() % 8);
+            } else if (kind == 1) {
+                b.resize(rng() % (b.size() + 1));
+            } else {
+                b.resize(16 + rng() % 128);
+                for (auto& x : b) x = (uint8_t)rng();
+            }
+            wire::Parsed p1;
+            wire::ParsedView p2;
+            bool ok1 = wire::deserialize(b.data(), b.size(), &p1);
+            bool ok2 = wire::deserialize_views(b.data(), b.size(), &p2);
+            CHECK(ok1 == ok2);
+            if (ok1) {
+                CHECK(p1.disc == p2.disc);
+                CHECK(p1.payload ==
+                      std::vector<uint8_t>(p2.payload, p2.payload + p2.payload_len));
+                CHECK(p1.recipient ==
+                      std::vector<uint8_t>(p2.recipient,
+                                           p2.recipient + p2.recipient_len));
+                CHECK(p1.topics ==
+                      std::vector<uint8_t>(p2.topics, p2.topics + p2.topics_len));
+                CHECK(p1.context == p2.context);
+            }
+        }
+        ++checks;
+    }
+    return 0;
+}
+
 int main() {
     if (test_wire_roundtrip()) return 1;
     if (test_wire_fuzz_no_crash()) return 1;
@@ -433,6 +535,7 @@ int main() {
     if (test_udp_stream_reliability()) return 1;
     if (test_wire_single_pass_equivalence()) return 1;
     if (test_udp_pump_datagram_fuzz()) return 1;
+    if (test_adversarial_parsers()) return 1;
     printf("native tests OK (%d checks)\n", checks);
     return 0;
 }
