@@ -47,27 +47,44 @@ class _Resp:
             await self.writer.drain()
             return await self._read_reply()
 
+    # a compromised/byzantine discovery server must not be able to OOM the
+    # broker (a "$<huge>" bulk length would readexactly() unboundedly) or
+    # escape the DiscoveryError contract with ValueError/UnicodeDecodeError
+    _MAX_BULK = 64 << 20
+    _MAX_ARRAY = 1 << 20
+
+    @staticmethod
+    def _int(rest: bytes) -> int:
+        try:
+            return int(rest)
+        except ValueError as e:
+            raise DiscoveryError(f"bad RESP integer {rest!r}") from e
+
     async def _read_reply(self):
         line = await self.reader.readline()
         if not line:
             raise DiscoveryError("redis connection closed")
         kind, rest = line[:1], line[1:].strip()
         if kind == b"+":
-            return rest.decode()
+            return rest.decode("utf-8", "replace")
         if kind == b"-":
-            raise DiscoveryError(f"redis error: {rest.decode()}")
+            raise DiscoveryError(f"redis error: {rest.decode('utf-8', 'replace')}")
         if kind == b":":
-            return int(rest)
+            return self._int(rest)
         if kind == b"$":
-            n = int(rest)
+            n = self._int(rest)
             if n == -1:
                 return None
+            if not 0 <= n <= self._MAX_BULK:
+                raise DiscoveryError(f"RESP bulk length {n} out of bounds")
             data = await self.reader.readexactly(n + 2)
             return data[:-2]
         if kind == b"*":
-            n = int(rest)
+            n = self._int(rest)
             if n == -1:
                 return None
+            if not 0 <= n <= self._MAX_ARRAY:
+                raise DiscoveryError(f"RESP array length {n} out of bounds")
             return [await self._read_reply() for _ in range(n)]
         raise DiscoveryError(f"bad RESP reply {line!r}")
 

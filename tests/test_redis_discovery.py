@@ -292,3 +292,61 @@ def test_redis_discovery_against_real_server(tmp_path):
     finally:
         proc.terminate()
         proc.wait(timeout=10)
+
+
+def test_resp_parser_rejects_byzantine_replies():
+    """A malicious/corrupted discovery server must surface DiscoveryError —
+    never OOM (unbounded bulk/array lengths), ValueError, or
+    UnicodeDecodeError."""
+    import asyncio
+
+    from pushcdn_amd.discovery.redis import _Resp
+    from pushcdn_amd.proto.errors import DiscoveryError
+
+    async def one(reply: bytes):
+        srv_reader = None
+
+        async def handler(reader, writer):
+            nonlocal srv_reader
+            srv_reader = reader
+            await reader.read(256)  # the command
+            writer.write(reply)
+            await writer.drain()
+
+        server = await asyncio.start_server(handler, "127.0.0.1", 0)
+        port = server.sockets[0].getsockname()[1]
+        r = _Resp("127.0.0.1", port)
+        try:
+            return await asyncio.wait_for(r.cmd("PING"), 5)
+        finally:
+            close = getattr(r, "close", None)
+            if close:
+                try:
+                    res = close()
+                    if asyncio.iscoroutine(res):
+                        await res
+                except Exception:
+                    pass
+            server.close()
+            await server.wait_closed()
+
+    async def go():
+        import pytest
+
+        # huge bulk length: bounded, not an attempted 1 TiB readexactly
+        with pytest.raises(DiscoveryError):
+            await one(b"$1099511627776\r\n")
+        # huge array length
+        with pytest.raises(DiscoveryError):
+            await one(b"*2147483647\r\n")
+        # non-integer length
+        with pytest.raises(DiscoveryError):
+            await one(b"$abc\r\n")
+        # invalid UTF-8 in a simple string must not raise UnicodeDecodeError
+        out = await one(b"+\xff\xfe\r\n")
+        assert isinstance(out, str)
+        # negative bulk other than -1
+        with pytest.raises(DiscoveryError):
+            await one(b"$-7\r\n")
+
+    asyncio.run(go())
