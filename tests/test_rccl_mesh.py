@@ -147,7 +147,7 @@ def test_mesh_interest_digest_routing_gloo(tmp_path):
         [
             sys.executable, "-m", "torch.distributed.run",
             "--nnodes=1", "--nproc-per-node=2",
-            "--master-addr", "127.0.0.1", "--master-port", "29531",
+            "--master-addr", "127.0.0.1", "--master-port", "29537",
             str(script),
         ],
         capture_output=True, text=True, timeout=300, env=env,
